@@ -1,0 +1,117 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: Lux-capability graph engine on MI355X.
+
+Headline config (BASELINE.json): PageRank pull on synthetic RMAT-27
+(|V|=2^27, |E|=2^31), GTEPS aggregate over N GPUs, strong scaling.
+Other configs via --app {pagerank,sssp,cc,cf} and --scale/--edges.
+
+Launched by the driver as
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+(one rank per GPU over RCCL); with no flags runs N=1 with defaults that
+finish in minutes. A "step" is one engine iteration over the whole graph;
+GTEPS counts ne edges per pull iteration (honest accounting).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from lux_amd import dist as dx  # noqa: E402
+from lux_amd.engine import DeviceCSC, GraphPart, PagerankEngine  # noqa: E402
+
+
+def build_engine(args, device):
+    if args.app == "pagerank":
+        full = DeviceCSC.rmat(args.scale, args.edges, seed=args.seed,
+                              device=device)
+        part = GraphPart(full, dx.world_size(), dx.rank())
+        return PagerankEngine(part), part
+    elif args.app == "cc":
+        from lux_amd.apps.cc import build_cc_bench
+        return build_cc_bench(args, device)
+    elif args.app == "sssp":
+        from lux_amd.apps.sssp import build_sssp_bench
+        return build_sssp_bench(args, device)
+    elif args.app == "cf":
+        from lux_amd.apps.cf import build_cf_bench
+        return build_cf_bench(args, device)
+    raise ValueError(args.app)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--app", default="pagerank")
+    ap.add_argument("--scale", type=int, default=27)
+    ap.add_argument("--edges", type=int, default=1 << 31)
+    ap.add_argument("--seed", type=int, default=1)
+    args = ap.parse_args()
+
+    dx.init_process_group("cuda")
+    rank = dx.rank()
+    world = dx.world_size()
+    local = dx.env_local_rank()
+    torch.cuda.set_device(local)
+    device = f"cuda:{local}"
+
+    engine, part = build_engine(args, device)
+    torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        engine.step()
+    dx.barrier()
+    torch.cuda.synchronize()
+
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        engine.step()
+    torch.cuda.synchronize()
+    dx.barrier()
+    t1 = time.perf_counter()
+    elapsed = t1 - t0
+
+    # max over ranks
+    et = torch.tensor([elapsed], device=device)
+    dx.all_reduce_max_(et)
+    elapsed = float(et.item())
+
+    edges_per_iter = args.edges
+    gteps = edges_per_iter * args.steps / elapsed / 1e9
+    ms_per_step = elapsed * 1000.0 / args.steps
+
+    if rank == 0:
+        out = {
+            "metric": "GTEPS",
+            "value": round(gteps, 3),
+            "unit": "GTEPS",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.app,
+                "graph": f"rmat{args.scale}" if args.app != "cf"
+                         else "netflix-shaped",
+                "nv": 1 << args.scale,
+                "ne": args.edges,
+                "parallelism": f"graph-partition x{world}",
+            },
+        }
+        print(json.dumps(out), flush=True)
+
+
+if __name__ == "__main__":
+    main()

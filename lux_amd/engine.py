@@ -1,0 +1,240 @@
+"""GPU engines: device graph construction + the pull-model engine (PageRank,
+dense SSSP/CC fallback). One process per GPU; partition slice per rank.
+
+Design (MI355X-first, not a Lux port): no task runtime — the per-iteration
+dependence chain (compute partition slice -> all-gather(v) slices) is
+expressed directly on one HIP stream per rank, with RCCL collectives over
+xGMI replacing the reference's zero-copy host staging
+(pull_app_task_impl, pagerank_gpu.cu:105-151).
+"""
+import math
+
+import torch
+
+from . import _native_gpu as ng
+from . import dist as dx
+
+U32 = torch.int32
+U64 = torch.int64
+F32 = torch.float32
+
+
+def _stream():
+    return torch.cuda.current_stream().cuda_stream
+
+
+def partition_bounds(col_end, ne, nparts):
+    """Edge-balanced contiguous ranges from a device col_end tensor (u64 end
+    offsets). Same greedy rule as the CPU partitioner
+    (src/core/luxio.cpp partition_edge_balanced)."""
+    nv = col_end.numel()
+    cap = (ne + nparts - 1) // nparts
+    targets = torch.tensor(
+        [min(cap * (p + 1), ne) for p in range(nparts)], dtype=U64,
+        device=col_end.device)
+    cuts = torch.searchsorted(col_end, targets, right=True)  # first v with end>target
+    cuts = cuts.cpu().tolist()  # small host sync at init time only
+    row_left, row_right = [], []
+    v = 0
+    for p in range(nparts):
+        rl = v
+        rr = max(cuts[p] - 1, rl) if p < nparts - 1 else nv - 1
+        rr = min(rr, nv - 1)
+        if rl >= nv:  # exhausted: empty partition
+            row_left.append(1)
+            row_right.append(0)
+            continue
+        row_left.append(rl)
+        row_right.append(rr)
+        v = rr + 1
+    return row_left, row_right
+
+
+class DeviceCSC:
+    """Full CSC graph resident on one GPU (col_end u64[nv], src u32[ne],
+    optional weight i32[ne])."""
+
+    def __init__(self, nv, ne, col_end, src, weight=None):
+        self.nv, self.ne = nv, ne
+        self.col_end, self.src, self.weight = col_end, src, weight
+
+    @classmethod
+    def rmat(cls, scale, ne, seed=1, device="cuda"):
+        nv = 1 << scale
+        s = _stream()
+        esrc = torch.empty(ne, dtype=U32, device=device)
+        edst = torch.empty(ne, dtype=U32, device=device)
+        ng.rmat_edges(s, seed, scale, ne, esrc, edst)
+        g = cls._from_device_edges(nv, ne, esrc, edst, None, device)
+        return g
+
+    @classmethod
+    def bipartite(cls, n_users, n_items, ne, seed=1, device="cuda"):
+        nv = n_users + n_items
+        s = _stream()
+        esrc = torch.empty(ne, dtype=U32, device=device)
+        edst = torch.empty(ne, dtype=U32, device=device)
+        ew = torch.empty(ne, dtype=U32, device=device)
+        ng.bipartite_edges(s, seed, n_users, n_items, ne, esrc, edst, ew)
+        return cls._from_device_edges(nv, ne, esrc, edst, ew, device)
+
+    @classmethod
+    def _from_device_edges(cls, nv, ne, esrc, edst, ew, device):
+        s = _stream()
+        col_end = torch.empty(nv, dtype=U64, device=device)
+        out_src = torch.empty(ne, dtype=U32, device=device)
+        out_w = torch.empty(ne, dtype=U32, device=device) if ew is not None \
+            else None
+        hist = torch.zeros(nv, dtype=U32, device=device)
+        cursor = torch.empty(nv, dtype=U64, device=device)
+        partials = torch.empty(ng.scan_partials_size(nv), dtype=U64,
+                               device=device)
+        ng.edges_to_csc(s, nv, ne, esrc, edst, ew, col_end, out_src, out_w,
+                        hist, cursor, partials)
+        torch.cuda.synchronize()
+        del esrc, edst, ew, hist, cursor, partials
+        return cls(nv, ne, col_end, out_src, out_w)
+
+    @classmethod
+    def from_host(cls, g, device="cuda"):
+        col_end = torch.from_numpy(g.col_end.view("int64")).to(device)
+        src = torch.from_numpy(g.src.view("int32")).to(device)
+        w = None
+        if g.weight is not None:
+            w = torch.from_numpy(g.weight).to(device)
+        return cls(g.nv, g.ne, col_end, src, w)
+
+
+class GraphPart:
+    """This rank's partition: local row_ptr (u64[vp+1], 0-based), local col
+    slice, plus the global partition table (all ranks' bounds)."""
+
+    def __init__(self, full: DeviceCSC, nparts, my_part, keep_full=False):
+        device = full.col_end.device
+        self.nv, self.ne = full.nv, full.ne
+        self.nparts, self.p = nparts, my_part
+        rl, rr = partition_bounds(full.col_end, full.ne, nparts)
+        self.row_left_all, self.row_right_all = rl, rr
+        self.verts_all = [max(rr[p] - rl[p] + 1, 0) for p in range(nparts)]
+        self.row_left, self.row_right = rl[my_part], rr[my_part]
+        self.vp = self.verts_all[my_part]
+        s = _stream()
+        # local edge range from col_end (small D2H)
+        ce = full.col_end
+        self.col_left = 0 if self.row_left == 0 else int(
+            ce[self.row_left - 1].item())
+        self.col_right = int(ce[self.row_right].item())
+        self.ep = self.col_right - self.col_left
+        self.row_ptr = torch.empty(self.vp + 1, dtype=U64, device=device)
+        ng.local_row_ptr(s, self.vp, self.col_left,
+                         ce.narrow(0, self.row_left, self.vp), self.row_ptr)
+        self.col = full.src.narrow(0, self.col_left, self.ep).clone()
+        self.weight = None
+        if full.weight is not None:
+            self.weight = full.weight.narrow(0, self.col_left,
+                                             self.ep).clone()
+        if not keep_full:
+            # release the full graph (each rank keeps only its slice)
+            full.col_end = full.src = full.weight = None
+            torch.cuda.empty_cache()
+        self.device = device
+
+    def build_bins(self):
+        device = self.device
+        s = _stream()
+        bin0 = torch.empty(self.vp, dtype=U32, device=device)
+        bin1 = torch.empty(self.vp, dtype=U32, device=device)
+        nbig_max = max(min(self.vp, self.ep // 2048 + 1), 1)
+        n2_max = self.ep // 8192 + nbig_max + 1
+        bin2 = torch.empty(n2_max * 2, dtype=U32, device=device)
+        bin2v = torch.empty(nbig_max, dtype=U32, device=device)
+        counters = torch.zeros(4, dtype=U32, device=device)
+        ng.build_bins(s, self.vp, self.row_ptr, bin0, bin1, bin2, bin2v,
+                      counters)
+        c = counters.cpu()
+        self.n0, self.n1, self.n2, self.nbig = (int(c[0]), int(c[1]),
+                                                int(c[2]), int(c[3]))
+        self.bin0, self.bin1, self.bin2, self.bin2v = bin0, bin1, bin2, bin2v
+
+
+class PagerankEngine:
+    """Distributed pull PageRank. State: old ranks replicated f32[nv]
+    (pre-divided by out-degree, the reference's stored form), new slice
+    f32[vp]; per iteration compute + all-gather(v)."""
+
+    def __init__(self, part: GraphPart):
+        self.part = part
+        device = part.device
+        part.build_bins()
+        s = _stream()
+        # global out-degrees: histogram over my edges, then sum-all-reduce
+        deg = torch.zeros(part.nv, dtype=U32, device=device)
+        ng.hist_u32(s, part.ep, part.col, deg)
+        dx.all_reduce_sum_(deg)
+        self.deg = deg
+        # init: rank/deg (deg==0 -> rank), pagerank_gpu.cu:255-259
+        rank0 = 1.0 / part.nv
+        degf = deg.to(F32)
+        self.old = torch.where(deg == 0, torch.full_like(degf, rank0),
+                               rank0 / degf.clamp(min=1.0))
+        self.new_part = torch.empty(part.vp, dtype=F32, device=device)
+        self.init_rank = (1.0 - 0.15) / part.nv
+
+    def step(self):
+        p = self.part
+        s = _stream()
+        ng.pull_iter(s, ng.PULL_PR, p.n0, p.bin0, p.n1, p.bin1, p.n2, p.bin2,
+                     p.nbig, p.bin2v, p.row_ptr, p.col, self.old,
+                     self.new_part, self.deg, p.row_left, self.init_rank)
+        dx.all_gather_slices(self.old, self.new_part, p.verts_all,
+                             p.row_left_all)
+
+    def ranks(self):
+        """Replicated stored ranks (pr/out_degree) as a torch tensor."""
+        return self.old
+
+
+class LabelPullEngine:
+    """Dense label-propagation pull engine (SSSP min / CC max) — both the
+    standalone dense path and the push engine's pull fallback."""
+
+    def __init__(self, part: GraphPart, mode, init_labels):
+        assert mode in (ng.PULL_MIN, ng.PULL_MAX)
+        self.part = part
+        self.mode = mode
+        if not hasattr(part, "bin0"):
+            part.build_bins()
+        self.old = init_labels  # u32[nv] replicated (as int32 tensor)
+        self.new_part = torch.empty(part.vp, dtype=U32, device=part.device)
+
+    def step(self):
+        p = self.part
+        s = _stream()
+        ng.pull_iter(s, self.mode, p.n0, p.bin0, p.n1, p.bin1, p.n2, p.bin2,
+                     p.nbig, p.bin2v, p.row_ptr, p.col, self.old,
+                     self.new_part, None, p.row_left, 0.0)
+        # changed count before the gather overwrites old
+        changed = (self.new_part
+                   != self.old.narrow(0, p.row_left, p.vp)).sum()
+        dx.all_gather_slices(self.old, self.new_part, p.verts_all,
+                             p.row_left_all)
+        return changed
+
+    def run_to_fixpoint(self, max_iters=None):
+        it = 0
+        pending = []
+        while True:
+            changed = self.step()
+            pending.append(changed)
+            it += 1
+            # lag convergence checks like the reference's 4-deep sliding
+            # window (sssp.cc:111-129) to avoid a host sync per iteration
+            if len(pending) >= 4 or (max_iters and it >= max_iters):
+                total = torch.stack(pending).sum()
+                dx.all_reduce_sum_(total)
+                if int(total.item()) == 0:
+                    break
+                pending = []
+            if max_iters and it >= max_iters:
+                break
+        return it

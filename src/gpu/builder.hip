@@ -1,0 +1,219 @@
+// GPU graph construction: R-MAT edge generation + CSC build (counting sort
+// by dst: histogram -> device-wide exclusive scan -> scatter), plus degree
+// histograms. All kernels gfx950-native; the device-wide scan replaces the
+// reference's serial single-thread prefix sum (sssp_gpu.cu:550-565 — a
+// deliberate wart SURVEY.md §7 says to fix).
+//
+// Edge generation is bit-identical to the CPU generator (src/include/lux/rmat.h
+// is compiled into both), so GPU-built graphs are validated against CPU ones.
+#include "gpu_common.h"
+#include "lux/rmat.h"
+
+namespace lux {
+
+// ---------------- edge generation ----------------
+
+__global__ void rmat_edges_kernel(uint64_t seed, int scale, uint64_t ne,
+                                  V_ID* src, V_ID* dst) {
+  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
+  for (uint64_t e = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; e < ne;
+       e += stride) {
+    rmat_edge(seed, e, scale, &src[e], &dst[e]);
+  }
+}
+
+__global__ void bipartite_edges_kernel(uint64_t seed, V_ID n_users,
+                                       V_ID n_items, int item_scale,
+                                       uint64_t ne, V_ID* src, V_ID* dst,
+                                       WeightType* w) {
+  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
+  for (uint64_t e = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; e < ne;
+       e += stride) {
+    bipartite_edge(seed, e, n_users, n_items, item_scale, &src[e], &dst[e]);
+    w[e] = rmat_weight(seed, e);
+  }
+}
+
+// ---------------- histogram ----------------
+
+__global__ void hist_u32_kernel(uint64_t n, const V_ID* ids, uint32_t* hist) {
+  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
+  for (uint64_t e = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; e < n;
+       e += stride)
+    atomicAdd(&hist[ids[e]], 1u);
+}
+
+// ---------------- device-wide scan u32 -> u64 (inclusive end-offsets) ----
+
+constexpr int SCAN_ITEMS = 16;
+constexpr int SCAN_TILE = BLOCK * SCAN_ITEMS;  // 4096
+
+__global__ void scan_reduce_kernel(uint32_t n, const uint32_t* in,
+                                   unsigned long long* partials) {
+  __shared__ unsigned long long lds[BLOCK / WAVE];
+  uint64_t base = (uint64_t)blockIdx.x * SCAN_TILE;
+  unsigned long long sum = 0;
+  for (int i = 0; i < SCAN_ITEMS; i++) {
+    uint64_t idx = base + threadIdx.x + (uint64_t)i * BLOCK;
+    if (idx < n) sum += in[idx];
+  }
+  sum = block_reduce_sum(sum, lds);
+  if (threadIdx.x == 0) partials[blockIdx.x] = sum;
+}
+
+__global__ void scan_partials_kernel(uint32_t nblocks,
+                                     unsigned long long* partials) {
+  // single block of 1024 threads, looped with carry
+  __shared__ unsigned long long lds[1024 / WAVE + 1];
+  unsigned long long carry = 0;
+  for (uint32_t base = 0; base < nblocks; base += 1024) {
+    uint32_t idx = base + threadIdx.x;
+    unsigned long long v = idx < nblocks ? partials[idx] : 0;
+    unsigned long long total;
+    unsigned long long ex = block_exscan<unsigned long long, 1024>(v, lds,
+                                                                   &total);
+    if (idx < nblocks) partials[idx] = carry + ex;
+    carry += total;
+    __syncthreads();
+  }
+}
+
+__global__ void scan_apply_kernel(uint32_t n, const uint32_t* in,
+                                  const unsigned long long* partials,
+                                  E_ID* out_end) {
+  __shared__ unsigned long long lds[BLOCK / WAVE + 1];
+  uint64_t base = (uint64_t)blockIdx.x * SCAN_TILE;
+  // per-thread sequential over its contiguous run of SCAN_ITEMS
+  uint32_t vals[SCAN_ITEMS];
+  unsigned long long mysum = 0;
+  uint64_t tbase = base + (uint64_t)threadIdx.x * SCAN_ITEMS;
+  for (int i = 0; i < SCAN_ITEMS; i++) {
+    uint64_t idx = tbase + i;
+    vals[i] = idx < n ? in[idx] : 0;
+    mysum += vals[i];
+  }
+  unsigned long long ex =
+      block_exscan<unsigned long long, BLOCK>(mysum, lds, nullptr);
+  unsigned long long run = partials[blockIdx.x] + ex;
+  for (int i = 0; i < SCAN_ITEMS; i++) {
+    uint64_t idx = tbase + i;
+    run += vals[i];
+    if (idx < n) out_end[idx] = run;  // inclusive end offset
+  }
+}
+
+void scan_u32_to_end_u64(hipStream_t s, uint32_t n, const uint32_t* in,
+                         E_ID* out_end, unsigned long long* partials) {
+  uint32_t nblocks = (uint32_t)((n + (uint64_t)SCAN_TILE - 1) / SCAN_TILE);
+  hipLaunchKernelGGL(scan_reduce_kernel, dim3(nblocks), dim3(BLOCK), 0, s, n,
+                     in, partials);
+  hipLaunchKernelGGL(scan_partials_kernel, dim3(1), dim3(1024), 0, s, nblocks,
+                     partials);
+  hipLaunchKernelGGL(scan_apply_kernel, dim3(nblocks), dim3(BLOCK), 0, s, n,
+                     in, partials, out_end);
+}
+
+// ---------------- scatter (counting sort by dst) ----------------
+
+__global__ void init_cursor_kernel(uint32_t nv, const E_ID* col_end,
+                                   unsigned long long* cursor) {
+  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
+  for (uint64_t v = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; v < nv;
+       v += stride)
+    cursor[v] = v == 0 ? 0ull : (unsigned long long)col_end[v - 1];
+}
+
+__global__ void scatter_kernel(uint64_t ne, const V_ID* src, const V_ID* dst,
+                               const WeightType* w,
+                               unsigned long long* cursor, V_ID* out_src,
+                               WeightType* out_w) {
+  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
+  for (uint64_t e = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; e < ne;
+       e += stride) {
+    unsigned long long pos = atomicAdd(&cursor[dst[e]], 1ull);
+    out_src[pos] = src[e];
+    if (w) out_w[pos] = w[e];
+  }
+}
+
+// ---------------- local row_ptr from global col_end slice ----------------
+
+__global__ void local_row_ptr_kernel(uint32_t vp, E_ID col_left,
+                                     const E_ID* col_end_slice,
+                                     E_ID* row_ptr_loc) {
+  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
+  for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; i <= vp;
+       i += stride) {
+    row_ptr_loc[i] = i == 0 ? 0 : col_end_slice[i - 1] - col_left;
+  }
+}
+
+}  // namespace lux
+
+// ---------------- C ABI ----------------
+
+using namespace lux;
+
+extern "C" {
+
+void lux_gpu_rmat_edges(uint64_t stream, uint64_t seed, int scale,
+                        uint64_t ne, V_ID* src, V_ID* dst) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(rmat_edges_kernel, dim3(grid_for(ne)), dim3(BLOCK), 0, s,
+                     seed, scale, ne, src, dst);
+}
+
+void lux_gpu_bipartite_edges(uint64_t stream, uint64_t seed, V_ID n_users,
+                             V_ID n_items, uint64_t ne, V_ID* src, V_ID* dst,
+                             WeightType* w) {
+  hipStream_t s = (hipStream_t)stream;
+  int item_scale = 0;
+  while (((V_ID)1 << item_scale) < n_items) item_scale++;
+  hipLaunchKernelGGL(bipartite_edges_kernel, dim3(grid_for(ne)), dim3(BLOCK),
+                     0, s, seed, n_users, n_items, item_scale, ne, src, dst,
+                     w);
+}
+
+void lux_gpu_hist_u32(uint64_t stream, uint64_t n, const V_ID* ids,
+                      uint32_t* hist /*pre-zeroed*/) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(hist_u32_kernel, dim3(grid_for(n)), dim3(BLOCK), 0, s, n,
+                     ids, hist);
+}
+
+uint32_t lux_gpu_scan_partials_size(uint32_t n) {
+  return (uint32_t)((n + (uint64_t)SCAN_TILE - 1) / SCAN_TILE);
+}
+
+void lux_gpu_scan_end_offsets(uint64_t stream, uint32_t n, const uint32_t* in,
+                              E_ID* out_end, unsigned long long* partials) {
+  scan_u32_to_end_u64((hipStream_t)stream, n, in, out_end, partials);
+}
+
+// Full CSC build from device edge lists. hist must be pre-zeroed u32[nv];
+// cursor u64[nv]; partials u64[lux_gpu_scan_partials_size(nv)].
+void lux_gpu_edges_to_csc(uint64_t stream, uint32_t nv, uint64_t ne,
+                          const V_ID* src, const V_ID* dst,
+                          const WeightType* w, E_ID* col_end, V_ID* out_src,
+                          WeightType* out_w, uint32_t* hist,
+                          unsigned long long* cursor,
+                          unsigned long long* partials) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(hist_u32_kernel, dim3(grid_for(ne)), dim3(BLOCK), 0, s,
+                     ne, dst, hist);
+  scan_u32_to_end_u64(s, nv, hist, col_end, partials);
+  hipLaunchKernelGGL(init_cursor_kernel, dim3(grid_for(nv)), dim3(BLOCK), 0,
+                     s, nv, col_end, cursor);
+  hipLaunchKernelGGL(scatter_kernel, dim3(grid_for(ne)), dim3(BLOCK), 0, s,
+                     ne, src, dst, w, cursor, out_src, out_w);
+}
+
+void lux_gpu_local_row_ptr(uint64_t stream, uint32_t vp, E_ID col_left,
+                           const E_ID* col_end_slice, E_ID* row_ptr_loc) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(local_row_ptr_kernel, dim3(grid_for((uint64_t)vp + 1)),
+                     dim3(BLOCK), 0, s, vp, col_left, col_end_slice,
+                     row_ptr_loc);
+}
+
+}  // extern "C"
