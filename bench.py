@@ -26,12 +26,25 @@ from lux_amd import dist as dx  # noqa: E402
 from lux_amd.engine import DeviceCSC, GraphPart, PagerankEngine  # noqa: E402
 
 
+def apply_app_defaults(args):
+    """BASELINE.md config shapes when --edges/--scale are left at default."""
+    defaults = {
+        "pagerank": dict(),                      # RMAT-27, 2^31 edges
+        "sssp": dict(),                          # RMAT-27, 2^31 edges
+        "cc": dict(edges=1468365182, nv=41652230),   # Twitter-2010-shaped
+        "cf": dict(edges=200961014, nv=497959),      # NetFlix-shaped
+    }
+    d = defaults[args.app]
+    if args.edges == (1 << 31) and "edges" in d:
+        args.edges = d["edges"]
+    args.nv = d.get("nv", 1 << args.scale)
+    return args
+
+
 def build_engine(args, device):
     if args.app == "pagerank":
-        full = DeviceCSC.rmat(args.scale, args.edges, seed=args.seed,
-                              device=device)
-        part = GraphPart(full, dx.world_size(), dx.rank())
-        return PagerankEngine(part), part
+        from lux_amd.apps.pagerank import build_pagerank_bench
+        return build_pagerank_bench(args, device)
     elif args.app == "cc":
         from lux_amd.apps.cc import build_cc_bench
         return build_cc_bench(args, device)
@@ -54,6 +67,7 @@ def main():
     ap.add_argument("--edges", type=int, default=1 << 31)
     ap.add_argument("--seed", type=int, default=1)
     args = ap.parse_args()
+    apply_app_defaults(args)
 
     dx.init_process_group("cuda")
     rank = dx.rank()
@@ -103,9 +117,11 @@ def main():
             "data": "synthetic",
             "config": {
                 "model": args.app,
-                "graph": f"rmat{args.scale}" if args.app != "cf"
-                         else "netflix-shaped",
-                "nv": 1 << args.scale,
+                "graph": {"pagerank": f"rmat{args.scale}",
+                          "sssp": f"rmat{args.scale}",
+                          "cc": "twitter2010-shaped",
+                          "cf": "netflix-shaped"}[args.app],
+                "nv": args.nv,
                 "ne": args.edges,
                 "parallelism": f"graph-partition x{world}",
             },

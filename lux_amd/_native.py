@@ -70,6 +70,15 @@ def rmat_edges(seed, scale, ne):
     return src, dst
 
 
+def rmat_edges_folded(seed, scale, nv, ne):
+    src = np.empty(ne, np.uint32)
+    dst = np.empty(ne, np.uint32)
+    cpu.lux_rmat_edges_folded(ctypes.c_uint64(seed), ctypes.c_int(scale),
+                              ctypes.c_uint32(nv), ctypes.c_uint64(ne),
+                              ptr(src, u32p), ptr(dst, u32p))
+    return src, dst
+
+
 def bipartite_edges(seed, n_users, n_items, ne):
     src = np.empty(ne, np.uint32)
     dst = np.empty(ne, np.uint32)

@@ -49,6 +49,20 @@ def rmat_edges(stream, seed, scale, ne, src, dst):
                              _u64(ne), dp(src), dp(dst))
 
 
+def rmat_edges_folded(stream, seed, scale, nv, ne, src, dst):
+    lib().lux_gpu_rmat_edges_folded(_u64(stream), _u64(seed),
+                                    ctypes.c_int(scale), _u32(nv), _u64(ne),
+                                    dp(src), dp(dst))
+
+
+def cf_iter(stream, n0, bin0, n1, bin1, n2, bin2, nbig, bin2v, row_ptr, col,
+            w, oldv, newv, row_left, K):
+    lib().lux_gpu_cf_iter(_u64(stream), _u32(n0), dp(bin0), _u32(n1),
+                          dp(bin1), _u32(n2), dp(bin2), _u32(nbig), dp(bin2v),
+                          dp(row_ptr), dp(col), dp(w), dp(oldv), dp(newv),
+                          _u32(row_left), ctypes.c_int(K))
+
+
 def bipartite_edges(stream, seed, n_users, n_items, ne, src, dst, w):
     lib().lux_gpu_bipartite_edges(_u64(stream), _u64(seed), _u32(n_users),
                                   _u32(n_items), _u64(ne), dp(src), dp(dst),
@@ -83,6 +97,40 @@ def local_row_ptr(stream, vp, col_left, col_end_slice, row_ptr_loc):
 def build_bins(stream, vp, row_ptr, bin0, bin1, bin2, bin2v, counters):
     lib().lux_gpu_build_bins(_u64(stream), _u32(vp), dp(row_ptr), dp(bin0),
                              dp(bin1), dp(bin2), dp(bin2v), dp(counters))
+
+
+def csr_scatter(stream, ep, col, row_ptr_loc, vp, row_left, cursor,
+                push_col):
+    lib().lux_gpu_csr_scatter(_u64(stream), _u64(ep), dp(col),
+                              dp(row_ptr_loc), _u32(vp), _u32(row_left),
+                              dp(cursor), dp(push_col))
+
+
+def push_scatter(stream, is_min, old_dense, new_dense, in_row_left, in_count,
+                 old_seg, push_row_ptr, push_col, old_labels, snapshot,
+                 new_labels, my_row_left, new_seg, capacity):
+    lib().lux_gpu_push_scatter(
+        _u64(stream), ctypes.c_int(is_min), ctypes.c_int(old_dense),
+        ctypes.c_int(new_dense), _u32(in_row_left), _u32(in_count),
+        dp(old_seg), dp(push_row_ptr), dp(push_col), dp(old_labels),
+        dp(snapshot), dp(new_labels), _u32(my_row_left), dp(new_seg),
+        _u32(capacity))
+
+
+def build_bitmap(stream, vp, snapshot, new_labels, seg):
+    lib().lux_gpu_build_bitmap(_u64(stream), _u32(vp), dp(snapshot),
+                               dp(new_labels), dp(seg))
+
+
+def d2s(stream, vp, row_left, dense_seg, sparse_seg):
+    lib().lux_gpu_d2s(_u64(stream), _u32(vp), _u32(row_left), dp(dense_seg),
+                      dp(sparse_seg))
+
+
+def check(stream, is_min, vp, row_left, row_ptr_loc, col, labels, mistakes):
+    lib().lux_gpu_check(_u64(stream), ctypes.c_int(is_min), _u32(vp),
+                        _u32(row_left), dp(row_ptr_loc), dp(col), dp(labels),
+                        dp(mistakes))
 
 
 PULL_PR = 0

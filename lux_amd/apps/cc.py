@@ -1,0 +1,56 @@
+"""Connected Components app driver (reference parity:
+components/components.cc; max-label propagation, push model)."""
+import sys
+
+from .. import dist as dx
+from ..engine import GraphPart
+from ..push_engine import PushEngine
+from .common import (ElapsedTimer, load_device_graph, parse_input_args,
+                     print_memory_estimate)
+
+
+class CCBench:
+    def __init__(self, part):
+        self.eng = PushEngine(part, PushEngine.MODE_MAX)
+
+    def step(self):
+        self.eng.reset()
+        self.eng.run()
+
+
+def build_cc_bench(args, device):
+    from ..engine import DeviceCSC
+    # Twitter-2010-shaped synthetic
+    nv = getattr(args, "nv", None) or 41652230
+    full = DeviceCSC.rmat_folded(nv, args.edges, seed=args.seed,
+                                 device=device)
+    part = GraphPart(full, dx.world_size(), dx.rank())
+    return CCBench(part), part
+
+
+def main(argv=None):
+    a = parse_input_args(sys.argv[1:] if argv is None else argv)
+    dx.init_process_group("cuda")
+    import torch
+    local = dx.env_local_rank()
+    torch.cuda.set_device(local)
+    device = f"cuda:{local}"
+    full = load_device_graph(a, device)
+    if dx.rank() == 0:
+        print_memory_estimate(full.nv, full.ne, dx.world_size())
+    part = GraphPart(full, dx.world_size(), dx.rank())
+    eng = PushEngine(part, PushEngine.MODE_MAX)
+    with ElapsedTimer():
+        iters = eng.run()
+    if dx.rank() == 0:
+        print(f"[lux] converged in {iters} iterations")
+    if a.check:
+        mistakes = eng.check()
+        tag = "PASS" if mistakes == 0 else "FAIL"
+        if dx.rank() == 0:
+            print(f"[{tag}] {mistakes} mistakes")
+    return eng
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,45 @@
+"""Collaborative filtering app driver (reference parity:
+col_filter/colfilter.cc; SGD matrix-factorization sweeps, rank -k)."""
+import sys
+
+from .. import dist as dx
+from ..cf_engine import CFEngine
+from ..engine import GraphPart
+from .common import (ElapsedTimer, load_device_graph, parse_input_args,
+                     print_memory_estimate)
+
+# NetFlix-prize shape (reference README.md:86: 497,959 V / 200,961,014 E)
+NETFLIX_USERS = 480189
+NETFLIX_ITEMS = 17770
+
+
+def build_cf_bench(args, device):
+    from ..engine import DeviceCSC
+    ne = args.edges if args.edges != (1 << 31) else 200961014
+    full = DeviceCSC.bipartite(NETFLIX_USERS, NETFLIX_ITEMS, ne,
+                               seed=args.seed, device=device)
+    part = GraphPart(full, dx.world_size(), dx.rank())
+    return CFEngine(part, K=64), part
+
+
+def main(argv=None):
+    a = parse_input_args(sys.argv[1:] if argv is None else argv)
+    dx.init_process_group("cuda")
+    import torch
+    local = dx.env_local_rank()
+    torch.cuda.set_device(local)
+    device = f"cuda:{local}"
+    full = load_device_graph(a, device, weighted=True)
+    if dx.rank() == 0:
+        print_memory_estimate(full.nv, full.ne, dx.world_size(),
+                              weighted=True, k=a.k)
+    part = GraphPart(full, dx.world_size(), dx.rank())
+    eng = CFEngine(part, K=a.k)
+    with ElapsedTimer():
+        for _ in range(a.num_iter):
+            eng.step()
+    return eng
+
+
+if __name__ == "__main__":
+    main()

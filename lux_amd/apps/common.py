@@ -1,0 +1,103 @@
+"""Lux-compatible CLI parsing + shared driver plumbing.
+
+Flag surface parity with the reference drivers (sssp/sssp.cc:148-180,
+pagerank/pagerank.cc:43-58, README.md:42-54): -ng/-ll:gpu N, -ni N,
+-file PATH, -start V, -verbose/-v, -check/-c; Legion's -ll:fsize/-ll:zsize
+(MB pool sizes) are accepted and ignored (HIP/torch pool sizing is
+automatic), other -ll:* flags are accepted for drop-in compatibility.
+"""
+import sys
+import time
+
+
+class AppArgs:
+    def __init__(self):
+        self.num_gpu = 1
+        self.num_iter = 10
+        self.file = None
+        self.start = 0
+        self.verbose = False
+        self.check = False
+        self.k = 64
+        self.synthetic = None  # e.g. "rmat:20:1000000"
+
+
+def parse_input_args(argv):
+    a = AppArgs()
+    i = 0
+    while i < len(argv):
+        f = argv[i]
+        if f in ("-ng", "-ll:gpu"):
+            a.num_gpu = int(argv[i + 1]); i += 2
+        elif f == "-ni":
+            a.num_iter = int(argv[i + 1]); i += 2
+        elif f == "-file":
+            a.file = argv[i + 1]; i += 2
+        elif f == "-start":
+            a.start = int(argv[i + 1]); i += 2
+        elif f in ("-verbose", "-v"):
+            a.verbose = True; i += 1
+        elif f in ("-check", "-c"):
+            a.check = True; i += 1
+        elif f == "-k":
+            a.k = int(argv[i + 1]); i += 2
+        elif f == "-synthetic":
+            a.synthetic = argv[i + 1]; i += 2
+        elif f.startswith("-ll:") or f.startswith("-lg:"):
+            # Legion runtime flags: accepted for CLI compatibility
+            i += 2 if i + 1 < len(argv) and not argv[i + 1].startswith("-") \
+                else 1
+        else:
+            print(f"warning: ignoring unknown flag {f}", file=sys.stderr)
+            i += 1
+    return a
+
+
+def load_device_graph(a, device, weighted=False):
+    """Load -file .lux (or -synthetic spec) onto the device."""
+    from ..engine import DeviceCSC
+    from ..graph import Graph
+    if a.file:
+        g = Graph.load(a.file, want_weights=weighted)
+        return DeviceCSC.from_host(g, device)
+    if a.synthetic:
+        parts = a.synthetic.split(":")
+        kind = parts[0]
+        if kind == "rmat":
+            return DeviceCSC.rmat(int(parts[1]), int(parts[2]), device=device)
+        if kind == "rmat_folded":
+            return DeviceCSC.rmat_folded(int(parts[1]), int(parts[2]),
+                                         device=device)
+        if kind == "bipartite":
+            return DeviceCSC.bipartite(int(parts[1]), int(parts[2]),
+                                       int(parts[3]), device=device)
+    raise SystemExit("need -file graph.lux or -synthetic kind:args")
+
+
+def print_memory_estimate(nv, ne, nparts, weighted=False, k=1):
+    """Startup FB-requirement printout, the analog of pagerank.cc:60-85 /
+    sssp.cc:59-90 against 288 GB HBM3E per MI355X."""
+    per_edge = 4 + (4 if weighted else 0)
+    fb = ne * per_edge // max(nparts, 1) + 8 * (nv // max(nparts, 1) + 1) \
+        + 2 * 4 * k * nv
+    print(f"[lux] estimated FB usage per GPU: {fb / (1 << 20):.0f} MB "
+          f"(of 294912 MB HBM3E)")
+
+
+class ElapsedTimer:
+    """Reference-format timing: wall clock around the iteration loop only,
+    printed as 'ELAPSED TIME = ... s' (pagerank.cc:118, sssp.cc:137)."""
+
+    def __enter__(self):
+        import torch
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+        self.t0 = time.perf_counter()
+        return self
+
+    def __exit__(self, *exc):
+        import torch
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+        self.seconds = time.perf_counter() - self.t0
+        print("ELAPSED TIME = %7.7f s" % self.seconds)

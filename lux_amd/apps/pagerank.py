@@ -1,0 +1,40 @@
+"""PageRank app driver (reference parity: pagerank/pagerank.cc)."""
+import sys
+
+from .. import dist as dx
+from ..engine import GraphPart, PagerankEngine
+from .common import (ElapsedTimer, load_device_graph, parse_input_args,
+                     print_memory_estimate)
+
+
+def build_pagerank_bench(args, device):
+    from ..engine import DeviceCSC
+    full = DeviceCSC.rmat(args.scale, args.edges, seed=args.seed,
+                          device=device)
+    part = GraphPart(full, dx.world_size(), dx.rank())
+    return PagerankEngine(part), part
+
+
+def main(argv=None):
+    a = parse_input_args(sys.argv[1:] if argv is None else argv)
+    dx.init_process_group("cuda")
+    import torch
+    local = dx.env_local_rank()
+    torch.cuda.set_device(local)
+    device = f"cuda:{local}"
+    full = load_device_graph(a, device)
+    if dx.rank() == 0:
+        print_memory_estimate(full.nv, full.ne, dx.world_size())
+    part = GraphPart(full, dx.world_size(), dx.rank())
+    eng = PagerankEngine(part)
+    with ElapsedTimer():
+        for _ in range(a.num_iter):
+            eng.step()
+    if a.verbose and dx.rank() == 0:
+        r = eng.ranks()[:5].cpu().tolist()
+        print("[lux] first ranks (pr/out_degree):", r)
+    return eng
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,57 @@
+"""SSSP app driver (reference parity: sssp/sssp.cc; unweighted hop
+distances, push model with adaptive frontiers)."""
+import sys
+
+from .. import dist as dx
+from ..engine import GraphPart
+from ..push_engine import PushEngine
+from .common import (ElapsedTimer, load_device_graph, parse_input_args,
+                     print_memory_estimate)
+
+
+class SSSPBench:
+    """One bench step = one full traversal to convergence from -start
+    (reset + run; GTEPS counts ne per traversal, Graph500-style)."""
+
+    def __init__(self, part, source):
+        self.eng = PushEngine(part, PushEngine.MODE_MIN, source=source)
+
+    def step(self):
+        self.eng.reset()
+        self.eng.run()
+
+
+def build_sssp_bench(args, device):
+    from ..engine import DeviceCSC
+    full = DeviceCSC.rmat(args.scale, args.edges, seed=args.seed,
+                          device=device)
+    part = GraphPart(full, dx.world_size(), dx.rank())
+    return SSSPBench(part, 0), part
+
+
+def main(argv=None):
+    a = parse_input_args(sys.argv[1:] if argv is None else argv)
+    dx.init_process_group("cuda")
+    import torch
+    local = dx.env_local_rank()
+    torch.cuda.set_device(local)
+    device = f"cuda:{local}"
+    full = load_device_graph(a, device)
+    if dx.rank() == 0:
+        print_memory_estimate(full.nv, full.ne, dx.world_size())
+    part = GraphPart(full, dx.world_size(), dx.rank())
+    eng = PushEngine(part, PushEngine.MODE_MIN, source=a.start)
+    with ElapsedTimer():
+        iters = eng.run()
+    if dx.rank() == 0:
+        print(f"[lux] converged in {iters} iterations")
+    if a.check:
+        mistakes = eng.check()
+        tag = "PASS" if mistakes == 0 else "FAIL"
+        if dx.rank() == 0:
+            print(f"[{tag}] {mistakes} mistakes")
+    return eng
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,44 @@
+"""Distributed collaborative-filtering engine (pull model, K-dim latent
+vectors). Exchange pattern identical to PageRank (all-gather(v) of vector
+slices); compute per sweep is src-vector gather bound (256 B per edge at
+K=64)."""
+import math
+
+import torch
+
+from . import _native_gpu as ng
+from . import dist as dx
+from .engine import GraphPart
+
+F32 = torch.float32
+
+
+def _stream():
+    return torch.cuda.current_stream().cuda_stream
+
+
+class CFEngine:
+    def __init__(self, part: GraphPart, K=64):
+        assert part.weight is not None, "CF needs a weighted graph"
+        assert K <= 256
+        self.part = part
+        self.K = K
+        part.build_bins()
+        device = part.device
+        # reference init: every component sqrt(1/K) (colfilter_gpu.cu:260-264)
+        v0 = math.sqrt(1.0 / K)
+        self.old = torch.full((part.nv * K,), v0, dtype=F32, device=device)
+        self.new_part = torch.empty(part.vp * K, dtype=F32, device=device)
+        self.verts_elems = [v * K for v in part.verts_all]
+        self.left_elems = [l * K for l in part.row_left_all]
+
+    def step(self):
+        p = self.part
+        ng.cf_iter(_stream(), p.n0, p.bin0, p.n1, p.bin1, p.n2, p.bin2,
+                   p.nbig, p.bin2v, p.row_ptr, p.col, p.weight, self.old,
+                   self.new_part, p.row_left, self.K)
+        dx.all_gather_slices(self.old, self.new_part, self.verts_elems,
+                             self.left_elems)
+
+    def vectors(self):
+        return self.old.view(self.part.nv, self.K)

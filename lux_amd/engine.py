@@ -69,6 +69,17 @@ class DeviceCSC:
         return g
 
     @classmethod
+    def rmat_folded(cls, nv, ne, seed=1, device="cuda"):
+        scale = 0
+        while (1 << scale) < nv:
+            scale += 1
+        s = _stream()
+        esrc = torch.empty(ne, dtype=U32, device=device)
+        edst = torch.empty(ne, dtype=U32, device=device)
+        ng.rmat_edges_folded(s, seed, scale, nv, ne, esrc, edst)
+        return cls._from_device_edges(nv, ne, esrc, edst, None, device)
+
+    @classmethod
     def bipartite(cls, n_users, n_items, ne, seed=1, device="cuda"):
         nv = n_users + n_items
         s = _stream()

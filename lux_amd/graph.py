@@ -53,6 +53,16 @@ class Graph:
         return cls(1 << scale, ne, col_end, csrc)
 
     @classmethod
+    def rmat_folded(cls, nv, ne, seed=1):
+        """RMAT skew with a non-power-of-two nv (Twitter-shaped synthetics)."""
+        scale = 0
+        while (1 << scale) < nv:
+            scale += 1
+        src, dst = nat.rmat_edges_folded(seed, scale, nv, ne)
+        col_end, csrc, _ = nat.edges_to_csc(nv, src, dst)
+        return cls(nv, ne, col_end, csrc)
+
+    @classmethod
     def bipartite(cls, n_users, n_items, ne, seed=1):
         src, dst, w = nat.bipartite_edges(seed, n_users, n_items, ne)
         col_end, csrc, cw = nat.edges_to_csc(n_users + n_items, src, dst, w)

@@ -1,0 +1,237 @@
+"""Distributed push-model engine (SSSP, Connected Components).
+
+The MI355X re-design of the reference's push pipeline
+(push_app_task_impl, sssp_gpu.cu:335-522):
+  - per-rank push CSR over ALL nv sources holding only edges into my
+    partition (core/push_model.inl:321-324), built with device-wide scans;
+  - adaptive per-partition frontier segments (dense bitmap / sparse queue,
+    FrontierHeader-compatible) with majority-vote output format
+    (sssp_gpu.cu:395-408), dense->sparse conversion and sparse-overflow
+    fallback (sssp_gpu.cu:462-491);
+  - pull fallback when the global frontier exceeds nv/16 (sssp_gpu.cu:414),
+    served by the degree-binned pull kernels;
+  - exchange: RCCL all-gather(v) of label slices + frontier segments over
+    xGMI replaces the reference's zero-copy host staging; termination is
+    read off the exchanged headers (no extra collective, replacing the
+    FutureMap vote of sssp.cc:116-124).
+"""
+import numpy as np
+import torch
+
+from . import _native_gpu as ng
+from . import dist as dx
+from .engine import GraphPart
+from .types import (DENSE_BITMAP, SPARSE_QUEUE, frontier_bytes,
+                    frontier_capacity)
+
+U8, U32, U64 = torch.uint8, torch.int32, torch.int64
+
+
+def _stream():
+    return torch.cuda.current_stream().cuda_stream
+
+
+def _align16(x):
+    return (x + 15) & ~15
+
+
+class PushEngine:
+    MODE_MIN = 1  # SSSP
+    MODE_MAX = 2  # CC
+
+    def __init__(self, part: GraphPart, mode, source=0):
+        self.part = part
+        self.mode = mode
+        self.is_min = mode == self.MODE_MIN
+        device = part.device
+        self.device = device
+        s = _stream()
+        p = part
+
+        # pull bins for the dense fallback
+        part.build_bins()
+
+        # ---- push CSR: all nv sources -> my-partition dsts ----
+        deg_src = torch.zeros(p.nv, dtype=U32, device=device)
+        ng.hist_u32(s, p.ep, p.col, deg_src)
+        ends = torch.empty(p.nv, dtype=U64, device=device)
+        partials = torch.empty(ng.scan_partials_size(p.nv), dtype=U64,
+                               device=device)
+        ng.scan_end_offsets(s, p.nv, deg_src, ends, partials)
+        self.push_row_ptr = torch.empty(p.nv + 1, dtype=U64, device=device)
+        ng.local_row_ptr(s, p.nv, 0, ends, self.push_row_ptr)
+        cursor = self.push_row_ptr[:p.nv].clone()
+        self.push_col = torch.empty(max(p.ep, 1), dtype=U32, device=device)
+        ng.csr_scatter(s, p.ep, p.col, p.row_ptr, p.vp, p.row_left, cursor,
+                       self.push_col)
+        torch.cuda.synchronize()
+        del deg_src, ends, partials, cursor
+
+        # ---- frontier buffers ----
+        self.seg_bytes = [_align16(frontier_bytes(v)) for v in p.verts_all]
+        self.seg_off = np.concatenate([[0], np.cumsum(self.seg_bytes)])
+        self.capacity = frontier_capacity(p.vp)
+        self.fq_all = torch.zeros(int(self.seg_off[-1]), dtype=U8,
+                                  device=device)
+        self.new_seg = torch.zeros(self.seg_bytes[p.p], dtype=U8,
+                                   device=device)
+        self.tmp_seg = torch.zeros_like(self.new_seg)
+        # header byte index for the post-exchange D2H gather
+        idx = []
+        for q in range(p.nparts):
+            idx.extend(range(int(self.seg_off[q]), int(self.seg_off[q]) + 8))
+        self.hdr_idx = torch.tensor(idx, dtype=torch.long, device=device)
+
+        # ---- labels + frontier state ----
+        self.labels = torch.empty(p.nv, dtype=U32, device=device)
+        self.labels_part = torch.empty(p.vp, dtype=U32, device=device)
+        self.snapshot = torch.empty_like(self.labels_part)
+        self._fq_init = None
+        self.reset(source)
+
+    def reset(self, source=None):
+        """(Re)initialise labels and the seed frontier
+        (sssp_gpu.cu:733-744 / components_gpu.cu:733-740). Structures
+        (CSRs, bins, buffers) are kept."""
+        p = self.part
+        if source is not None:
+            self.source = source
+        source = self.source
+        if self.is_min:
+            self.labels.fill_(-1)
+            self.labels[source] = 0
+        else:
+            torch.arange(p.nv, dtype=U32, device=self.device,
+                         out=self.labels)
+        self.labels_part.copy_(self.labels.narrow(0, p.row_left, p.vp))
+
+        if self._fq_init is None:
+            fq_host = np.zeros(int(self.seg_off[-1]), np.uint8)
+            headers = []
+            for q in range(p.nparts):
+                off = int(self.seg_off[q])
+                hv = fq_host[off:off + 8].view(np.uint32)
+                if self.is_min:
+                    hv[0] = SPARSE_QUEUE
+                    owner = (p.row_left_all[q] <= source
+                             <= p.row_right_all[q]) if p.verts_all[q] \
+                        else False
+                    if owner:
+                        hv[1] = 1
+                        fq_host[off + 8:off + 12].view(np.uint32)[0] = source
+                    headers.append((SPARSE_QUEUE, int(hv[1])))
+                else:
+                    hv[0] = DENSE_BITMAP
+                    hv[1] = p.verts_all[q]
+                    nbytes = (p.verts_all[q] + 7) // 8
+                    fq_host[off + 8:off + 8 + nbytes] = 0xFF
+                    headers.append((DENSE_BITMAP, p.verts_all[q]))
+            self._fq_init = (torch.from_numpy(fq_host).to(self.device),
+                             headers)
+        self.fq_all.copy_(self._fq_init[0])
+        self.headers = list(self._fq_init[1])
+        self.iterations = 0
+        self.stats = []
+
+    # -------- helpers --------
+    def _my_seg_i32(self):
+        return self.new_seg.view(U32)
+
+    def _read_my_count(self):
+        return int(self.new_seg[4:8].view(U32).cpu().item())
+
+    def step(self):
+        """One push iteration. Returns my partition's new-frontier count."""
+        p = self.part
+        s = _stream()
+        nparts = p.nparts
+        old_fq_size = sum(h[1] for h in self.headers)
+        dense_votes = sum(1 for h in self.headers if h[0] == DENSE_BITMAP)
+        new_dense = dense_votes >= nparts - dense_votes
+        self.snapshot.copy_(self.labels_part)
+        # zero my new header (type patched at the end)
+        self._my_seg_i32()[0] = 0
+        self._my_seg_i32()[1] = 0
+
+        pull_fallback = old_fq_size > p.nv // 16
+        if pull_fallback:
+            new_dense = True
+            mode = ng.PULL_MIN if self.is_min else ng.PULL_MAX
+            ng.pull_iter(s, mode, p.n0, p.bin0, p.n1, p.bin1, p.n2, p.bin2,
+                         p.nbig, p.bin2v, p.row_ptr, p.col, self.labels,
+                         self.labels_part, None, p.row_left, 0.0)
+        else:
+            for q in range(nparts):
+                typ, num = self.headers[q]
+                if p.verts_all[q] == 0:
+                    continue
+                seg = self.fq_all.narrow(0, int(self.seg_off[q]),
+                                         self.seg_bytes[q])
+                if typ == DENSE_BITMAP:
+                    ng.push_scatter(s, int(self.is_min), 1, int(new_dense),
+                                    p.row_left_all[q], p.verts_all[q], seg,
+                                    self.push_row_ptr, self.push_col,
+                                    self.labels, self.snapshot,
+                                    self.labels_part, p.row_left,
+                                    self.new_seg, self.capacity)
+                else:
+                    if num == 0:
+                        continue
+                    ng.push_scatter(s, int(self.is_min), 0, int(new_dense),
+                                    0, num, seg, self.push_row_ptr,
+                                    self.push_col, self.labels,
+                                    self.snapshot, self.labels_part,
+                                    p.row_left, self.new_seg, self.capacity)
+
+        # ---- frontier format fix-ups (sssp_gpu.cu:462-491) ----
+        if new_dense:
+            ng.build_bitmap(s, p.vp, self.snapshot, self.labels_part,
+                            self.new_seg)
+            my_count = self._read_my_count()
+            if my_count < self.capacity:
+                # dense result fits the sparse capacity: convert
+                self.tmp_seg.copy_(self.new_seg)
+                self._my_seg_i32()[1] = 0
+                ng.d2s(s, p.vp, p.row_left, self.tmp_seg, self.new_seg)
+                new_dense = False
+        else:
+            my_count = self._read_my_count()
+            if my_count >= self.capacity:
+                # sparse overflow: rebuild as bitmap
+                new_dense = True
+                self._my_seg_i32()[1] = 0
+                ng.build_bitmap(s, p.vp, self.snapshot, self.labels_part,
+                                self.new_seg)
+                my_count = self._read_my_count()
+        self._my_seg_i32()[0] = DENSE_BITMAP if new_dense else SPARSE_QUEUE
+
+        # ---- exchange: labels + frontier segments ----
+        dx.all_gather_slices(self.labels, self.labels_part, p.verts_all,
+                             p.row_left_all)
+        dx.all_gather_slices(self.fq_all, self.new_seg, self.seg_bytes,
+                             [int(o) for o in self.seg_off[:-1]])
+        hdr = self.fq_all[self.hdr_idx].cpu().numpy().view(np.uint32)
+        self.headers = [(int(hdr[2 * q]), int(hdr[2 * q + 1]))
+                        for q in range(nparts)]
+        self.iterations += 1
+        return sum(h[1] for h in self.headers)
+
+    def run(self, max_iters=None):
+        """Iterate to convergence (every partition reports an empty new
+        frontier)."""
+        while True:
+            total = self.step()
+            if total == 0:
+                break
+            if max_iters and self.iterations >= max_iters:
+                break
+        return self.iterations
+
+    def check(self):
+        """Device check oracle; returns global violation count."""
+        p = self.part
+        mistakes = torch.zeros(1, dtype=U64, device=self.device)
+        ng.check(_stream(), int(self.is_min), p.vp, p.row_left, p.row_ptr,
+                 p.col, self.labels, mistakes)
+        dx.all_reduce_sum_(mistakes)
+        return int(mistakes.cpu().item())

@@ -55,6 +55,11 @@ void lux_rmat_edges(uint64_t seed, int scale, uint64_t ne, uint32_t* src,
                     uint32_t* dst) {
   for (uint64_t e = 0; e < ne; e++) rmat_edge(seed, e, scale, &src[e], &dst[e]);
 }
+void lux_rmat_edges_folded(uint64_t seed, int scale, uint32_t nv,
+                           uint64_t ne, uint32_t* src, uint32_t* dst) {
+  for (uint64_t e = 0; e < ne; e++)
+    rmat_edge_folded(seed, e, scale, nv, &src[e], &dst[e]);
+}
 void lux_bipartite_edges(uint64_t seed, uint32_t n_users, uint32_t n_items,
                          uint64_t ne, uint32_t* src, uint32_t* dst,
                          int32_t* w) {

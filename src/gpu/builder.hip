@@ -22,6 +22,15 @@ __global__ void rmat_edges_kernel(uint64_t seed, int scale, uint64_t ne,
   }
 }
 
+__global__ void rmat_edges_folded_kernel(uint64_t seed, int scale, V_ID nv,
+                                         uint64_t ne, V_ID* src, V_ID* dst) {
+  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
+  for (uint64_t e = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; e < ne;
+       e += stride) {
+    rmat_edge_folded(seed, e, scale, nv, &src[e], &dst[e]);
+  }
+}
+
 __global__ void bipartite_edges_kernel(uint64_t seed, V_ID n_users,
                                        V_ID n_items, int item_scale,
                                        uint64_t ne, V_ID* src, V_ID* dst,
@@ -161,6 +170,13 @@ void lux_gpu_rmat_edges(uint64_t stream, uint64_t seed, int scale,
   hipStream_t s = (hipStream_t)stream;
   hipLaunchKernelGGL(rmat_edges_kernel, dim3(grid_for(ne)), dim3(BLOCK), 0, s,
                      seed, scale, ne, src, dst);
+}
+
+void lux_gpu_rmat_edges_folded(uint64_t stream, uint64_t seed, int scale,
+                               V_ID nv, uint64_t ne, V_ID* src, V_ID* dst) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(rmat_edges_folded_kernel, dim3(grid_for(ne)),
+                     dim3(BLOCK), 0, s, seed, scale, nv, ne, src, dst);
 }
 
 void lux_gpu_bipartite_edges(uint64_t stream, uint64_t seed, V_ID n_users,

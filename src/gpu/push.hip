@@ -1,0 +1,331 @@
+// Push-model engine: frontier-driven scatter along out-edges, with the
+// reference's adaptive dense-bitmap / sparse-queue frontier representation
+// (FrontierHeader, core/graph.h:100-106) and first-improvement enqueue
+// (process_edge_sparse, sssp_gpu.cu:63-82) re-validated on CDNA4 atomics.
+//
+// Structures per rank: a "push CSR" mapping EVERY source vertex to its
+// out-edges that land in THIS rank's partition (the reference's
+// push_row_ptr/push_col_idx of size nv per GPU, core/push_model.inl:321-324),
+// built with a device-wide scan instead of the reference's serial
+// single-thread prefix sum (sssp_gpu.cu:550-565).
+//
+// Per iteration the Python engine: snapshots labels, reads all P segment
+// headers, picks push vs pull fallback (oldFqSize > nv/16, sssp_gpu.cu:414),
+// launches one scatter per source segment, then bitmap/d2s fix-ups — all on
+// one HIP stream; exchange is RCCL all-gather(v) (lux_amd/push_engine.py).
+#include "gpu_common.h"
+
+namespace lux {
+
+// min (SSSP hop) / max (CC label) relaxation semantics.
+template <bool IS_MIN> struct LabOp;
+template <> struct LabOp<true> {
+  static __device__ __forceinline__ uint32_t map(uint32_t src_label) {
+    return src_label + 1;  // hop relaxation; sources in a frontier are finite
+  }
+  static __device__ __forceinline__ bool better(uint32_t a, uint32_t b) {
+    return a < b;
+  }
+  static __device__ __forceinline__ uint32_t atom(uint32_t* p, uint32_t v) {
+    return atomicMin(p, v);
+  }
+};
+template <> struct LabOp<false> {
+  static __device__ __forceinline__ uint32_t map(uint32_t src_label) {
+    return src_label;  // CC propagates the label itself
+  }
+  static __device__ __forceinline__ bool better(uint32_t a, uint32_t b) {
+    return a > b;
+  }
+  static __device__ __forceinline__ uint32_t atom(uint32_t* p, uint32_t v) {
+    return atomicMax(p, v);
+  }
+};
+
+// ---------------- push CSR build ----------------
+
+__global__ void csr_scatter_kernel(uint64_t ep, const V_ID* col,
+                                   const E_ID* row_ptr_loc, V_ID vp,
+                                   V_ID row_left,
+                                   unsigned long long* cursor,
+                                   V_ID* push_col) {
+  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
+  for (uint64_t j = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; j < ep;
+       j += stride) {
+    // binary search: dst row v with row_ptr[v] <= j < row_ptr[v+1]
+    V_ID lo = 0, hi = vp - 1;
+    while (lo < hi) {
+      V_ID mid = (lo + hi + 1) >> 1;
+      if (row_ptr_loc[mid] <= j) lo = mid;
+      else hi = mid - 1;
+    }
+    V_ID src = col[j];
+    unsigned long long pos = atomicAdd(&cursor[src], 1ull);
+    push_col[pos] = lo + row_left;  // global dst id
+  }
+}
+
+// ---------------- frontier scatter (the hot push kernel) ----------------
+
+// One launch per (my partition x source segment). Block = 256 threads
+// (4 waves) cooperatively drains the block's 256 frontier vertices' edges:
+// per-thread out-degree -> block exclusive scan (LDS, wave64 ladder) ->
+// block-stride edge loop with a forward srcIdx walk (amortized O(1) per
+// edge, matching sssp_gpu.cu:198-244).
+template <bool IS_MIN, bool OLD_DENSE, bool NEW_DENSE>
+__global__ void push_scatter_kernel(
+    V_ID in_row_left, V_ID in_count,        // dense: segment vert range base/count; sparse: queue count
+    const uint8_t* old_seg,                 // segment bytes incl header
+    const E_ID* push_row_ptr,               // u64[nv+1]
+    const V_ID* push_col,
+    const uint32_t* old_labels,             // u32[nv] replicated
+    const uint32_t* snapshot,               // u32[vp] labels at iter start
+    uint32_t* new_labels,                   // u32[vp]
+    V_ID my_row_left,
+    uint8_t* new_seg,                       // my segment (sparse out) or null
+    V_ID capacity) {
+  using OP = LabOp<IS_MIN>;
+  __shared__ uint32_t src_label[BLOCK];
+  __shared__ unsigned long long ex_off[BLOCK / WAVE + 1];
+  __shared__ unsigned long long edge_base[BLOCK];
+  __shared__ unsigned long long prefix[BLOCK];
+  __shared__ uint32_t lds_scan[BLOCK / WAVE + 1];
+  __shared__ uint32_t queue_base;
+
+  const uint8_t* old_bitmap = old_seg + sizeof(FrontierHeader);
+  const V_ID* old_queue = (const V_ID*)(old_seg + sizeof(FrontierHeader));
+  V_ID* new_queue = nullptr;
+  uint32_t* num_nodes = nullptr;
+  if (!NEW_DENSE) {
+    num_nodes = &((FrontierHeader*)new_seg)->numNodes;
+    new_queue = (V_ID*)(new_seg + sizeof(FrontierHeader));
+  }
+
+  for (V_ID blk_start = blockIdx.x * blockDim.x; blk_start < in_count;
+       blk_start += blockDim.x * gridDim.x) {
+    V_ID idx = blk_start + threadIdx.x;
+    unsigned long long my_edges = 0, my_base = 0;
+    uint32_t lab = 0;
+    if (idx < in_count) {
+      V_ID u;
+      bool active;
+      if (OLD_DENSE) {
+        u = in_row_left + idx;
+        active = (old_bitmap[idx >> 3] >> (idx & 7)) & 1;
+      } else {
+        u = old_queue[idx];
+        active = true;
+      }
+      if (active) {
+        my_base = push_row_ptr[u];
+        my_edges = push_row_ptr[u + 1] - my_base;
+        lab = old_labels[u];
+      }
+    }
+    src_label[threadIdx.x] = lab;
+    __syncthreads();
+    unsigned long long total;
+    unsigned long long ex =
+        block_exscan<unsigned long long, BLOCK>(my_edges, ex_off, &total);
+    prefix[threadIdx.x] = ex;
+    edge_base[threadIdx.x] = my_base;
+    __syncthreads();
+
+    unsigned long long done = 0;
+    int si = 0;
+    while (done < total) {
+      unsigned long long k = done + threadIdx.x;
+      uint32_t flag = 0;
+      V_ID dstv = 0;
+      if (k < total) {
+        while (si + 1 < BLOCK && k >= prefix[si + 1]) si++;
+        E_ID e = edge_base[si] + (k - prefix[si]);
+        V_ID v = push_col[e];
+        uint32_t new_lab = OP::map(src_label[si]);
+        uint32_t* slot = &new_labels[v - my_row_left];
+        uint32_t cur = __hip_atomic_load(slot, __ATOMIC_RELAXED,
+                                         __HIP_MEMORY_SCOPE_AGENT);
+        if (OP::better(new_lab, cur)) {
+          if (NEW_DENSE) {
+            OP::atom(slot, new_lab);
+          } else {
+            // first-improvement enqueue (sssp_gpu.cu:63-82): the thread
+            // whose atomic moved the label off its iteration-start value
+            // owns the enqueue — exactly once per vertex per iteration.
+            uint32_t last = snapshot[v - my_row_left];
+            uint32_t act = OP::atom(slot, new_lab);
+            if (act == last) {
+              flag = 1;
+              dstv = v;
+            }
+          }
+        }
+      }
+      if (!NEW_DENSE) {
+        __syncthreads();
+        uint32_t q_total;
+        uint32_t q_off = block_exscan<uint32_t, BLOCK>(flag, lds_scan,
+                                                       &q_total);
+        if (threadIdx.x == 0 && q_total)
+          queue_base = atomicAdd(num_nodes, q_total);
+        __syncthreads();
+        if (flag && q_total) {
+          uint32_t pos = queue_base + q_off;
+          if (pos < capacity) new_queue[pos] = dstv;
+        }
+        __syncthreads();
+      }
+      done += blockDim.x;
+    }
+    __syncthreads();
+  }
+}
+
+// ---------------- frontier representation fix-ups ----------------
+
+// Build my dense output bitmap + count: bit v <=> snapshot[v]!=new[v]
+// (bitmap_kernel, sssp_gpu.cu:248-281).
+__global__ void build_bitmap_kernel(V_ID vp, const uint32_t* snapshot,
+                                    const uint32_t* new_labels,
+                                    uint8_t* seg) {
+  __shared__ uint32_t lds[BLOCK / WAVE];
+  uint32_t* num_nodes = &((FrontierHeader*)seg)->numNodes;
+  uint8_t* bitmap = seg + sizeof(FrontierHeader);
+  V_ID nbytes = (vp + 7) / 8;
+  uint32_t cnt = 0;
+  for (V_ID i = blockIdx.x * blockDim.x + threadIdx.x; i < nbytes;
+       i += blockDim.x * gridDim.x) {
+    uint8_t byte = 0;
+    for (int b = 0; b < 8; b++) {
+      V_ID v = i * 8 + b;
+      if (v < vp && snapshot[v] != new_labels[v]) {
+        byte |= (1u << b);
+        cnt++;
+      }
+    }
+    bitmap[i] = byte;
+  }
+  cnt = block_reduce_sum(cnt, lds);
+  if (threadIdx.x == 0 && cnt) atomicAdd(num_nodes, cnt);
+}
+
+// Dense bitmap -> sparse queue (convert_d2s_kernel, sssp_gpu.cu:283-315).
+// Queue entries are GLOBAL vertex ids.
+__global__ void d2s_kernel(V_ID vp, V_ID row_left, const uint8_t* dense_seg,
+                           uint8_t* sparse_seg) {
+  __shared__ uint32_t lds[BLOCK / WAVE + 1];
+  __shared__ uint32_t qbase;
+  const uint8_t* bitmap = dense_seg + sizeof(FrontierHeader);
+  uint32_t* num_nodes = &((FrontierHeader*)sparse_seg)->numNodes;
+  V_ID* queue = (V_ID*)(sparse_seg + sizeof(FrontierHeader));
+  for (V_ID blk = blockIdx.x * blockDim.x; blk < vp;
+       blk += blockDim.x * gridDim.x) {
+    V_ID v = blk + threadIdx.x;
+    uint32_t flag = 0;
+    if (v < vp) flag = (bitmap[v >> 3] >> (v & 7)) & 1;
+    __syncthreads();
+    uint32_t total;
+    uint32_t off = block_exscan<uint32_t, BLOCK>(flag, lds, &total);
+    if (threadIdx.x == 0 && total) qbase = atomicAdd(num_nodes, total);
+    __syncthreads();
+    if (flag) queue[qbase + off] = v + row_left;
+    __syncthreads();
+  }
+}
+
+// Check oracles on device (check_kernel, sssp_gpu.cu:773-798 /
+// components_gpu.cu:767-791): count violations over my pull-CSC partition.
+template <bool IS_MIN>
+__global__ void check_kernel(V_ID vp, V_ID row_left, const E_ID* row_ptr_loc,
+                             const V_ID* col, const uint32_t* labels,
+                             unsigned long long* mistakes) {
+  __shared__ unsigned long long lds[BLOCK / WAVE];
+  unsigned long long cnt = 0;
+  for (V_ID v = blockIdx.x * blockDim.x + threadIdx.x; v < vp;
+       v += blockDim.x * gridDim.x) {
+    uint32_t lab = labels[row_left + v];
+    for (E_ID j = row_ptr_loc[v]; j < row_ptr_loc[v + 1]; j++) {
+      uint32_t sl = labels[col[j]];
+      if (IS_MIN) {
+        uint32_t cand = sl == INF_LABEL ? INF_LABEL : sl + 1;
+        if (lab > cand) cnt++;
+      } else {
+        if (lab < sl) cnt++;
+      }
+    }
+  }
+  cnt = block_reduce_sum(cnt, lds);
+  if (threadIdx.x == 0 && cnt) atomicAdd(mistakes, cnt);
+}
+
+}  // namespace lux
+
+// ---------------- C ABI ----------------
+
+using namespace lux;
+
+extern "C" {
+
+void lux_gpu_csr_scatter(uint64_t stream, uint64_t ep, const V_ID* col,
+                         const E_ID* row_ptr_loc, V_ID vp, V_ID row_left,
+                         unsigned long long* cursor, V_ID* push_col) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(csr_scatter_kernel, dim3(grid_for(ep)), dim3(BLOCK), 0,
+                     s, ep, col, row_ptr_loc, vp, row_left, cursor, push_col);
+}
+
+void lux_gpu_push_scatter(uint64_t stream, int is_min, int old_dense,
+                          int new_dense, V_ID in_row_left, V_ID in_count,
+                          const uint8_t* old_seg, const E_ID* push_row_ptr,
+                          const V_ID* push_col, const uint32_t* old_labels,
+                          const uint32_t* snapshot, uint32_t* new_labels,
+                          V_ID my_row_left, uint8_t* new_seg, V_ID capacity) {
+  hipStream_t s = (hipStream_t)stream;
+  if (in_count == 0) return;
+  int grid = grid_for(in_count);
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3(grid), dim3(BLOCK), 0, s, in_row_left,
+                       in_count, old_seg, push_row_ptr, push_col, old_labels,
+                       snapshot, new_labels, my_row_left, new_seg, capacity);
+  };
+  int key = (is_min ? 4 : 0) | (old_dense ? 2 : 0) | (new_dense ? 1 : 0);
+  switch (key) {
+    case 0: launch(push_scatter_kernel<false, false, false>); break;
+    case 1: launch(push_scatter_kernel<false, false, true>); break;
+    case 2: launch(push_scatter_kernel<false, true, false>); break;
+    case 3: launch(push_scatter_kernel<false, true, true>); break;
+    case 4: launch(push_scatter_kernel<true, false, false>); break;
+    case 5: launch(push_scatter_kernel<true, false, true>); break;
+    case 6: launch(push_scatter_kernel<true, true, false>); break;
+    case 7: launch(push_scatter_kernel<true, true, true>); break;
+  }
+}
+
+void lux_gpu_build_bitmap(uint64_t stream, V_ID vp, const uint32_t* snapshot,
+                          const uint32_t* new_labels, uint8_t* seg) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(build_bitmap_kernel, dim3(grid_for((vp + 7) / 8)),
+                     dim3(BLOCK), 0, s, vp, snapshot, new_labels, seg);
+}
+
+void lux_gpu_d2s(uint64_t stream, V_ID vp, V_ID row_left,
+                 const uint8_t* dense_seg, uint8_t* sparse_seg) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(d2s_kernel, dim3(grid_for(vp)), dim3(BLOCK), 0, s, vp,
+                     row_left, dense_seg, sparse_seg);
+}
+
+void lux_gpu_check(uint64_t stream, int is_min, V_ID vp, V_ID row_left,
+                   const E_ID* row_ptr_loc, const V_ID* col,
+                   const uint32_t* labels, unsigned long long* mistakes) {
+  hipStream_t s = (hipStream_t)stream;
+  if (is_min)
+    hipLaunchKernelGGL(check_kernel<true>, dim3(grid_for(vp)), dim3(BLOCK), 0,
+                       s, vp, row_left, row_ptr_loc, col, labels, mistakes);
+  else
+    hipLaunchKernelGGL(check_kernel<false>, dim3(grid_for(vp)), dim3(BLOCK),
+                       0, s, vp, row_left, row_ptr_loc, col, labels,
+                       mistakes);
+}
+
+}  // extern "C"

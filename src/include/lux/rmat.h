@@ -65,6 +65,16 @@ LUX_HD void rmat_edge(uint64_t seed, uint64_t e, int scale, V_ID* src,
 // up and ids folded by modulo — see uniform_fold below).
 LUX_HD V_ID fold_id(V_ID x, V_ID nv) { return (nv & (nv - 1)) == 0 ? (x & (nv - 1)) : (x % nv); }
 
+// RMAT with a non-power-of-two vertex count (Twitter-/NetFlix-shaped
+// synthetic graphs): generate at the ceiling scale, fold ids by modulo.
+// Skew is preserved (fold maps the hot low-id region onto itself).
+LUX_HD void rmat_edge_folded(uint64_t seed, uint64_t e, int scale, V_ID nv,
+                             V_ID* src, V_ID* dst) {
+  rmat_edge(seed, e, scale, src, dst);
+  *src = fold_id(*src, nv);
+  *dst = fold_id(*dst, nv);
+}
+
 // Deterministic edge weight for weighted synthetic graphs (CF): int in
 // [1, 5] like a ratings matrix.
 LUX_HD WeightType rmat_weight(uint64_t seed, uint64_t e) {
