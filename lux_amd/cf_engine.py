@@ -38,7 +38,7 @@ class CFEngine:
                    p.nbig, p.bin2v, p.row_ptr, p.col, p.weight, self.old,
                    self.new_part, p.row_left, self.K)
         dx.all_gather_slices(self.old, self.new_part, self.verts_elems,
-                             self.left_elems)
+                             self.left_elems, my_index=p.p)
 
     def vectors(self):
         return self.old.view(self.part.nv, self.K)

@@ -56,7 +56,7 @@ def barrier():
         dist.barrier()
 
 
-def all_gather_slices(full, my_slice, verts, row_left):
+def all_gather_slices(full, my_slice, verts, row_left, my_index=None):
     """All-gather(v) of per-rank slices into the replicated `full` tensor.
 
     full:     flat tensor covering all ranks' slices concatenated
@@ -64,10 +64,13 @@ def all_gather_slices(full, my_slice, verts, row_left):
     my_slice: this rank's contiguous slice (must equal
               full[row_left[r] : row_left[r]+verts[r]] layout-wise).
     verts:    per-rank element counts (list of ints, in elements).
+    my_index: which slice is mine (defaults to this process's rank; passed
+              explicitly by single-process multi-partition simulations).
     """
     ws = world_size()
     if ws == 1:
-        dst = full.narrow(0, row_left[0], verts[0])
+        i = 0 if my_index is None else my_index
+        dst = full.narrow(0, row_left[i], verts[i])
         if dst.data_ptr() != my_slice.data_ptr():
             dst.copy_(my_slice)
         return

@@ -23,6 +23,50 @@ def _stream():
     return torch.cuda.current_stream().cuda_stream
 
 
+LLC_BLOCK_SHIFT = 25  # 2^25 verts * 4 B = 128 MB gather window (LLC 256 MB)
+
+
+def _bins_for(row_ptr, vp, ep, device, compact=False):
+    """Run the bin-build kernel over one row_ptr; returns counts + lists."""
+    s = _stream()
+    bin0 = torch.empty(max(vp, 1), dtype=U32, device=device)
+    bin1 = torch.empty(max(vp, 1), dtype=U32, device=device)
+    nbig_max = max(min(vp, ep // 2048 + 1), 1)
+    n2_max = ep // 8192 + nbig_max + 1
+    bin2 = torch.empty(n2_max * 2, dtype=U32, device=device)
+    bin2v = torch.empty(nbig_max, dtype=U32, device=device)
+    counters = torch.zeros(4, dtype=U32, device=device)
+    ng.build_bins(s, vp, row_ptr, bin0, bin1, bin2, bin2v, counters)
+    c = counters.cpu()
+    n0, n1, n2, nbig = int(c[0]), int(c[1]), int(c[2]), int(c[3])
+    if compact:  # free slack (per-block lists would otherwise be SB*vp)
+        bin0 = bin0[:max(n0, 1)].clone()
+        bin1 = bin1[:max(n1, 1)].clone()
+        bin2 = bin2[:max(n2 * 2, 1)].clone()
+        bin2v = bin2v[:max(nbig, 1)].clone()
+    return n0, n1, n2, nbig, bin0, bin1, bin2, bin2v
+
+
+def run_pull(part, mode, oldv, newv, deg, init_rank):
+    """One pull sweep over this rank's partition — via the src-blocked CSC
+    (phased accumulation, LLC-resident gathers) when built, else the plain
+    single-sweep path."""
+    s = _stream()
+    blocks = getattr(part, "blocks", None)
+    if blocks:
+        nb = len(blocks)
+        for i, blk in enumerate(blocks):
+            phase = (1 if i == 0 else 0) | (2 if i == nb - 1 else 0)
+            ng.pull_iter(s, mode, blk["n0"], blk["bin0"], blk["n1"],
+                         blk["bin1"], blk["n2"], blk["bin2"], blk["nbig"],
+                         blk["bin2v"], blk["row_ptr"], blk["col"], oldv,
+                         newv, deg, part.row_left, init_rank, phase)
+    else:
+        ng.pull_iter(s, mode, part.n0, part.bin0, part.n1, part.bin1,
+                     part.n2, part.bin2, part.nbig, part.bin2v, part.row_ptr,
+                     part.col, oldv, newv, deg, part.row_left, init_rank, 3)
+
+
 def partition_bounds(col_end, ne, nparts):
     """Edge-balanced contiguous ranges from a device col_end tensor (u64 end
     offsets). Same greedy rule as the CPU partitioner
@@ -151,21 +195,62 @@ class GraphPart:
         self.device = device
 
     def build_bins(self):
+        if hasattr(self, "bin0"):
+            return
+        (self.n0, self.n1, self.n2, self.nbig, self.bin0, self.bin1,
+         self.bin2, self.bin2v) = _bins_for(self.row_ptr, self.vp, self.ep,
+                                            self.device)
+
+    def prepare_pull(self, force_shift=None):
+        """Build degree bins and, when the gather window (nv*4 B) exceeds
+        the 256 MiB Infinity Cache, the src-blocked CSC (edges regrouped by
+        src>>shift so each sweep's random gathers stay LLC-resident)."""
+        self.build_bins()
+        if getattr(self, "blocks", None) is not None:
+            return
+        shift = force_shift if force_shift is not None else LLC_BLOCK_SHIFT
+        if force_shift is None and (self.nv <= (1 << shift) or self.ep == 0):
+            self.blocks = None
+            return
+        self.build_blocked(shift)
+
+    def build_blocked(self, shift):
         device = self.device
         s = _stream()
-        bin0 = torch.empty(self.vp, dtype=U32, device=device)
-        bin1 = torch.empty(self.vp, dtype=U32, device=device)
-        nbig_max = max(min(self.vp, self.ep // 2048 + 1), 1)
-        n2_max = self.ep // 8192 + nbig_max + 1
-        bin2 = torch.empty(n2_max * 2, dtype=U32, device=device)
-        bin2v = torch.empty(nbig_max, dtype=U32, device=device)
-        counters = torch.zeros(4, dtype=U32, device=device)
-        ng.build_bins(s, self.vp, self.row_ptr, bin0, bin1, bin2, bin2v,
-                      counters)
-        c = counters.cpu()
-        self.n0, self.n1, self.n2, self.nbig = (int(c[0]), int(c[1]),
-                                                int(c[2]), int(c[3]))
-        self.bin0, self.bin1, self.bin2, self.bin2v = bin0, bin1, bin2, bin2v
+        vp, ep = self.vp, self.ep
+        sb = (self.nv + (1 << shift) - 1) >> shift
+        if sb <= 1:
+            self.blocks = None
+            return
+        n = sb * vp
+        counts = torch.zeros(n, dtype=U32, device=device)
+        ng.blocked_count(s, ep, self.col, self.row_ptr, vp, shift, counts)
+        ends = torch.empty(n, dtype=U64, device=device)
+        partials = torch.empty(ng.scan_partials_size(n), dtype=U64,
+                               device=device)
+        ng.scan_end_offsets(s, n, counts, ends, partials)
+        cursor = torch.empty(n + 1, dtype=U64, device=device)
+        ng.local_row_ptr(s, n, 0, ends, cursor)
+        blk_col = torch.empty(max(ep, 1), dtype=U32, device=device)
+        ng.blocked_scatter(s, ep, self.col, self.row_ptr, vp, shift, cursor,
+                           blk_col)
+        del counts, partials, cursor
+        self.blocks = []
+        begin = 0
+        for b in range(sb):
+            end = int(ends[(b + 1) * vp - 1].item())
+            row_ptr_b = torch.empty(vp + 1, dtype=U64, device=device)
+            ng.local_row_ptr(s, vp, begin, ends.narrow(0, b * vp, vp),
+                             row_ptr_b)
+            col_b = blk_col.narrow(0, begin, max(end - begin, 1))
+            n0, n1, n2, nbig, b0, b1, b2, b2v = _bins_for(
+                row_ptr_b, vp, end - begin, device, compact=True)
+            self.blocks.append(dict(row_ptr=row_ptr_b, col=col_b, n0=n0,
+                                    n1=n1, n2=n2, nbig=nbig, bin0=b0,
+                                    bin1=b1, bin2=b2, bin2v=b2v))
+            begin = end
+        self._blk_col = blk_col  # keep the narrow()s' base alive
+        del ends
 
 
 class PagerankEngine:
@@ -176,7 +261,7 @@ class PagerankEngine:
     def __init__(self, part: GraphPart):
         self.part = part
         device = part.device
-        part.build_bins()
+        part.prepare_pull()
         s = _stream()
         # global out-degrees: histogram over my edges, then sum-all-reduce
         deg = torch.zeros(part.nv, dtype=U32, device=device)
@@ -193,12 +278,10 @@ class PagerankEngine:
 
     def step(self):
         p = self.part
-        s = _stream()
-        ng.pull_iter(s, ng.PULL_PR, p.n0, p.bin0, p.n1, p.bin1, p.n2, p.bin2,
-                     p.nbig, p.bin2v, p.row_ptr, p.col, self.old,
-                     self.new_part, self.deg, p.row_left, self.init_rank)
+        run_pull(p, ng.PULL_PR, self.old, self.new_part, self.deg,
+                 self.init_rank)
         dx.all_gather_slices(self.old, self.new_part, p.verts_all,
-                             p.row_left_all)
+                             p.row_left_all, my_index=p.p)
 
     def ranks(self):
         """Replicated stored ranks (pr/out_degree) as a torch tensor."""
@@ -213,22 +296,18 @@ class LabelPullEngine:
         assert mode in (ng.PULL_MIN, ng.PULL_MAX)
         self.part = part
         self.mode = mode
-        if not hasattr(part, "bin0"):
-            part.build_bins()
+        part.prepare_pull()
         self.old = init_labels  # u32[nv] replicated (as int32 tensor)
         self.new_part = torch.empty(part.vp, dtype=U32, device=part.device)
 
     def step(self):
         p = self.part
-        s = _stream()
-        ng.pull_iter(s, self.mode, p.n0, p.bin0, p.n1, p.bin1, p.n2, p.bin2,
-                     p.nbig, p.bin2v, p.row_ptr, p.col, self.old,
-                     self.new_part, None, p.row_left, 0.0)
+        run_pull(p, self.mode, self.old, self.new_part, None, 0.0)
         # changed count before the gather overwrites old
         changed = (self.new_part
                    != self.old.narrow(0, p.row_left, p.vp)).sum()
         dx.all_gather_slices(self.old, self.new_part, p.verts_all,
-                             p.row_left_all)
+                             p.row_left_all, my_index=p.p)
         return changed
 
     def run_to_fixpoint(self, max_iters=None):

@@ -20,7 +20,7 @@ import torch
 
 from . import _native_gpu as ng
 from . import dist as dx
-from .engine import GraphPart
+from .engine import GraphPart, run_pull
 from .types import (DENSE_BITMAP, SPARSE_QUEUE, frontier_bytes,
                     frontier_capacity)
 
@@ -48,8 +48,9 @@ class PushEngine:
         s = _stream()
         p = part
 
-        # pull bins for the dense fallback
-        part.build_bins()
+        # pull bins (+ src-blocked CSC when nv is LLC-large) for the
+        # dense fallback
+        part.prepare_pull()
 
         # ---- push CSR: all nv sources -> my-partition dsts ----
         deg_src = torch.zeros(p.nv, dtype=U32, device=device)
@@ -157,9 +158,7 @@ class PushEngine:
         if pull_fallback:
             new_dense = True
             mode = ng.PULL_MIN if self.is_min else ng.PULL_MAX
-            ng.pull_iter(s, mode, p.n0, p.bin0, p.n1, p.bin1, p.n2, p.bin2,
-                         p.nbig, p.bin2v, p.row_ptr, p.col, self.labels,
-                         self.labels_part, None, p.row_left, 0.0)
+            run_pull(p, mode, self.labels, self.labels_part, None, 0.0)
         else:
             for q in range(nparts):
                 typ, num = self.headers[q]
@@ -207,9 +206,10 @@ class PushEngine:
 
         # ---- exchange: labels + frontier segments ----
         dx.all_gather_slices(self.labels, self.labels_part, p.verts_all,
-                             p.row_left_all)
+                             p.row_left_all, my_index=p.p)
         dx.all_gather_slices(self.fq_all, self.new_seg, self.seg_bytes,
-                             [int(o) for o in self.seg_off[:-1]])
+                             [int(o) for o in self.seg_off[:-1]],
+                             my_index=p.p)
         hdr = self.fq_all[self.hdr_idx].cpu().numpy().view(np.uint32)
         self.headers = [(int(hdr[2 * q]), int(hdr[2 * q + 1]))
                         for q in range(nparts)]

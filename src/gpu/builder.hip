@@ -145,6 +145,52 @@ __global__ void scatter_kernel(uint64_t ne, const V_ID* src, const V_ID* dst,
   }
 }
 
+// ---------------- src-blocked CSC build ----------------
+// Re-groups a partition's CSC edges by (src >> shift, dst): per iteration
+// the pull kernels then sweep one src block at a time, keeping the random
+// old-property gather window (2^shift * 4 B) resident in the 256 MiB
+// Infinity Cache. This is the key MI355X bandwidth lever for nv*4B > LLC
+// graphs (RMAT-27: 512 MB of ranks); the reference has no equivalent (its
+// gathers run cold over ZC/host memory).
+
+__device__ __forceinline__ V_ID row_of_edge(const E_ID* row_ptr_loc, V_ID vp,
+                                            uint64_t j) {
+  V_ID lo = 0, hi = vp - 1;
+  while (lo < hi) {
+    V_ID mid = (lo + hi + 1) >> 1;
+    if (row_ptr_loc[mid] <= j) lo = mid;
+    else hi = mid - 1;
+  }
+  return lo;
+}
+
+__global__ void blocked_count_kernel(uint64_t ep, const V_ID* col,
+                                     const E_ID* row_ptr_loc, V_ID vp,
+                                     int shift, uint32_t* counts) {
+  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
+  for (uint64_t j = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; j < ep;
+       j += stride) {
+    V_ID v = row_of_edge(row_ptr_loc, vp, j);
+    uint32_t b = col[j] >> shift;
+    atomicAdd(&counts[(uint64_t)b * vp + v], 1u);
+  }
+}
+
+__global__ void blocked_scatter_kernel(uint64_t ep, const V_ID* col,
+                                       const E_ID* row_ptr_loc, V_ID vp,
+                                       int shift,
+                                       unsigned long long* cursor,
+                                       V_ID* out_col) {
+  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
+  for (uint64_t j = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; j < ep;
+       j += stride) {
+    V_ID v = row_of_edge(row_ptr_loc, vp, j);
+    uint32_t b = col[j] >> shift;
+    unsigned long long pos = atomicAdd(&cursor[(uint64_t)b * vp + v], 1ull);
+    out_col[pos] = col[j];
+  }
+}
+
 // ---------------- local row_ptr from global col_end slice ----------------
 
 __global__ void local_row_ptr_kernel(uint32_t vp, E_ID col_left,
@@ -222,6 +268,22 @@ void lux_gpu_edges_to_csc(uint64_t stream, uint32_t nv, uint64_t ne,
                      s, nv, col_end, cursor);
   hipLaunchKernelGGL(scatter_kernel, dim3(grid_for(ne)), dim3(BLOCK), 0, s,
                      ne, src, dst, w, cursor, out_src, out_w);
+}
+
+void lux_gpu_blocked_count(uint64_t stream, uint64_t ep, const V_ID* col,
+                           const E_ID* row_ptr_loc, V_ID vp, int shift,
+                           uint32_t* counts /*pre-zeroed u32[SB*vp]*/) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(blocked_count_kernel, dim3(grid_for(ep)), dim3(BLOCK),
+                     0, s, ep, col, row_ptr_loc, vp, shift, counts);
+}
+
+void lux_gpu_blocked_scatter(uint64_t stream, uint64_t ep, const V_ID* col,
+                             const E_ID* row_ptr_loc, V_ID vp, int shift,
+                             unsigned long long* cursor, V_ID* out_col) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(blocked_scatter_kernel, dim3(grid_for(ep)), dim3(BLOCK),
+                     0, s, ep, col, row_ptr_loc, vp, shift, cursor, out_col);
 }
 
 void lux_gpu_local_row_ptr(uint64_t stream, uint32_t vp, E_ID col_left,

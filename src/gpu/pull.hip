@@ -25,24 +25,52 @@ enum PullMode { PR_SUM = 0, LAB_MIN = 1, LAB_MAX = 2, CF_SGD = 3 };
 
 // ---------------- bin building ----------------
 
+// LDS-aggregated bin build: per-tile LDS counters, ONE global atomic per
+// bin per tile (global-atomic contention on 3 words made the naive version
+// ~250x slower at nv=2^27 — guideline 12: reduce first, atomic once).
 __global__ void build_bins_kernel(V_ID vp, const E_ID* row_ptr,
                                   V_ID* bin0, V_ID* bin1, uint2* bin2,
                                   V_ID* bin2v, uint32_t* counters) {
+  __shared__ uint32_t lcnt[4], lbase[4];
   uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
-  for (uint64_t v = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; v < vp;
-       v += stride) {
-    E_ID deg = row_ptr[v + 1] - row_ptr[v];
-    if (deg < T1) {
-      bin0[atomicAdd(&counters[0], 1u)] = (V_ID)v;
-    } else if (deg < T2) {
-      bin1[atomicAdd(&counters[1], 1u)] = (V_ID)v;
-    } else {
-      bin2v[atomicAdd(&counters[3], 1u)] = (V_ID)v;
-      uint32_t nchunks = (uint32_t)((deg + CHUNK_EDGES - 1) / CHUNK_EDGES);
-      uint32_t base = atomicAdd(&counters[2], nchunks);
+  for (uint64_t tile = (uint64_t)blockIdx.x * blockDim.x; tile < vp;
+       tile += stride) {
+    uint64_t v = tile + threadIdx.x;
+    if (threadIdx.x < 4) lcnt[threadIdx.x] = 0;
+    __syncthreads();
+    int my_bin = -1;
+    uint32_t lpos = 0, nchunks = 0, lpos_c = 0;
+    if (v < vp) {
+      E_ID deg = row_ptr[v + 1] - row_ptr[v];
+      if (deg < T1) {
+        my_bin = 0;
+        lpos = atomicAdd(&lcnt[0], 1u);
+      } else if (deg < T2) {
+        my_bin = 1;
+        lpos = atomicAdd(&lcnt[1], 1u);
+      } else {
+        my_bin = 3;
+        lpos = atomicAdd(&lcnt[3], 1u);
+        nchunks = (uint32_t)((deg + CHUNK_EDGES - 1) / CHUNK_EDGES);
+        lpos_c = atomicAdd(&lcnt[2], nchunks);
+      }
+    }
+    __syncthreads();
+    if (threadIdx.x < 4)
+      lbase[threadIdx.x] = lcnt[threadIdx.x]
+                               ? atomicAdd(&counters[threadIdx.x],
+                                           lcnt[threadIdx.x])
+                               : 0;
+    __syncthreads();
+    if (my_bin == 0) bin0[lbase[0] + lpos] = (V_ID)v;
+    else if (my_bin == 1) bin1[lbase[1] + lpos] = (V_ID)v;
+    else if (my_bin == 3) {
+      bin2v[lbase[3] + lpos] = (V_ID)v;
+      uint32_t base = lbase[2] + lpos_c;
       for (uint32_t c = 0; c < nchunks; c++)
         bin2[base + c] = make_uint2((V_ID)v, c);
     }
+    __syncthreads();
   }
 }
 
@@ -103,7 +131,24 @@ struct PullArgs {
   const V_ID* deg;       // u32[nv] out-degrees (PR) or null
   V_ID row_left;
   float init_rank;
+  int phase;             // bit0 FIRST (init/seed), bit1 LAST (finish); 3 =
+                         // single-sweep. MID sweeps (0) fold the partial
+                         // already in newv (src-blocked CSC iteration).
 };
+
+// Phased result store for the non-atomic bins: FIRST writes the raw partial
+// (labels seeded with the dst's own old label), MID folds into the existing
+// partial, LAST applies finish().
+template <PullMode M>
+__device__ __forceinline__ void store_result(typename Val<M>::T* slot,
+                                             typename Val<M>::T acc,
+                                             typename Val<M>::T own,
+                                             float init_rank, V_ID deg,
+                                             int phase) {
+  if (!(phase & 1)) acc = Val<M>::comb(acc, *slot);
+  else if (M != PR_SUM) acc = Val<M>::comb(acc, own);
+  *slot = (phase & 2) ? finish<M>(acc, own, init_rank, deg) : acc;
+}
 
 // ---- bin0: thread per vertex ----
 template <PullMode M>
@@ -120,8 +165,8 @@ __global__ void pull_thread_kernel(uint32_t n0, const V_ID* bin0,
     E_ID b = a.row_ptr[v], e = a.row_ptr[v + 1];
     T acc = V::ident();
     for (E_ID j = b; j < e; j++) acc = V::comb(acc, V::map(oldv[a.col[j]]));
-    newv[v] = finish<M>(acc, oldv[a.row_left + v], a.init_rank,
-                        a.deg ? a.deg[a.row_left + v] : 0);
+    store_result<M>(&newv[v], acc, oldv[a.row_left + v], a.init_rank,
+                    a.deg ? a.deg[a.row_left + v] : 0, a.phase);
   }
 }
 
@@ -143,8 +188,8 @@ __global__ void pull_wave_kernel(uint32_t n1, const V_ID* bin1, PullArgs a) {
       acc = V::comb(acc, V::map(oldv[a.col[j]]));
     acc = V::reduce_wave(acc);
     if (lane == 0)
-      newv[v] = finish<M>(acc, oldv[a.row_left + v], a.init_rank,
-                          a.deg ? a.deg[a.row_left + v] : 0);
+      store_result<M>(&newv[v], acc, oldv[a.row_left + v], a.init_rank,
+                      a.deg ? a.deg[a.row_left + v] : 0, a.phase);
   }
 }
 
@@ -221,12 +266,13 @@ static void pull_iter(hipStream_t s, uint32_t n0, const V_ID* bin0,
                       const uint2* bin2, uint32_t nbig, const V_ID* bin2v,
                       const PullArgs& a) {
   if (nbig) {
-    hipLaunchKernelGGL(pull_prep_kernel<M>, dim3(ceil_div_u32(nbig, BLOCK)),
-                       dim3(BLOCK), 0, s, nbig, bin2v, a);
+    if (a.phase & 1)
+      hipLaunchKernelGGL(pull_prep_kernel<M>, dim3(ceil_div_u32(nbig, BLOCK)),
+                         dim3(BLOCK), 0, s, nbig, bin2v, a);
     hipLaunchKernelGGL(pull_chunk_kernel<M>,
                        dim3(n2 > MAX_GRID ? MAX_GRID : n2), dim3(BLOCK), 0, s,
                        n2, bin2, a);
-    if (M == PR_SUM)
+    if (M == PR_SUM && (a.phase & 2))
       hipLaunchKernelGGL(pull_epilogue_kernel<M>,
                          dim3(ceil_div_u32(nbig, BLOCK)), dim3(BLOCK), 0, s,
                          nbig, bin2v, a);
@@ -262,9 +308,10 @@ void lux_gpu_pull_iter(uint64_t stream, int mode, uint32_t n0,
                        uint32_t n2, const uint2* bin2, uint32_t nbig,
                        const V_ID* bin2v, const E_ID* row_ptr,
                        const V_ID* col, const void* oldv, void* newv,
-                       const V_ID* deg, V_ID row_left, float init_rank) {
+                       const V_ID* deg, V_ID row_left, float init_rank,
+                       int phase) {
   hipStream_t s = (hipStream_t)stream;
-  PullArgs a{row_ptr, col, oldv, newv, deg, row_left, init_rank};
+  PullArgs a{row_ptr, col, oldv, newv, deg, row_left, init_rank, phase};
   switch (mode) {
     case 0:
       pull_iter<PR_SUM>(s, n0, bin0, n1, bin1, n2, bin2, nbig, bin2v, a);

@@ -1,0 +1,91 @@
+// C ABI of liblux_gpu.so (gfx950 kernels). Consumed by the Python ctypes
+// bindings (lux_amd/_native_gpu.py) and the native C++ runtime
+// (src/runtime/). Pointers are device pointers; `stream` is a hipStream_t.
+#pragma once
+#include <cstdint>
+
+#include "types.h"
+
+// HIP vector type fwd-compat for non-HIP translation units
+#ifndef __HIP_PLATFORM_AMD__
+struct lux_uint2 { uint32_t x, y; };
+#else
+using lux_uint2 = uint2;
+#endif
+
+extern "C" {
+
+// builder.hip
+void lux_gpu_rmat_edges(uint64_t stream, uint64_t seed, int scale,
+                        uint64_t ne, lux::V_ID* src, lux::V_ID* dst);
+void lux_gpu_rmat_edges_folded(uint64_t stream, uint64_t seed, int scale,
+                               lux::V_ID nv, uint64_t ne, lux::V_ID* src,
+                               lux::V_ID* dst);
+void lux_gpu_bipartite_edges(uint64_t stream, uint64_t seed,
+                             lux::V_ID n_users, lux::V_ID n_items,
+                             uint64_t ne, lux::V_ID* src, lux::V_ID* dst,
+                             lux::WeightType* w);
+void lux_gpu_hist_u32(uint64_t stream, uint64_t n, const lux::V_ID* ids,
+                      uint32_t* hist);
+uint32_t lux_gpu_scan_partials_size(uint32_t n);
+void lux_gpu_scan_end_offsets(uint64_t stream, uint32_t n,
+                              const uint32_t* in, lux::E_ID* out_end,
+                              unsigned long long* partials);
+void lux_gpu_edges_to_csc(uint64_t stream, uint32_t nv, uint64_t ne,
+                          const lux::V_ID* src, const lux::V_ID* dst,
+                          const lux::WeightType* w, lux::E_ID* col_end,
+                          lux::V_ID* out_src, lux::WeightType* out_w,
+                          uint32_t* hist, unsigned long long* cursor,
+                          unsigned long long* partials);
+void lux_gpu_local_row_ptr(uint64_t stream, uint32_t vp, lux::E_ID col_left,
+                           const lux::E_ID* col_end_slice,
+                           lux::E_ID* row_ptr_loc);
+
+// pull.hip
+void lux_gpu_build_bins(uint64_t stream, uint32_t vp,
+                        const lux::E_ID* row_ptr, lux::V_ID* bin0,
+                        lux::V_ID* bin1, lux_uint2* bin2, lux::V_ID* bin2v,
+                        uint32_t* counters);
+void lux_gpu_pull_iter(uint64_t stream, int mode, uint32_t n0,
+                       const lux::V_ID* bin0, uint32_t n1,
+                       const lux::V_ID* bin1, uint32_t n2,
+                       const lux_uint2* bin2, uint32_t nbig,
+                       const lux::V_ID* bin2v, const lux::E_ID* row_ptr,
+                       const lux::V_ID* col, const void* oldv, void* newv,
+                       const lux::V_ID* deg, lux::V_ID row_left,
+                       float init_rank);
+
+// push.hip
+void lux_gpu_csr_scatter(uint64_t stream, uint64_t ep, const lux::V_ID* col,
+                         const lux::E_ID* row_ptr_loc, lux::V_ID vp,
+                         lux::V_ID row_left, unsigned long long* cursor,
+                         lux::V_ID* push_col);
+void lux_gpu_push_scatter(uint64_t stream, int is_min, int old_dense,
+                          int new_dense, lux::V_ID in_row_left,
+                          lux::V_ID in_count, const uint8_t* old_seg,
+                          const lux::E_ID* push_row_ptr,
+                          const lux::V_ID* push_col,
+                          const uint32_t* old_labels,
+                          const uint32_t* snapshot, uint32_t* new_labels,
+                          lux::V_ID my_row_left, uint8_t* new_seg,
+                          lux::V_ID capacity);
+void lux_gpu_build_bitmap(uint64_t stream, lux::V_ID vp,
+                          const uint32_t* snapshot,
+                          const uint32_t* new_labels, uint8_t* seg);
+void lux_gpu_d2s(uint64_t stream, lux::V_ID vp, lux::V_ID row_left,
+                 const uint8_t* dense_seg, uint8_t* sparse_seg);
+void lux_gpu_check(uint64_t stream, int is_min, lux::V_ID vp,
+                   lux::V_ID row_left, const lux::E_ID* row_ptr_loc,
+                   const lux::V_ID* col, const uint32_t* labels,
+                   unsigned long long* mistakes);
+
+// cf.hip
+void lux_gpu_cf_iter(uint64_t stream, uint32_t n0, const lux::V_ID* bin0,
+                     uint32_t n1, const lux::V_ID* bin1, uint32_t n2,
+                     const lux_uint2* bin2, uint32_t nbig,
+                     const lux::V_ID* bin2v, const lux::E_ID* row_ptr,
+                     const lux::V_ID* col, const lux::WeightType* w,
+                     const float* oldv, float* newv, lux::V_ID row_left,
+                     int K);
+
+}  // extern "C"

@@ -1,0 +1,319 @@
+#include "single_gpu.h"
+
+#include <cassert>
+#include <cstdio>
+#include <cstring>
+
+#define CHECK_HIP(cmd)                                                      \
+  do {                                                                      \
+    hipError_t e_ = (cmd);                                                  \
+    if (e_ != hipSuccess) {                                                 \
+      fprintf(stderr, "HIP error %s:%d: %s\n", __FILE__, __LINE__,          \
+              hipGetErrorString(e_));                                       \
+      abort();                                                              \
+    }                                                                       \
+  } while (0)
+
+namespace lux {
+
+DeviceArena::DeviceArena(size_t bytes) : cap_(bytes) {
+  CHECK_HIP(hipMalloc(&base_, bytes));
+}
+DeviceArena::~DeviceArena() { hipFree(base_); }
+void* DeviceArena::alloc(size_t bytes) {
+  size_t off = (used_ + 255) & ~size_t(255);
+  if (off + bytes > cap_) {
+    fprintf(stderr, "lux: FB arena exhausted (%zu + %zu > %zu)\n", off,
+            bytes, cap_);
+    abort();
+  }
+  used_ = off + bytes;
+  return base_ + off;
+}
+
+DeviceGraph DeviceGraph::upload(const HostCSC& g, DeviceArena& arena,
+                                hipStream_t s) {
+  DeviceGraph d;
+  d.nv = g.nv;
+  d.ne = g.ne;
+  d.col_end = arena.alloc_n<E_ID>(g.nv);
+  d.src = arena.alloc_n<V_ID>(g.ne);
+  CHECK_HIP(hipMemcpyAsync(d.col_end, g.col_end.data(),
+                           sizeof(E_ID) * g.nv, hipMemcpyHostToDevice, s));
+  CHECK_HIP(hipMemcpyAsync(d.src, g.src.data(), sizeof(V_ID) * g.ne,
+                           hipMemcpyHostToDevice, s));
+  if (g.weighted()) {
+    d.weight = arena.alloc_n<WeightType>(g.ne);
+    CHECK_HIP(hipMemcpyAsync(d.weight, g.weight.data(),
+                             sizeof(WeightType) * g.ne,
+                             hipMemcpyHostToDevice, s));
+  }
+  CHECK_HIP(hipStreamSynchronize(s));
+  return d;
+}
+
+DeviceGraph DeviceGraph::rmat(int scale, E_ID ne, uint64_t seed,
+                              DeviceArena& arena, hipStream_t s) {
+  DeviceGraph d;
+  d.nv = (V_ID)1 << scale;
+  d.ne = ne;
+  d.col_end = arena.alloc_n<E_ID>(d.nv);
+  d.src = arena.alloc_n<V_ID>(ne);
+  // temporaries via hipMalloc (freed after build, not arena-held)
+  V_ID *esrc, *edst;
+  uint32_t* hist;
+  unsigned long long *cursor, *partials;
+  CHECK_HIP(hipMalloc(&esrc, sizeof(V_ID) * ne));
+  CHECK_HIP(hipMalloc(&edst, sizeof(V_ID) * ne));
+  CHECK_HIP(hipMalloc(&hist, sizeof(uint32_t) * d.nv));
+  CHECK_HIP(hipMalloc(&cursor, sizeof(uint64_t) * d.nv));
+  CHECK_HIP(
+      hipMalloc(&partials, sizeof(uint64_t) * lux_gpu_scan_partials_size(d.nv)));
+  CHECK_HIP(hipMemsetAsync(hist, 0, sizeof(uint32_t) * d.nv, s));
+  lux_gpu_rmat_edges((uint64_t)s, seed, scale, ne, esrc, edst);
+  lux_gpu_edges_to_csc((uint64_t)s, d.nv, ne, esrc, edst, nullptr, d.col_end,
+                       d.src, nullptr, hist, cursor, partials);
+  CHECK_HIP(hipStreamSynchronize(s));
+  hipFree(esrc);
+  hipFree(edst);
+  hipFree(hist);
+  hipFree(cursor);
+  hipFree(partials);
+  return d;
+}
+
+void Bins::build(const E_ID* row_ptr_loc, V_ID vp, E_ID ep,
+                 DeviceArena& arena, hipStream_t s) {
+  bin0 = arena.alloc_n<V_ID>(vp);
+  bin1 = arena.alloc_n<V_ID>(vp);
+  uint32_t nbig_max = (uint32_t)std::min<uint64_t>(vp, ep / 2048 + 1);
+  uint32_t n2_max = (uint32_t)(ep / 8192 + nbig_max + 1);
+  bin2 = arena.alloc_n<lux_uint2>(n2_max);
+  bin2v = arena.alloc_n<V_ID>(nbig_max);
+  uint32_t* counters;
+  CHECK_HIP(hipMalloc(&counters, 4 * sizeof(uint32_t)));
+  CHECK_HIP(hipMemsetAsync(counters, 0, 16, s));
+  lux_gpu_build_bins((uint64_t)s, vp, row_ptr_loc, bin0, bin1, bin2, bin2v,
+                     counters);
+  uint32_t c[4];
+  CHECK_HIP(hipMemcpyAsync(c, counters, 16, hipMemcpyDeviceToHost, s));
+  CHECK_HIP(hipStreamSynchronize(s));
+  hipFree(counters);
+  n0 = c[0];
+  n1 = c[1];
+  n2 = c[2];
+  nbig = c[3];
+}
+
+// ---------------- PageRank ----------------
+
+SingleGpuPagerank::SingleGpuPagerank(const DeviceGraph& g, DeviceArena& arena,
+                                     hipStream_t s)
+    : g_(g), s_(s) {
+  row_ptr_ = arena.alloc_n<E_ID>(g.nv + 1);
+  lux_gpu_local_row_ptr((uint64_t)s, g.nv, 0, g.col_end, row_ptr_);
+  bins_.build(row_ptr_, g.nv, g.ne, arena, s);
+  deg_ = arena.alloc_n<V_ID>(g.nv);
+  CHECK_HIP(hipMemsetAsync(deg_, 0, sizeof(V_ID) * g.nv, s));
+  lux_gpu_hist_u32((uint64_t)s, g.ne, g.src, deg_);
+  old_ = arena.alloc_n<float>(g.nv);
+  new_ = arena.alloc_n<float>(g.nv);
+  // init on host: rank/deg (pagerank_gpu.cu:255-259)
+  std::vector<V_ID> hdeg(g.nv);
+  CHECK_HIP(hipMemcpyAsync(hdeg.data(), deg_, sizeof(V_ID) * g.nv,
+                           hipMemcpyDeviceToHost, s));
+  CHECK_HIP(hipStreamSynchronize(s));
+  std::vector<float> hpr(g.nv);
+  float rank = 1.0f / g.nv;
+  for (V_ID v = 0; v < g.nv; v++)
+    hpr[v] = hdeg[v] == 0 ? rank : rank / hdeg[v];
+  CHECK_HIP(hipMemcpyAsync(old_, hpr.data(), sizeof(float) * g.nv,
+                           hipMemcpyHostToDevice, s));
+  CHECK_HIP(hipStreamSynchronize(s));
+}
+
+void SingleGpuPagerank::iterate(int iters) {
+  float init_rank = (1.0f - PR_ALPHA) / g_.nv;
+  for (int it = 0; it < iters; it++) {
+    lux_gpu_pull_iter((uint64_t)s_, 0, bins_.n0, bins_.bin0, bins_.n1,
+                      bins_.bin1, bins_.n2, bins_.bin2, bins_.nbig,
+                      bins_.bin2v, row_ptr_, g_.src, old_, new_, deg_, 0,
+                      init_rank);
+    std::swap(old_, new_);
+  }
+  CHECK_HIP(hipStreamSynchronize(s_));
+}
+
+// ---------------- Push (SSSP / CC) ----------------
+
+SingleGpuPush::SingleGpuPush(const DeviceGraph& g, bool is_min, V_ID source,
+                             DeviceArena& arena, hipStream_t s, bool verbose)
+    : g_(g), s_(s), is_min_(is_min), verbose_(verbose) {
+  row_ptr_ = arena.alloc_n<E_ID>(g.nv + 1);
+  lux_gpu_local_row_ptr((uint64_t)s, g.nv, 0, g.col_end, row_ptr_);
+  bins_.build(row_ptr_, g.nv, g.ne, arena, s);
+  // push CSR (single partition: transpose over all nv)
+  push_row_ptr_ = arena.alloc_n<E_ID>(g.nv + 1);
+  push_col_ = arena.alloc_n<V_ID>(g.ne);
+  {
+    uint32_t* degs;
+    E_ID* ends;
+    unsigned long long *cursor, *partials;
+    CHECK_HIP(hipMalloc(&degs, sizeof(uint32_t) * g.nv));
+    CHECK_HIP(hipMalloc(&ends, sizeof(E_ID) * g.nv));
+    CHECK_HIP(hipMalloc(&cursor, sizeof(uint64_t) * g.nv));
+    CHECK_HIP(hipMalloc(&partials,
+                        sizeof(uint64_t) * lux_gpu_scan_partials_size(g.nv)));
+    CHECK_HIP(hipMemsetAsync(degs, 0, sizeof(uint32_t) * g.nv, s));
+    lux_gpu_hist_u32((uint64_t)s, g.ne, g.src, degs);
+    lux_gpu_scan_end_offsets((uint64_t)s, g.nv, degs, ends, partials);
+    lux_gpu_local_row_ptr((uint64_t)s, g.nv, 0, ends, push_row_ptr_);
+    CHECK_HIP(hipMemcpyAsync(cursor, push_row_ptr_, sizeof(E_ID) * g.nv,
+                             hipMemcpyDeviceToDevice, s));
+    lux_gpu_csr_scatter((uint64_t)s, g.ne, g.src, row_ptr_, g.nv, 0, cursor,
+                        push_col_);
+    CHECK_HIP(hipStreamSynchronize(s));
+    hipFree(degs);
+    hipFree(ends);
+    hipFree(cursor);
+    hipFree(partials);
+  }
+  labels_ = arena.alloc_n<uint32_t>(g.nv);
+  snapshot_ = arena.alloc_n<uint32_t>(g.nv);
+  uint64_t fq_bytes = frontier_bytes(g.nv);
+  fq_ = arena.alloc_n<uint8_t>(fq_bytes);
+  new_fq_ = arena.alloc_n<uint8_t>(fq_bytes);
+  tmp_fq_ = arena.alloc_n<uint8_t>(fq_bytes);
+  capacity_ = frontier_capacity(g.nv);
+  // seed labels + frontier (sssp_gpu.cu:733-744, components_gpu.cu:733-740)
+  std::vector<uint32_t> hl(g.nv);
+  std::vector<uint8_t> hfq(fq_bytes, 0);
+  FrontierHeader* hdr = (FrontierHeader*)hfq.data();
+  if (is_min) {
+    for (V_ID v = 0; v < g.nv; v++) hl[v] = INF_LABEL;
+    hl[source] = 0;
+    hdr->type = FrontierHeader::SPARSE_QUEUE;
+    hdr->numNodes = 1;
+    *(V_ID*)(hfq.data() + sizeof(FrontierHeader)) = source;
+  } else {
+    for (V_ID v = 0; v < g.nv; v++) hl[v] = v;
+    hdr->type = FrontierHeader::DENSE_BITMAP;
+    hdr->numNodes = g.nv;
+    memset(hfq.data() + sizeof(FrontierHeader), 0xFF, (g.nv + 7) / 8);
+  }
+  fq_type_ = hdr->type;
+  fq_num_ = hdr->numNodes;
+  CHECK_HIP(hipMemcpyAsync(labels_, hl.data(), sizeof(uint32_t) * g.nv,
+                           hipMemcpyHostToDevice, s));
+  CHECK_HIP(hipMemcpyAsync(fq_, hfq.data(), fq_bytes, hipMemcpyHostToDevice,
+                           s));
+  CHECK_HIP(hipStreamSynchronize(s));
+}
+
+V_ID SingleGpuPush::step() {
+  CHECK_HIP(hipMemcpyAsync(snapshot_, labels_, sizeof(uint32_t) * g_.nv,
+                           hipMemcpyDeviceToDevice, s_));
+  CHECK_HIP(hipMemsetAsync(new_fq_, 0, sizeof(FrontierHeader), s_));
+  bool new_dense = fq_type_ == FrontierHeader::DENSE_BITMAP;
+  bool pull_fallback = fq_num_ > g_.nv / SPARSE_THRESHOLD;
+  if (pull_fallback) {
+    new_dense = true;
+    // dense pull iteration; labels_ serves as both old (all) and new (slice)
+    lux_gpu_pull_iter((uint64_t)s_, is_min_ ? 1 : 2, bins_.n0, bins_.bin0,
+                      bins_.n1, bins_.bin1, bins_.n2, bins_.bin2, bins_.nbig,
+                      bins_.bin2v, row_ptr_, g_.src, snapshot_, labels_,
+                      nullptr, 0, 0.0f);
+  } else {
+    lux_gpu_push_scatter((uint64_t)s_, is_min_ ? 1 : 0,
+                         fq_type_ == FrontierHeader::DENSE_BITMAP ? 1 : 0,
+                         new_dense ? 1 : 0, 0,
+                         fq_type_ == FrontierHeader::DENSE_BITMAP ? g_.nv
+                                                                  : fq_num_,
+                         fq_, push_row_ptr_, push_col_, snapshot_, snapshot_,
+                         labels_, 0, new_fq_, capacity_);
+  }
+  FrontierHeader hh;
+  if (new_dense) {
+    lux_gpu_build_bitmap((uint64_t)s_, g_.nv, snapshot_, labels_, new_fq_);
+    CHECK_HIP(hipMemcpyAsync(&hh, new_fq_, 8, hipMemcpyDeviceToHost, s_));
+    CHECK_HIP(hipStreamSynchronize(s_));
+    if (hh.numNodes < capacity_) {
+      CHECK_HIP(hipMemcpyAsync(tmp_fq_, new_fq_, frontier_bytes(g_.nv),
+                               hipMemcpyDeviceToDevice, s_));
+      CHECK_HIP(hipMemsetAsync(new_fq_, 0, sizeof(FrontierHeader), s_));
+      lux_gpu_d2s((uint64_t)s_, g_.nv, 0, tmp_fq_, new_fq_);
+      CHECK_HIP(hipMemcpyAsync(&hh, new_fq_, 8, hipMemcpyDeviceToHost, s_));
+      CHECK_HIP(hipStreamSynchronize(s_));
+      new_dense = false;
+    }
+  } else {
+    CHECK_HIP(hipMemcpyAsync(&hh, new_fq_, 8, hipMemcpyDeviceToHost, s_));
+    CHECK_HIP(hipStreamSynchronize(s_));
+    if (hh.numNodes >= capacity_) {
+      new_dense = true;
+      CHECK_HIP(hipMemsetAsync(new_fq_, 0, sizeof(FrontierHeader), s_));
+      lux_gpu_build_bitmap((uint64_t)s_, g_.nv, snapshot_, labels_, new_fq_);
+      CHECK_HIP(hipMemcpyAsync(&hh, new_fq_, 8, hipMemcpyDeviceToHost, s_));
+      CHECK_HIP(hipStreamSynchronize(s_));
+    }
+  }
+  fq_type_ = new_dense ? FrontierHeader::DENSE_BITMAP
+                       : FrontierHeader::SPARSE_QUEUE;
+  fq_num_ = hh.numNodes;
+  std::swap(fq_, new_fq_);
+  if (verbose_)
+    printf("iter %d: activeNodes(%u) %s\n", iters_, fq_num_,
+           new_dense ? "dense" : "sparse");
+  return fq_num_;
+}
+
+int SingleGpuPush::run(int max_iters) {
+  while (true) {
+    V_ID n = step();
+    iters_++;
+    if (n == 0) break;
+    if (max_iters && iters_ >= max_iters) break;
+  }
+  return iters_;
+}
+
+uint64_t SingleGpuPush::check() {
+  unsigned long long* mistakes;
+  CHECK_HIP(hipMalloc(&mistakes, 8));
+  CHECK_HIP(hipMemsetAsync(mistakes, 0, 8, s_));
+  lux_gpu_check((uint64_t)s_, is_min_ ? 1 : 0, g_.nv, 0, row_ptr_, g_.src,
+                labels_, mistakes);
+  unsigned long long h;
+  CHECK_HIP(hipMemcpyAsync(&h, mistakes, 8, hipMemcpyDeviceToHost, s_));
+  CHECK_HIP(hipStreamSynchronize(s_));
+  hipFree(mistakes);
+  return h;
+}
+
+// ---------------- CF ----------------
+
+SingleGpuCF::SingleGpuCF(const DeviceGraph& g, int K, DeviceArena& arena,
+                         hipStream_t s)
+    : g_(g), s_(s), K_(K) {
+  row_ptr_ = arena.alloc_n<E_ID>(g.nv + 1);
+  lux_gpu_local_row_ptr((uint64_t)s, g.nv, 0, g.col_end, row_ptr_);
+  bins_.build(row_ptr_, g.nv, g.ne, arena, s);
+  old_ = arena.alloc_n<float>((size_t)g.nv * K);
+  new_ = arena.alloc_n<float>((size_t)g.nv * K);
+  std::vector<float> h((size_t)g.nv * K, sqrtf(1.0f / K));
+  CHECK_HIP(hipMemcpyAsync(old_, h.data(), sizeof(float) * h.size(),
+                           hipMemcpyHostToDevice, s));
+  CHECK_HIP(hipStreamSynchronize(s));
+}
+
+void SingleGpuCF::iterate(int iters) {
+  for (int it = 0; it < iters; it++) {
+    lux_gpu_cf_iter((uint64_t)s_, bins_.n0, bins_.bin0, bins_.n1, bins_.bin1,
+                    bins_.n2, bins_.bin2, bins_.nbig, bins_.bin2v, row_ptr_,
+                    g_.src, g_.weight, old_, new_, 0, K_);
+    std::swap(old_, new_);
+  }
+  CHECK_HIP(hipStreamSynchronize(s_));
+}
+
+}  // namespace lux

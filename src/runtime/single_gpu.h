@@ -1,0 +1,117 @@
+// Native single-GPU runtime: FB arena allocator + engine drivers that run
+// the gfx950 kernels without Python. Used by the CLI app binaries
+// (apps/*.cpp) — the MI355X equivalents of the reference's per-app
+// top_level_task drivers (pagerank/pagerank.cc:32-119 etc.), with the
+// Legion region machinery replaced by explicit device buffers on one HIP
+// stream. (Multi-GPU runs go through the torch.distributed/RCCL Python
+// engine — one process per GPU.)
+#pragma once
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <vector>
+
+#include "lux/graph.h"
+#include "lux/gpu_api.h"
+#include "lux/types.h"
+
+namespace lux {
+
+// FB memory pool: one upfront hipMalloc, bump allocation, 256-B alignment
+// (we own our arena — no Realm-internal back-doors, SURVEY.md §7 "fix").
+class DeviceArena {
+ public:
+  explicit DeviceArena(size_t bytes);
+  ~DeviceArena();
+  void* alloc(size_t bytes);
+  template <typename T>
+  T* alloc_n(size_t n) {
+    return (T*)alloc(n * sizeof(T));
+  }
+  size_t used() const { return used_; }
+  size_t capacity() const { return cap_; }
+
+ private:
+  char* base_ = nullptr;
+  size_t cap_ = 0, used_ = 0;
+};
+
+struct DeviceGraph {
+  V_ID nv = 0;
+  E_ID ne = 0;
+  E_ID* col_end = nullptr;   // u64[nv] end offsets
+  V_ID* src = nullptr;       // u32[ne]
+  WeightType* weight = nullptr;
+
+  static DeviceGraph upload(const HostCSC& g, DeviceArena& arena,
+                            hipStream_t s);
+  // Synthetic generation fully on-device.
+  static DeviceGraph rmat(int scale, E_ID ne, uint64_t seed,
+                          DeviceArena& arena, hipStream_t s);
+};
+
+// Degree bins shared by the pull/CF engines (built once; see pull.hip).
+struct Bins {
+  V_ID *bin0, *bin1, *bin2v;
+  lux_uint2* bin2;
+  uint32_t n0, n1, n2, nbig;
+  void build(const E_ID* row_ptr_loc, V_ID vp, E_ID ep, DeviceArena& arena,
+             hipStream_t s);
+};
+
+class SingleGpuPagerank {
+ public:
+  SingleGpuPagerank(const DeviceGraph& g, DeviceArena& arena, hipStream_t s);
+  void iterate(int iters);
+  const float* ranks() const { return old_; }  // device ptr, stored pr/deg
+
+ private:
+  const DeviceGraph& g_;
+  hipStream_t s_;
+  E_ID* row_ptr_;
+  Bins bins_;
+  V_ID* deg_;
+  float *old_, *new_;
+};
+
+class SingleGpuPush {
+ public:
+  SingleGpuPush(const DeviceGraph& g, bool is_min, V_ID source,
+                DeviceArena& arena, hipStream_t s, bool verbose = false);
+  // returns iterations to convergence
+  int run(int max_iters = 0);
+  uint64_t check();  // violation count (check oracle)
+  const uint32_t* labels() const { return labels_; }
+
+ private:
+  V_ID step();  // returns new frontier size
+  const DeviceGraph& g_;
+  hipStream_t s_;
+  bool is_min_, verbose_;
+  E_ID *row_ptr_, *push_row_ptr_;
+  V_ID* push_col_;
+  Bins bins_;
+  uint32_t *labels_, *snapshot_;
+  uint8_t *fq_, *new_fq_, *tmp_fq_;
+  V_ID capacity_;
+  uint32_t fq_type_, fq_num_;
+  int iters_ = 0;
+};
+
+class SingleGpuCF {
+ public:
+  SingleGpuCF(const DeviceGraph& g, int K, DeviceArena& arena,
+              hipStream_t s);
+  void iterate(int iters);
+  const float* vectors() const { return old_; }
+
+ private:
+  const DeviceGraph& g_;
+  hipStream_t s_;
+  int K_;
+  E_ID* row_ptr_;
+  Bins bins_;
+  float *old_, *new_;
+};
+
+}  // namespace lux
