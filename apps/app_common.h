@@ -1,0 +1,76 @@
+// Shared CLI parsing for the native app binaries — flag surface parity with
+// the reference drivers (sssp/sssp.cc:148-180, README.md:42-54): -ng/-ll:gpu,
+// -ni, -file, -start, -verbose/-v, -check/-c; -ll:fsize/-ll:zsize and other
+// Legion flags are accepted and ignored. Extra (ours): -synthetic spec, -k.
+//
+// These binaries run the single-GPU native runtime (src/runtime/); the
+// multi-GPU path is `torchrun ... python -m lux_amd.apps.<app>` (one process
+// per GPU over RCCL). -ng > 1 here prints a pointer to that path.
+#pragma once
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <string>
+
+#include "lux/graph.h"
+
+struct AppArgs {
+  int num_gpu = 1;
+  int num_iter = 10;
+  const char* file = nullptr;
+  lux::V_ID start = 0;
+  bool verbose = false;
+  bool check = false;
+  int k = 64;
+  const char* synthetic = nullptr;
+};
+
+inline AppArgs parse_input_args(int argc, char** argv) {
+  AppArgs a;
+  for (int i = 1; i < argc; i++) {
+    std::string f = argv[i];
+    auto next = [&]() { return argv[++i]; };
+    if (f == "-ng" || f == "-ll:gpu") a.num_gpu = atoi(next());
+    else if (f == "-ni") a.num_iter = atoi(next());
+    else if (f == "-file") a.file = next();
+    else if (f == "-start") a.start = (lux::V_ID)atoll(next());
+    else if (f == "-verbose" || f == "-v") a.verbose = true;
+    else if (f == "-check" || f == "-c") a.check = true;
+    else if (f == "-k") a.k = atoi(next());
+    else if (f == "-synthetic") a.synthetic = next();
+    else if (f.rfind("-ll:", 0) == 0 || f.rfind("-lg:", 0) == 0) {
+      if (i + 1 < argc && argv[i + 1][0] != '-') i++;  // value-flag: skip
+    } else {
+      fprintf(stderr, "warning: ignoring unknown flag %s\n", f.c_str());
+    }
+  }
+  return a;
+}
+
+inline bool load_graph(const AppArgs& a, lux::HostCSC* g, bool weighted) {
+  if (a.file) return lux::lux_read(a.file, g, weighted);
+  if (a.synthetic) {
+    int scale;
+    long long ne;
+    if (sscanf(a.synthetic, "rmat:%d:%lld", &scale, &ne) == 2) {
+      *g = lux::rmat_csc_cpu(scale, (lux::E_ID)ne, 1);
+      return true;
+    }
+    long long nu, ni;
+    if (sscanf(a.synthetic, "bipartite:%lld:%lld:%lld", &nu, &ni, &ne) == 3) {
+      *g = lux::bipartite_csc_cpu((lux::V_ID)nu, (lux::V_ID)ni,
+                                  (lux::E_ID)ne, 1);
+      return true;
+    }
+  }
+  fprintf(stderr, "usage: need -file graph.lux or -synthetic rmat:S:NE\n");
+  return false;
+}
+
+inline void print_memory_estimate(lux::V_ID nv, lux::E_ID ne, bool weighted,
+                                  int k) {
+  double fb = double(ne) * (weighted ? 8 : 4) + 48.0 * nv +
+              8.0 * k * nv;
+  printf("[lux] estimated FB usage: %.0f MB (of 294912 MB HBM3E)\n",
+         fb / (1 << 20));
+}

@@ -1,0 +1,43 @@
+// Native collaborative-filtering driver (single GPU) — reference parity:
+// /root/reference/col_filter/colfilter.cc (weighted graph, -ni SGD sweeps,
+// latent rank -k; reference K=20, our default 64).
+#include <chrono>
+#include <cstdio>
+
+#include "../src/runtime/single_gpu.h"
+#include "app_common.h"
+
+using namespace lux;
+
+int main(int argc, char** argv) {
+  AppArgs a = parse_input_args(argc, argv);
+  if (a.num_gpu > 1) {
+    fprintf(stderr,
+            "[lux] multi-GPU runs use the RCCL engine: torchrun "
+            "--nproc-per-node %d -m lux_amd.apps.cf ...\n", a.num_gpu);
+    return 2;
+  }
+  HostCSC g;
+  if (!load_graph(a, &g, true)) return 1;
+  print_memory_estimate(g.nv, g.ne, true, a.k);
+
+  hipStream_t s;
+  LUX_OK(hipStreamCreate(&s));
+  size_t arena_bytes = 8ull * g.nv + 8ull * g.ne          // graph + weights
+                       + 8ull * (g.nv + 1)
+                       + 12ull * g.nv + (64ull << 20)
+                       + 8ull * (uint64_t)g.nv * a.k      // old/new vectors
+                       + 8ull * (g.ne / 8192 + g.nv / 16);
+  DeviceArena arena(arena_bytes);
+  DeviceGraph dg = DeviceGraph::upload(g, arena, s);
+  SingleGpuCF engine(dg, a.k, arena, s);
+
+  auto t0 = std::chrono::steady_clock::now();
+  engine.iterate(a.num_iter);
+  auto t1 = std::chrono::steady_clock::now();
+  double secs = std::chrono::duration<double>(t1 - t0).count();
+  printf("ELAPSED TIME = %7.7f s\n", secs);
+  printf("[lux] %.3f GTEPS (%d sweeps, rank %d)\n",
+         double(g.ne) * a.num_iter / secs / 1e9, a.num_iter, a.k);
+  return 0;
+}

@@ -159,3 +159,8 @@ def blocked_scatter(stream, ep, col, row_ptr_loc, vp, shift, cursor,
                                   dp(row_ptr_loc), _u32(vp),
                                   ctypes.c_int(shift), dp(cursor),
                                   dp(out_col))
+
+
+def pull_finish_pr(stream, vp, newv, deg, row_left, init_rank):
+    lib().lux_gpu_pull_finish_pr(_u64(stream), _u32(vp), dp(newv), dp(deg),
+                                 _u32(row_left), ctypes.c_float(init_rank))

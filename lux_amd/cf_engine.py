@@ -34,6 +34,10 @@ class CFEngine:
 
     def step(self):
         p = self.part
+        # seed: old*(1 - GAMMA*LAMBDA); sweeps add GAMMA*acc (see cf.hip)
+        self.new_part.copy_(
+            self.old.narrow(0, p.row_left * self.K, p.vp * self.K))
+        self.new_part.mul_(1.0 - 0.00000035 * 0.001)
         ng.cf_iter(_stream(), p.n0, p.bin0, p.n1, p.bin1, p.n2, p.bin2,
                    p.nbig, p.bin2v, p.row_ptr, p.col, p.weight, self.old,
                    self.new_part, p.row_left, self.K)

@@ -47,11 +47,17 @@ def _bins_for(row_ptr, vp, ep, device, compact=False):
     return n0, n1, n2, nbig, bin0, bin1, bin2, bin2v
 
 
-def run_pull(part, mode, oldv, newv, deg, init_rank):
-    """One pull sweep over this rank's partition — via the src-blocked CSC
-    (phased accumulation, LLC-resident gathers) when built, else the plain
-    single-sweep path."""
+def run_pull(part, mode, oldv, newv, deg, init_rank, seed=True):
+    """One pull iteration over this rank's partition. Contract: seed newv
+    (PR: zeros, labels: the old label slice), fold-sweep the (blocked) CSC,
+    then for PR apply the epilogue once (pull_finish_pr). Uses the
+    src-blocked CSC (LLC-resident gathers) when built."""
     s = _stream()
+    if seed:
+        if mode == ng.PULL_PR:
+            newv.zero_()
+        else:
+            newv.copy_(oldv.narrow(0, part.row_left, part.vp))
     blocks = getattr(part, "blocks", None)
     if blocks:
         nb = len(blocks)
@@ -64,7 +70,9 @@ def run_pull(part, mode, oldv, newv, deg, init_rank):
     else:
         ng.pull_iter(s, mode, part.n0, part.bin0, part.n1, part.bin1,
                      part.n2, part.bin2, part.nbig, part.bin2v, part.row_ptr,
-                     part.col, oldv, newv, deg, part.row_left, init_rank, 3)
+                     part.col, oldv, newv, deg, part.row_left, init_rank, 0)
+    if mode == ng.PULL_PR:
+        ng.pull_finish_pr(s, part.vp, newv, deg, part.row_left, init_rank)
 
 
 def partition_bounds(col_end, ne, nparts):

@@ -216,6 +216,7 @@ void lux_gpu_rmat_edges(uint64_t stream, uint64_t seed, int scale,
   hipStream_t s = (hipStream_t)stream;
   hipLaunchKernelGGL(rmat_edges_kernel, dim3(grid_for(ne)), dim3(BLOCK), 0, s,
                      seed, scale, ne, src, dst);
+  LUX_POST_LAUNCH(stream);
 }
 
 void lux_gpu_rmat_edges_folded(uint64_t stream, uint64_t seed, int scale,
@@ -223,6 +224,7 @@ void lux_gpu_rmat_edges_folded(uint64_t stream, uint64_t seed, int scale,
   hipStream_t s = (hipStream_t)stream;
   hipLaunchKernelGGL(rmat_edges_folded_kernel, dim3(grid_for(ne)),
                      dim3(BLOCK), 0, s, seed, scale, nv, ne, src, dst);
+  LUX_POST_LAUNCH(stream);
 }
 
 void lux_gpu_bipartite_edges(uint64_t stream, uint64_t seed, V_ID n_users,
@@ -234,6 +236,7 @@ void lux_gpu_bipartite_edges(uint64_t stream, uint64_t seed, V_ID n_users,
   hipLaunchKernelGGL(bipartite_edges_kernel, dim3(grid_for(ne)), dim3(BLOCK),
                      0, s, seed, n_users, n_items, item_scale, ne, src, dst,
                      w);
+  LUX_POST_LAUNCH(stream);
 }
 
 void lux_gpu_hist_u32(uint64_t stream, uint64_t n, const V_ID* ids,
@@ -241,6 +244,7 @@ void lux_gpu_hist_u32(uint64_t stream, uint64_t n, const V_ID* ids,
   hipStream_t s = (hipStream_t)stream;
   hipLaunchKernelGGL(hist_u32_kernel, dim3(grid_for(n)), dim3(BLOCK), 0, s, n,
                      ids, hist);
+  LUX_POST_LAUNCH(stream);
 }
 
 uint32_t lux_gpu_scan_partials_size(uint32_t n) {
@@ -250,6 +254,7 @@ uint32_t lux_gpu_scan_partials_size(uint32_t n) {
 void lux_gpu_scan_end_offsets(uint64_t stream, uint32_t n, const uint32_t* in,
                               E_ID* out_end, unsigned long long* partials) {
   scan_u32_to_end_u64((hipStream_t)stream, n, in, out_end, partials);
+  LUX_POST_LAUNCH(stream);
 }
 
 // Full CSC build from device edge lists. hist must be pre-zeroed u32[nv];
@@ -268,6 +273,7 @@ void lux_gpu_edges_to_csc(uint64_t stream, uint32_t nv, uint64_t ne,
                      s, nv, col_end, cursor);
   hipLaunchKernelGGL(scatter_kernel, dim3(grid_for(ne)), dim3(BLOCK), 0, s,
                      ne, src, dst, w, cursor, out_src, out_w);
+  LUX_POST_LAUNCH(stream);
 }
 
 void lux_gpu_blocked_count(uint64_t stream, uint64_t ep, const V_ID* col,
@@ -276,6 +282,7 @@ void lux_gpu_blocked_count(uint64_t stream, uint64_t ep, const V_ID* col,
   hipStream_t s = (hipStream_t)stream;
   hipLaunchKernelGGL(blocked_count_kernel, dim3(grid_for(ep)), dim3(BLOCK),
                      0, s, ep, col, row_ptr_loc, vp, shift, counts);
+  LUX_POST_LAUNCH(stream);
 }
 
 void lux_gpu_blocked_scatter(uint64_t stream, uint64_t ep, const V_ID* col,
@@ -284,6 +291,7 @@ void lux_gpu_blocked_scatter(uint64_t stream, uint64_t ep, const V_ID* col,
   hipStream_t s = (hipStream_t)stream;
   hipLaunchKernelGGL(blocked_scatter_kernel, dim3(grid_for(ep)), dim3(BLOCK),
                      0, s, ep, col, row_ptr_loc, vp, shift, cursor, out_col);
+  LUX_POST_LAUNCH(stream);
 }
 
 void lux_gpu_local_row_ptr(uint64_t stream, uint32_t vp, E_ID col_left,
@@ -292,6 +300,7 @@ void lux_gpu_local_row_ptr(uint64_t stream, uint32_t vp, E_ID col_left,
   hipLaunchKernelGGL(local_row_ptr_kernel, dim3(grid_for((uint64_t)vp + 1)),
                      dim3(BLOCK), 0, s, vp, col_left, col_end_slice,
                      row_ptr_loc);
+  LUX_POST_LAUNCH(stream);
 }
 
 }  // extern "C"

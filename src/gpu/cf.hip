@@ -65,23 +65,15 @@ __global__ void cf_wave_kernel(uint32_t n, const V_ID* binlist, CFArgs a) {
 #pragma unroll
       for (int c = 0; c < CF_MAXC; c++) acc[c] += err * sv[c];
     }
+    // output pre-seeded by the engine with old*(1 - GAMMA*LAMBDA); sweeps
+    // add GAMMA*acc (identical update to colfilter_gpu.cu:96-102, split so
+    // deg-0 rows and hub chunks need no separate epilogue)
     float* out = a.newv + (uint64_t)v * a.K;
 #pragma unroll
     for (int c = 0; c < CF_MAXC; c++) {
       int k = c * WAVE + lane;
-      if (c < nc && k < a.K)
-        out[k] = dv[c] + CF_GAMMA * (acc[c] - CF_LAMBDA * dv[c]);
+      if (c < nc && k < a.K) out[k] += CF_GAMMA * acc[c];
     }
-  }
-}
-
-// hub path: zero the acc area, chunk-accumulate, vector epilogue
-__global__ void cf_prep_kernel(uint32_t nbig, const V_ID* bin2v, CFArgs a) {
-  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  uint64_t total = (uint64_t)nbig * a.K;
-  for (; i < total; i += (uint64_t)blockDim.x * gridDim.x) {
-    V_ID v = bin2v[i / a.K];
-    a.newv[(uint64_t)v * a.K + i % a.K] = 0.0f;
   }
 }
 
@@ -125,23 +117,11 @@ __global__ void cf_chunk_kernel(uint32_t n2, const uint2* bin2,
 #pragma unroll
     for (int c = 0; c < CF_MAXC; c++) {
       int k = c * WAVE + lane;
-      if (c < nc && k < a.K) atomicAdd(&out[k], acc[c]);
+      if (c < nc && k < a.K) atomicAdd(&out[k], CF_GAMMA * acc[c]);
     }
   }
 }
 
-__global__ void cf_epilogue_kernel(uint32_t nbig, const V_ID* bin2v,
-                                   CFArgs a) {
-  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  uint64_t total = (uint64_t)nbig * a.K;
-  for (; i < total; i += (uint64_t)blockDim.x * gridDim.x) {
-    V_ID v = bin2v[i / a.K];
-    int k = i % a.K;
-    float dv = a.oldv[(uint64_t)(a.row_left + v) * a.K + k];
-    float acc = a.newv[(uint64_t)v * a.K + k];
-    a.newv[(uint64_t)v * a.K + k] = dv + CF_GAMMA * (acc - CF_LAMBDA * dv);
-  }
-}
 
 }  // namespace lux
 
@@ -160,14 +140,8 @@ void lux_gpu_cf_iter(uint64_t stream, uint32_t n0, const V_ID* bin0,
   hipStream_t s = (hipStream_t)stream;
   CFArgs a{row_ptr, col, w, oldv, newv, row_left, K};
   if (nbig) {
-    hipLaunchKernelGGL(cf_prep_kernel,
-                       dim3(grid_for((uint64_t)nbig * K)), dim3(BLOCK), 0, s,
-                       nbig, bin2v, a);
     hipLaunchKernelGGL(cf_chunk_kernel, dim3(n2 > MAX_GRID ? MAX_GRID : n2),
                        dim3(BLOCK), 0, s, n2, bin2, (V_ID)8192, a);
-    hipLaunchKernelGGL(cf_epilogue_kernel,
-                       dim3(grid_for((uint64_t)nbig * K)), dim3(BLOCK), 0, s,
-                       nbig, bin2v, a);
   }
   if (n1)
     hipLaunchKernelGGL(cf_wave_kernel, dim3(grid_for((uint64_t)n1 * WAVE)),
@@ -175,6 +149,7 @@ void lux_gpu_cf_iter(uint64_t stream, uint32_t n0, const V_ID* bin0,
   if (n0)
     hipLaunchKernelGGL(cf_wave_kernel, dim3(grid_for((uint64_t)n0 * WAVE)),
                        dim3(BLOCK), 0, s, n0, bin0, a);
+  LUX_POST_LAUNCH(stream);
 }
 
 }  // extern "C"

@@ -22,6 +22,34 @@
     }                                                                       \
   } while (0)
 
+// Post-launch check for every C-ABI entry point: catches bad launch configs
+// immediately; with LUX_SYNC_CHECK=1 in the env also synchronises and
+// surfaces async faults at the offending call (debug mode).
+// NOTE: hipGetLastError() is process-global sticky state — other libraries
+// (torch probing) can leave a stale error behind, so the non-sync path only
+// warns (once). LUX_SYNC_CHECK=1 turns every entry point into a synchronous
+// checkpoint that aborts at the offending call (debug mode).
+#define LUX_POST_LAUNCH(stream)                                             \
+  do {                                                                      \
+    hipError_t e_ = hipGetLastError();                                      \
+    if (e_ != hipSuccess) {                                                 \
+      static int warned_ = 0;                                               \
+      if (!warned_++)                                                       \
+        fprintf(stderr,                                                     \
+                "[lux] warning: HIP error state at %s: %s (may be "         \
+                "pre-existing; set LUX_SYNC_CHECK=1 to localise)\n",        \
+                __func__, hipGetErrorString(e_));                           \
+    }                                                                       \
+    if (getenv("LUX_SYNC_CHECK")) {                                         \
+      e_ = hipStreamSynchronize((hipStream_t)(stream));                     \
+      if (e_ != hipSuccess) {                                               \
+        fprintf(stderr, "HIP async error %s:%d (%s): %s\n", __FILE__,       \
+                __LINE__, __func__, hipGetErrorString(e_));                 \
+        abort();                                                            \
+      }                                                                     \
+    }                                                                       \
+  } while (0)
+
 namespace lux {
 
 constexpr int WAVE = 64;

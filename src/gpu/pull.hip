@@ -42,7 +42,10 @@ __global__ void build_bins_kernel(V_ID vp, const E_ID* row_ptr,
     uint32_t lpos = 0, nchunks = 0, lpos_c = 0;
     if (v < vp) {
       E_ID deg = row_ptr[v + 1] - row_ptr[v];
-      if (deg < T1) {
+      if (deg == 0) {
+        // not binned: rows with no edges in this (block-)CSC are covered by
+        // the seed/finish pass, so sweeps touch only active rows
+      } else if (deg < T1) {
         my_bin = 0;
         lpos = atomicAdd(&lcnt[0], 1u);
       } else if (deg < T2) {
@@ -136,18 +139,19 @@ struct PullArgs {
                          // already in newv (src-blocked CSC iteration).
 };
 
-// Phased result store for the non-atomic bins: FIRST writes the raw partial
-// (labels seeded with the dst's own old label), MID folds into the existing
-// partial, LAST applies finish().
+// Result store for the non-atomic bins. The iteration contract: newv is
+// pre-seeded (PR: zeros; labels: the old label slice) before the sweep(s),
+// every sweep — blocked or not — folds its partial in, and PR's epilogue
+// runs once at the end (pull_finish_kernel). One writer per row per sweep,
+// so the fold needs no atomics.
 template <PullMode M>
 __device__ __forceinline__ void store_result(typename Val<M>::T* slot,
                                              typename Val<M>::T acc,
                                              typename Val<M>::T own,
                                              float init_rank, V_ID deg,
                                              int phase) {
-  if (!(phase & 1)) acc = Val<M>::comb(acc, *slot);
-  else if (M != PR_SUM) acc = Val<M>::comb(acc, own);
-  *slot = (phase & 2) ? finish<M>(acc, own, init_rank, deg) : acc;
+  (void)own; (void)init_rank; (void)deg; (void)phase;
+  *slot = Val<M>::comb(acc, *slot);
 }
 
 // ---- bin0: thread per vertex ----
@@ -193,25 +197,7 @@ __global__ void pull_wave_kernel(uint32_t n1, const V_ID* bin1, PullArgs a) {
   }
 }
 
-// ---- bin2: prep + chunk + epilogue ----
-template <PullMode M>
-__global__ void pull_prep_kernel(uint32_t nbig, const V_ID* bin2v,
-                                 PullArgs a) {
-  using V = Val<M>;
-  using T = typename V::T;
-  const T* oldv = (const T*)a.oldv;
-  T* newv = (T*)a.newv;
-  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < nbig) {
-    V_ID v = bin2v[i];
-    // PR accumulates raw sums (epilogue finishes); labels accumulate the
-    // final value directly, seeded with the dst's own old label.
-    newv[v] = (M == PR_SUM) ? V::ident()
-                            : Val<M>::comb(V::ident(),
-                                           oldv[a.row_left + v]);
-  }
-}
-
+// ---- bin2: chunk accumulation (hubs) ----
 template <PullMode M>
 __global__ void pull_chunk_kernel(uint32_t n2, const uint2* bin2,
                                   PullArgs a) {
@@ -245,20 +231,6 @@ __global__ void pull_chunk_kernel(uint32_t n2, const uint2* bin2,
   }
 }
 
-template <PullMode M>
-__global__ void pull_epilogue_kernel(uint32_t nbig, const V_ID* bin2v,
-                                     PullArgs a) {
-  using V = Val<M>;
-  using T = typename V::T;
-  const T* oldv = (const T*)a.oldv;
-  T* newv = (T*)a.newv;
-  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < nbig) {
-    V_ID v = bin2v[i];
-    newv[v] = finish<M>(newv[v], oldv[a.row_left + v], a.init_rank,
-                        a.deg ? a.deg[a.row_left + v] : 0);
-  }
-}
 
 template <PullMode M>
 static void pull_iter(hipStream_t s, uint32_t n0, const V_ID* bin0,
@@ -266,16 +238,9 @@ static void pull_iter(hipStream_t s, uint32_t n0, const V_ID* bin0,
                       const uint2* bin2, uint32_t nbig, const V_ID* bin2v,
                       const PullArgs& a) {
   if (nbig) {
-    if (a.phase & 1)
-      hipLaunchKernelGGL(pull_prep_kernel<M>, dim3(ceil_div_u32(nbig, BLOCK)),
-                         dim3(BLOCK), 0, s, nbig, bin2v, a);
     hipLaunchKernelGGL(pull_chunk_kernel<M>,
                        dim3(n2 > MAX_GRID ? MAX_GRID : n2), dim3(BLOCK), 0, s,
                        n2, bin2, a);
-    if (M == PR_SUM && (a.phase & 2))
-      hipLaunchKernelGGL(pull_epilogue_kernel<M>,
-                         dim3(ceil_div_u32(nbig, BLOCK)), dim3(BLOCK), 0, s,
-                         nbig, bin2v, a);
   }
   if (n1)
     hipLaunchKernelGGL(pull_wave_kernel<M>,
@@ -284,6 +249,21 @@ static void pull_iter(hipStream_t s, uint32_t n0, const V_ID* bin0,
   if (n0)
     hipLaunchKernelGGL(pull_thread_kernel<M>, dim3(grid_for(n0)), dim3(BLOCK),
                        0, s, n0, bin0, a);
+}
+
+// PR epilogue over the whole partition: pr = (1-a)/nv + a*sum, stored
+// divided by out-degree (pagerank_gpu.cu:97-100, :255-259). Covers rows
+// with no in-edges too (sum stays 0 from the seed).
+__global__ void pull_finish_pr_kernel(V_ID vp, float* newv,
+                                      const V_ID* deg, V_ID row_left,
+                                      float init_rank) {
+  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
+  for (uint64_t v = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; v < vp;
+       v += stride) {
+    float pr = init_rank + PR_ALPHA * newv[v];
+    V_ID d = deg[row_left + v];
+    newv[v] = d != 0 ? pr / (float)d : pr;
+  }
 }
 
 }  // namespace lux
@@ -300,6 +280,7 @@ void lux_gpu_build_bins(uint64_t stream, uint32_t vp, const E_ID* row_ptr,
   hipStream_t s = (hipStream_t)stream;
   hipLaunchKernelGGL(build_bins_kernel, dim3(grid_for(vp)), dim3(BLOCK), 0, s,
                      vp, row_ptr, bin0, bin1, bin2, bin2v, counters);
+  LUX_POST_LAUNCH(stream);
 }
 
 // mode: 0 = PageRank float-sum, 1 = u32 min (SSSP dense), 2 = u32 max (CC).
@@ -323,6 +304,16 @@ void lux_gpu_pull_iter(uint64_t stream, int mode, uint32_t n0,
       pull_iter<LAB_MAX>(s, n0, bin0, n1, bin1, n2, bin2, nbig, bin2v, a);
       break;
   }
+  LUX_POST_LAUNCH(stream);
+}
+
+void lux_gpu_pull_finish_pr(uint64_t stream, V_ID vp, float* newv,
+                            const V_ID* deg, V_ID row_left,
+                            float init_rank) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(pull_finish_pr_kernel, dim3(grid_for(vp)), dim3(BLOCK),
+                     0, s, vp, newv, deg, row_left, init_rank);
+  LUX_POST_LAUNCH(stream);
 }
 
 }  // extern "C"

@@ -272,6 +272,7 @@ void lux_gpu_csr_scatter(uint64_t stream, uint64_t ep, const V_ID* col,
   hipStream_t s = (hipStream_t)stream;
   hipLaunchKernelGGL(csr_scatter_kernel, dim3(grid_for(ep)), dim3(BLOCK), 0,
                      s, ep, col, row_ptr_loc, vp, row_left, cursor, push_col);
+  LUX_POST_LAUNCH(stream);
 }
 
 void lux_gpu_push_scatter(uint64_t stream, int is_min, int old_dense,
@@ -299,6 +300,7 @@ void lux_gpu_push_scatter(uint64_t stream, int is_min, int old_dense,
     case 6: launch(push_scatter_kernel<true, true, false>); break;
     case 7: launch(push_scatter_kernel<true, true, true>); break;
   }
+  LUX_POST_LAUNCH(stream);
 }
 
 void lux_gpu_build_bitmap(uint64_t stream, V_ID vp, const uint32_t* snapshot,
@@ -306,6 +308,7 @@ void lux_gpu_build_bitmap(uint64_t stream, V_ID vp, const uint32_t* snapshot,
   hipStream_t s = (hipStream_t)stream;
   hipLaunchKernelGGL(build_bitmap_kernel, dim3(grid_for((vp + 7) / 8)),
                      dim3(BLOCK), 0, s, vp, snapshot, new_labels, seg);
+  LUX_POST_LAUNCH(stream);
 }
 
 void lux_gpu_d2s(uint64_t stream, V_ID vp, V_ID row_left,
@@ -313,6 +316,7 @@ void lux_gpu_d2s(uint64_t stream, V_ID vp, V_ID row_left,
   hipStream_t s = (hipStream_t)stream;
   hipLaunchKernelGGL(d2s_kernel, dim3(grid_for(vp)), dim3(BLOCK), 0, s, vp,
                      row_left, dense_seg, sparse_seg);
+  LUX_POST_LAUNCH(stream);
 }
 
 void lux_gpu_check(uint64_t stream, int is_min, V_ID vp, V_ID row_left,
@@ -326,6 +330,7 @@ void lux_gpu_check(uint64_t stream, int is_min, V_ID vp, V_ID row_left,
     hipLaunchKernelGGL(check_kernel<false>, dim3(grid_for(vp)), dim3(BLOCK),
                        0, s, vp, row_left, row_ptr_loc, col, labels,
                        mistakes);
+  LUX_POST_LAUNCH(stream);
 }
 
 }  // extern "C"

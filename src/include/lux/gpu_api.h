@@ -40,6 +40,14 @@ void lux_gpu_edges_to_csc(uint64_t stream, uint32_t nv, uint64_t ne,
 void lux_gpu_local_row_ptr(uint64_t stream, uint32_t vp, lux::E_ID col_left,
                            const lux::E_ID* col_end_slice,
                            lux::E_ID* row_ptr_loc);
+void lux_gpu_blocked_count(uint64_t stream, uint64_t ep, const lux::V_ID* col,
+                           const lux::E_ID* row_ptr_loc, lux::V_ID vp,
+                           int shift, uint32_t* counts);
+void lux_gpu_blocked_scatter(uint64_t stream, uint64_t ep,
+                             const lux::V_ID* col,
+                             const lux::E_ID* row_ptr_loc, lux::V_ID vp,
+                             int shift, unsigned long long* cursor,
+                             lux::V_ID* out_col);
 
 // pull.hip
 void lux_gpu_build_bins(uint64_t stream, uint32_t vp,
@@ -53,7 +61,10 @@ void lux_gpu_pull_iter(uint64_t stream, int mode, uint32_t n0,
                        const lux::V_ID* bin2v, const lux::E_ID* row_ptr,
                        const lux::V_ID* col, const void* oldv, void* newv,
                        const lux::V_ID* deg, lux::V_ID row_left,
-                       float init_rank);
+                       float init_rank, int phase);
+void lux_gpu_pull_finish_pr(uint64_t stream, lux::V_ID vp, float* newv,
+                            const lux::V_ID* deg, lux::V_ID row_left,
+                            float init_rank);
 
 // push.hip
 void lux_gpu_csr_scatter(uint64_t stream, uint64_t ep, const lux::V_ID* col,

@@ -1,23 +1,15 @@
 #include "single_gpu.h"
 
+#include <algorithm>
 #include <cassert>
+#include <cmath>
 #include <cstdio>
 #include <cstring>
-
-#define CHECK_HIP(cmd)                                                      \
-  do {                                                                      \
-    hipError_t e_ = (cmd);                                                  \
-    if (e_ != hipSuccess) {                                                 \
-      fprintf(stderr, "HIP error %s:%d: %s\n", __FILE__, __LINE__,          \
-              hipGetErrorString(e_));                                       \
-      abort();                                                              \
-    }                                                                       \
-  } while (0)
 
 namespace lux {
 
 DeviceArena::DeviceArena(size_t bytes) : cap_(bytes) {
-  CHECK_HIP(hipMalloc(&base_, bytes));
+  LUX_OK(hipMalloc(&base_, bytes));
 }
 DeviceArena::~DeviceArena() { hipFree(base_); }
 void* DeviceArena::alloc(size_t bytes) {
@@ -38,17 +30,17 @@ DeviceGraph DeviceGraph::upload(const HostCSC& g, DeviceArena& arena,
   d.ne = g.ne;
   d.col_end = arena.alloc_n<E_ID>(g.nv);
   d.src = arena.alloc_n<V_ID>(g.ne);
-  CHECK_HIP(hipMemcpyAsync(d.col_end, g.col_end.data(),
+  LUX_OK(hipMemcpyAsync(d.col_end, g.col_end.data(),
                            sizeof(E_ID) * g.nv, hipMemcpyHostToDevice, s));
-  CHECK_HIP(hipMemcpyAsync(d.src, g.src.data(), sizeof(V_ID) * g.ne,
+  LUX_OK(hipMemcpyAsync(d.src, g.src.data(), sizeof(V_ID) * g.ne,
                            hipMemcpyHostToDevice, s));
   if (g.weighted()) {
     d.weight = arena.alloc_n<WeightType>(g.ne);
-    CHECK_HIP(hipMemcpyAsync(d.weight, g.weight.data(),
+    LUX_OK(hipMemcpyAsync(d.weight, g.weight.data(),
                              sizeof(WeightType) * g.ne,
                              hipMemcpyHostToDevice, s));
   }
-  CHECK_HIP(hipStreamSynchronize(s));
+  LUX_OK(hipStreamSynchronize(s));
   return d;
 }
 
@@ -63,17 +55,17 @@ DeviceGraph DeviceGraph::rmat(int scale, E_ID ne, uint64_t seed,
   V_ID *esrc, *edst;
   uint32_t* hist;
   unsigned long long *cursor, *partials;
-  CHECK_HIP(hipMalloc(&esrc, sizeof(V_ID) * ne));
-  CHECK_HIP(hipMalloc(&edst, sizeof(V_ID) * ne));
-  CHECK_HIP(hipMalloc(&hist, sizeof(uint32_t) * d.nv));
-  CHECK_HIP(hipMalloc(&cursor, sizeof(uint64_t) * d.nv));
-  CHECK_HIP(
+  LUX_OK(hipMalloc(&esrc, sizeof(V_ID) * ne));
+  LUX_OK(hipMalloc(&edst, sizeof(V_ID) * ne));
+  LUX_OK(hipMalloc(&hist, sizeof(uint32_t) * d.nv));
+  LUX_OK(hipMalloc(&cursor, sizeof(uint64_t) * d.nv));
+  LUX_OK(
       hipMalloc(&partials, sizeof(uint64_t) * lux_gpu_scan_partials_size(d.nv)));
-  CHECK_HIP(hipMemsetAsync(hist, 0, sizeof(uint32_t) * d.nv, s));
+  LUX_OK(hipMemsetAsync(hist, 0, sizeof(uint32_t) * d.nv, s));
   lux_gpu_rmat_edges((uint64_t)s, seed, scale, ne, esrc, edst);
   lux_gpu_edges_to_csc((uint64_t)s, d.nv, ne, esrc, edst, nullptr, d.col_end,
                        d.src, nullptr, hist, cursor, partials);
-  CHECK_HIP(hipStreamSynchronize(s));
+  LUX_OK(hipStreamSynchronize(s));
   hipFree(esrc);
   hipFree(edst);
   hipFree(hist);
@@ -91,13 +83,13 @@ void Bins::build(const E_ID* row_ptr_loc, V_ID vp, E_ID ep,
   bin2 = arena.alloc_n<lux_uint2>(n2_max);
   bin2v = arena.alloc_n<V_ID>(nbig_max);
   uint32_t* counters;
-  CHECK_HIP(hipMalloc(&counters, 4 * sizeof(uint32_t)));
-  CHECK_HIP(hipMemsetAsync(counters, 0, 16, s));
+  LUX_OK(hipMalloc(&counters, 4 * sizeof(uint32_t)));
+  LUX_OK(hipMemsetAsync(counters, 0, 16, s));
   lux_gpu_build_bins((uint64_t)s, vp, row_ptr_loc, bin0, bin1, bin2, bin2v,
                      counters);
   uint32_t c[4];
-  CHECK_HIP(hipMemcpyAsync(c, counters, 16, hipMemcpyDeviceToHost, s));
-  CHECK_HIP(hipStreamSynchronize(s));
+  LUX_OK(hipMemcpyAsync(c, counters, 16, hipMemcpyDeviceToHost, s));
+  LUX_OK(hipStreamSynchronize(s));
   hipFree(counters);
   n0 = c[0];
   n1 = c[1];
@@ -114,34 +106,36 @@ SingleGpuPagerank::SingleGpuPagerank(const DeviceGraph& g, DeviceArena& arena,
   lux_gpu_local_row_ptr((uint64_t)s, g.nv, 0, g.col_end, row_ptr_);
   bins_.build(row_ptr_, g.nv, g.ne, arena, s);
   deg_ = arena.alloc_n<V_ID>(g.nv);
-  CHECK_HIP(hipMemsetAsync(deg_, 0, sizeof(V_ID) * g.nv, s));
+  LUX_OK(hipMemsetAsync(deg_, 0, sizeof(V_ID) * g.nv, s));
   lux_gpu_hist_u32((uint64_t)s, g.ne, g.src, deg_);
   old_ = arena.alloc_n<float>(g.nv);
   new_ = arena.alloc_n<float>(g.nv);
   // init on host: rank/deg (pagerank_gpu.cu:255-259)
   std::vector<V_ID> hdeg(g.nv);
-  CHECK_HIP(hipMemcpyAsync(hdeg.data(), deg_, sizeof(V_ID) * g.nv,
+  LUX_OK(hipMemcpyAsync(hdeg.data(), deg_, sizeof(V_ID) * g.nv,
                            hipMemcpyDeviceToHost, s));
-  CHECK_HIP(hipStreamSynchronize(s));
+  LUX_OK(hipStreamSynchronize(s));
   std::vector<float> hpr(g.nv);
   float rank = 1.0f / g.nv;
   for (V_ID v = 0; v < g.nv; v++)
     hpr[v] = hdeg[v] == 0 ? rank : rank / hdeg[v];
-  CHECK_HIP(hipMemcpyAsync(old_, hpr.data(), sizeof(float) * g.nv,
+  LUX_OK(hipMemcpyAsync(old_, hpr.data(), sizeof(float) * g.nv,
                            hipMemcpyHostToDevice, s));
-  CHECK_HIP(hipStreamSynchronize(s));
+  LUX_OK(hipStreamSynchronize(s));
 }
 
 void SingleGpuPagerank::iterate(int iters) {
   float init_rank = (1.0f - PR_ALPHA) / g_.nv;
   for (int it = 0; it < iters; it++) {
+    LUX_OK(hipMemsetAsync(new_, 0, sizeof(float) * g_.nv, s_));
     lux_gpu_pull_iter((uint64_t)s_, 0, bins_.n0, bins_.bin0, bins_.n1,
                       bins_.bin1, bins_.n2, bins_.bin2, bins_.nbig,
                       bins_.bin2v, row_ptr_, g_.src, old_, new_, deg_, 0,
-                      init_rank);
+                      init_rank, 0);
+    lux_gpu_pull_finish_pr((uint64_t)s_, g_.nv, new_, deg_, 0, init_rank);
     std::swap(old_, new_);
   }
-  CHECK_HIP(hipStreamSynchronize(s_));
+  LUX_OK(hipStreamSynchronize(s_));
 }
 
 // ---------------- Push (SSSP / CC) ----------------
@@ -159,20 +153,20 @@ SingleGpuPush::SingleGpuPush(const DeviceGraph& g, bool is_min, V_ID source,
     uint32_t* degs;
     E_ID* ends;
     unsigned long long *cursor, *partials;
-    CHECK_HIP(hipMalloc(&degs, sizeof(uint32_t) * g.nv));
-    CHECK_HIP(hipMalloc(&ends, sizeof(E_ID) * g.nv));
-    CHECK_HIP(hipMalloc(&cursor, sizeof(uint64_t) * g.nv));
-    CHECK_HIP(hipMalloc(&partials,
+    LUX_OK(hipMalloc(&degs, sizeof(uint32_t) * g.nv));
+    LUX_OK(hipMalloc(&ends, sizeof(E_ID) * g.nv));
+    LUX_OK(hipMalloc(&cursor, sizeof(uint64_t) * g.nv));
+    LUX_OK(hipMalloc(&partials,
                         sizeof(uint64_t) * lux_gpu_scan_partials_size(g.nv)));
-    CHECK_HIP(hipMemsetAsync(degs, 0, sizeof(uint32_t) * g.nv, s));
+    LUX_OK(hipMemsetAsync(degs, 0, sizeof(uint32_t) * g.nv, s));
     lux_gpu_hist_u32((uint64_t)s, g.ne, g.src, degs);
     lux_gpu_scan_end_offsets((uint64_t)s, g.nv, degs, ends, partials);
     lux_gpu_local_row_ptr((uint64_t)s, g.nv, 0, ends, push_row_ptr_);
-    CHECK_HIP(hipMemcpyAsync(cursor, push_row_ptr_, sizeof(E_ID) * g.nv,
+    LUX_OK(hipMemcpyAsync(cursor, push_row_ptr_, sizeof(E_ID) * g.nv,
                              hipMemcpyDeviceToDevice, s));
     lux_gpu_csr_scatter((uint64_t)s, g.ne, g.src, row_ptr_, g.nv, 0, cursor,
                         push_col_);
-    CHECK_HIP(hipStreamSynchronize(s));
+    LUX_OK(hipStreamSynchronize(s));
     hipFree(degs);
     hipFree(ends);
     hipFree(cursor);
@@ -203,17 +197,17 @@ SingleGpuPush::SingleGpuPush(const DeviceGraph& g, bool is_min, V_ID source,
   }
   fq_type_ = hdr->type;
   fq_num_ = hdr->numNodes;
-  CHECK_HIP(hipMemcpyAsync(labels_, hl.data(), sizeof(uint32_t) * g.nv,
+  LUX_OK(hipMemcpyAsync(labels_, hl.data(), sizeof(uint32_t) * g.nv,
                            hipMemcpyHostToDevice, s));
-  CHECK_HIP(hipMemcpyAsync(fq_, hfq.data(), fq_bytes, hipMemcpyHostToDevice,
+  LUX_OK(hipMemcpyAsync(fq_, hfq.data(), fq_bytes, hipMemcpyHostToDevice,
                            s));
-  CHECK_HIP(hipStreamSynchronize(s));
+  LUX_OK(hipStreamSynchronize(s));
 }
 
 V_ID SingleGpuPush::step() {
-  CHECK_HIP(hipMemcpyAsync(snapshot_, labels_, sizeof(uint32_t) * g_.nv,
+  LUX_OK(hipMemcpyAsync(snapshot_, labels_, sizeof(uint32_t) * g_.nv,
                            hipMemcpyDeviceToDevice, s_));
-  CHECK_HIP(hipMemsetAsync(new_fq_, 0, sizeof(FrontierHeader), s_));
+  LUX_OK(hipMemsetAsync(new_fq_, 0, sizeof(FrontierHeader), s_));
   bool new_dense = fq_type_ == FrontierHeader::DENSE_BITMAP;
   bool pull_fallback = fq_num_ > g_.nv / SPARSE_THRESHOLD;
   if (pull_fallback) {
@@ -222,7 +216,7 @@ V_ID SingleGpuPush::step() {
     lux_gpu_pull_iter((uint64_t)s_, is_min_ ? 1 : 2, bins_.n0, bins_.bin0,
                       bins_.n1, bins_.bin1, bins_.n2, bins_.bin2, bins_.nbig,
                       bins_.bin2v, row_ptr_, g_.src, snapshot_, labels_,
-                      nullptr, 0, 0.0f);
+                      nullptr, 0, 0.0f, 0);
   } else {
     lux_gpu_push_scatter((uint64_t)s_, is_min_ ? 1 : 0,
                          fq_type_ == FrontierHeader::DENSE_BITMAP ? 1 : 0,
@@ -235,26 +229,26 @@ V_ID SingleGpuPush::step() {
   FrontierHeader hh;
   if (new_dense) {
     lux_gpu_build_bitmap((uint64_t)s_, g_.nv, snapshot_, labels_, new_fq_);
-    CHECK_HIP(hipMemcpyAsync(&hh, new_fq_, 8, hipMemcpyDeviceToHost, s_));
-    CHECK_HIP(hipStreamSynchronize(s_));
+    LUX_OK(hipMemcpyAsync(&hh, new_fq_, 8, hipMemcpyDeviceToHost, s_));
+    LUX_OK(hipStreamSynchronize(s_));
     if (hh.numNodes < capacity_) {
-      CHECK_HIP(hipMemcpyAsync(tmp_fq_, new_fq_, frontier_bytes(g_.nv),
+      LUX_OK(hipMemcpyAsync(tmp_fq_, new_fq_, frontier_bytes(g_.nv),
                                hipMemcpyDeviceToDevice, s_));
-      CHECK_HIP(hipMemsetAsync(new_fq_, 0, sizeof(FrontierHeader), s_));
+      LUX_OK(hipMemsetAsync(new_fq_, 0, sizeof(FrontierHeader), s_));
       lux_gpu_d2s((uint64_t)s_, g_.nv, 0, tmp_fq_, new_fq_);
-      CHECK_HIP(hipMemcpyAsync(&hh, new_fq_, 8, hipMemcpyDeviceToHost, s_));
-      CHECK_HIP(hipStreamSynchronize(s_));
+      LUX_OK(hipMemcpyAsync(&hh, new_fq_, 8, hipMemcpyDeviceToHost, s_));
+      LUX_OK(hipStreamSynchronize(s_));
       new_dense = false;
     }
   } else {
-    CHECK_HIP(hipMemcpyAsync(&hh, new_fq_, 8, hipMemcpyDeviceToHost, s_));
-    CHECK_HIP(hipStreamSynchronize(s_));
+    LUX_OK(hipMemcpyAsync(&hh, new_fq_, 8, hipMemcpyDeviceToHost, s_));
+    LUX_OK(hipStreamSynchronize(s_));
     if (hh.numNodes >= capacity_) {
       new_dense = true;
-      CHECK_HIP(hipMemsetAsync(new_fq_, 0, sizeof(FrontierHeader), s_));
+      LUX_OK(hipMemsetAsync(new_fq_, 0, sizeof(FrontierHeader), s_));
       lux_gpu_build_bitmap((uint64_t)s_, g_.nv, snapshot_, labels_, new_fq_);
-      CHECK_HIP(hipMemcpyAsync(&hh, new_fq_, 8, hipMemcpyDeviceToHost, s_));
-      CHECK_HIP(hipStreamSynchronize(s_));
+      LUX_OK(hipMemcpyAsync(&hh, new_fq_, 8, hipMemcpyDeviceToHost, s_));
+      LUX_OK(hipStreamSynchronize(s_));
     }
   }
   fq_type_ = new_dense ? FrontierHeader::DENSE_BITMAP
@@ -279,13 +273,13 @@ int SingleGpuPush::run(int max_iters) {
 
 uint64_t SingleGpuPush::check() {
   unsigned long long* mistakes;
-  CHECK_HIP(hipMalloc(&mistakes, 8));
-  CHECK_HIP(hipMemsetAsync(mistakes, 0, 8, s_));
+  LUX_OK(hipMalloc(&mistakes, 8));
+  LUX_OK(hipMemsetAsync(mistakes, 0, 8, s_));
   lux_gpu_check((uint64_t)s_, is_min_ ? 1 : 0, g_.nv, 0, row_ptr_, g_.src,
                 labels_, mistakes);
   unsigned long long h;
-  CHECK_HIP(hipMemcpyAsync(&h, mistakes, 8, hipMemcpyDeviceToHost, s_));
-  CHECK_HIP(hipStreamSynchronize(s_));
+  LUX_OK(hipMemcpyAsync(&h, mistakes, 8, hipMemcpyDeviceToHost, s_));
+  LUX_OK(hipStreamSynchronize(s_));
   hipFree(mistakes);
   return h;
 }
@@ -301,9 +295,9 @@ SingleGpuCF::SingleGpuCF(const DeviceGraph& g, int K, DeviceArena& arena,
   old_ = arena.alloc_n<float>((size_t)g.nv * K);
   new_ = arena.alloc_n<float>((size_t)g.nv * K);
   std::vector<float> h((size_t)g.nv * K, sqrtf(1.0f / K));
-  CHECK_HIP(hipMemcpyAsync(old_, h.data(), sizeof(float) * h.size(),
+  LUX_OK(hipMemcpyAsync(old_, h.data(), sizeof(float) * h.size(),
                            hipMemcpyHostToDevice, s));
-  CHECK_HIP(hipStreamSynchronize(s));
+  LUX_OK(hipStreamSynchronize(s));
 }
 
 void SingleGpuCF::iterate(int iters) {
@@ -313,7 +307,7 @@ void SingleGpuCF::iterate(int iters) {
                     g_.src, g_.weight, old_, new_, 0, K_);
     std::swap(old_, new_);
   }
-  CHECK_HIP(hipStreamSynchronize(s_));
+  LUX_OK(hipStreamSynchronize(s_));
 }
 
 }  // namespace lux

@@ -15,6 +15,16 @@
 #include "lux/gpu_api.h"
 #include "lux/types.h"
 
+#define LUX_OK(cmd)                                                         \
+  do {                                                                      \
+    hipError_t e_ = (cmd);                                                  \
+    if (e_ != hipSuccess) {                                                 \
+      fprintf(stderr, "HIP error %s:%d: %s\n", __FILE__, __LINE__,          \
+              hipGetErrorString(e_));                                       \
+      abort();                                                              \
+    }                                                                       \
+  } while (0)
+
 namespace lux {
 
 // FB memory pool: one upfront hipMalloc, bump allocation, 256-B alignment
