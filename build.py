@@ -68,21 +68,35 @@ def build_gpu(force=False):
 
 
 def build_apps(force=False):
-    """Lux-CLI-compatible standalone binaries (link the CPU lib; GPU path is
-    driven through the Python engine — see apps/*.cpp)."""
+    """Lux-CLI-compatible standalone binaries: tools/ are CPU (g++ +
+    liblux_cpu); apps/ are single-GPU native drivers (hipcc + runtime +
+    liblux_gpu). Multi-GPU runs go through the Python RCCL engine."""
     app_srcs = sorted(glob.glob(os.path.join(ROOT, "apps", "*.cpp")))
     tool_srcs = sorted(glob.glob(os.path.join(ROOT, "tools", "*.cpp")))
+    runtime = sorted(glob.glob(os.path.join(ROOT, "src", "runtime", "*.cpp")))
     bin_dir = os.path.join(ROOT, "bin")
     os.makedirs(bin_dir, exist_ok=True)
-    hdrs = glob.glob(os.path.join(INC, "lux", "*.h"))
+    hdrs = glob.glob(os.path.join(INC, "lux", "*.h")) + glob.glob(
+        os.path.join(ROOT, "src", "runtime", "*.h")) + glob.glob(
+        os.path.join(ROOT, "apps", "*.h"))
+    rpath = f"-Wl,-rpath,{os.path.dirname(CPU_LIB)}"
     out = []
-    for s in app_srcs + tool_srcs:
+    for s in tool_srcs:
         name = os.path.splitext(os.path.basename(s))[0]
         exe = os.path.join(bin_dir, name)
         if force or _newer(exe, [s, CPU_LIB] + hdrs):
-            _run(["g++", *CXXFLAGS, s, CPU_LIB, "-o", exe,
-                  f"-Wl,-rpath,{os.path.dirname(CPU_LIB)}"])
+            _run(["g++", *CXXFLAGS, s, CPU_LIB, "-o", exe, rpath])
         out.append(exe)
+    if os.path.exists(GPU_LIB):
+        for s in app_srcs:
+            name = os.path.splitext(os.path.basename(s))[0]
+            exe = os.path.join(bin_dir, name)
+            if force or _newer(exe, [s, CPU_LIB, GPU_LIB] + runtime + hdrs):
+                libdir = os.path.dirname(CPU_LIB)
+                _run([HIPCC, *HIPFLAGS, "-Wno-unused-value", s, *runtime,
+                      f"-L{libdir}", "-llux_cpu", "-llux_gpu", "-o", exe,
+                      rpath])
+            out.append(exe)
     return out
 
 

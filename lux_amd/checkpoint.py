@@ -1,0 +1,78 @@
+"""Vertex-state checkpoint/resume.
+
+The reference has no checkpointing; its only persistence is
+the .lux graph format. We add a vertex-state dump in the same binary-header
+style, so a long PageRank/CF run (or a converged SSSP/CC labelling) can be
+saved and resumed/analysed:
+
+    u32 magic 'LUXS' (0x5358554C) | u32 dtype (0=f32, 1=u32) | u32 K |
+    u32 nv | u64 iteration | payload nv*K elems
+
+Distributed use: each rank saves its slice via `save_state(..., row_left,
+data)` into per-rank files, or rank 0 saves the replicated array.
+"""
+import struct
+
+import numpy as np
+
+MAGIC = 0x5358554C
+_DTYPES = {0: np.float32, 1: np.uint32}
+_CODES = {np.dtype(np.float32): 0, np.dtype(np.uint32): 1,
+          np.dtype(np.int32): 1}
+
+
+def save_state(path, data, iteration=0):
+    """data: numpy array [nv] or [nv, K] (f32 or u32/i32)."""
+    arr = np.ascontiguousarray(data)
+    k = 1 if arr.ndim == 1 else arr.shape[1]
+    nv = arr.shape[0]
+    code = _CODES[arr.dtype]
+    with open(path, "wb") as f:
+        f.write(struct.pack("<IIIIQ", MAGIC, code, k, nv, iteration))
+        f.write(arr.tobytes())
+
+
+def load_state(path):
+    """Returns (array, iteration). Shape [nv] when K==1 else [nv, K]."""
+    with open(path, "rb") as f:
+        magic, code, k, nv, iteration = struct.unpack("<IIIIQ", f.read(24))
+        if magic != MAGIC:
+            raise IOError(f"{path}: not a lux state file")
+        arr = np.frombuffer(f.read(nv * k * 4), _DTYPES[code]).copy()
+    if k > 1:
+        arr = arr.reshape(nv, k)
+    return arr, iteration
+
+
+def save_engine(path, engine, iteration=None):
+    """Dump an engine's replicated state (rank 0 only in distributed runs)."""
+    from . import dist as dx
+    if dx.rank() != 0:
+        return
+    if hasattr(engine, "ranks"):  # PagerankEngine
+        save_state(path, engine.ranks().cpu().numpy(), iteration or 0)
+    elif hasattr(engine, "labels"):  # PushEngine
+        save_state(path, engine.labels.cpu().numpy().view(np.uint32),
+                   iteration if iteration is not None else engine.iterations)
+    elif hasattr(engine, "vectors"):  # CFEngine
+        save_state(path, engine.vectors().cpu().numpy(), iteration or 0)
+    else:
+        raise TypeError(f"unknown engine {type(engine)}")
+
+
+def resume_engine(path, engine):
+    """Load saved state back into an engine's replicated array (all ranks)."""
+    import torch
+    arr, iteration = load_state(path)
+    if hasattr(engine, "ranks"):
+        engine.old.copy_(torch.from_numpy(arr).to(engine.old.device))
+    elif hasattr(engine, "labels"):
+        t = torch.from_numpy(arr.view(np.int32)).to(engine.labels.device)
+        engine.labels.copy_(t)
+        p = engine.part
+        engine.labels_part.copy_(engine.labels.narrow(0, p.row_left, p.vp))
+        engine.iterations = iteration
+    elif hasattr(engine, "vectors"):
+        t = torch.from_numpy(arr.reshape(-1)).to(engine.old.device)
+        engine.old.copy_(t)
+    return iteration

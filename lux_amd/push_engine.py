@@ -214,6 +214,11 @@ class PushEngine:
         self.headers = [(int(hdr[2 * q]), int(hdr[2 * q + 1]))
                         for q in range(nparts)]
         self.iterations += 1
+        # per-iteration trace row (reference -verbose parity,
+        # sssp_gpu.cu:516-518: activeNodes + phase info per partition)
+        self.stats.append(dict(iter=self.iterations, old_frontier=old_fq_size,
+                               pull_fallback=bool(pull_fallback),
+                               out_dense=bool(new_dense), my_new=my_count))
         return sum(h[1] for h in self.headers)
 
     def run(self, max_iters=None):

@@ -1,0 +1,46 @@
+"""Checkpoint/resume + tracing tests (CPU)."""
+import numpy as np
+
+from lux_amd import checkpoint as ckpt
+from lux_amd.trace import IterTrace
+
+
+def test_state_roundtrip_f32(tmp_path):
+    p = str(tmp_path / "s.luxstate")
+    data = np.random.default_rng(1).random(1000).astype(np.float32)
+    ckpt.save_state(p, data, iteration=7)
+    got, it = ckpt.load_state(p)
+    assert it == 7
+    assert np.array_equal(got, data)
+
+
+def test_state_roundtrip_u32_2d(tmp_path):
+    p = str(tmp_path / "s2.luxstate")
+    data = np.arange(64 * 8, dtype=np.uint32).reshape(64, 8)
+    ckpt.save_state(p, data, iteration=3)
+    got, it = ckpt.load_state(p)
+    assert got.shape == (64, 8) and it == 3
+    assert np.array_equal(got, data)
+
+
+def test_bad_magic(tmp_path):
+    p = str(tmp_path / "bad")
+    open(p, "wb").write(b"\x00" * 64)
+    try:
+        ckpt.load_state(p)
+        assert False
+    except IOError:
+        pass
+
+
+def test_trace_csv_and_summary():
+    t = IterTrace()
+    t.record(iter=1, ms=2.0, frontier=10)
+    t.record(iter=2, ms=4.0, frontier=20)
+    csv_text = t.to_csv()
+    assert "frontier" in csv_text.splitlines()[0]
+    assert len(csv_text.splitlines()) == 3
+    s = t.summary(ne=6_000_000)
+    assert s["iterations"] == 2
+    assert s["ms_per_iter"] == 3.0
+    assert s["gteps"] == 2.0
