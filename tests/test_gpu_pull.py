@@ -113,15 +113,13 @@ def test_pagerank_multipart_single_process():
                       rank0 / degf.clamp(min=1.0))
     new = torch.empty_like(old)
     init_rank = (1 - 0.15) / nv
+    from lux_amd.engine import run_pull
     for _ in range(iters):
         for pt in parts:
             if pt.vp == 0:
                 continue
-            ng.pull_iter(stream(), ng.PULL_PR, pt.n0, pt.bin0, pt.n1,
-                         pt.bin1, pt.n2, pt.bin2, pt.nbig, pt.bin2v,
-                         pt.row_ptr, pt.col, old,
-                         new.narrow(0, pt.row_left, pt.vp), deg,
-                         pt.row_left, init_rank)
+            run_pull(pt, ng.PULL_PR, old,
+                     new.narrow(0, pt.row_left, pt.vp), deg, init_rank)
         old, new = new, old
     g = Graph.rmat(scale, ne, seed=13)
     want = cpu_ref.pagerank(g, iters)
