@@ -23,7 +23,9 @@ GPU_SRCS = sorted(glob.glob(os.path.join(ROOT, "src", "gpu", "*.hip")))
 CPU_LIB = os.path.join(ROOT, "lux_amd", "liblux_cpu.so")
 GPU_LIB = os.path.join(ROOT, "lux_amd", "liblux_gpu.so")
 
-CXXFLAGS = ["-O2", "-std=c++17", "-fPIC", "-Wall", "-fopenmp", f"-I{INC}"]
+# NOTE: no -fopenmp — loading libgomp before torch breaks ROCm device
+# detection in the same process (found via tests/ import bisection).
+CXXFLAGS = ["-O2", "-std=c++17", "-fPIC", "-Wall", f"-I{INC}"]
 HIPFLAGS = [
     "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC", f"-I{INC}",
     "-Wall", "-Wno-unused-function",
