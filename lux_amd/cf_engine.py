@@ -46,3 +46,53 @@ class CFEngine:
 
     def vectors(self):
         return self.old.view(self.part.nv, self.K)
+
+
+class CFALSEngine:
+    """ALS sweeps via MFMA Gram accumulation + per-wave Cholesky
+    (src/gpu/cf_als.hip). Same data movement and exchange as CFEngine;
+    an alternative optimizer that reaches the SGD fixed point in far fewer
+    sweeps. K <= 64 (the MFMA tile grid is 4x4 of 16x16)."""
+
+    def __init__(self, part: GraphPart, K=64):
+        assert part.weight is not None, "CF needs a weighted graph"
+        assert K <= 64, "ALS MFMA path covers K <= 64"
+        self.part = part
+        self.K = K
+        part.build_bins()
+        device = part.device
+        v0 = math.sqrt(1.0 / K)
+        self.old = torch.full((part.nv * K,), v0, dtype=F32, device=device)
+        self.new_part = torch.empty(part.vp * K, dtype=F32, device=device)
+        self.verts_elems = [v * K for v in part.verts_all]
+        self.left_elems = [l * K for l in part.row_left_all]
+        if part.nbig:
+            self.hubidx = torch.full((part.vp,), -1, dtype=torch.int32,
+                                     device=device)
+            hubs = part.bin2v[:part.nbig].long()
+            self.hubidx[hubs] = torch.arange(part.nbig, dtype=torch.int32,
+                                             device=device)
+            self.gram = torch.empty(part.nbig * 64 * 64, dtype=F32,
+                                    device=device)
+            self.rhs_h = torch.empty(part.nbig * 64, dtype=F32,
+                                     device=device)
+        else:
+            self.hubidx = self.gram = self.rhs_h = None
+
+    def step(self):
+        p = self.part
+        # seed with old slice: vertices with no in-edges keep their vector
+        self.new_part.copy_(
+            self.old.narrow(0, p.row_left * self.K, p.vp * self.K))
+        if p.nbig:
+            self.gram.zero_()
+            self.rhs_h.zero_()
+        ng.cf_als_iter(_stream(), p.n0, p.bin0, p.n1, p.bin1, p.n2, p.bin2,
+                       p.nbig, p.bin2v, self.hubidx, self.gram, self.rhs_h,
+                       p.row_ptr, p.col, p.weight, self.old, self.new_part,
+                       p.row_left, self.K)
+        dx.all_gather_slices(self.old, self.new_part, self.verts_elems,
+                             self.left_elems, my_index=p.p)
+
+    def vectors(self):
+        return self.old.view(self.part.nv, self.K)

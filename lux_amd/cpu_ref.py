@@ -82,3 +82,28 @@ def sssp_partitioned(g: Graph, nparts: int, source: int):
         if changed == 0:
             break
     return old, iters
+
+
+def cf_als(g: Graph, K: int, iters: int, lam: float = 0.001) -> np.ndarray:
+    """Plain-numpy ALS reference: per sweep, for each vertex with in-edges,
+    solve (S^T S + lam I) d = S^T w exactly against the OLD vectors (S =
+    src vectors of the in-edges). Vertices with no in-edges keep their old
+    vector. Same fixed point as the reference SGD sweep (cf.hip docstring);
+    solved in float64 here, compared with tolerance against the fp32 GPU
+    Cholesky path (src/gpu/cf_als.hip)."""
+    import math as _math
+    vec = np.full((g.nv, K), _math.sqrt(1.0 / K), dtype=np.float32)
+    eye = lam * np.eye(K, dtype=np.float64)
+    for _ in range(iters):
+        new = vec.copy()
+        b = 0
+        for v in range(g.nv):
+            e = int(g.col_end[v])
+            if e > b:
+                S = vec[g.src[b:e]].astype(np.float64)
+                w = g.weight[b:e].astype(np.float64)
+                G = S.T @ S + eye
+                new[v] = np.linalg.solve(G, S.T @ w).astype(np.float32)
+            b = e
+        vec = new
+    return vec

@@ -19,6 +19,14 @@ namespace lux {
 
 constexpr int CF_MAXC = 4;  // K <= 4*64
 
+// Wave-private LDS ordering: make this wave's ds_writes visible to its own
+// cross-lane ds_reads without a block barrier (waves in a block work on
+// independent vertices, so __syncthreads() would mismatch). Lockstep wave +
+// lgkmcnt(0) is sufficient; "memory" stops compiler reordering.
+__device__ __forceinline__ void wave_lds_sync() {
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+}
+
 struct CFArgs {
   const E_ID* row_ptr;     // u64[vp+1] local
   const V_ID* col;         // u32[ep]
@@ -74,6 +82,126 @@ __global__ void cf_wave_kernel(uint32_t n, const V_ID* binlist, CFArgs a) {
       int k = c * WAVE + lane;
       if (c < nc && k < a.K) out[k] += CF_GAMMA * acc[c];
     }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K<=64 fast path: LDS-tiled wave kernel.
+//
+// The generic wave kernel above serialises a ~500-cycle dependent chain per
+// edge (gather -> dot -> 6-step shuffle reduce -> broadcast -> axpy). Here a
+// wave stages CF_TILE=64 edges' src vectors in an LDS tile S[64][65] with a
+// coalesced lane=dim load, then switches lane mapping twice:
+//   dot phase    lane=edge: each lane walks its OWN row serially against the
+//                dst vector (LDS broadcast reads) -> 64 independent FMA
+//                chains instead of 64 serialised wave reduces;
+//   update phase lane=dim:  acc_k += err_e * S[e][k], err_e broadcast.
+// Row stride 65 words makes both access patterns bank-conflict-free
+// ((e+k) mod 32 distinct within each 32-lane group).
+// ---------------------------------------------------------------------------
+constexpr int CF_TILE = 64;
+constexpr int CF_ROW = CF_TILE + 1;
+constexpr int CF_TB = 64;  // one wave per workgroup: LDS 17.2 KB -> 9 wg/CU
+
+struct CFTileLds {
+  float S[CF_TILE * CF_ROW];
+  float dv[CF_TILE];
+  float err[CF_TILE];
+};
+
+// Stage up to CF_TILE edges [t, t+rem) into lds->S and per-lane col/weight;
+// returns this lane's weight (lane e holds edge t+e's weight).
+__device__ __forceinline__ float cf_stage_tile(CFTileLds* lds, E_ID t,
+                                               int rem, const CFArgs& a,
+                                               int lane) {
+  uint32_t mycol = 0;
+  float myw = 0.0f;
+  if (lane < rem) {
+    mycol = a.col[t + lane];
+    myw = (float)a.w[t + lane];
+  }
+  for (int r = 0; r < rem; r++) {
+    uint32_t src = __shfl(mycol, r, WAVE);
+    lds->S[r * CF_ROW + lane] =
+        lane < a.K ? a.oldv[(uint64_t)src * a.K + lane] : 0.0f;
+  }
+  return myw;
+}
+
+// dot+err+update over a staged tile; returns updated acc (lane=dim).
+__device__ __forceinline__ float cf_tile_pass(CFTileLds* lds, int rem,
+                                              float myw, int K, float acc,
+                                              int lane) {
+  wave_lds_sync();  // S rows visible
+  float d0 = 0, d1 = 0, d2 = 0, d3 = 0;
+  const float* row = &lds->S[lane * CF_ROW];
+  int k = 0;
+  for (; k + 3 < K; k += 4) {
+    d0 += row[k] * lds->dv[k];
+    d1 += row[k + 1] * lds->dv[k + 1];
+    d2 += row[k + 2] * lds->dv[k + 2];
+    d3 += row[k + 3] * lds->dv[k + 3];
+  }
+  for (; k < K; k++) d0 += row[k] * lds->dv[k];
+  float dot = (d0 + d1) + (d2 + d3);
+  lds->err[lane] = lane < rem ? myw - dot : 0.0f;
+  wave_lds_sync();  // err visible
+  float a0 = 0, a1 = 0, a2 = 0, a3 = 0;
+  int r = 0;
+  for (; r + 3 < rem; r += 4) {
+    a0 += lds->err[r] * lds->S[r * CF_ROW + lane];
+    a1 += lds->err[r + 1] * lds->S[(r + 1) * CF_ROW + lane];
+    a2 += lds->err[r + 2] * lds->S[(r + 2) * CF_ROW + lane];
+    a3 += lds->err[r + 3] * lds->S[(r + 3) * CF_ROW + lane];
+  }
+  for (; r < rem; r++) a0 += lds->err[r] * lds->S[r * CF_ROW + lane];
+  wave_lds_sync();  // tile consumed before the next stage overwrites it
+  return acc + ((a0 + a1) + (a2 + a3));
+}
+
+// one wave per dst vertex over a bin list (deg < T2)
+__global__ __launch_bounds__(CF_TB) void cf_tile_kernel(uint32_t n,
+                                                        const V_ID* binlist,
+                                                        CFArgs a) {
+  __shared__ CFTileLds lds;
+  int lane = threadIdx.x;
+  uint64_t nwaves = gridDim.x;
+  for (uint64_t i = blockIdx.x; i < n; i += nwaves) {
+    V_ID v = binlist[i];
+    E_ID b = a.row_ptr[v], e = a.row_ptr[v + 1];
+    lds.dv[lane] =
+        lane < a.K ? a.oldv[(uint64_t)(a.row_left + v) * a.K + lane] : 0.0f;
+    float acc = 0.0f;
+    for (E_ID t = b; t < e; t += CF_TILE) {
+      int rem = (int)(e - t < CF_TILE ? e - t : (E_ID)CF_TILE);
+      float myw = cf_stage_tile(&lds, t, rem, a, lane);
+      acc = cf_tile_pass(&lds, rem, myw, a.K, acc, lane);
+    }
+    if (lane < a.K) a.newv[(uint64_t)v * a.K + lane] += CF_GAMMA * acc;
+  }
+}
+
+// one wave per hub chunk (deg >= T2), atomic epilogue
+__global__ __launch_bounds__(CF_TB) void cf_tile_chunk_kernel(
+    uint32_t n2, const uint2* bin2, V_ID chunk_edges, CFArgs a) {
+  __shared__ CFTileLds lds;
+  int lane = threadIdx.x;
+  for (uint32_t i = blockIdx.x; i < n2; i += gridDim.x) {
+    uint2 ent = bin2[i];
+    V_ID v = ent.x;
+    E_ID b = a.row_ptr[v] + (E_ID)ent.y * chunk_edges;
+    E_ID e = a.row_ptr[v + 1];
+    if (e > b + chunk_edges) e = b + chunk_edges;
+    lds.dv[lane] =
+        lane < a.K ? a.oldv[(uint64_t)(a.row_left + v) * a.K + lane] : 0.0f;
+    float acc = 0.0f;
+    for (E_ID t = b; t < e; t += CF_TILE) {
+      int rem = (int)(e - t < CF_TILE ? e - t : (E_ID)CF_TILE);
+      float myw = cf_stage_tile(&lds, t, rem, a, lane);
+      acc = cf_tile_pass(&lds, rem, myw, a.K, acc, lane);
+    }
+    if (lane < a.K)
+      atomicAdd(&a.newv[(uint64_t)v * a.K + lane], CF_GAMMA * acc);
   }
 }
 
@@ -139,6 +267,22 @@ void lux_gpu_cf_iter(uint64_t stream, uint32_t n0, const V_ID* bin0,
                      V_ID row_left, int K) {
   hipStream_t s = (hipStream_t)stream;
   CFArgs a{row_ptr, col, w, oldv, newv, row_left, K};
+  if (K <= CF_TILE) {
+    // LDS-tiled fast path (benchmark config K=64; reference K=20)
+    if (nbig)
+      hipLaunchKernelGGL(cf_tile_chunk_kernel,
+                         dim3(n2 > MAX_GRID ? MAX_GRID : n2), dim3(CF_TB), 0,
+                         s, n2, bin2, (V_ID)8192, a);
+    if (n1)
+      hipLaunchKernelGGL(cf_tile_kernel,
+                         dim3(n1 > MAX_GRID ? MAX_GRID : n1), dim3(CF_TB), 0,
+                         s, n1, bin1, a);
+    if (n0)
+      hipLaunchKernelGGL(cf_wave_kernel, dim3(grid_for((uint64_t)n0 * WAVE)),
+                         dim3(BLOCK), 0, s, n0, bin0, a);
+    LUX_POST_LAUNCH(stream);
+    return;
+  }
   if (nbig) {
     hipLaunchKernelGGL(cf_chunk_kernel, dim3(n2 > MAX_GRID ? MAX_GRID : n2),
                        dim3(BLOCK), 0, s, n2, bin2, (V_ID)8192, a);

@@ -1,0 +1,291 @@
+// Collaborative filtering — ALS sweep with MFMA Gram accumulation (gfx950).
+//
+// A second CF optimizer beside the reference-parity SGD sweep (cf.hip;
+// reference colfilter_gpu.cu:32-104). One ALS sweep solves, per dst vertex,
+// the exact normal equations the SGD sweep only steps toward:
+//     (S^T S + lambda I) d = S^T w
+// where S (deg x K) stacks the OLD src vectors of v's in-edges and w the
+// edge weights. This is the SGD fixed point (grad = S^T(w - S d) - lambda d
+// = 0), so both optimizers share semantics; ALS just converges in far fewer
+// sweeps. Data access per sweep is identical to the SGD sweep (gather src
+// vectors over in-edges, publish my slice), so the distributed exchange
+// (all-gather of vector slices) is unchanged.
+//
+// MI355X mapping: the Gram matrix G = S^T S is the dense hot loop —
+// K^2 * deg MACs per vertex — and is exactly MFMA-shaped. One 64-lane wave
+// owns one dst vertex, stages 64-edge tiles of src vectors in LDS, and
+// accumulates the upper-triangular 16x16 tiles of G (K <= 64 -> 4x4 tile
+// grid, 10 upper tiles) with v_mfma_f32_16x16x4_f32: for each 4-edge group,
+// fragment f[t] = S[e0 + (lane>>4)][t*16 + (lane&15)] serves as operand A of
+// tile-row t AND operand B of tile-col t (A[i][k]=S[e0+k][16ti+i],
+// B[k][j]=S[e0+k][16tj+j] per the gfx950 f32 MFMA lane maps), so 4 LDS reads
+// feed 10 MFMAs. fp32-input MFMA runs at the fp32 vector rate on gfx950 but
+// packs the 64x64 outer-product accumulation into 40 accumulator VGPRs with
+// no cross-lane reduction — the vector formulation needs either 64 serial
+// wave reduces per 4 edges or an LDS round-trip per rank-1 update.
+// The 64x64 SPD solve (Cholesky + two triangular solves) runs in the same
+// wave on the LDS copy of G — no global scratch for the common case.
+// Hub vertices (deg >= T2, bin2 chunk lists from pull.hip) accumulate
+// partial G/rhs into global scratch with atomics, then a second kernel
+// solves per hub.
+#include "gpu_common.h"
+
+namespace lux {
+
+constexpr int ALS_K = 64;        // max latent rank of the MFMA path
+constexpr int ALS_ROW = 65;      // LDS row pitch (bank-conflict pad)
+constexpr int ALS_TILE = 64;     // edges staged per LDS tile
+constexpr int ALS_TB = 64;       // one wave per workgroup
+
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+__device__ __forceinline__ void als_lds_sync() {
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+}
+
+struct CFAlsArgs {
+  const E_ID* row_ptr;   // u64[vp+1] local
+  const V_ID* col;       // u32[ep]
+  const WeightType* w;   // i32[ep]
+  const float* oldv;     // f32[nv*K]
+  float* newv;           // f32[vp*K] (pre-seeded with old slice)
+  V_ID row_left;
+  int K;                 // <= 64
+};
+
+struct AlsLds {
+  float S[ALS_TILE * ALS_ROW];  // staged src vectors, row = edge
+  float G[ALS_K * ALS_ROW];     // Gram -> Cholesky factor (in place)
+  float W[ALS_TILE];            // staged edge weights
+};
+
+// ---- Gram accumulation over one vertex's edge range [b, e) ----
+// acc[10] are the upper tiles (ti<=tj) in order (0,0)(0,1)(0,2)(0,3)
+// (1,1)(1,2)(1,3)(2,2)(2,3)(3,3); rhs is lane=dim.
+__device__ __forceinline__ void als_gram_range(const CFAlsArgs& a, E_ID b,
+                                               E_ID e, int lane,
+                                               AlsLds* lds, f32x4 acc[10],
+                                               float* rhs) {
+  for (E_ID t = b; t < e; t += ALS_TILE) {
+    int rem = (int)(e - t < ALS_TILE ? e - t : (E_ID)ALS_TILE);
+    uint32_t mycol = 0;
+    float myw = 0.0f;
+    if (lane < rem) {
+      mycol = a.col[t + lane];
+      myw = (float)a.w[t + lane];
+    }
+    lds->W[lane] = myw;
+    for (int r = 0; r < rem; r++) {
+      uint32_t src = __shfl(mycol, r, WAVE);
+      lds->S[r * ALS_ROW + lane] =
+          lane < a.K ? a.oldv[(uint64_t)src * a.K + lane] : 0.0f;
+    }
+    int rem4 = (rem + 3) & ~3;
+    for (int r = rem; r < rem4; r++) lds->S[r * ALS_ROW + lane] = 0.0f;
+    als_lds_sync();
+    // rhs += sum_r w_r * S[r][lane]
+    float r0 = 0;
+    for (int r = 0; r < rem; r++)
+      r0 += lds->W[r] * lds->S[r * ALS_ROW + lane];
+    *rhs += r0;
+    // MFMA over 4-edge groups
+    int erow = lane >> 4, ecol = lane & 15;
+    for (int kk = 0; kk < rem4; kk += 4) {
+      const float* base = &lds->S[(kk + erow) * ALS_ROW + ecol];
+      float f0 = base[0], f1 = base[16], f2 = base[32], f3 = base[48];
+      acc[0] = __builtin_amdgcn_mfma_f32_16x16x4f32(f0, f0, acc[0], 0, 0, 0);
+      acc[1] = __builtin_amdgcn_mfma_f32_16x16x4f32(f0, f1, acc[1], 0, 0, 0);
+      acc[2] = __builtin_amdgcn_mfma_f32_16x16x4f32(f0, f2, acc[2], 0, 0, 0);
+      acc[3] = __builtin_amdgcn_mfma_f32_16x16x4f32(f0, f3, acc[3], 0, 0, 0);
+      acc[4] = __builtin_amdgcn_mfma_f32_16x16x4f32(f1, f1, acc[4], 0, 0, 0);
+      acc[5] = __builtin_amdgcn_mfma_f32_16x16x4f32(f1, f2, acc[5], 0, 0, 0);
+      acc[6] = __builtin_amdgcn_mfma_f32_16x16x4f32(f1, f3, acc[6], 0, 0, 0);
+      acc[7] = __builtin_amdgcn_mfma_f32_16x16x4f32(f2, f2, acc[7], 0, 0, 0);
+      acc[8] = __builtin_amdgcn_mfma_f32_16x16x4f32(f2, f3, acc[8], 0, 0, 0);
+      acc[9] = __builtin_amdgcn_mfma_f32_16x16x4f32(f3, f3, acc[9], 0, 0, 0);
+    }
+    als_lds_sync();  // tile consumed before next stage overwrites
+  }
+}
+
+// Scatter MFMA accumulators into the LDS Gram (mirroring the lower half)
+// using the 16x16x4 C/D lane map: element (row=(lane>>4)*4+reg, col=lane&15).
+__device__ __forceinline__ void als_dump_gram(AlsLds* lds, f32x4 acc[10],
+                                              int lane, int K) {
+  static constexpr int TI[10] = {0, 0, 0, 0, 1, 1, 1, 2, 2, 3};
+  static constexpr int TJ[10] = {0, 1, 2, 3, 1, 2, 3, 2, 3, 3};
+  int row = (lane >> 4) * 4, col = lane & 15;
+#pragma unroll
+  for (int tidx = 0; tidx < 10; tidx++) {
+#pragma unroll
+    for (int reg = 0; reg < 4; reg++) {
+      int gr = TI[tidx] * 16 + row + reg;
+      int gc = TJ[tidx] * 16 + col;
+      float v = acc[tidx][reg];
+      lds->G[gr * ALS_ROW + gc] = v;
+      if (TI[tidx] != TJ[tidx]) lds->G[gc * ALS_ROW + gr] = v;
+    }
+  }
+  als_lds_sync();
+  // regularizer + identity padding for unused dims (keeps G SPD for K<64)
+  lds->G[lane * ALS_ROW + lane] =
+      lane < K ? lds->G[lane * ALS_ROW + lane] + CF_LAMBDA : 1.0f;
+  als_lds_sync();
+}
+
+// In-place lower Cholesky of the 64x64 LDS matrix (one wave; lane = row).
+__device__ __forceinline__ void wave_cholesky64(float* G, int lane) {
+  for (int k = 0; k < ALS_K; k++) {
+    float dkk = sqrtf(G[k * ALS_ROW + k]);
+    float lik = lane > k ? G[lane * ALS_ROW + k] / dkk : 0.0f;
+    if (lane == k) G[k * ALS_ROW + k] = dkk;
+    if (lane > k) G[lane * ALS_ROW + k] = lik;
+    als_lds_sync();
+    for (int j = k + 1; j <= lane; j++)
+      G[lane * ALS_ROW + j] -= lik * G[j * ALS_ROW + k];
+    als_lds_sync();
+  }
+}
+
+// Solve L L^T d = rhs; rhs/result live lane=dim in a register.
+__device__ __forceinline__ float wave_spd_solve64(const float* G, float r,
+                                                  int lane) {
+  for (int k = 0; k < ALS_K; k++) {  // forward: L y = r
+    float yk = __shfl(r, k, WAVE) / G[k * ALS_ROW + k];
+    if (lane == k) r = yk;
+    else if (lane > k) r -= G[lane * ALS_ROW + k] * yk;
+  }
+  for (int k = ALS_K - 1; k >= 0; k--) {  // backward: L^T d = y
+    float dk = __shfl(r, k, WAVE) / G[k * ALS_ROW + k];
+    if (lane == k) r = dk;
+    else if (lane < k) r -= G[k * ALS_ROW + lane] * dk;
+  }
+  return r;
+}
+
+// ---- one wave per vertex: gram + solve fused (deg < T2 bin lists) ----
+__global__ __launch_bounds__(ALS_TB) void cf_als_solve_kernel(
+    uint32_t n, const V_ID* binlist, CFAlsArgs a) {
+  __shared__ AlsLds lds;
+  int lane = threadIdx.x;
+  for (uint64_t i = blockIdx.x; i < n; i += gridDim.x) {
+    V_ID v = binlist[i];
+    E_ID b = a.row_ptr[v], e = a.row_ptr[v + 1];
+    f32x4 acc[10];
+#pragma unroll
+    for (int t = 0; t < 10; t++) acc[t] = {0, 0, 0, 0};
+    float rhs = 0.0f;
+    als_gram_range(a, b, e, lane, &lds, acc, &rhs);
+    als_dump_gram(&lds, acc, lane, a.K);
+    wave_cholesky64(lds.G, lane);
+    float d = wave_spd_solve64(lds.G, rhs, lane);
+    if (lane < a.K) a.newv[(uint64_t)v * a.K + lane] = d;
+    als_lds_sync();  // G reads done before next vertex's MFMA dump
+  }
+}
+
+// ---- hub path: chunk-parallel Gram into global scratch ----
+// gram_scratch: f32[nbig * 64 * 64] (upper-triangular elements only),
+// rhs_scratch: f32[nbig * 64]; both pre-zeroed by the engine each sweep.
+__global__ __launch_bounds__(ALS_TB) void cf_als_gram_chunk_kernel(
+    uint32_t n2, const uint2* bin2, V_ID chunk_edges, const int* hubidx,
+    float* gram_scratch, float* rhs_scratch, CFAlsArgs a) {
+  __shared__ AlsLds lds;  // G part unused here
+  int lane = threadIdx.x;
+  for (uint32_t i = blockIdx.x; i < n2; i += gridDim.x) {
+    uint2 ent = bin2[i];
+    V_ID v = ent.x;
+    E_ID b = a.row_ptr[v] + (E_ID)ent.y * chunk_edges;
+    E_ID e = a.row_ptr[v + 1];
+    if (e > b + chunk_edges) e = b + chunk_edges;
+    f32x4 acc[10];
+#pragma unroll
+    for (int t = 0; t < 10; t++) acc[t] = {0, 0, 0, 0};
+    float rhs = 0.0f;
+    als_gram_range(a, b, e, lane, &lds, acc, &rhs);
+    int idx = hubidx[v];
+    float* Gg = gram_scratch + (uint64_t)idx * ALS_K * ALS_K;
+    atomicAdd(&rhs_scratch[(uint64_t)idx * ALS_K + lane], rhs);
+    static constexpr int TI[10] = {0, 0, 0, 0, 1, 1, 1, 2, 2, 3};
+    static constexpr int TJ[10] = {0, 1, 2, 3, 1, 2, 3, 2, 3, 3};
+    int row = (lane >> 4) * 4, col = lane & 15;
+#pragma unroll
+    for (int tidx = 0; tidx < 10; tidx++) {
+#pragma unroll
+      for (int reg = 0; reg < 4; reg++) {
+        int gr = TI[tidx] * 16 + row + reg;
+        int gc = TJ[tidx] * 16 + col;
+        if (gc >= gr) atomicAdd(&Gg[gr * ALS_K + gc], acc[tidx][reg]);
+      }
+    }
+  }
+}
+
+// ---- hub path: per-vertex solve from global scratch ----
+__global__ __launch_bounds__(ALS_TB) void cf_als_hub_solve_kernel(
+    uint32_t nbig, const V_ID* bin2v, const float* gram_scratch,
+    const float* rhs_scratch, CFAlsArgs a) {
+  __shared__ AlsLds lds;
+  int lane = threadIdx.x;
+  for (uint32_t i = blockIdx.x; i < nbig; i += gridDim.x) {
+    V_ID v = bin2v[i];
+    const float* Gg = gram_scratch + (uint64_t)i * ALS_K * ALS_K;
+    for (int r = 0; r < ALS_K; r++) {
+      float x = Gg[r * ALS_K + lane];
+      lds.G[r * ALS_ROW + lane] = x;  // upper half valid (lane >= r)
+    }
+    als_lds_sync();
+    for (int r = 0; r < ALS_K; r++)  // mirror upper -> lower
+      if (lane > r) lds.G[lane * ALS_ROW + r] = lds.G[r * ALS_ROW + lane];
+    als_lds_sync();
+    lds.G[lane * ALS_ROW + lane] =
+        lane < a.K ? lds.G[lane * ALS_ROW + lane] + CF_LAMBDA : 1.0f;
+    float rhs = rhs_scratch[(uint64_t)i * ALS_K + lane];
+    als_lds_sync();
+    wave_cholesky64(lds.G, lane);
+    float d = wave_spd_solve64(lds.G, rhs, lane);
+    if (lane < a.K) a.newv[(uint64_t)v * a.K + lane] = d;
+    als_lds_sync();
+  }
+}
+
+}  // namespace lux
+
+using namespace lux;
+
+extern "C" {
+
+// One ALS sweep over my partition. Reuses the pull.hip degree-bin lists;
+// hubidx maps a hub vertex (local id) to its scratch slot [0, nbig).
+// gram_scratch/rhs_scratch must be zeroed before each sweep when nbig > 0.
+void lux_gpu_cf_als_iter(uint64_t stream, uint32_t n0, const V_ID* bin0,
+                         uint32_t n1, const V_ID* bin1, uint32_t n2,
+                         const uint2* bin2, uint32_t nbig, const V_ID* bin2v,
+                         const int* hubidx, float* gram_scratch,
+                         float* rhs_scratch, const E_ID* row_ptr,
+                         const V_ID* col, const WeightType* w,
+                         const float* oldv, float* newv, V_ID row_left,
+                         int K) {
+  hipStream_t s = (hipStream_t)stream;
+  CFAlsArgs a{row_ptr, col, w, oldv, newv, row_left, K};
+  if (nbig) {
+    hipLaunchKernelGGL(cf_als_gram_chunk_kernel,
+                       dim3(n2 > MAX_GRID ? MAX_GRID : n2), dim3(ALS_TB), 0,
+                       s, n2, bin2, (V_ID)8192, hubidx, gram_scratch,
+                       rhs_scratch, a);
+    hipLaunchKernelGGL(cf_als_hub_solve_kernel,
+                       dim3(nbig > MAX_GRID ? MAX_GRID : nbig), dim3(ALS_TB),
+                       0, s, nbig, bin2v, gram_scratch, rhs_scratch, a);
+  }
+  if (n1)
+    hipLaunchKernelGGL(cf_als_solve_kernel,
+                       dim3(n1 > MAX_GRID ? MAX_GRID : n1), dim3(ALS_TB), 0,
+                       s, n1, bin1, a);
+  if (n0)
+    hipLaunchKernelGGL(cf_als_solve_kernel,
+                       dim3(n0 > MAX_GRID ? MAX_GRID : n0), dim3(ALS_TB), 0,
+                       s, n0, bin0, a);
+  LUX_POST_LAUNCH(stream);
+}
+
+}  // extern "C"

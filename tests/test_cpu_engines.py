@@ -164,3 +164,28 @@ def test_cf_loss_decreases():
     v5 = cpu_ref.cf(g, K, 5)
     l5 = cpu_ref.cf_loss(g, K, v5)
     assert l5 < l0
+
+
+def test_cf_als_reference_beats_sgd():
+    """The ALS normal-equations reference reaches lower loss than the same
+    number of SGD sweeps, from the same init (it solves each sweep's
+    subproblem exactly)."""
+    from lux_amd.graph import Graph
+    g = Graph.bipartite(300, 60, 6000, seed=21)
+    K, sweeps = 16, 2
+    sgd = cpu_ref.cf(g, K, sweeps)
+    als = cpu_ref.cf_als(g, K, sweeps)
+    assert cpu_ref.cf_loss(g, K, als) < cpu_ref.cf_loss(g, K, sgd)
+
+
+def test_cf_als_deg0_keeps_vector():
+    import math
+    from lux_amd.graph import Graph
+    g = Graph.bipartite(50, 10, 300, seed=5)
+    K = 8
+    out = cpu_ref.cf_als(g, K, 1)
+    deg = np.diff(np.concatenate([[0], g.col_end]))
+    v0 = math.sqrt(1.0 / K)
+    want = np.full(K, v0, dtype=np.float32)
+    for v in np.nonzero(deg == 0)[0][:5]:
+        np.testing.assert_array_equal(out[v], want)

@@ -66,3 +66,53 @@ def test_rmat_folded_gpu_matches_cpu():
     assert np.array_equal(gs.cpu().numpy().view(np.uint32), cs)
     assert np.array_equal(gd.cpu().numpy().view(np.uint32), cd)
     assert cs.max() < nv and cd.max() < nv
+
+
+@pytest.mark.parametrize("K", [20, 64])
+def test_cf_als_vs_numpy(K):
+    """MFMA ALS sweep (cf_als.hip) vs the float64 numpy normal-equations
+    reference: one sweep from the common init must agree to fp32-Cholesky
+    tolerance."""
+    from lux_amd.cf_engine import CFALSEngine
+    nu, ni, ne = 400, 100, 15000
+    full = DeviceCSC.bipartite(nu, ni, ne, seed=11)
+    part = GraphPart(full, 1, 0)
+    eng = CFALSEngine(part, K=K)
+    for _ in range(2):
+        eng.step()
+    got = eng.vectors().cpu().numpy()
+    g = Graph.bipartite(nu, ni, ne, seed=11)
+    want = cpu_ref.cf_als(g, K, 2)
+    np.testing.assert_allclose(got, want, rtol=3e-3, atol=3e-4)
+
+
+def test_cf_als_hub_path():
+    """Extreme-degree items exercise the chunked Gram + hub-solve path."""
+    from lux_amd.cf_engine import CFALSEngine
+    nu, ni, ne = 2000, 4, 60000
+    full = DeviceCSC.bipartite(nu, ni, ne, seed=13)
+    part = GraphPart(full, 1, 0)
+    part.build_bins()
+    assert part.nbig > 0
+    eng = CFALSEngine(part, K=64)
+    eng.step()
+    got = eng.vectors().cpu().numpy()
+    g = Graph.bipartite(nu, ni, ne, seed=13)
+    want = cpu_ref.cf_als(g, 64, 1)
+    np.testing.assert_allclose(got, want, rtol=3e-3, atol=3e-4)
+
+
+def test_cf_als_beats_sgd_loss():
+    """ALS reaches a lower loss than the same number of SGD sweeps."""
+    from lux_amd.cf_engine import CFALSEngine
+    nu, ni, ne, K = 5000, 512, 200000, 64
+    full = DeviceCSC.bipartite(nu, ni, ne, seed=9)
+    g = Graph.bipartite(nu, ni, ne, seed=9)
+    sgd = CFEngine(GraphPart(full, 1, 0), K=K)
+    als = CFALSEngine(GraphPart(full, 1, 0), K=K)
+    for _ in range(3):
+        sgd.step()
+        als.step()
+    l_sgd = cpu_ref.cf_loss(g, K, sgd.vectors().cpu().numpy())
+    l_als = cpu_ref.cf_loss(g, K, als.vectors().cpu().numpy())
+    assert l_als < l_sgd
