@@ -33,6 +33,7 @@ def apply_app_defaults(args):
         "sssp": dict(),                          # RMAT-27, 2^31 edges
         "cc": dict(edges=1468365182, nv=41652230),   # Twitter-2010-shaped
         "cf": dict(edges=200961014, nv=497959),      # NetFlix-shaped
+        "cf_als": dict(edges=200961014, nv=497959),  # NetFlix-shaped, MFMA
     }
     d = defaults[args.app]
     if args.edges == (1 << 31) and "edges" in d:
@@ -51,8 +52,9 @@ def build_engine(args, device):
     elif args.app == "sssp":
         from lux_amd.apps.sssp import build_sssp_bench
         return build_sssp_bench(args, device)
-    elif args.app == "cf":
+    elif args.app in ("cf", "cf_als"):
         from lux_amd.apps.cf import build_cf_bench
+        args.als = args.app == "cf_als"
         return build_cf_bench(args, device)
     raise ValueError(args.app)
 
@@ -120,7 +122,8 @@ def main():
                 "graph": {"pagerank": f"rmat{args.scale}",
                           "sssp": f"rmat{args.scale}",
                           "cc": "twitter2010-shaped",
-                          "cf": "netflix-shaped"}[args.app],
+                          "cf": "netflix-shaped",
+                          "cf_als": "netflix-shaped"}[args.app],
                 "nv": args.nv,
                 "ne": args.edges,
                 "parallelism": f"graph-partition x{world}",

@@ -3,7 +3,7 @@ col_filter/colfilter.cc; SGD matrix-factorization sweeps, rank -k)."""
 import sys
 
 from .. import dist as dx
-from ..cf_engine import CFEngine
+from ..cf_engine import CFALSEngine, CFEngine
 from ..engine import GraphPart
 from .common import (ElapsedTimer, load_device_graph, parse_input_args,
                      print_memory_estimate)
@@ -19,7 +19,8 @@ def build_cf_bench(args, device):
     full = DeviceCSC.bipartite(NETFLIX_USERS, NETFLIX_ITEMS, ne,
                                seed=args.seed, device=device)
     part = GraphPart(full, dx.world_size(), dx.rank())
-    return CFEngine(part, K=64), part
+    cls = CFALSEngine if getattr(args, "als", False) else CFEngine
+    return cls(part, K=64), part
 
 
 def main(argv=None):
@@ -34,7 +35,7 @@ def main(argv=None):
         print_memory_estimate(full.nv, full.ne, dx.world_size(),
                               weighted=True, k=a.k)
     part = GraphPart(full, dx.world_size(), dx.rank())
-    eng = CFEngine(part, K=a.k)
+    eng = (CFALSEngine if a.als else CFEngine)(part, K=a.k)
     with ElapsedTimer():
         for _ in range(a.num_iter):
             eng.step()

@@ -19,6 +19,7 @@ class AppArgs:
         self.verbose = False
         self.check = False
         self.k = 64
+        self.als = False  # col_filter: ALS (MFMA) optimizer instead of SGD
         self.synthetic = None  # e.g. "rmat:20:1000000"
 
 
@@ -41,6 +42,8 @@ def parse_input_args(argv):
             a.check = True; i += 1
         elif f == "-k":
             a.k = int(argv[i + 1]); i += 2
+        elif f == "-als":
+            a.als = True; i += 1
         elif f == "-synthetic":
             a.synthetic = argv[i + 1]; i += 2
         elif f.startswith("-ll:") or f.startswith("-lg:"):
