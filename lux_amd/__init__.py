@@ -7,6 +7,15 @@ xGMI, one process per GPU.
 import os
 import sys
 
+# torch MUST be imported (and its bundled ROCm runtime stack loaded) before
+# any other native library in this process: loading liblux_cpu.so / numpy's
+# BLAS before torch breaks ROCm device detection process-wide ("no
+# ROCm-capable device is detected" from HIP while torch.cuda still works) —
+# bisected twice on MI355X boxes (gpurun_scripts/bisect_imports.sh,
+# bisect2.sh). Importing torch here guarantees the order for every user of
+# the package.
+import torch  # noqa: F401
+
 # repo root importable (for build.py lazy builds)
 _ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 if _ROOT not in sys.path:

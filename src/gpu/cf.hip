@@ -111,6 +111,13 @@ struct CFTileLds {
 
 // Stage up to CF_TILE edges [t, t+rem) into lds->S and per-lane col/weight;
 // returns this lane's weight (lane e holds edge t+e's weight).
+//
+// The load loop MUST be statically unrolled in depth-16 batches: with a
+// dynamic trip count the compiler keeps it rolled and each ds_write waits
+// for its own global load (one outstanding ~500-cycle LLC access per edge —
+// measured 35.9 ms/sweep on the NetFlix config, WORSE than the un-tiled v1).
+// Batching 16 loads into registers before the LDS writes keeps 16 reads in
+// flight per wave.
 __device__ __forceinline__ float cf_stage_tile(CFTileLds* lds, E_ID t,
                                                int rem, const CFArgs& a,
                                                int lane) {
@@ -120,12 +127,65 @@ __device__ __forceinline__ float cf_stage_tile(CFTileLds* lds, E_ID t,
     mycol = a.col[t + lane];
     myw = (float)a.w[t + lane];
   }
-  for (int r = 0; r < rem; r++) {
+  if (rem == CF_TILE) {
+#pragma unroll
+    for (int r0 = 0; r0 < CF_TILE; r0 += 16) {
+      float tmp[16];
+#pragma unroll
+      for (int r = 0; r < 16; r++) {
+        uint32_t src = __shfl(mycol, r0 + r, WAVE);
+        tmp[r] = lane < a.K ? a.oldv[(uint64_t)src * a.K + lane] : 0.0f;
+      }
+#pragma unroll
+      for (int r = 0; r < 16; r++)
+        lds->S[(r0 + r) * CF_ROW + lane] = tmp[r];
+    }
+    return myw;
+  }
+  int r = 0;
+  for (; r + 4 <= rem; r += 4) {
+    float tmp[4];
+#pragma unroll
+    for (int q = 0; q < 4; q++) {
+      uint32_t src = __shfl(mycol, r + q, WAVE);
+      tmp[q] = lane < a.K ? a.oldv[(uint64_t)src * a.K + lane] : 0.0f;
+    }
+#pragma unroll
+    for (int q = 0; q < 4; q++) lds->S[(r + q) * CF_ROW + lane] = tmp[q];
+  }
+  for (; r < rem; r++) {
     uint32_t src = __shfl(mycol, r, WAVE);
     lds->S[r * CF_ROW + lane] =
         lane < a.K ? a.oldv[(uint64_t)src * a.K + lane] : 0.0f;
   }
   return myw;
+}
+
+// Full-tile (rem==64), K==64 pass: every loop statically unrolled.
+__device__ __forceinline__ float cf_tile_pass_fast(CFTileLds* lds, float myw,
+                                                   float acc, int lane) {
+  wave_lds_sync();
+  const float* row = &lds->S[lane * CF_ROW];
+  float d0 = 0, d1 = 0, d2 = 0, d3 = 0;
+#pragma unroll
+  for (int k = 0; k < CF_TILE; k += 4) {
+    d0 += row[k] * lds->dv[k];
+    d1 += row[k + 1] * lds->dv[k + 1];
+    d2 += row[k + 2] * lds->dv[k + 2];
+    d3 += row[k + 3] * lds->dv[k + 3];
+  }
+  lds->err[lane] = myw - ((d0 + d1) + (d2 + d3));
+  wave_lds_sync();
+  float a0 = 0, a1 = 0, a2 = 0, a3 = 0;
+#pragma unroll
+  for (int r = 0; r < CF_TILE; r += 4) {
+    a0 += lds->err[r] * lds->S[r * CF_ROW + lane];
+    a1 += lds->err[r + 1] * lds->S[(r + 1) * CF_ROW + lane];
+    a2 += lds->err[r + 2] * lds->S[(r + 2) * CF_ROW + lane];
+    a3 += lds->err[r + 3] * lds->S[(r + 3) * CF_ROW + lane];
+  }
+  wave_lds_sync();
+  return acc + ((a0 + a1) + (a2 + a3));
 }
 
 // dot+err+update over a staged tile; returns updated acc (lane=dim).
@@ -159,6 +219,25 @@ __device__ __forceinline__ float cf_tile_pass(CFTileLds* lds, int rem,
   return acc + ((a0 + a1) + (a2 + a3));
 }
 
+// accumulate gradient over edge range [b, e); lane = dim
+__device__ __forceinline__ float cf_range_acc(CFTileLds* lds, E_ID b, E_ID e,
+                                              const CFArgs& a, int lane) {
+  float acc = 0.0f;
+  E_ID t = b;
+  if (a.K == CF_TILE) {
+    for (; t + CF_TILE <= e; t += CF_TILE) {
+      float myw = cf_stage_tile(lds, t, CF_TILE, a, lane);
+      acc = cf_tile_pass_fast(lds, myw, acc, lane);
+    }
+  }
+  for (; t < e; t += CF_TILE) {
+    int rem = (int)(e - t < CF_TILE ? e - t : (E_ID)CF_TILE);
+    float myw = cf_stage_tile(lds, t, rem, a, lane);
+    acc = cf_tile_pass(lds, rem, myw, a.K, acc, lane);
+  }
+  return acc;
+}
+
 // one wave per dst vertex over a bin list (deg < T2)
 __global__ __launch_bounds__(CF_TB) void cf_tile_kernel(uint32_t n,
                                                         const V_ID* binlist,
@@ -171,12 +250,7 @@ __global__ __launch_bounds__(CF_TB) void cf_tile_kernel(uint32_t n,
     E_ID b = a.row_ptr[v], e = a.row_ptr[v + 1];
     lds.dv[lane] =
         lane < a.K ? a.oldv[(uint64_t)(a.row_left + v) * a.K + lane] : 0.0f;
-    float acc = 0.0f;
-    for (E_ID t = b; t < e; t += CF_TILE) {
-      int rem = (int)(e - t < CF_TILE ? e - t : (E_ID)CF_TILE);
-      float myw = cf_stage_tile(&lds, t, rem, a, lane);
-      acc = cf_tile_pass(&lds, rem, myw, a.K, acc, lane);
-    }
+    float acc = cf_range_acc(&lds, b, e, a, lane);
     if (lane < a.K) a.newv[(uint64_t)v * a.K + lane] += CF_GAMMA * acc;
   }
 }
@@ -194,12 +268,7 @@ __global__ __launch_bounds__(CF_TB) void cf_tile_chunk_kernel(
     if (e > b + chunk_edges) e = b + chunk_edges;
     lds.dv[lane] =
         lane < a.K ? a.oldv[(uint64_t)(a.row_left + v) * a.K + lane] : 0.0f;
-    float acc = 0.0f;
-    for (E_ID t = b; t < e; t += CF_TILE) {
-      int rem = (int)(e - t < CF_TILE ? e - t : (E_ID)CF_TILE);
-      float myw = cf_stage_tile(&lds, t, rem, a, lane);
-      acc = cf_tile_pass(&lds, rem, myw, a.K, acc, lane);
-    }
+    float acc = cf_range_acc(&lds, b, e, a, lane);
     if (lane < a.K)
       atomicAdd(&a.newv[(uint64_t)v * a.K + lane], CF_GAMMA * acc);
   }

@@ -75,19 +75,43 @@ __device__ __forceinline__ void als_gram_range(const CFAlsArgs& a, E_ID b,
       myw = (float)a.w[t + lane];
     }
     lds->W[lane] = myw;
-    for (int r = 0; r < rem; r++) {
-      uint32_t src = __shfl(mycol, r, WAVE);
-      lds->S[r * ALS_ROW + lane] =
-          lane < a.K ? a.oldv[(uint64_t)src * a.K + lane] : 0.0f;
+    // depth-16 batched loads: a dynamic-trip rolled loop would leave one
+    // outstanding global read per edge (see cf.hip cf_stage_tile note)
+    if (rem == ALS_TILE) {
+#pragma unroll
+      for (int r0 = 0; r0 < ALS_TILE; r0 += 16) {
+        float tmp[16];
+#pragma unroll
+        for (int r = 0; r < 16; r++) {
+          uint32_t src = __shfl(mycol, r0 + r, WAVE);
+          tmp[r] = lane < a.K ? a.oldv[(uint64_t)src * a.K + lane] : 0.0f;
+        }
+#pragma unroll
+        for (int r = 0; r < 16; r++)
+          lds->S[(r0 + r) * ALS_ROW + lane] = tmp[r];
+      }
+    } else {
+      for (int r = 0; r < rem; r++) {
+        uint32_t src = __shfl(mycol, r, WAVE);
+        lds->S[r * ALS_ROW + lane] =
+            lane < a.K ? a.oldv[(uint64_t)src * a.K + lane] : 0.0f;
+      }
     }
     int rem4 = (rem + 3) & ~3;
     for (int r = rem; r < rem4; r++) lds->S[r * ALS_ROW + lane] = 0.0f;
     als_lds_sync();
-    // rhs += sum_r w_r * S[r][lane]
-    float r0 = 0;
-    for (int r = 0; r < rem; r++)
-      r0 += lds->W[r] * lds->S[r * ALS_ROW + lane];
-    *rhs += r0;
+    // rhs += sum_r w_r * S[r][lane] (4 independent LDS-read chains)
+    float r0 = 0, r1 = 0, r2 = 0, r3 = 0;
+    int rr = 0;
+    for (; rr + 4 <= rem; rr += 4) {
+      r0 += lds->W[rr] * lds->S[rr * ALS_ROW + lane];
+      r1 += lds->W[rr + 1] * lds->S[(rr + 1) * ALS_ROW + lane];
+      r2 += lds->W[rr + 2] * lds->S[(rr + 2) * ALS_ROW + lane];
+      r3 += lds->W[rr + 3] * lds->S[(rr + 3) * ALS_ROW + lane];
+    }
+    for (; rr < rem; rr++)
+      r0 += lds->W[rr] * lds->S[rr * ALS_ROW + lane];
+    *rhs += ((r0 + r1) + (r2 + r3));
     // MFMA over 4-edge groups
     int erow = lane >> 4, ecol = lane & 15;
     for (int kk = 0; kk < rem4; kk += 4) {

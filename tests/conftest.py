@@ -1,6 +1,10 @@
 import os
 import sys
 
+# Load torch's ROCm runtime stack before any test module imports lux_amd /
+# numpy native libs (see lux_amd/__init__.py note: other load orders break
+# HIP device detection process-wide on the GPU boxes).
+import torch  # noqa: F401
 import pytest
 
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
