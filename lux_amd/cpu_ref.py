@@ -84,7 +84,8 @@ def sssp_partitioned(g: Graph, nparts: int, source: int):
     return old, iters
 
 
-def cf_als(g: Graph, K: int, iters: int, lam: float = 0.001) -> np.ndarray:
+def cf_als(g: Graph, K: int, iters: int, lam: float = 0.001,
+           init: "np.ndarray | None" = None) -> np.ndarray:
     """Plain-numpy ALS reference: per sweep, for each vertex with in-edges,
     solve (S^T S + lam I) d = S^T w exactly against the OLD vectors (S =
     src vectors of the in-edges). Vertices with no in-edges keep their old
@@ -92,7 +93,10 @@ def cf_als(g: Graph, K: int, iters: int, lam: float = 0.001) -> np.ndarray:
     solved in float64 here, compared with tolerance against the fp32 GPU
     Cholesky path (src/gpu/cf_als.hip)."""
     import math as _math
-    vec = np.full((g.nv, K), _math.sqrt(1.0 / K), dtype=np.float32)
+    if init is not None:
+        vec = np.array(init, dtype=np.float32).reshape(g.nv, K).copy()
+    else:
+        vec = np.full((g.nv, K), _math.sqrt(1.0 / K), dtype=np.float32)
     eye = lam * np.eye(K, dtype=np.float64)
     for _ in range(iters):
         new = vec.copy()

@@ -68,26 +68,39 @@ def test_rmat_folded_gpu_matches_cpu():
     assert cs.max() < nv and cd.max() < nv
 
 
+def _rand_init(nv, K, seed):
+    """Well-conditioned random init. The reference's constant init
+    (sqrt(1/K) everywhere) makes every sweep-1 Gram matrix rank-1 with
+    cond ~ deg/lambda (up to 1e7) — there the fp32 GPU Cholesky and the
+    f64 numpy reference legitimately diverge, so accuracy tests run from
+    the realistic regime instead (loss tests keep the parity init)."""
+    rng = np.random.default_rng(seed)
+    return (rng.standard_normal((nv, K)) * 0.1 + 0.1).astype(np.float32)
+
+
 @pytest.mark.parametrize("K", [20, 64])
 def test_cf_als_vs_numpy(K):
     """MFMA ALS sweep (cf_als.hip) vs the float64 numpy normal-equations
-    reference: one sweep from the common init must agree to fp32-Cholesky
-    tolerance."""
+    reference from a common random init."""
+    import torch
     from lux_amd.cf_engine import CFALSEngine
     nu, ni, ne = 400, 100, 15000
     full = DeviceCSC.bipartite(nu, ni, ne, seed=11)
     part = GraphPart(full, 1, 0)
     eng = CFALSEngine(part, K=K)
+    init = _rand_init(part.nv, K, seed=42)
+    eng.old.copy_(torch.from_numpy(init.ravel()))
     for _ in range(2):
         eng.step()
     got = eng.vectors().cpu().numpy()
     g = Graph.bipartite(nu, ni, ne, seed=11)
-    want = cpu_ref.cf_als(g, K, 2)
+    want = cpu_ref.cf_als(g, K, 2, init=init)
     np.testing.assert_allclose(got, want, rtol=3e-3, atol=3e-4)
 
 
 def test_cf_als_hub_path():
     """Extreme-degree items exercise the chunked Gram + hub-solve path."""
+    import torch
     from lux_amd.cf_engine import CFALSEngine
     nu, ni, ne = 2000, 4, 60000
     full = DeviceCSC.bipartite(nu, ni, ne, seed=13)
@@ -95,20 +108,51 @@ def test_cf_als_hub_path():
     part.build_bins()
     assert part.nbig > 0
     eng = CFALSEngine(part, K=64)
+    init = _rand_init(part.nv, 64, seed=43)
+    eng.old.copy_(torch.from_numpy(init.ravel()))
     eng.step()
     got = eng.vectors().cpu().numpy()
     g = Graph.bipartite(nu, ni, ne, seed=13)
-    want = cpu_ref.cf_als(g, 64, 1)
+    want = cpu_ref.cf_als(g, 64, 1, init=init)
     np.testing.assert_allclose(got, want, rtol=3e-3, atol=3e-4)
 
 
+def test_cf_als_normal_equation_residual():
+    """From the reference's constant init (worst conditioning: sweep-1 Gram
+    is rank-1 + lambda I), the fp32 solve must still satisfy its own normal
+    equations to fp32 backward error — the condition-independent check."""
+    from lux_amd.cf_engine import CFALSEngine
+    nu, ni, ne, K = 400, 100, 15000, 64
+    full = DeviceCSC.bipartite(nu, ni, ne, seed=11)
+    part = GraphPart(full, 1, 0, keep_full=True)
+    eng = CFALSEngine(part, K=K)
+    old = eng.vectors().cpu().numpy().copy()
+    eng.step()
+    new = eng.vectors().cpu().numpy()
+    g = Graph.bipartite(nu, ni, ne, seed=11)
+    lam = 0.001
+    b = 0
+    for v in range(g.nv):
+        e = int(g.col_end[v])
+        if e > b:
+            S = old[g.src[b:e]].astype(np.float64)
+            w = g.weight[b:e].astype(np.float64)
+            G = S.T @ S + lam * np.eye(K)
+            r = G @ new[v] - S.T @ w
+            scale = np.linalg.norm(G, ord=np.inf) * np.linalg.norm(new[v]) \
+                + np.linalg.norm(S.T @ w) + 1e-30
+            assert np.linalg.norm(r) / scale < 5e-5, f"vertex {v}"
+        b = e
+
+
 def test_cf_als_beats_sgd_loss():
-    """ALS reaches a lower loss than the same number of SGD sweeps."""
+    """ALS reaches a lower loss than the same number of SGD sweeps (both
+    from the reference parity init)."""
     from lux_amd.cf_engine import CFALSEngine
     nu, ni, ne, K = 5000, 512, 200000, 64
     full = DeviceCSC.bipartite(nu, ni, ne, seed=9)
     g = Graph.bipartite(nu, ni, ne, seed=9)
-    sgd = CFEngine(GraphPart(full, 1, 0), K=K)
+    sgd = CFEngine(GraphPart(full, 1, 0, keep_full=True), K=K)
     als = CFALSEngine(GraphPart(full, 1, 0), K=K)
     for _ in range(3):
         sgd.step()
