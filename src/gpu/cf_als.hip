@@ -53,10 +53,14 @@ struct CFAlsArgs {
   int K;                 // <= 64
 };
 
-struct AlsLds {
+struct AlsGramLds {
   float S[ALS_TILE * ALS_ROW];  // staged src vectors, row = edge
-  float G[ALS_K * ALS_ROW];     // Gram -> Cholesky factor (in place)
   float W[ALS_TILE];            // staged edge weights
+};
+
+struct AlsLds {
+  AlsGramLds g;                 // staging (S/W)
+  float G[ALS_K * ALS_ROW];     // Gram -> Cholesky factor (in place)
 };
 
 // ---- Gram accumulation over one vertex's edge range [b, e) ----
@@ -64,8 +68,8 @@ struct AlsLds {
 // (1,1)(1,2)(1,3)(2,2)(2,3)(3,3); rhs is lane=dim.
 __device__ __forceinline__ void als_gram_range(const CFAlsArgs& a, E_ID b,
                                                E_ID e, int lane,
-                                               AlsLds* lds, f32x4 acc[10],
-                                               float* rhs) {
+                                               AlsGramLds* lds,
+                                               f32x4 acc[10], float* rhs) {
   for (E_ID t = b; t < e; t += ALS_TILE) {
     int rem = (int)(e - t < ALS_TILE ? e - t : (E_ID)ALS_TILE);
     uint32_t mycol = 0;
@@ -199,7 +203,7 @@ __global__ __launch_bounds__(ALS_TB) void cf_als_solve_kernel(
 #pragma unroll
     for (int t = 0; t < 10; t++) acc[t] = {0, 0, 0, 0};
     float rhs = 0.0f;
-    als_gram_range(a, b, e, lane, &lds, acc, &rhs);
+    als_gram_range(a, b, e, lane, &lds.g, acc, &rhs);
     als_dump_gram(&lds, acc, lane, a.K);
     wave_cholesky64(lds.G, lane);
     float d = wave_spd_solve64(lds.G, rhs, lane);
@@ -211,12 +215,17 @@ __global__ __launch_bounds__(ALS_TB) void cf_als_solve_kernel(
 // ---- hub path: chunk-parallel Gram into global scratch ----
 // gram_scratch: f32[nbig * 64 * 64] (upper-triangular elements only),
 // rhs_scratch: f32[nbig * 64]; both pre-zeroed by the engine each sweep.
-__global__ __launch_bounds__(ALS_TB) void cf_als_gram_chunk_kernel(
+constexpr int ALS_CHUNK_TB = 128;  // 2 independent waves per workgroup
+
+__global__ __launch_bounds__(ALS_CHUNK_TB) void cf_als_gram_chunk_kernel(
     uint32_t n2, const uint2* bin2, V_ID chunk_edges, const int* hubidx,
     float* gram_scratch, float* rhs_scratch, CFAlsArgs a) {
-  __shared__ AlsLds lds;  // G part unused here
-  int lane = threadIdx.x;
-  for (uint32_t i = blockIdx.x; i < n2; i += gridDim.x) {
+  __shared__ AlsGramLds lds2[ALS_CHUNK_TB / WAVE];
+  AlsGramLds& lds = lds2[threadIdx.x >> 6];
+  int lane = threadIdx.x & (WAVE - 1);
+  uint32_t wave = (blockIdx.x * (ALS_CHUNK_TB / WAVE)) + (threadIdx.x >> 6);
+  uint32_t nwaves = gridDim.x * (ALS_CHUNK_TB / WAVE);
+  for (uint32_t i = wave; i < n2; i += nwaves) {
     uint2 ent = bin2[i];
     V_ID v = ent.x;
     E_ID b = a.row_ptr[v] + (E_ID)ent.y * chunk_edges;
@@ -293,10 +302,11 @@ void lux_gpu_cf_als_iter(uint64_t stream, uint32_t n0, const V_ID* bin0,
   hipStream_t s = (hipStream_t)stream;
   CFAlsArgs a{row_ptr, col, w, oldv, newv, row_left, K};
   if (nbig) {
+    uint32_t gw = (n2 + 1) / 2;
     hipLaunchKernelGGL(cf_als_gram_chunk_kernel,
-                       dim3(n2 > MAX_GRID ? MAX_GRID : n2), dim3(ALS_TB), 0,
-                       s, n2, bin2, (V_ID)8192, hubidx, gram_scratch,
-                       rhs_scratch, a);
+                       dim3(gw > MAX_GRID ? MAX_GRID : gw),
+                       dim3(ALS_CHUNK_TB), 0, s, n2, bin2, (V_ID)8192,
+                       hubidx, gram_scratch, rhs_scratch, a);
     hipLaunchKernelGGL(cf_als_hub_solve_kernel,
                        dim3(nbig > MAX_GRID ? MAX_GRID : nbig), dim3(ALS_TB),
                        0, s, nbig, bin2v, gram_scratch, rhs_scratch, a);
