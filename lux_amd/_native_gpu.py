@@ -30,9 +30,11 @@ def lib():
 
 
 def dp(t):
-    """torch tensor -> device pointer (as c_void_p)."""
+    """torch tensor (or raw int device address) -> c_void_p."""
     if t is None:
         return ctypes.c_void_p(0)
+    if isinstance(t, int):
+        return ctypes.c_void_p(t)
     return ctypes.c_void_p(t.data_ptr())
 
 
@@ -147,17 +149,17 @@ def pull_iter(stream, mode, n0, bin0, n1, bin1, n2, bin2, nbig, bin2v,
                             ctypes.c_float(init_rank), ctypes.c_int(phase))
 
 
-def blocked_count(stream, ep, col, row_ptr_loc, vp, shift, counts):
+def blocked_count(stream, ep, col, row_ptr_loc, vp, bounds, nb, counts):
     lib().lux_gpu_blocked_count(_u64(stream), _u64(ep), dp(col),
-                                dp(row_ptr_loc), _u32(vp),
-                                ctypes.c_int(shift), dp(counts))
+                                dp(row_ptr_loc), _u32(vp), dp(bounds),
+                                ctypes.c_int(nb), dp(counts))
 
 
-def blocked_scatter(stream, ep, col, row_ptr_loc, vp, shift, cursor,
+def blocked_scatter(stream, ep, col, row_ptr_loc, vp, bounds, nb, cursor,
                     out_col):
     lib().lux_gpu_blocked_scatter(_u64(stream), _u64(ep), dp(col),
-                                  dp(row_ptr_loc), _u32(vp),
-                                  ctypes.c_int(shift), dp(cursor),
+                                  dp(row_ptr_loc), _u32(vp), dp(bounds),
+                                  ctypes.c_int(nb), dp(cursor),
                                   dp(out_col))
 
 

@@ -82,6 +82,29 @@ def all_gather_slices(full, my_slice, verts, row_left, my_index=None):
                            input_split_sizes=in_splits)
 
 
+class _DoneWork:
+    def wait(self):
+        return True
+
+
+def all_gather_slices_async(full, my_slice, verts, row_left, my_index=None):
+    """Async all_gather_slices: returns a Work-like handle whose .wait()
+    makes the CURRENT stream wait for completion (NCCL semantics — a
+    GPU-side dependency, not a host block). The pipelined pull engines
+    launch this right after computing their slice and overlap the xGMI
+    exchange with the next iteration's rank-local sweep."""
+    ws = world_size()
+    if ws == 1:
+        all_gather_slices(full, my_slice, verts, row_left, my_index)
+        return _DoneWork()
+    inp = my_slice.repeat(ws)
+    in_splits = [my_slice.numel()] * ws
+    return dist.all_to_all_single(full, inp,
+                                  output_split_sizes=list(verts),
+                                  input_split_sizes=in_splits,
+                                  async_op=True)
+
+
 def all_reduce_sum_(t):
     if dist.is_initialized():
         dist.all_reduce(t, op=dist.ReduceOp.SUM)

@@ -47,32 +47,49 @@ def _bins_for(row_ptr, vp, ep, device, compact=False):
     return n0, n1, n2, nbig, bin0, bin1, bin2, bin2v
 
 
-def run_pull(part, mode, oldv, newv, deg, init_rank, seed=True):
-    """One pull iteration over this rank's partition. Contract: seed newv
-    (PR: zeros, labels: the old label slice), fold-sweep the (blocked) CSC,
-    then for PR apply the epilogue once (pull_finish_pr). Uses the
-    src-blocked CSC (LLC-resident gathers) when built."""
+def run_pull_sweeps(part, mode, oldv, newv, deg, init_rank, subset=None):
+    """Fold-sweep the (blocked) CSC into newv. `oldv` may be a tensor or a
+    raw device address (int) — the pipelined engines sweep the rank-LOCAL
+    src block against `cur_part.data_ptr() - row_left*4` while the
+    all-gather of remote slices is still in flight. subset: None = all
+    blocks, "local" / "remote" = blocks inside / outside this rank's own
+    vertex range (requires the rank-aligned blocked build)."""
     s = _stream()
-    if seed:
-        if mode == ng.PULL_PR:
-            newv.zero_()
-        else:
-            newv.copy_(oldv.narrow(0, part.row_left, part.vp))
     blocks = getattr(part, "blocks", None)
     if blocks:
-        nb = len(blocks)
-        for i, blk in enumerate(blocks):
-            phase = (1 if i == 0 else 0) | (2 if i == nb - 1 else 0)
+        for blk in blocks:
+            if subset == "local" and not blk["local"]:
+                continue
+            if subset == "remote" and blk["local"]:
+                continue
             ng.pull_iter(s, mode, blk["n0"], blk["bin0"], blk["n1"],
                          blk["bin1"], blk["n2"], blk["bin2"], blk["nbig"],
                          blk["bin2v"], blk["row_ptr"], blk["col"], oldv,
-                         newv, deg, part.row_left, init_rank, phase)
+                         newv, deg, part.row_left, init_rank, 0)
     else:
+        assert subset is None, "subset sweeps need the blocked CSC"
         ng.pull_iter(s, mode, part.n0, part.bin0, part.n1, part.bin1,
                      part.n2, part.bin2, part.nbig, part.bin2v, part.row_ptr,
                      part.col, oldv, newv, deg, part.row_left, init_rank, 0)
+
+
+def seed_pull(part, mode, oldv, newv):
     if mode == ng.PULL_PR:
-        ng.pull_finish_pr(s, part.vp, newv, deg, part.row_left, init_rank)
+        newv.zero_()
+    else:
+        newv.copy_(oldv.narrow(0, part.row_left, part.vp))
+
+
+def run_pull(part, mode, oldv, newv, deg, init_rank, seed=True):
+    """One full synchronous pull iteration over this rank's partition.
+    Contract: seed newv (PR: zeros, labels: the old label slice), fold-sweep
+    the (blocked) CSC, then for PR apply the epilogue once."""
+    if seed:
+        seed_pull(part, mode, oldv, newv)
+    run_pull_sweeps(part, mode, oldv, newv, deg, init_rank)
+    if mode == ng.PULL_PR:
+        ng.pull_finish_pr(_stream(), part.vp, newv, deg, part.row_left,
+                          init_rank)
 
 
 def partition_bounds(col_end, ne, nparts):
@@ -209,30 +226,45 @@ class GraphPart:
          self.bin2, self.bin2v) = _bins_for(self.row_ptr, self.vp, self.ep,
                                             self.device)
 
+    def pull_bounds(self, shift):
+        """Src-block boundary list: 2^shift-sized windows (LLC residency),
+        with this rank's own [row_left, row_right+1) range aligned on
+        boundaries when partitioned — the pipelined engines sweep the
+        rank-local block(s) while the all-gather of remote slices flies."""
+        bounds = set(range(0, self.nv, 1 << shift)) | {self.nv}
+        if self.nparts > 1 and self.vp > 0:
+            bounds |= {self.row_left, self.row_right + 1}
+        return sorted(bounds)
+
     def prepare_pull(self, force_shift=None):
-        """Build degree bins and, when the gather window (nv*4 B) exceeds
-        the 256 MiB Infinity Cache, the src-blocked CSC (edges regrouped by
-        src>>shift so each sweep's random gathers stay LLC-resident)."""
+        """Build degree bins and the src-blocked CSC where it pays: when
+        the gather window (nv*4 B) exceeds the 256 MiB Infinity Cache
+        (LLC-resident sweeps), or when partitioned (rank-local block for
+        comm/compute overlap)."""
         self.build_bins()
         if getattr(self, "blocks", None) is not None:
             return
         shift = force_shift if force_shift is not None else LLC_BLOCK_SHIFT
-        if force_shift is None and (self.nv <= (1 << shift) or self.ep == 0):
+        bounds = self.pull_bounds(shift)
+        if self.ep == 0 or len(bounds) <= 2:
             self.blocks = None
             return
-        self.build_blocked(shift)
+        if force_shift is None and self.nv <= (1 << shift) \
+                and self.nparts == 1:
+            self.blocks = None
+            return
+        self.build_blocked(bounds)
 
-    def build_blocked(self, shift):
+    def build_blocked(self, bounds):
         device = self.device
         s = _stream()
         vp, ep = self.vp, self.ep
-        sb = (self.nv + (1 << shift) - 1) >> shift
-        if sb <= 1:
-            self.blocks = None
-            return
+        sb = len(bounds) - 1
+        bounds_t = torch.tensor(bounds, dtype=U32, device=device)
         n = sb * vp
         counts = torch.zeros(n, dtype=U32, device=device)
-        ng.blocked_count(s, ep, self.col, self.row_ptr, vp, shift, counts)
+        ng.blocked_count(s, ep, self.col, self.row_ptr, vp, bounds_t, sb + 1,
+                         counts)
         ends = torch.empty(n, dtype=U64, device=device)
         partials = torch.empty(ng.scan_partials_size(n), dtype=U64,
                                device=device)
@@ -240,8 +272,8 @@ class GraphPart:
         cursor = torch.empty(n + 1, dtype=U64, device=device)
         ng.local_row_ptr(s, n, 0, ends, cursor)
         blk_col = torch.empty(max(ep, 1), dtype=U32, device=device)
-        ng.blocked_scatter(s, ep, self.col, self.row_ptr, vp, shift, cursor,
-                           blk_col)
+        ng.blocked_scatter(s, ep, self.col, self.row_ptr, vp, bounds_t,
+                           sb + 1, cursor, blk_col)
         del counts, partials, cursor
         self.blocks = []
         begin = 0
@@ -253,9 +285,12 @@ class GraphPart:
             col_b = blk_col.narrow(0, begin, max(end - begin, 1))
             n0, n1, n2, nbig, b0, b1, b2, b2v = _bins_for(
                 row_ptr_b, vp, end - begin, device, compact=True)
+            local = self.nparts > 1 and bounds[b] >= self.row_left \
+                and bounds[b + 1] <= self.row_right + 1
             self.blocks.append(dict(row_ptr=row_ptr_b, col=col_b, n0=n0,
                                     n1=n1, n2=n2, nbig=nbig, bin0=b0,
-                                    bin1=b1, bin2=b2, bin2v=b2v))
+                                    bin1=b1, bin2=b2, bin2v=b2v,
+                                    local=local))
             begin = end
         self._blk_col = blk_col  # keep the narrow()s' base alive
         del ends
@@ -283,16 +318,49 @@ class PagerankEngine:
                                rank0 / degf.clamp(min=1.0))
         self.new_part = torch.empty(part.vp, dtype=F32, device=device)
         self.init_rank = (1.0 - 0.15) / part.nv
+        # pipelined state: cur_part = my slice's latest values; the
+        # all-gather publishing them into `old` may still be in flight
+        self.cur_part = self.old.narrow(0, part.row_left,
+                                        part.vp).clone()
+        self._handle = None
+
+    def _pipelined(self):
+        p = self.part
+        return dx.world_size() > 1 and p.blocks is not None and p.vp > 0
 
     def step(self):
         p = self.part
-        run_pull(p, ng.PULL_PR, self.old, self.new_part, self.deg,
-                 self.init_rank)
-        dx.all_gather_slices(self.old, self.new_part, p.verts_all,
-                             p.row_left_all, my_index=p.p)
+        if not self._pipelined():
+            run_pull(p, ng.PULL_PR, self.old, self.new_part, self.deg,
+                     self.init_rank)
+            dx.all_gather_slices(self.old, self.new_part, p.verts_all,
+                                 p.row_left_all, my_index=p.p)
+            return
+        # overlap: sweep the rank-local src block against cur_part (offset
+        # base so global src ids index it) while the gather of remote
+        # slices from the previous step is still in flight (SURVEY.md §7
+        # M2: xGMI exchange hidden under local compute)
+        self.new_part.zero_()
+        local_base = self.cur_part.data_ptr() - p.row_left * 4
+        run_pull_sweeps(p, ng.PULL_PR, local_base, self.new_part, self.deg,
+                        self.init_rank, subset="local")
+        if self._handle is not None:
+            self._handle.wait()
+            self._handle = None
+        run_pull_sweeps(p, ng.PULL_PR, self.old, self.new_part, self.deg,
+                        self.init_rank, subset="remote")
+        ng.pull_finish_pr(_stream(), p.vp, self.new_part, self.deg,
+                          p.row_left, self.init_rank)
+        self.cur_part, self.new_part = self.new_part, self.cur_part
+        self._handle = dx.all_gather_slices_async(
+            self.old, self.cur_part, p.verts_all, p.row_left_all,
+            my_index=p.p)
 
     def ranks(self):
         """Replicated stored ranks (pr/out_degree) as a torch tensor."""
+        if self._handle is not None:
+            self._handle.wait()
+            self._handle = None
         return self.old
 
 
@@ -307,16 +375,46 @@ class LabelPullEngine:
         part.prepare_pull()
         self.old = init_labels  # u32[nv] replicated (as int32 tensor)
         self.new_part = torch.empty(part.vp, dtype=U32, device=part.device)
+        self.cur_part = init_labels.narrow(0, part.row_left,
+                                           part.vp).clone()
+        self._handle = None
+
+    def _pipelined(self):
+        p = self.part
+        return dx.world_size() > 1 and p.blocks is not None and p.vp > 0
 
     def step(self):
         p = self.part
-        run_pull(p, self.mode, self.old, self.new_part, None, 0.0)
-        # changed count before the gather overwrites old
-        changed = (self.new_part
-                   != self.old.narrow(0, p.row_left, p.vp)).sum()
-        dx.all_gather_slices(self.old, self.new_part, p.verts_all,
-                             p.row_left_all, my_index=p.p)
+        if not self._pipelined():
+            run_pull(p, self.mode, self.old, self.new_part, None, 0.0)
+            changed = (self.new_part
+                       != self.old.narrow(0, p.row_left, p.vp)).sum()
+            dx.all_gather_slices(self.old, self.new_part, p.verts_all,
+                                 p.row_left_all, my_index=p.p)
+            return changed
+        # pipelined (see PagerankEngine.step): local block vs cur_part
+        # while the previous publish is in flight
+        self.new_part.copy_(self.cur_part)  # label seed: own old labels
+        local_base = self.cur_part.data_ptr() - p.row_left * 4
+        run_pull_sweeps(p, self.mode, local_base, self.new_part, None, 0.0,
+                        subset="local")
+        if self._handle is not None:
+            self._handle.wait()
+            self._handle = None
+        run_pull_sweeps(p, self.mode, self.old, self.new_part, None, 0.0,
+                        subset="remote")
+        changed = (self.new_part != self.cur_part).sum()
+        self.cur_part, self.new_part = self.new_part, self.cur_part
+        self._handle = dx.all_gather_slices_async(
+            self.old, self.cur_part, p.verts_all, p.row_left_all,
+            my_index=p.p)
         return changed
+
+    def labels(self):
+        if self._handle is not None:
+            self._handle.wait()
+            self._handle = None
+        return self.old
 
     def run_to_fixpoint(self, max_iters=None):
         it = 0

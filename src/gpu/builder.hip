@@ -164,28 +164,45 @@ __device__ __forceinline__ V_ID row_of_edge(const E_ID* row_ptr_loc, V_ID vp,
   return lo;
 }
 
+// Block boundaries are an explicit sorted list (bounds[0]=0 ... bounds[nb]
+// = nv) rather than a uniform shift: the distributed pull engines align one
+// boundary pair with this rank's own vertex range so the rank-local src
+// block can be swept while the RCCL all-gather of remote slices is still in
+// flight (comm/compute overlap over xGMI).
+__device__ __forceinline__ uint32_t block_of(V_ID src, const V_ID* bounds,
+                                             int nb) {
+  int lo = 0, hi = nb - 1;
+  while (lo < hi) {
+    int mid = (lo + hi + 1) >> 1;
+    if (src >= bounds[mid]) lo = mid;
+    else hi = mid - 1;
+  }
+  return (uint32_t)lo;
+}
+
 __global__ void blocked_count_kernel(uint64_t ep, const V_ID* col,
                                      const E_ID* row_ptr_loc, V_ID vp,
-                                     int shift, uint32_t* counts) {
+                                     const V_ID* bounds, int nb,
+                                     uint32_t* counts) {
   uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
   for (uint64_t j = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; j < ep;
        j += stride) {
     V_ID v = row_of_edge(row_ptr_loc, vp, j);
-    uint32_t b = col[j] >> shift;
+    uint32_t b = block_of(col[j], bounds, nb);
     atomicAdd(&counts[(uint64_t)b * vp + v], 1u);
   }
 }
 
 __global__ void blocked_scatter_kernel(uint64_t ep, const V_ID* col,
                                        const E_ID* row_ptr_loc, V_ID vp,
-                                       int shift,
+                                       const V_ID* bounds, int nb,
                                        unsigned long long* cursor,
                                        V_ID* out_col) {
   uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
   for (uint64_t j = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; j < ep;
        j += stride) {
     V_ID v = row_of_edge(row_ptr_loc, vp, j);
-    uint32_t b = col[j] >> shift;
+    uint32_t b = block_of(col[j], bounds, nb);
     unsigned long long pos = atomicAdd(&cursor[(uint64_t)b * vp + v], 1ull);
     out_col[pos] = col[j];
   }
@@ -277,20 +294,23 @@ void lux_gpu_edges_to_csc(uint64_t stream, uint32_t nv, uint64_t ne,
 }
 
 void lux_gpu_blocked_count(uint64_t stream, uint64_t ep, const V_ID* col,
-                           const E_ID* row_ptr_loc, V_ID vp, int shift,
-                           uint32_t* counts /*pre-zeroed u32[SB*vp]*/) {
+                           const E_ID* row_ptr_loc, V_ID vp,
+                           const V_ID* bounds /*device u32[nb+1]*/, int nb,
+                           uint32_t* counts /*pre-zeroed u32[nb*vp]*/) {
   hipStream_t s = (hipStream_t)stream;
   hipLaunchKernelGGL(blocked_count_kernel, dim3(grid_for(ep)), dim3(BLOCK),
-                     0, s, ep, col, row_ptr_loc, vp, shift, counts);
+                     0, s, ep, col, row_ptr_loc, vp, bounds, nb, counts);
   LUX_POST_LAUNCH(stream);
 }
 
 void lux_gpu_blocked_scatter(uint64_t stream, uint64_t ep, const V_ID* col,
-                             const E_ID* row_ptr_loc, V_ID vp, int shift,
+                             const E_ID* row_ptr_loc, V_ID vp,
+                             const V_ID* bounds, int nb,
                              unsigned long long* cursor, V_ID* out_col) {
   hipStream_t s = (hipStream_t)stream;
   hipLaunchKernelGGL(blocked_scatter_kernel, dim3(grid_for(ep)), dim3(BLOCK),
-                     0, s, ep, col, row_ptr_loc, vp, shift, cursor, out_col);
+                     0, s, ep, col, row_ptr_loc, vp, bounds, nb, cursor,
+                     out_col);
   LUX_POST_LAUNCH(stream);
 }
 

@@ -119,3 +119,62 @@ def test_dist_sssp_matches_single(world):
     g = Graph.rmat(scale, ne, seed=seed)
     want, _ = cpu_ref.sssp(g, 0)
     assert np.array_equal(got, want)
+
+
+def _dist_pagerank_pipelined_worker(rank, world, port, scale, ne, seed,
+                                    iters, outq):
+    """Publish-early schedule with the async all-gather handle — the
+    collective mechanics the pipelined GPU engines use (engine.py
+    PagerankEngine.step), with the CPU reference compute."""
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from lux_amd import _native as nat
+        from lux_amd import dist as dx
+        from lux_amd.graph import Graph
+
+        g = Graph.rmat(scale, ne, seed=seed)
+        part = g.partition(world)
+        rl, rr, cl, ce, src, _w = g.slice(part, rank)
+        vp = part.verts(rank)
+        deg_t = torch.zeros(g.nv, dtype=torch.int32)
+        deg_t += torch.from_numpy(
+            np.bincount(src, minlength=g.nv).astype(np.int32))
+        dx.all_reduce_sum_(deg_t)
+        deg = deg_t.numpy().view(np.uint32)
+        old_t = torch.from_numpy(nat.pagerank_init(g.nv, deg))
+        new_t = torch.empty(vp, dtype=torch.float32)
+        cur_t = old_t.narrow(0, int(part.row_left[rank]), vp).clone()
+        verts = [part.verts(p) for p in range(world)]
+        lefts = [int(part.row_left[p]) for p in range(world)]
+        handle = None
+        for _ in range(iters):
+            if handle is not None:
+                handle.wait()
+            nat.pagerank_iter_part(g.nv, rl, rr, cl, ce, src, deg,
+                                   old_t.numpy(), new_t.numpy())
+            cur_t.copy_(new_t)
+            handle = dx.all_gather_slices_async(old_t, cur_t, verts, lefts)
+        handle.wait()
+        if rank == 0:
+            outq.put(old_t.numpy().copy())
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("world", [2])
+def test_dist_pagerank_async_pipeline(world):
+    from lux_amd import cpu_ref
+    from lux_amd.graph import Graph
+    scale, ne, seed, iters = 10, 20000, 47, 4
+    ctx = mp.get_context("spawn")
+    outq = ctx.SimpleQueue()
+    mp.spawn(_dist_pagerank_pipelined_worker,
+             args=(world, _find_port(), scale, ne, seed, iters, outq),
+             nprocs=world, join=True)
+    got = outq.get()
+    g = Graph.rmat(scale, ne, seed=seed)
+    want = cpu_ref.pagerank(g, iters)
+    np.testing.assert_allclose(got, want, rtol=1e-6)
