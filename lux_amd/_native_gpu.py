@@ -176,3 +176,21 @@ def cf_als_iter(stream, n0, bin0, n1, bin1, n2, bin2, nbig, bin2v, hubidx,
         dp(bin2), _u32(nbig), dp(bin2v), dp(hubidx), dp(gram_scratch),
         dp(rhs_scratch), dp(row_ptr), dp(col), dp(w), dp(oldv), dp(newv),
         _u32(row_left), ctypes.c_int(K))
+
+
+def frontier_expand(stream, old_dense, in_row_left, in_count, old_seg,
+                    push_row_ptr, items, counter, max_items):
+    lib().lux_gpu_frontier_expand(
+        _u64(stream), ctypes.c_int(old_dense), _u32(in_row_left),
+        _u32(in_count), dp(old_seg), dp(push_row_ptr), dp(items),
+        dp(counter), _u32(max_items))
+
+
+def push_chunk_scatter(stream, is_min, new_dense, items, counter, max_items,
+                       push_row_ptr, push_col, old_labels, snapshot,
+                       new_labels, my_row_left, new_seg, capacity):
+    lib().lux_gpu_push_chunk_scatter(
+        _u64(stream), ctypes.c_int(is_min), ctypes.c_int(new_dense),
+        dp(items), dp(counter), _u32(max_items), dp(push_row_ptr),
+        dp(push_col), dp(old_labels), dp(snapshot), dp(new_labels),
+        _u32(my_row_left), dp(new_seg), _u32(capacity))

@@ -83,6 +83,14 @@ class PushEngine:
             idx.extend(range(int(self.seg_off[q]), int(self.seg_off[q]) + 8))
         self.hdr_idx = torch.tensor(idx, dtype=torch.long, device=device)
 
+        # ---- edge-balanced scatter work items (push.hip expand+chunk):
+        # push runs only while total frontier <= nv/16, so items are bounded
+        # by nv/16 active vertices + ep/8192 extra chunks ----
+        self.max_items = p.nv // 16 + p.ep // 8192 + 1024
+        self.items = torch.empty(self.max_items * 2, dtype=U32,
+                                 device=device)
+        self.item_counter = torch.zeros(1, dtype=U32, device=device)
+
         # ---- labels + frontier state ----
         self.labels = torch.empty(p.nv, dtype=U32, device=device)
         self.labels_part = torch.empty(p.vp, dtype=U32, device=device)
@@ -160,6 +168,9 @@ class PushEngine:
             mode = ng.PULL_MIN if self.is_min else ng.PULL_MAX
             run_pull(p, mode, self.labels, self.labels_part, None, 0.0)
         else:
+            # expand all source segments into <=8192-edge work items, then
+            # one edge-balanced scatter launch over every item
+            self.item_counter.zero_()
             for q in range(nparts):
                 typ, num = self.headers[q]
                 if p.verts_all[q] == 0:
@@ -167,20 +178,20 @@ class PushEngine:
                 seg = self.fq_all.narrow(0, int(self.seg_off[q]),
                                          self.seg_bytes[q])
                 if typ == DENSE_BITMAP:
-                    ng.push_scatter(s, int(self.is_min), 1, int(new_dense),
-                                    p.row_left_all[q], p.verts_all[q], seg,
-                                    self.push_row_ptr, self.push_col,
-                                    self.labels, self.snapshot,
-                                    self.labels_part, p.row_left,
-                                    self.new_seg, self.capacity)
-                else:
-                    if num == 0:
-                        continue
-                    ng.push_scatter(s, int(self.is_min), 0, int(new_dense),
-                                    0, num, seg, self.push_row_ptr,
-                                    self.push_col, self.labels,
-                                    self.snapshot, self.labels_part,
-                                    p.row_left, self.new_seg, self.capacity)
+                    ng.frontier_expand(s, 1, p.row_left_all[q],
+                                       p.verts_all[q], seg,
+                                       self.push_row_ptr, self.items,
+                                       self.item_counter, self.max_items)
+                elif num:
+                    ng.frontier_expand(s, 0, 0, num, seg,
+                                       self.push_row_ptr, self.items,
+                                       self.item_counter, self.max_items)
+            ng.push_chunk_scatter(s, int(self.is_min), int(new_dense),
+                                  self.items, self.item_counter,
+                                  self.max_items, self.push_row_ptr,
+                                  self.push_col, self.labels, self.snapshot,
+                                  self.labels_part, p.row_left,
+                                  self.new_seg, self.capacity)
 
         # ---- frontier format fix-ups (sssp_gpu.cu:462-491) ----
         if new_dense:

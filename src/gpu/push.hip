@@ -181,6 +181,128 @@ __global__ void push_scatter_kernel(
   }
 }
 
+// ---------------- edge-balanced scatter (expand + chunk) ----------------
+// The block-cooperative kernel above (reference parity, sssp_gpu.cu:132-246)
+// drains each 256-frontier-vertex group in ONE block: a hub's whole
+// out-edge range serializes there (measured: the 1-vertex first SSSP
+// iteration from RMAT-27's root took 58 ms — one block walked ~4M edges
+// while 255 CUs idled; the 902K-vertex second iteration 104 ms on its
+// worst block). The MI355X path splits every active vertex's range into
+// <=PUSH_CHUNK-edge work items first (LDS-aggregated append), then one
+// grid covers all items — edge-balanced regardless of degree skew, and
+// one launch per iteration instead of one per source segment.
+
+constexpr V_ID PUSH_CHUNK = 8192;
+
+__global__ void frontier_expand_kernel(int old_dense, V_ID in_row_left,
+                                       V_ID in_count,
+                                       const uint8_t* old_seg,
+                                       const E_ID* push_row_ptr,
+                                       uint2* items, uint32_t* counter,
+                                       uint32_t max_items) {
+  __shared__ uint32_t lds_scan[BLOCK / WAVE + 1];
+  __shared__ uint32_t blk_base;
+  const uint8_t* bitmap = old_seg + sizeof(FrontierHeader);
+  const V_ID* queue = (const V_ID*)(old_seg + sizeof(FrontierHeader));
+  for (V_ID blk = blockIdx.x * blockDim.x; blk < in_count;
+       blk += blockDim.x * gridDim.x) {
+    V_ID idx = blk + threadIdx.x;
+    V_ID u = 0;
+    uint32_t nch = 0;
+    if (idx < in_count) {
+      bool active;
+      if (old_dense) {
+        u = in_row_left + idx;
+        active = (bitmap[idx >> 3] >> (idx & 7)) & 1;
+      } else {
+        u = queue[idx];
+        active = true;
+      }
+      if (active) {
+        E_ID deg = push_row_ptr[u + 1] - push_row_ptr[u];
+        nch = (uint32_t)((deg + PUSH_CHUNK - 1) / PUSH_CHUNK);
+      }
+    }
+    uint32_t total;
+    uint32_t ex = block_exscan<uint32_t, BLOCK>(nch, lds_scan, &total);
+    if (threadIdx.x == 0) blk_base = total ? atomicAdd(counter, total) : 0;
+    __syncthreads();
+    for (uint32_t c = 0; c < nch; c++) {
+      uint32_t pos = blk_base + ex + c;
+      if (pos < max_items) items[pos] = make_uint2(u, c);
+    }
+    __syncthreads();
+  }
+}
+
+template <bool IS_MIN, bool NEW_DENSE>
+__global__ void push_chunk_scatter_kernel(
+    const uint2* items, const uint32_t* counter, uint32_t max_items,
+    const E_ID* push_row_ptr, const V_ID* push_col,
+    const uint32_t* old_labels, const uint32_t* snapshot,
+    uint32_t* new_labels, V_ID my_row_left, uint8_t* new_seg,
+    V_ID capacity) {
+  using OP = LabOp<IS_MIN>;
+  __shared__ uint32_t lds_scan[BLOCK / WAVE + 1];
+  __shared__ uint32_t queue_base;
+  V_ID* new_queue = nullptr;
+  uint32_t* num_nodes = nullptr;
+  if (!NEW_DENSE) {
+    num_nodes = &((FrontierHeader*)new_seg)->numNodes;
+    new_queue = (V_ID*)(new_seg + sizeof(FrontierHeader));
+  }
+  uint32_t n = *counter;
+  if (n > max_items) n = max_items;
+  for (uint32_t i = blockIdx.x; i < n; i += gridDim.x) {
+    uint2 it = items[i];
+    V_ID u = it.x;
+    uint32_t new_lab = OP::map(old_labels[u]);
+    E_ID b = push_row_ptr[u] + (E_ID)it.y * PUSH_CHUNK;
+    E_ID e = push_row_ptr[u + 1];
+    if (e > b + PUSH_CHUNK) e = b + PUSH_CHUNK;
+    uint32_t nstripes =
+        (uint32_t)((e - b + blockDim.x - 1) / blockDim.x);
+    for (uint32_t s = 0; s < nstripes; s++) {
+      E_ID k = b + (E_ID)s * blockDim.x + threadIdx.x;
+      uint32_t flag = 0;
+      V_ID dstv = 0;
+      if (k < e) {
+        V_ID v = push_col[k];
+        uint32_t* slot = &new_labels[v - my_row_left];
+        uint32_t cur = __hip_atomic_load(slot, __ATOMIC_RELAXED,
+                                         __HIP_MEMORY_SCOPE_AGENT);
+        if (OP::better(new_lab, cur)) {
+          if (NEW_DENSE) {
+            OP::atom(slot, new_lab);
+          } else {
+            // first-improvement enqueue (sssp_gpu.cu:63-82)
+            uint32_t last = snapshot[v - my_row_left];
+            uint32_t act = OP::atom(slot, new_lab);
+            if (act == last) {
+              flag = 1;
+              dstv = v;
+            }
+          }
+        }
+      }
+      if (!NEW_DENSE) {
+        __syncthreads();
+        uint32_t q_total;
+        uint32_t q_off = block_exscan<uint32_t, BLOCK>(flag, lds_scan,
+                                                       &q_total);
+        if (threadIdx.x == 0 && q_total)
+          queue_base = atomicAdd(num_nodes, q_total);
+        __syncthreads();
+        if (flag) {
+          uint32_t pos = queue_base + q_off;
+          if (pos < capacity) new_queue[pos] = dstv;
+        }
+        __syncthreads();
+      }
+    }
+  }
+}
+
 // ---------------- frontier representation fix-ups ----------------
 
 // Build my dense output bitmap + count: bit v <=> snapshot[v]!=new[v]
@@ -299,6 +421,59 @@ void lux_gpu_push_scatter(uint64_t stream, int is_min, int old_dense,
     case 5: launch(push_scatter_kernel<true, false, true>); break;
     case 6: launch(push_scatter_kernel<true, true, false>); break;
     case 7: launch(push_scatter_kernel<true, true, true>); break;
+  }
+  LUX_POST_LAUNCH(stream);
+}
+
+void lux_gpu_frontier_expand(uint64_t stream, int old_dense,
+                             V_ID in_row_left, V_ID in_count,
+                             const uint8_t* old_seg,
+                             const E_ID* push_row_ptr, uint2* items,
+                             uint32_t* counter /*pre-zeroed u32[1]*/,
+                             uint32_t max_items) {
+  hipStream_t s = (hipStream_t)stream;
+  if (in_count == 0) return;
+  hipLaunchKernelGGL(frontier_expand_kernel, dim3(grid_for(in_count)),
+                     dim3(BLOCK), 0, s, old_dense, in_row_left, in_count,
+                     old_seg, push_row_ptr, items, counter, max_items);
+  LUX_POST_LAUNCH(stream);
+}
+
+void lux_gpu_push_chunk_scatter(uint64_t stream, int is_min, int new_dense,
+                                const uint2* items, const uint32_t* counter,
+                                uint32_t max_items,
+                                const E_ID* push_row_ptr,
+                                const V_ID* push_col,
+                                const uint32_t* old_labels,
+                                const uint32_t* snapshot,
+                                uint32_t* new_labels, V_ID my_row_left,
+                                uint8_t* new_seg, V_ID capacity) {
+  hipStream_t s = (hipStream_t)stream;
+  dim3 grid(MAX_GRID), blk(BLOCK);
+  // grid-strided over the device-side item count: no host sync needed;
+  // surplus blocks read the counter and exit
+  if (is_min) {
+    if (new_dense)
+      hipLaunchKernelGGL((push_chunk_scatter_kernel<true, true>), grid, blk,
+                         0, s, items, counter, max_items, push_row_ptr,
+                         push_col, old_labels, snapshot, new_labels,
+                         my_row_left, new_seg, capacity);
+    else
+      hipLaunchKernelGGL((push_chunk_scatter_kernel<true, false>), grid,
+                         blk, 0, s, items, counter, max_items, push_row_ptr,
+                         push_col, old_labels, snapshot, new_labels,
+                         my_row_left, new_seg, capacity);
+  } else {
+    if (new_dense)
+      hipLaunchKernelGGL((push_chunk_scatter_kernel<false, true>), grid,
+                         blk, 0, s, items, counter, max_items, push_row_ptr,
+                         push_col, old_labels, snapshot, new_labels,
+                         my_row_left, new_seg, capacity);
+    else
+      hipLaunchKernelGGL((push_chunk_scatter_kernel<false, false>), grid,
+                         blk, 0, s, items, counter, max_items, push_row_ptr,
+                         push_col, old_labels, snapshot, new_labels,
+                         my_row_left, new_seg, capacity);
   }
   LUX_POST_LAUNCH(stream);
 }
