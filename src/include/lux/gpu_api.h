@@ -80,6 +80,23 @@ void lux_gpu_push_scatter(uint64_t stream, int is_min, int old_dense,
                           const uint32_t* snapshot, uint32_t* new_labels,
                           lux::V_ID my_row_left, uint8_t* new_seg,
                           lux::V_ID capacity);
+
+void lux_gpu_frontier_expand(uint64_t stream, int old_dense,
+                             lux::V_ID in_row_left, lux::V_ID in_count,
+                             const uint8_t* old_seg,
+                             const lux::E_ID* push_row_ptr,
+                             lux_uint2* items, uint32_t* counter,
+                             uint32_t max_items);
+void lux_gpu_push_chunk_scatter(uint64_t stream, int is_min, int new_dense,
+                                const lux_uint2* items,
+                                const uint32_t* counter, uint32_t max_items,
+                                const lux::E_ID* push_row_ptr,
+                                const lux::V_ID* push_col,
+                                const uint32_t* old_labels,
+                                const uint32_t* snapshot,
+                                uint32_t* new_labels, lux::V_ID my_row_left,
+                                uint8_t* new_seg, lux::V_ID capacity);
+
 void lux_gpu_build_bitmap(uint64_t stream, lux::V_ID vp,
                           const uint32_t* snapshot,
                           const uint32_t* new_labels, uint8_t* seg);

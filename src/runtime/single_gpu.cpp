@@ -178,6 +178,11 @@ SingleGpuPush::SingleGpuPush(const DeviceGraph& g, bool is_min, V_ID source,
   fq_ = arena.alloc_n<uint8_t>(fq_bytes);
   new_fq_ = arena.alloc_n<uint8_t>(fq_bytes);
   tmp_fq_ = arena.alloc_n<uint8_t>(fq_bytes);
+  // edge-balanced scatter work items (push.hip expand+chunk); bounded by
+  // the nv/16 push threshold + per-8192-edge chunk splits
+  max_items_ = (uint32_t)(g.nv / SPARSE_THRESHOLD + g.ne / 8192 + 1024);
+  items_ = arena.alloc_n<lux_uint2>(max_items_);
+  item_counter_ = arena.alloc_n<uint32_t>(1);
   capacity_ = frontier_capacity(g.nv);
   // seed labels + frontier (sssp_gpu.cu:733-744, components_gpu.cu:733-740)
   std::vector<uint32_t> hl(g.nv);
@@ -218,13 +223,16 @@ V_ID SingleGpuPush::step() {
                       bins_.bin2v, row_ptr_, g_.src, snapshot_, labels_,
                       nullptr, 0, 0.0f, 0);
   } else {
-    lux_gpu_push_scatter((uint64_t)s_, is_min_ ? 1 : 0,
-                         fq_type_ == FrontierHeader::DENSE_BITMAP ? 1 : 0,
-                         new_dense ? 1 : 0, 0,
-                         fq_type_ == FrontierHeader::DENSE_BITMAP ? g_.nv
-                                                                  : fq_num_,
-                         fq_, push_row_ptr_, push_col_, snapshot_, snapshot_,
-                         labels_, 0, new_fq_, capacity_);
+    LUX_OK(hipMemsetAsync(item_counter_, 0, 4, s_));
+    lux_gpu_frontier_expand(
+        (uint64_t)s_, fq_type_ == FrontierHeader::DENSE_BITMAP ? 1 : 0, 0,
+        fq_type_ == FrontierHeader::DENSE_BITMAP ? g_.nv : fq_num_, fq_,
+        push_row_ptr_, items_, item_counter_, max_items_);
+    lux_gpu_push_chunk_scatter((uint64_t)s_, is_min_ ? 1 : 0,
+                               new_dense ? 1 : 0, items_, item_counter_,
+                               max_items_, push_row_ptr_, push_col_,
+                               snapshot_, snapshot_, labels_, 0, new_fq_,
+                               capacity_);
   }
   FrontierHeader hh;
   if (new_dense) {

@@ -103,6 +103,9 @@ class SingleGpuPush {
   Bins bins_;
   uint32_t *labels_, *snapshot_;
   uint8_t *fq_, *new_fq_, *tmp_fq_;
+  lux_uint2* items_;
+  uint32_t* item_counter_;
+  uint32_t max_items_;
   V_ID capacity_;
   uint32_t fq_type_, fq_num_;
   int iters_ = 0;
