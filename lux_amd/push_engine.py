@@ -163,13 +163,8 @@ class PushEngine:
         self._my_seg_i32()[1] = 0
 
         pull_fallback = old_fq_size > p.nv // 16
-        if pull_fallback:
-            new_dense = True
-            mode = ng.PULL_MIN if self.is_min else ng.PULL_MAX
-            run_pull(p, mode, self.labels, self.labels_part, None, 0.0)
-        else:
-            # expand all source segments into <=8192-edge work items, then
-            # one edge-balanced scatter launch over every item
+        if not pull_fallback:
+            # expand all source segments into <=8192-edge work items
             self.item_counter.zero_()
             for q in range(nparts):
                 typ, num = self.headers[q]
@@ -186,6 +181,20 @@ class PushEngine:
                     ng.frontier_expand(s, 0, 0, num, seg,
                                        self.push_row_ptr, self.items,
                                        self.item_counter, self.max_items)
+            # second adaptivity axis (ours, not the reference's): the
+            # vertex-count threshold misses RMAT's hub explosion — a 902K-
+            # vertex frontier can cover ~half of all edges. The item count
+            # (chunks of <=8192 edges) is a free edge-volume estimate; a
+            # dense pull sweep (identical labels, src-blocked LLC-resident
+            # gathers) is faster beyond ~ep/8 traversed edges.
+            n_items = int(self.item_counter.cpu().item())
+            if (n_items - old_fq_size) * 8192 > p.ep // 8:
+                pull_fallback = True
+        if pull_fallback:
+            new_dense = True
+            mode = ng.PULL_MIN if self.is_min else ng.PULL_MAX
+            run_pull(p, mode, self.labels, self.labels_part, None, 0.0)
+        else:
             ng.push_chunk_scatter(s, int(self.is_min), int(new_dense),
                                   self.items, self.item_counter,
                                   self.max_items, self.push_row_ptr,
