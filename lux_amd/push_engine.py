@@ -89,7 +89,7 @@ class PushEngine:
         self.max_items = p.nv // 16 + p.ep // 8192 + 1024
         self.items = torch.empty(self.max_items * 2, dtype=U32,
                                  device=device)
-        self.item_counter = torch.zeros(1, dtype=U32, device=device)
+        self.item_counter = torch.zeros(2, dtype=U32, device=device)
 
         # ---- labels + frontier state ----
         self.labels = torch.empty(p.nv, dtype=U32, device=device)
@@ -183,12 +183,12 @@ class PushEngine:
                                        self.item_counter, self.max_items)
             # second adaptivity axis (ours, not the reference's): the
             # vertex-count threshold misses RMAT's hub explosion — a 902K-
-            # vertex frontier can cover ~half of all edges. The item count
-            # (chunks of <=8192 edges) is a free edge-volume estimate; a
-            # dense pull sweep (identical labels, src-blocked LLC-resident
-            # gathers) is faster beyond ~ep/8 traversed edges.
-            n_items = int(self.item_counter.cpu().item())
-            if (n_items - old_fq_size) * 8192 > p.ep // 8:
+            # vertex frontier can cover ~half of all edges. The expand
+            # kernel counts the frontier's out-edges (counter[1]); a dense
+            # pull sweep (identical labels per iteration, src-blocked
+            # LLC-resident gathers) is faster beyond ~ep/8 edges.
+            n_edges = int(self.item_counter[1].cpu().item())
+            if n_edges > p.ep // 8:
                 pull_fallback = True
         if pull_fallback:
             new_dense = True

@@ -194,6 +194,9 @@ __global__ void push_scatter_kernel(
 
 constexpr V_ID PUSH_CHUNK = 8192;
 
+// counter[0] = work items; counter[1] = total out-edges of the segment's
+// active vertices (the engine's free traversed-edge estimate for its
+// push-vs-pull decision; per-rank edges < 2^32 by construction).
 __global__ void frontier_expand_kernel(int old_dense, V_ID in_row_left,
                                        V_ID in_count,
                                        const uint8_t* old_seg,
@@ -201,6 +204,7 @@ __global__ void frontier_expand_kernel(int old_dense, V_ID in_row_left,
                                        uint2* items, uint32_t* counter,
                                        uint32_t max_items) {
   __shared__ uint32_t lds_scan[BLOCK / WAVE + 1];
+  __shared__ unsigned long long lds_red[BLOCK / WAVE];
   __shared__ uint32_t blk_base;
   const uint8_t* bitmap = old_seg + sizeof(FrontierHeader);
   const V_ID* queue = (const V_ID*)(old_seg + sizeof(FrontierHeader));
@@ -209,6 +213,7 @@ __global__ void frontier_expand_kernel(int old_dense, V_ID in_row_left,
     V_ID idx = blk + threadIdx.x;
     V_ID u = 0;
     uint32_t nch = 0;
+    E_ID deg = 0;
     if (idx < in_count) {
       bool active;
       if (old_dense) {
@@ -219,13 +224,18 @@ __global__ void frontier_expand_kernel(int old_dense, V_ID in_row_left,
         active = true;
       }
       if (active) {
-        E_ID deg = push_row_ptr[u + 1] - push_row_ptr[u];
+        deg = push_row_ptr[u + 1] - push_row_ptr[u];
         nch = (uint32_t)((deg + PUSH_CHUNK - 1) / PUSH_CHUNK);
       }
     }
+    unsigned long long esum =
+        block_reduce_sum((unsigned long long)deg, lds_red);
     uint32_t total;
     uint32_t ex = block_exscan<uint32_t, BLOCK>(nch, lds_scan, &total);
-    if (threadIdx.x == 0) blk_base = total ? atomicAdd(counter, total) : 0;
+    if (threadIdx.x == 0) {
+      blk_base = total ? atomicAdd(counter, total) : 0;
+      if (esum) atomicAdd(&counter[1], (uint32_t)esum);
+    }
     __syncthreads();
     for (uint32_t c = 0; c < nch; c++) {
       uint32_t pos = blk_base + ex + c;

@@ -182,7 +182,7 @@ SingleGpuPush::SingleGpuPush(const DeviceGraph& g, bool is_min, V_ID source,
   // the nv/16 push threshold + per-8192-edge chunk splits
   max_items_ = (uint32_t)(g.nv / SPARSE_THRESHOLD + g.ne / 8192 + 1024);
   items_ = arena.alloc_n<lux_uint2>(max_items_);
-  item_counter_ = arena.alloc_n<uint32_t>(1);
+  item_counter_ = arena.alloc_n<uint32_t>(2);
   capacity_ = frontier_capacity(g.nv);
   // seed labels + frontier (sssp_gpu.cu:733-744, components_gpu.cu:733-740)
   std::vector<uint32_t> hl(g.nv);
@@ -223,7 +223,7 @@ V_ID SingleGpuPush::step() {
                       bins_.bin2v, row_ptr_, g_.src, snapshot_, labels_,
                       nullptr, 0, 0.0f, 0);
   } else {
-    LUX_OK(hipMemsetAsync(item_counter_, 0, 4, s_));
+    LUX_OK(hipMemsetAsync(item_counter_, 0, 8, s_));
     lux_gpu_frontier_expand(
         (uint64_t)s_, fq_type_ == FrontierHeader::DENSE_BITMAP ? 1 : 0, 0,
         fq_type_ == FrontierHeader::DENSE_BITMAP ? g_.nv : fq_num_, fq_,
