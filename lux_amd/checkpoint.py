@@ -44,6 +44,11 @@ def load_state(path):
     return arr, iteration
 
 
+def _labels_tensor(engine):
+    lab = engine.labels
+    return lab() if callable(lab) else lab  # LabelPullEngine method / tensor
+
+
 def save_engine(path, engine, iteration=None):
     """Dump an engine's replicated state (rank 0 only in distributed runs)."""
     from . import dist as dx
@@ -51,28 +56,52 @@ def save_engine(path, engine, iteration=None):
         return
     if hasattr(engine, "ranks"):  # PagerankEngine
         save_state(path, engine.ranks().cpu().numpy(), iteration or 0)
-    elif hasattr(engine, "labels"):  # PushEngine
-        save_state(path, engine.labels.cpu().numpy().view(np.uint32),
-                   iteration if iteration is not None else engine.iterations)
-    elif hasattr(engine, "vectors"):  # CFEngine
+    elif hasattr(engine, "labels"):  # PushEngine / LabelPullEngine
+        save_state(path, _labels_tensor(engine).cpu().numpy().view(np.uint32),
+                   iteration if iteration is not None
+                   else getattr(engine, "iterations", 0))
+    elif hasattr(engine, "vectors"):  # CFEngine / CFALSEngine
         save_state(path, engine.vectors().cpu().numpy(), iteration or 0)
     else:
         raise TypeError(f"unknown engine {type(engine)}")
 
 
 def resume_engine(path, engine):
-    """Load saved state back into an engine's replicated array (all ranks)."""
+    """Load saved state back into an engine's replicated array (all ranks).
+
+    Restores vertex state (ranks / labels / latent vectors). For PushEngine
+    the frontier is NOT part of the state (the reference's only persistence
+    is the .lux graph); resume is for converged labellings and analysis —
+    call reset() to restart a traversal."""
     import torch
     arr, iteration = load_state(path)
     if hasattr(engine, "ranks"):
         engine.old.copy_(torch.from_numpy(arr).to(engine.old.device))
+        _refresh_slice(engine)
     elif hasattr(engine, "labels"):
-        t = torch.from_numpy(arr.view(np.int32)).to(engine.labels.device)
-        engine.labels.copy_(t)
+        lab = _labels_tensor(engine)
+        t = torch.from_numpy(arr.view(np.int32)).to(lab.device)
         p = engine.part
-        engine.labels_part.copy_(engine.labels.narrow(0, p.row_left, p.vp))
-        engine.iterations = iteration
+        if hasattr(engine, "labels_part"):  # PushEngine
+            engine.labels.copy_(t)
+            engine.labels_part.copy_(engine.labels.narrow(0, p.row_left,
+                                                          p.vp))
+            engine.iterations = iteration
+        else:  # LabelPullEngine
+            engine.old.copy_(t)
+            _refresh_slice(engine)
     elif hasattr(engine, "vectors"):
         t = torch.from_numpy(arr.reshape(-1)).to(engine.old.device)
         engine.old.copy_(t)
     return iteration
+
+
+def _refresh_slice(engine):
+    """Pipelined pull engines mirror their slice in cur_part and may hold an
+    in-flight publish handle — resync after overwriting `old`."""
+    if getattr(engine, "_handle", None) is not None:
+        engine._handle.wait()
+        engine._handle = None
+    if hasattr(engine, "cur_part"):
+        p = engine.part
+        engine.cur_part.copy_(engine.old.narrow(0, p.row_left, p.vp))
