@@ -1,5 +1,6 @@
-"""Checkpoint/resume + tracing tests (CPU)."""
+"""Checkpoint/resume + tracing tests (CPU + one GPU roundtrip)."""
 import numpy as np
+import pytest
 
 from lux_amd import checkpoint as ckpt
 from lux_amd.trace import IterTrace
@@ -44,3 +45,25 @@ def test_trace_csv_and_summary():
     assert s["iterations"] == 2
     assert s["ms_per_iter"] == 3.0
     assert s["gteps"] == 2.0
+
+
+@pytest.mark.gpu
+def test_engine_checkpoint_roundtrip_gpu(tmp_path):
+    """save_engine/resume_engine through a real engine: 3 steps + save +
+    resume into a fresh engine + 2 steps == 5 uninterrupted steps."""
+    import numpy as np
+    from lux_amd import checkpoint as ck
+    from lux_amd.engine import DeviceCSC, GraphPart, PagerankEngine
+    path = str(tmp_path / "pr.luxs")
+    a = PagerankEngine(GraphPart(DeviceCSC.rmat(12, 100000, seed=3), 1, 0))
+    for _ in range(3):
+        a.step()
+    ck.save_engine(path, a, iteration=3)
+    b = PagerankEngine(GraphPart(DeviceCSC.rmat(12, 100000, seed=3), 1, 0))
+    it = ck.resume_engine(path, b)
+    assert it == 3
+    for _ in range(2):
+        a.step()
+        b.step()
+    np.testing.assert_array_equal(a.ranks().cpu().numpy(),
+                                  b.ranks().cpu().numpy())
