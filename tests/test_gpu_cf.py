@@ -168,3 +168,47 @@ def test_cf_als_beats_sgd_loss():
     l_sgd = cpu_ref.cf_loss(g, K, sgd.vectors().cpu().numpy())
     l_als = cpu_ref.cf_loss(g, K, als.vectors().cpu().numpy())
     assert l_als < l_sgd
+
+
+def _sync_parts(engines, K):
+    """Emulate the all-gather: copy every engine's own new slice into every
+    replica (single-process stand-in for dist.all_gather_slices)."""
+    for src in engines:
+        p = src.part
+        sl = src.old.narrow(0, p.row_left * K, p.vp * K)
+        for dst in engines:
+            if dst is not src:
+                dst.old.narrow(0, p.row_left * K, p.vp * K).copy_(sl)
+
+
+@pytest.mark.parametrize("cls_name", ["CFEngine", "CFALSEngine"])
+def test_cf_multipart_single_process(cls_name):
+    """2 partitions stepped in one process match the whole-graph reference
+    (the N-GPU equivalence shape from SURVEY.md §4(d) for the CF family)."""
+    import lux_amd.cf_engine as cfe
+    cls = getattr(cfe, cls_name)
+    nu, ni, ne, K = 1000, 256, 40000, 32
+    full = DeviceCSC.bipartite(nu, ni, ne, seed=19)
+    pa = GraphPart(full, 2, 0, keep_full=True)
+    pb = GraphPart(full, 2, 1)
+    ea, eb = cls(pa, K=K), cls(pb, K=K)
+    init = None
+    if cls_name == "CFALSEngine":  # cf_als ref takes an explicit init;
+        init = _rand_init(pa.nv, K, seed=77)  # cf (SGD) uses parity init
+        for e in (ea, eb):
+            e.old.copy_(torch.from_numpy(init.ravel()))
+    for _ in range(2):
+        ea.step()
+        eb.step()
+        _sync_parts((ea, eb), K)
+    got = ea.vectors().cpu().numpy()
+    g = Graph.bipartite(nu, ni, ne, seed=19)
+    ref = cpu_ref.cf if cls_name == "CFEngine" else cpu_ref.cf_als
+    kw = {} if cls_name == "CFEngine" else {"init": init}
+    want = ref(g, K, 2, **kw)
+    if cls_name == "CFEngine":
+        np.testing.assert_allclose(got, want, rtol=2e-3, atol=1e-4)
+    else:
+        scale = np.maximum(np.abs(want).max(axis=1, keepdims=True), 1.0)
+        np.testing.assert_allclose(got / scale, want / scale, rtol=0,
+                                   atol=2e-3)
