@@ -65,5 +65,8 @@ def test_engine_checkpoint_roundtrip_gpu(tmp_path):
     for _ in range(2):
         a.step()
         b.step()
-    np.testing.assert_array_equal(a.ranks().cpu().numpy(),
-                                  b.ranks().cpu().numpy())
+    # float atomicAdd order in the hub-chunk kernel varies run to run:
+    # tolerance, not bit equality (documented fp32 PR nondeterminism)
+    np.testing.assert_allclose(a.ranks().cpu().numpy(),
+                               b.ranks().cpu().numpy(), rtol=1e-5,
+                               atol=1e-12)
