@@ -194,3 +194,17 @@ def push_chunk_scatter(stream, is_min, new_dense, items, counter, max_items,
         dp(items), dp(counter), _u32(max_items), dp(push_row_ptr),
         dp(push_col), dp(old_labels), dp(snapshot), dp(new_labels),
         _u32(my_row_left), dp(new_seg), _u32(capacity))
+
+
+def uf_union_edges(stream, ep, col, row_ptr_loc, vp, row_left, parent):
+    lib().lux_gpu_uf_union_edges(_u64(stream), _u64(ep), dp(col),
+                                 dp(row_ptr_loc), _u32(vp), _u32(row_left),
+                                 dp(parent))
+
+
+def uf_union_star(stream, nv, star, parent):
+    lib().lux_gpu_uf_union_star(_u64(stream), _u32(nv), dp(star), dp(parent))
+
+
+def uf_flatten(stream, nv, parent, labels):
+    lib().lux_gpu_uf_flatten(_u64(stream), _u32(nv), dp(parent), dp(labels))

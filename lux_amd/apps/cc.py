@@ -1,8 +1,11 @@
-"""Connected Components app driver (reference parity:
-components/components.cc; max-label propagation, push model)."""
+"""Connected Components app driver (reference CLI parity:
+components/components.cc). Default engine: union-find (cc_engine.py —
+one edge pass, identical max-label output); `-labelprop` selects the
+reference-parity iterated label propagation (push model)."""
 import sys
 
 from .. import dist as dx
+from ..cc_engine import CCUnionFindEngine
 from ..engine import GraphPart
 from ..push_engine import PushEngine
 from .common import (ElapsedTimer, load_device_graph, parse_input_args,
@@ -10,11 +13,16 @@ from .common import (ElapsedTimer, load_device_graph, parse_input_args,
 
 
 class CCBench:
-    def __init__(self, part):
-        self.eng = PushEngine(part, PushEngine.MODE_MAX)
+    def __init__(self, part, labelprop=False):
+        self.labelprop = labelprop
+        if labelprop:
+            self.eng = PushEngine(part, PushEngine.MODE_MAX)
+        else:
+            self.eng = CCUnionFindEngine(part)
 
     def step(self):
-        self.eng.reset()
+        if self.labelprop:
+            self.eng.reset()
         self.eng.run()
 
 
@@ -25,7 +33,7 @@ def build_cc_bench(args, device):
     full = DeviceCSC.rmat_folded(nv, args.edges, seed=args.seed,
                                  device=device)
     part = GraphPart(full, dx.world_size(), dx.rank())
-    return CCBench(part), part
+    return CCBench(part, labelprop=getattr(args, "labelprop", False)), part
 
 
 def main(argv=None):
@@ -39,7 +47,8 @@ def main(argv=None):
     if dx.rank() == 0:
         print_memory_estimate(full.nv, full.ne, dx.world_size())
     part = GraphPart(full, dx.world_size(), dx.rank())
-    eng = PushEngine(part, PushEngine.MODE_MAX)
+    eng = PushEngine(part, PushEngine.MODE_MAX) if a.labelprop \
+        else CCUnionFindEngine(part)
     with ElapsedTimer():
         iters = eng.run()
     if dx.rank() == 0:

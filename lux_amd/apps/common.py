@@ -20,6 +20,8 @@ class AppArgs:
         self.check = False
         self.k = 64
         self.als = False  # col_filter: ALS (MFMA) optimizer instead of SGD
+        self.labelprop = False  # components: reference-parity label prop
+        #                         (default is the union-find fast path)
         self.synthetic = None  # e.g. "rmat:20:1000000"
 
 
@@ -44,6 +46,8 @@ def parse_input_args(argv):
             a.k = int(argv[i + 1]); i += 2
         elif f == "-als":
             a.als = True; i += 1
+        elif f == "-labelprop":
+            a.labelprop = True; i += 1
         elif f == "-synthetic":
             a.synthetic = argv[i + 1]; i += 2
         elif f.startswith("-ll:") or f.startswith("-lg:"):
