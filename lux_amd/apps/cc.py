@@ -28,10 +28,12 @@ class CCBench:
 
 def build_cc_bench(args, device):
     from ..engine import DeviceCSC
-    # Twitter-2010-shaped synthetic
+    # Twitter-2010-shaped synthetic, symmetrized:
+    # connected components is defined on undirected graphs, so the bench
+    # graph stores both directions of ne/2 generated pairs (ne total edges)
     nv = getattr(args, "nv", None) or 41652230
     full = DeviceCSC.rmat_folded(nv, args.edges, seed=args.seed,
-                                 device=device)
+                                 device=device, sym=True)
     part = GraphPart(full, dx.world_size(), dx.rank())
     return CCBench(part, labelprop=getattr(args, "labelprop", False)), part
 

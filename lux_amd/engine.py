@@ -128,25 +128,33 @@ class DeviceCSC:
         self.col_end, self.src, self.weight = col_end, src, weight
 
     @classmethod
-    def rmat(cls, scale, ne, seed=1, device="cuda"):
+    def rmat(cls, scale, ne, seed=1, device="cuda", sym=False):
         nv = 1 << scale
         s = _stream()
-        esrc = torch.empty(ne, dtype=U32, device=device)
-        edst = torch.empty(ne, dtype=U32, device=device)
-        ng.rmat_edges(s, seed, scale, ne, esrc, edst)
-        g = cls._from_device_edges(nv, ne, esrc, edst, None, device)
+        npairs = ne // 2 if sym else ne
+        esrc = torch.empty(npairs, dtype=U32, device=device)
+        edst = torch.empty(npairs, dtype=U32, device=device)
+        ng.rmat_edges(s, seed, scale, npairs, esrc, edst)
+        if sym:  # undirected: store both directions (CC's standard input)
+            esrc, edst = (torch.cat([esrc, edst]), torch.cat([edst, esrc]))
+        g = cls._from_device_edges(nv, npairs * 2 if sym else ne, esrc,
+                                   edst, None, device)
         return g
 
     @classmethod
-    def rmat_folded(cls, nv, ne, seed=1, device="cuda"):
+    def rmat_folded(cls, nv, ne, seed=1, device="cuda", sym=False):
         scale = 0
         while (1 << scale) < nv:
             scale += 1
         s = _stream()
-        esrc = torch.empty(ne, dtype=U32, device=device)
-        edst = torch.empty(ne, dtype=U32, device=device)
-        ng.rmat_edges_folded(s, seed, scale, nv, ne, esrc, edst)
-        return cls._from_device_edges(nv, ne, esrc, edst, None, device)
+        npairs = ne // 2 if sym else ne
+        esrc = torch.empty(npairs, dtype=U32, device=device)
+        edst = torch.empty(npairs, dtype=U32, device=device)
+        ng.rmat_edges_folded(s, seed, scale, nv, npairs, esrc, edst)
+        if sym:
+            esrc, edst = (torch.cat([esrc, edst]), torch.cat([edst, esrc]))
+        return cls._from_device_edges(nv, npairs * 2 if sym else ne, esrc,
+                                      edst, None, device)
 
     @classmethod
     def bipartite(cls, n_users, n_items, ne, seed=1, device="cuda"):

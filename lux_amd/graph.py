@@ -47,20 +47,28 @@ class Graph:
         return cls(nv, ne, col_end, src, weight)
 
     @classmethod
-    def rmat(cls, scale, ne, seed=1):
-        src, dst = nat.rmat_edges(seed, scale, ne)
+    def rmat(cls, scale, ne, seed=1, sym=False):
+        npairs = ne // 2 if sym else ne
+        src, dst = nat.rmat_edges(seed, scale, npairs)
+        if sym:  # undirected: both directions (matches DeviceCSC.rmat sym)
+            src, dst = (np.concatenate([src, dst]),
+                        np.concatenate([dst, src]))
         col_end, csrc, _ = nat.edges_to_csc(1 << scale, src, dst)
-        return cls(1 << scale, ne, col_end, csrc)
+        return cls(1 << scale, len(src), col_end, csrc)
 
     @classmethod
-    def rmat_folded(cls, nv, ne, seed=1):
+    def rmat_folded(cls, nv, ne, seed=1, sym=False):
         """RMAT skew with a non-power-of-two nv (Twitter-shaped synthetics)."""
         scale = 0
         while (1 << scale) < nv:
             scale += 1
-        src, dst = nat.rmat_edges_folded(seed, scale, nv, ne)
+        npairs = ne // 2 if sym else ne
+        src, dst = nat.rmat_edges_folded(seed, scale, nv, npairs)
+        if sym:
+            src, dst = (np.concatenate([src, dst]),
+                        np.concatenate([dst, src]))
         col_end, csrc, _ = nat.edges_to_csc(nv, src, dst)
-        return cls(nv, ne, col_end, csrc)
+        return cls(nv, len(src), col_end, csrc)
 
     @classmethod
     def bipartite(cls, n_users, n_items, ne, seed=1):

@@ -1,6 +1,10 @@
-"""Union-find CC engine (src/gpu/cc_uf.hip) vs the CPU reference: the
-labelling must be IDENTICAL to converged max-label propagation (max vertex
-id per component)."""
+"""Union-find CC engine (src/gpu/cc_uf.hip) vs the CPU reference.
+
+On SYMMETRIC graphs (connected components' standard input) the UF labelling
+is IDENTICAL to converged max-label propagation (max vertex id per
+component). On directed inputs label propagation computes directional
+max-reachability instead — the components app keeps `-labelprop` for that
+exact reference behavior, and these tests use sym=True graphs."""
 import numpy as np
 import pytest
 import torch
@@ -21,11 +25,11 @@ def stream():
 @pytest.mark.parametrize("scale,ne,seed", [(12, 60000, 3), (14, 400000, 9),
                                            (10, 2000, 5)])
 def test_cc_uf_matches_labelprop_reference(scale, ne, seed):
-    full = DeviceCSC.rmat(scale, ne, seed=seed)
+    full = DeviceCSC.rmat(scale, ne, seed=seed, sym=True)
     eng = CCUnionFindEngine(GraphPart(full, 1, 0))
     eng.run()
     got = eng.labels.cpu().numpy().view(np.uint32)
-    g = Graph.rmat(scale, ne, seed=seed)
+    g = Graph.rmat(scale, ne, seed=seed, sym=True)
     want, _ = cpu_ref.cc(g)
     np.testing.assert_array_equal(got, want)
     assert eng.check() == 0
@@ -37,7 +41,7 @@ def test_cc_uf_multipart_star_exchange():
     are unioned as stars until stable — must match the whole-graph result."""
     scale, ne, seed = 13, 150000, 21
     nv = 1 << scale
-    full = DeviceCSC.rmat(scale, ne, seed=seed)
+    full = DeviceCSC.rmat(scale, ne, seed=seed, sym=True)
     pa = GraphPart(full, 2, 0, keep_full=True)
     pb = GraphPart(full, 2, 1)
     ea, eb = CCUnionFindEngine(pa), CCUnionFindEngine(pb)
@@ -54,7 +58,7 @@ def test_cc_uf_multipart_star_exchange():
         if bool((ea.labels_t == la).all()) and bool(
                 (eb.labels_t == lb).all()):
             break
-    g = Graph.rmat(scale, ne, seed=seed)
+    g = Graph.rmat(scale, ne, seed=seed, sym=True)
     want, _ = cpu_ref.cc(g)
     np.testing.assert_array_equal(ea.labels.cpu().numpy().view(np.uint32),
                                   want)
