@@ -208,3 +208,11 @@ def uf_union_star(stream, nv, star, parent):
 
 def uf_flatten(stream, nv, parent, labels):
     lib().lux_gpu_uf_flatten(_u64(stream), _u32(nv), dp(parent), dp(labels))
+
+
+def uf_union_binned(stream, n0, bin0, n1, bin1, n2, bin2, row_ptr, col,
+                    row_left, parent):
+    lib().lux_gpu_uf_union_binned(_u64(stream), _u32(n0), dp(bin0),
+                                  _u32(n1), dp(bin1), _u32(n2), dp(bin2),
+                                  dp(row_ptr), dp(col), _u32(row_left),
+                                  dp(parent))

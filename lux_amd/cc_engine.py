@@ -38,9 +38,10 @@ class CCUnionFindEngine:
         """Compute components; returns the number of exchange rounds."""
         p = self.part
         s = _stream()
+        p.build_bins()
         torch.arange(p.nv, dtype=U32, device=self.device, out=self.parent)
-        ng.uf_union_edges(s, p.ep, p.col, p.row_ptr, p.vp, p.row_left,
-                          self.parent)
+        ng.uf_union_binned(s, p.n0, p.bin0, p.n1, p.bin1, p.n2, p.bin2,
+                           p.row_ptr, p.col, p.row_left, self.parent)
         ng.uf_flatten(s, p.nv, self.parent, self.labels_t)
         self.iterations = 1
         ws = dx.world_size()

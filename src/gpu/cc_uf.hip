@@ -80,6 +80,60 @@ __global__ void uf_union_edges_kernel(uint64_t ep, const V_ID* col,
   }
 }
 
+// Degree-binned variants (reuse the pull engine's bin lists): no per-edge
+// row binary search, and the dst side's find amortizes to once per vertex
+// (uf_union re-finds internally, so a stale dst root stays correct).
+__global__ void uf_union_thread_kernel(uint32_t n0, const V_ID* bin0,
+                                       const E_ID* row_ptr, const V_ID* col,
+                                       V_ID row_left, V_ID* parent) {
+  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
+  for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; i < n0;
+       i += stride) {
+    V_ID v = bin0[i];
+    E_ID b = row_ptr[v], e = row_ptr[v + 1];
+    V_ID g = v + row_left;
+    for (E_ID j = b; j < e; j++) {
+      uf_union(col[j], g, parent);
+      g = uf_load(&parent[g]);  // ride toward the root as it moves
+    }
+  }
+}
+
+__global__ void uf_union_wave_kernel(uint32_t n1, const V_ID* bin1,
+                                     const E_ID* row_ptr, const V_ID* col,
+                                     V_ID row_left, V_ID* parent) {
+  int lane = threadIdx.x & (WAVE - 1);
+  uint64_t wave_id = ((uint64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  uint64_t nwaves = ((uint64_t)gridDim.x * blockDim.x) / WAVE;
+  for (uint64_t i = wave_id; i < n1; i += nwaves) {
+    V_ID v = bin1[i];
+    E_ID b = row_ptr[v], e = row_ptr[v + 1];
+    V_ID g = v + row_left;
+    for (E_ID j = b + lane; j < e; j += WAVE) {
+      uf_union(col[j], g, parent);
+      g = uf_load(&parent[g]);
+    }
+  }
+}
+
+__global__ void uf_union_chunk_kernel(uint32_t n2, const uint2* bin2,
+                                      V_ID chunk_edges, const E_ID* row_ptr,
+                                      const V_ID* col, V_ID row_left,
+                                      V_ID* parent) {
+  for (uint32_t i = blockIdx.x; i < n2; i += gridDim.x) {
+    uint2 ent = bin2[i];
+    V_ID v = ent.x;
+    E_ID b = row_ptr[v] + (E_ID)ent.y * chunk_edges;
+    E_ID e = row_ptr[v + 1];
+    if (e > b + chunk_edges) e = b + chunk_edges;
+    V_ID g = v + row_left;
+    for (E_ID j = b + threadIdx.x; j < e; j += blockDim.x) {
+      uf_union(col[j], g, parent);
+      g = uf_load(&parent[g]);
+    }
+  }
+}
+
 // Union a star forest: every v is connected to star[v] (a peer rank's
 // flattened label vector).
 __global__ void uf_union_star_kernel(V_ID nv, const V_ID* star,
@@ -116,6 +170,27 @@ void lux_gpu_uf_union_edges(uint64_t stream, uint64_t ep, const V_ID* col,
   if (ep == 0) return;
   hipLaunchKernelGGL(uf_union_edges_kernel, dim3(grid_for(ep)), dim3(BLOCK),
                      0, s, ep, col, row_ptr_loc, vp, row_left, parent);
+  LUX_POST_LAUNCH(stream);
+}
+
+void lux_gpu_uf_union_binned(uint64_t stream, uint32_t n0, const V_ID* bin0,
+                             uint32_t n1, const V_ID* bin1, uint32_t n2,
+                             const uint2* bin2, const E_ID* row_ptr,
+                             const V_ID* col, V_ID row_left, V_ID* parent) {
+  hipStream_t s = (hipStream_t)stream;
+  if (n2)
+    hipLaunchKernelGGL(uf_union_chunk_kernel,
+                       dim3(n2 > MAX_GRID ? MAX_GRID : n2), dim3(BLOCK), 0,
+                       s, n2, bin2, (V_ID)8192, row_ptr, col, row_left,
+                       parent);
+  if (n1)
+    hipLaunchKernelGGL(uf_union_wave_kernel,
+                       dim3(grid_for((uint64_t)n1 * WAVE)), dim3(BLOCK), 0,
+                       s, n1, bin1, row_ptr, col, row_left, parent);
+  if (n0)
+    hipLaunchKernelGGL(uf_union_thread_kernel, dim3(grid_for(n0)),
+                       dim3(BLOCK), 0, s, n0, bin0, row_ptr, col, row_left,
+                       parent);
   LUX_POST_LAUNCH(stream);
 }
 
