@@ -82,6 +82,7 @@ __global__ void build_bins_kernel(V_ID vp, const E_ID* row_ptr,
 template <PullMode M> struct Val;
 template <> struct Val<PR_SUM> {
   using T = float;
+  static constexpr bool SKIP_SETTLED = false;
   static __device__ __forceinline__ T ident() { return 0.0f; }
   static __device__ __forceinline__ T map(T s) { return s; }
   static __device__ __forceinline__ T comb(T a, T b) { return a + b; }
@@ -92,6 +93,11 @@ template <> struct Val<PR_SUM> {
 };
 template <> struct Val<LAB_MIN> {
   using T = uint32_t;
+  // hop-SSSP (BFS) invariant: a finite label is the true depth and can
+  // never improve under synchronized iterations — settled rows skip their
+  // whole edge range (newv keeps the seeded own label). Only valid for the
+  // +1 hop metric; LAB_MAX (CC) labels keep moving and never skip.
+  static constexpr bool SKIP_SETTLED = true;
   static __device__ __forceinline__ T ident() { return INF_LABEL; }
   static __device__ __forceinline__ T map(T s) {
     return s == INF_LABEL ? INF_LABEL : s + 1;  // hop relaxation (+1)
@@ -104,6 +110,7 @@ template <> struct Val<LAB_MIN> {
 };
 template <> struct Val<LAB_MAX> {
   using T = uint32_t;
+  static constexpr bool SKIP_SETTLED = false;
   static __device__ __forceinline__ T ident() { return 0; }
   static __device__ __forceinline__ T map(T s) { return s; }
   static __device__ __forceinline__ T comb(T a, T b) { return a > b ? a : b; }
@@ -190,6 +197,8 @@ __global__ void pull_thread_kernel(uint32_t n0, const V_ID* bin0,
   for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; i < n0;
        i += stride) {
     V_ID v = bin0[i];
+    if (V::SKIP_SETTLED && oldv[a.row_left + v] != (T)INF_LABEL)
+      continue;  // settled hop label: newv keeps the seed
     E_ID b = a.row_ptr[v], e = a.row_ptr[v + 1];
     T acc = gather_range<M>(oldv, a.col, b, e, 1);
     store_result<M>(&newv[v], acc, oldv[a.row_left + v], a.init_rank,
@@ -209,6 +218,8 @@ __global__ void pull_wave_kernel(uint32_t n1, const V_ID* bin1, PullArgs a) {
   uint64_t nwaves = ((uint64_t)gridDim.x * blockDim.x) / WAVE;
   for (uint64_t i = wave_id; i < n1; i += nwaves) {
     V_ID v = bin1[i];
+    if (V::SKIP_SETTLED && oldv[a.row_left + v] != (T)INF_LABEL)
+      continue;
     E_ID b = a.row_ptr[v], e = a.row_ptr[v + 1];
     T acc = gather_range<M>(oldv, a.col, b + lane, e, WAVE);
     acc = V::reduce_wave(acc);
@@ -230,6 +241,8 @@ __global__ void pull_chunk_kernel(uint32_t n2, const uint2* bin2,
   for (uint32_t i = blockIdx.x; i < n2; i += gridDim.x) {
     uint2 ent = bin2[i];
     V_ID v = ent.x;
+    if (V::SKIP_SETTLED && oldv[a.row_left + v] != (T)INF_LABEL)
+      continue;
     E_ID b = a.row_ptr[v] + (E_ID)ent.y * CHUNK_EDGES;
     E_ID e = a.row_ptr[v + 1];
     if (e > b + CHUNK_EDGES) e = b + CHUNK_EDGES;
