@@ -211,8 +211,18 @@ def uf_flatten(stream, nv, parent, labels):
 
 
 def uf_union_binned(stream, n0, bin0, n1, bin1, n2, bin2, row_ptr, col,
-                    row_left, parent):
+                    row_left, parent, gbits=None):
     lib().lux_gpu_uf_union_binned(_u64(stream), _u32(n0), dp(bin0),
                                   _u32(n1), dp(bin1), _u32(n2), dp(bin2),
                                   dp(row_ptr), dp(col), _u32(row_left),
-                                  dp(parent))
+                                  dp(parent), dp(gbits))
+
+
+def uf_union_kth(stream, vp, row_ptr, col, row_left, parent, k):
+    lib().lux_gpu_uf_union_kth(_u64(stream), _u32(vp), dp(row_ptr), dp(col),
+                               _u32(row_left), dp(parent), _u32(k))
+
+
+def cc_giant_bits(stream, nv, labels, giant, bits):
+    lib().lux_gpu_cc_giant_bits(_u64(stream), _u32(nv), dp(labels),
+                                _u32(giant), dp(bits))

@@ -40,8 +40,22 @@ class CCUnionFindEngine:
         s = _stream()
         p.build_bins()
         torch.arange(p.nv, dtype=U32, device=self.device, out=self.parent)
+        # Afforest-style: sample-hook 2 edges/vertex, find the giant
+        # component by sampling, then sweep remaining edges skipping
+        # both-endpoints-in-giant via an L2-resident bitmap (cc_uf.hip)
+        for k in range(2):
+            ng.uf_union_kth(s, p.vp, p.row_ptr, p.col, p.row_left,
+                            self.parent, k)
+        ng.uf_flatten(s, p.nv, self.parent, self.labels_t)
+        stride = max(1, p.nv // 65536)
+        giant = int(self.labels_t[::stride].mode().values.item())
+        nwords = (p.nv + 31) // 32
+        if not hasattr(self, "_gbits") or self._gbits.numel() < nwords:
+            self._gbits = torch.empty(nwords, dtype=U32, device=self.device)
+        ng.cc_giant_bits(s, p.nv, self.labels_t, giant, self._gbits)
         ng.uf_union_binned(s, p.n0, p.bin0, p.n1, p.bin1, p.n2, p.bin2,
-                           p.row_ptr, p.col, p.row_left, self.parent)
+                           p.row_ptr, p.col, p.row_left, self.parent,
+                           gbits=self._gbits)
         ng.uf_flatten(s, p.nv, self.parent, self.labels_t)
         self.iterations = 1
         ws = dx.world_size()

@@ -83,25 +83,36 @@ __global__ void uf_union_edges_kernel(uint64_t ep, const V_ID* col,
 // Degree-binned variants (reuse the pull engine's bin lists): no per-edge
 // row binary search, and the dst side's find amortizes to once per vertex
 // (uf_union re-finds internally, so a stale dst root stays correct).
+__device__ __forceinline__ bool giant_bit_pre(const uint32_t* bits,
+                                              V_ID v) {
+  return (bits[v >> 5] >> (v & 31)) & 1;
+}
+
+template <bool GB>
 __global__ void uf_union_thread_kernel(uint32_t n0, const V_ID* bin0,
                                        const E_ID* row_ptr, const V_ID* col,
-                                       V_ID row_left, V_ID* parent) {
+                                       V_ID row_left, V_ID* parent,
+                                       const uint32_t* gbits) {
   uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
   for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; i < n0;
        i += stride) {
     V_ID v = bin0[i];
     E_ID b = row_ptr[v], e = row_ptr[v + 1];
     V_ID g = v + row_left;
+    bool gv = GB && giant_bit_pre(gbits, g);
     for (E_ID j = b; j < e; j++) {
+      if (GB && gv && giant_bit_pre(gbits, col[j])) continue;
       uf_union(col[j], g, parent);
       g = uf_load(&parent[g]);  // ride toward the root as it moves
     }
   }
 }
 
+template <bool GB>
 __global__ void uf_union_wave_kernel(uint32_t n1, const V_ID* bin1,
                                      const E_ID* row_ptr, const V_ID* col,
-                                     V_ID row_left, V_ID* parent) {
+                                     V_ID row_left, V_ID* parent,
+                                     const uint32_t* gbits) {
   int lane = threadIdx.x & (WAVE - 1);
   uint64_t wave_id = ((uint64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
   uint64_t nwaves = ((uint64_t)gridDim.x * blockDim.x) / WAVE;
@@ -109,17 +120,20 @@ __global__ void uf_union_wave_kernel(uint32_t n1, const V_ID* bin1,
     V_ID v = bin1[i];
     E_ID b = row_ptr[v], e = row_ptr[v + 1];
     V_ID g = v + row_left;
+    bool gv = GB && giant_bit_pre(gbits, g);
     for (E_ID j = b + lane; j < e; j += WAVE) {
+      if (GB && gv && giant_bit_pre(gbits, col[j])) continue;
       uf_union(col[j], g, parent);
       g = uf_load(&parent[g]);
     }
   }
 }
 
+template <bool GB>
 __global__ void uf_union_chunk_kernel(uint32_t n2, const uint2* bin2,
                                       V_ID chunk_edges, const E_ID* row_ptr,
                                       const V_ID* col, V_ID row_left,
-                                      V_ID* parent) {
+                                      V_ID* parent, const uint32_t* gbits) {
   for (uint32_t i = blockIdx.x; i < n2; i += gridDim.x) {
     uint2 ent = bin2[i];
     V_ID v = ent.x;
@@ -127,12 +141,51 @@ __global__ void uf_union_chunk_kernel(uint32_t n2, const uint2* bin2,
     E_ID e = row_ptr[v + 1];
     if (e > b + chunk_edges) e = b + chunk_edges;
     V_ID g = v + row_left;
+    bool gv = GB && giant_bit_pre(gbits, g);
     for (E_ID j = b + threadIdx.x; j < e; j += blockDim.x) {
+      if (GB && gv && giant_bit_pre(gbits, col[j])) continue;
       uf_union(col[j], g, parent);
       g = uf_load(&parent[g]);
     }
   }
 }
+
+// ---- Afforest-style sampling (Sutton et al.'s idea, re-implemented) ----
+// Phase 1 unions only each vertex's k-th in-edge (2 rounds collapse most of
+// a power-law graph into one giant component); the engine then flattens,
+// finds the giant root by sampling, and phase 2 walks the remaining edges
+// skipping any whose BOTH endpoints are already in the giant component —
+// a 2-bit test against a packed nv/8-byte bitmap (L2-resident) instead of
+// two parent-chain walks in the 4*nv-byte array.
+
+__global__ void uf_union_kth_kernel(V_ID vp, const E_ID* row_ptr,
+                                    const V_ID* col, V_ID row_left,
+                                    V_ID* parent, uint32_t k) {
+  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
+  for (uint64_t v = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; v < vp;
+       v += stride) {
+    E_ID b = row_ptr[v], e = row_ptr[v + 1];
+    if (b + k < e) uf_union(col[b + k], (V_ID)v + row_left, parent);
+  }
+}
+
+// bits[w] bit i <=> labels[32w+i] == giant (word-parallel, no atomics)
+__global__ void cc_giant_bits_kernel(V_ID nv, const V_ID* labels,
+                                     V_ID giant, uint32_t* bits) {
+  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
+  V_ID nw = (nv + 31) / 32;
+  for (uint64_t w = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; w < nw;
+       w += stride) {
+    uint32_t word = 0;
+    V_ID base = (V_ID)w * 32;
+    int n = nv - base < 32 ? (int)(nv - base) : 32;
+    for (int i = 0; i < n; i++)
+      if (labels[base + i] == giant) word |= 1u << i;
+    bits[w] = word;
+  }
+}
+
+
 
 // Union a star forest: every v is connected to star[v] (a peer rank's
 // flattened label vector).
@@ -176,21 +229,56 @@ void lux_gpu_uf_union_edges(uint64_t stream, uint64_t ep, const V_ID* col,
 void lux_gpu_uf_union_binned(uint64_t stream, uint32_t n0, const V_ID* bin0,
                              uint32_t n1, const V_ID* bin1, uint32_t n2,
                              const uint2* bin2, const E_ID* row_ptr,
-                             const V_ID* col, V_ID row_left, V_ID* parent) {
+                             const V_ID* col, V_ID row_left, V_ID* parent,
+                             const uint32_t* gbits /*nullable*/) {
   hipStream_t s = (hipStream_t)stream;
-  if (n2)
-    hipLaunchKernelGGL(uf_union_chunk_kernel,
-                       dim3(n2 > MAX_GRID ? MAX_GRID : n2), dim3(BLOCK), 0,
-                       s, n2, bin2, (V_ID)8192, row_ptr, col, row_left,
-                       parent);
-  if (n1)
-    hipLaunchKernelGGL(uf_union_wave_kernel,
-                       dim3(grid_for((uint64_t)n1 * WAVE)), dim3(BLOCK), 0,
-                       s, n1, bin1, row_ptr, col, row_left, parent);
-  if (n0)
-    hipLaunchKernelGGL(uf_union_thread_kernel, dim3(grid_for(n0)),
-                       dim3(BLOCK), 0, s, n0, bin0, row_ptr, col, row_left,
-                       parent);
+  if (n2) {
+    dim3 g2(n2 > MAX_GRID ? MAX_GRID : n2);
+    if (gbits)
+      hipLaunchKernelGGL(uf_union_chunk_kernel<true>, g2, dim3(BLOCK), 0, s,
+                         n2, bin2, (V_ID)8192, row_ptr, col, row_left,
+                         parent, gbits);
+    else
+      hipLaunchKernelGGL(uf_union_chunk_kernel<false>, g2, dim3(BLOCK), 0,
+                         s, n2, bin2, (V_ID)8192, row_ptr, col, row_left,
+                         parent, gbits);
+  }
+  if (n1) {
+    dim3 g1(grid_for((uint64_t)n1 * WAVE));
+    if (gbits)
+      hipLaunchKernelGGL(uf_union_wave_kernel<true>, g1, dim3(BLOCK), 0, s,
+                         n1, bin1, row_ptr, col, row_left, parent, gbits);
+    else
+      hipLaunchKernelGGL(uf_union_wave_kernel<false>, g1, dim3(BLOCK), 0, s,
+                         n1, bin1, row_ptr, col, row_left, parent, gbits);
+  }
+  if (n0) {
+    dim3 g0(grid_for(n0));
+    if (gbits)
+      hipLaunchKernelGGL(uf_union_thread_kernel<true>, g0, dim3(BLOCK), 0,
+                         s, n0, bin0, row_ptr, col, row_left, parent, gbits);
+    else
+      hipLaunchKernelGGL(uf_union_thread_kernel<false>, g0, dim3(BLOCK), 0,
+                         s, n0, bin0, row_ptr, col, row_left, parent, gbits);
+  }
+  LUX_POST_LAUNCH(stream);
+}
+
+void lux_gpu_uf_union_kth(uint64_t stream, V_ID vp, const E_ID* row_ptr,
+                          const V_ID* col, V_ID row_left, V_ID* parent,
+                          uint32_t k) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(uf_union_kth_kernel, dim3(grid_for(vp)), dim3(BLOCK),
+                     0, s, vp, row_ptr, col, row_left, parent, k);
+  LUX_POST_LAUNCH(stream);
+}
+
+void lux_gpu_cc_giant_bits(uint64_t stream, V_ID nv, const V_ID* labels,
+                           V_ID giant, uint32_t* bits) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(cc_giant_bits_kernel,
+                     dim3(grid_for((uint64_t)(nv + 31) / 32)), dim3(BLOCK),
+                     0, s, nv, labels, giant, bits);
   LUX_POST_LAUNCH(stream);
 }
 
