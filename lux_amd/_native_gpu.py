@@ -108,17 +108,6 @@ def csr_scatter(stream, ep, col, row_ptr_loc, vp, row_left, cursor,
                               dp(cursor), dp(push_col))
 
 
-def push_scatter(stream, is_min, old_dense, new_dense, in_row_left, in_count,
-                 old_seg, push_row_ptr, push_col, old_labels, snapshot,
-                 new_labels, my_row_left, new_seg, capacity):
-    lib().lux_gpu_push_scatter(
-        _u64(stream), ctypes.c_int(is_min), ctypes.c_int(old_dense),
-        ctypes.c_int(new_dense), _u32(in_row_left), _u32(in_count),
-        dp(old_seg), dp(push_row_ptr), dp(push_col), dp(old_labels),
-        dp(snapshot), dp(new_labels), _u32(my_row_left), dp(new_seg),
-        _u32(capacity))
-
-
 def build_bitmap(stream, vp, snapshot, new_labels, seg):
     lib().lux_gpu_build_bitmap(_u64(stream), _u32(vp), dp(snapshot),
                                dp(new_labels), dp(seg))
@@ -194,12 +183,6 @@ def push_chunk_scatter(stream, is_min, new_dense, items, counter, max_items,
         dp(items), dp(counter), _u32(max_items), dp(push_row_ptr),
         dp(push_col), dp(old_labels), dp(snapshot), dp(new_labels),
         _u32(my_row_left), dp(new_seg), _u32(capacity))
-
-
-def uf_union_edges(stream, ep, col, row_ptr_loc, vp, row_left, parent):
-    lib().lux_gpu_uf_union_edges(_u64(stream), _u64(ep), dp(col),
-                                 dp(row_ptr_loc), _u32(vp), _u32(row_left),
-                                 dp(parent))
 
 
 def uf_union_star(stream, nv, star, parent):

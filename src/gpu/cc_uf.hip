@@ -62,24 +62,6 @@ __device__ __forceinline__ void uf_union(V_ID a, V_ID b, V_ID* parent) {
   }
 }
 
-// Edge-parallel union over this rank's local CSC slice: edge j belongs to
-// dst row v (binary search over the local row_ptr, as csr_scatter_kernel).
-__global__ void uf_union_edges_kernel(uint64_t ep, const V_ID* col,
-                                      const E_ID* row_ptr_loc, V_ID vp,
-                                      V_ID row_left, V_ID* parent) {
-  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
-  for (uint64_t j = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; j < ep;
-       j += stride) {
-    V_ID lo = 0, hi = vp - 1;
-    while (lo < hi) {
-      V_ID mid = (lo + hi + 1) >> 1;
-      if (row_ptr_loc[mid] <= j) lo = mid;
-      else hi = mid - 1;
-    }
-    uf_union(col[j], lo + row_left, parent);
-  }
-}
-
 // Degree-binned variants (reuse the pull engine's bin lists): no per-edge
 // row binary search, and the dst side's find amortizes to once per vertex
 // (uf_union re-finds internally, so a stale dst root stays correct).
@@ -215,16 +197,6 @@ __global__ void uf_flatten_kernel(V_ID nv, V_ID* parent, V_ID* labels) {
 using namespace lux;
 
 extern "C" {
-
-void lux_gpu_uf_union_edges(uint64_t stream, uint64_t ep, const V_ID* col,
-                            const E_ID* row_ptr_loc, V_ID vp, V_ID row_left,
-                            V_ID* parent) {
-  hipStream_t s = (hipStream_t)stream;
-  if (ep == 0) return;
-  hipLaunchKernelGGL(uf_union_edges_kernel, dim3(grid_for(ep)), dim3(BLOCK),
-                     0, s, ep, col, row_ptr_loc, vp, row_left, parent);
-  LUX_POST_LAUNCH(stream);
-}
 
 void lux_gpu_uf_union_binned(uint64_t stream, uint32_t n0, const V_ID* bin0,
                              uint32_t n1, const V_ID* bin1, uint32_t n2,

@@ -65,129 +65,13 @@ __global__ void csr_scatter_kernel(uint64_t ep, const V_ID* col,
   }
 }
 
-// ---------------- frontier scatter (the hot push kernel) ----------------
-
-// One launch per (my partition x source segment). Block = 256 threads
-// (4 waves) cooperatively drains the block's 256 frontier vertices' edges:
-// per-thread out-degree -> block exclusive scan (LDS, wave64 ladder) ->
-// block-stride edge loop with a forward srcIdx walk (amortized O(1) per
-// edge, matching sssp_gpu.cu:198-244).
-template <bool IS_MIN, bool OLD_DENSE, bool NEW_DENSE>
-__global__ void push_scatter_kernel(
-    V_ID in_row_left, V_ID in_count,        // dense: segment vert range base/count; sparse: queue count
-    const uint8_t* old_seg,                 // segment bytes incl header
-    const E_ID* push_row_ptr,               // u64[nv+1]
-    const V_ID* push_col,
-    const uint32_t* old_labels,             // u32[nv] replicated
-    const uint32_t* snapshot,               // u32[vp] labels at iter start
-    uint32_t* new_labels,                   // u32[vp]
-    V_ID my_row_left,
-    uint8_t* new_seg,                       // my segment (sparse out) or null
-    V_ID capacity) {
-  using OP = LabOp<IS_MIN>;
-  __shared__ uint32_t src_label[BLOCK];
-  __shared__ unsigned long long ex_off[BLOCK / WAVE + 1];
-  __shared__ unsigned long long edge_base[BLOCK];
-  __shared__ unsigned long long prefix[BLOCK];
-  __shared__ uint32_t lds_scan[BLOCK / WAVE + 1];
-  __shared__ uint32_t queue_base;
-
-  const uint8_t* old_bitmap = old_seg + sizeof(FrontierHeader);
-  const V_ID* old_queue = (const V_ID*)(old_seg + sizeof(FrontierHeader));
-  V_ID* new_queue = nullptr;
-  uint32_t* num_nodes = nullptr;
-  if (!NEW_DENSE) {
-    num_nodes = &((FrontierHeader*)new_seg)->numNodes;
-    new_queue = (V_ID*)(new_seg + sizeof(FrontierHeader));
-  }
-
-  for (V_ID blk_start = blockIdx.x * blockDim.x; blk_start < in_count;
-       blk_start += blockDim.x * gridDim.x) {
-    V_ID idx = blk_start + threadIdx.x;
-    unsigned long long my_edges = 0, my_base = 0;
-    uint32_t lab = 0;
-    if (idx < in_count) {
-      V_ID u;
-      bool active;
-      if (OLD_DENSE) {
-        u = in_row_left + idx;
-        active = (old_bitmap[idx >> 3] >> (idx & 7)) & 1;
-      } else {
-        u = old_queue[idx];
-        active = true;
-      }
-      if (active) {
-        my_base = push_row_ptr[u];
-        my_edges = push_row_ptr[u + 1] - my_base;
-        lab = old_labels[u];
-      }
-    }
-    src_label[threadIdx.x] = lab;
-    __syncthreads();
-    unsigned long long total;
-    unsigned long long ex =
-        block_exscan<unsigned long long, BLOCK>(my_edges, ex_off, &total);
-    prefix[threadIdx.x] = ex;
-    edge_base[threadIdx.x] = my_base;
-    __syncthreads();
-
-    unsigned long long done = 0;
-    int si = 0;
-    while (done < total) {
-      unsigned long long k = done + threadIdx.x;
-      uint32_t flag = 0;
-      V_ID dstv = 0;
-      if (k < total) {
-        while (si + 1 < BLOCK && k >= prefix[si + 1]) si++;
-        E_ID e = edge_base[si] + (k - prefix[si]);
-        V_ID v = push_col[e];
-        uint32_t new_lab = OP::map(src_label[si]);
-        uint32_t* slot = &new_labels[v - my_row_left];
-        uint32_t cur = __hip_atomic_load(slot, __ATOMIC_RELAXED,
-                                         __HIP_MEMORY_SCOPE_AGENT);
-        if (OP::better(new_lab, cur)) {
-          if (NEW_DENSE) {
-            OP::atom(slot, new_lab);
-          } else {
-            // first-improvement enqueue (sssp_gpu.cu:63-82): the thread
-            // whose atomic moved the label off its iteration-start value
-            // owns the enqueue — exactly once per vertex per iteration.
-            uint32_t last = snapshot[v - my_row_left];
-            uint32_t act = OP::atom(slot, new_lab);
-            if (act == last) {
-              flag = 1;
-              dstv = v;
-            }
-          }
-        }
-      }
-      if (!NEW_DENSE) {
-        __syncthreads();
-        uint32_t q_total;
-        uint32_t q_off = block_exscan<uint32_t, BLOCK>(flag, lds_scan,
-                                                       &q_total);
-        if (threadIdx.x == 0 && q_total)
-          queue_base = atomicAdd(num_nodes, q_total);
-        __syncthreads();
-        if (flag && q_total) {
-          uint32_t pos = queue_base + q_off;
-          if (pos < capacity) new_queue[pos] = dstv;
-        }
-        __syncthreads();
-      }
-      done += blockDim.x;
-    }
-    __syncthreads();
-  }
-}
-
 // ---------------- edge-balanced scatter (expand + chunk) ----------------
-// The block-cooperative kernel above (reference parity, sssp_gpu.cu:132-246)
-// drains each 256-frontier-vertex group in ONE block: a hub's whole
-// out-edge range serializes there (measured: the 1-vertex first SSSP
+// The reference's kernel (sssp_gpu.cu:132-246) drains each
+// 256-frontier-vertex group in ONE block, so a hub's whole out-edge range
+// serializes there (measured with a faithful port: the 1-vertex first SSSP
 // iteration from RMAT-27's root took 58 ms — one block walked ~4M edges
 // while 255 CUs idled; the 902K-vertex second iteration 104 ms on its
-// worst block). The MI355X path splits every active vertex's range into
+// worst block). The MI355X design splits every active vertex's range into
 // <=PUSH_CHUNK-edge work items first (LDS-aggregated append), then one
 // grid covers all items — edge-balanced regardless of degree skew, and
 // one launch per iteration instead of one per source segment.
@@ -404,34 +288,6 @@ void lux_gpu_csr_scatter(uint64_t stream, uint64_t ep, const V_ID* col,
   hipStream_t s = (hipStream_t)stream;
   hipLaunchKernelGGL(csr_scatter_kernel, dim3(grid_for(ep)), dim3(BLOCK), 0,
                      s, ep, col, row_ptr_loc, vp, row_left, cursor, push_col);
-  LUX_POST_LAUNCH(stream);
-}
-
-void lux_gpu_push_scatter(uint64_t stream, int is_min, int old_dense,
-                          int new_dense, V_ID in_row_left, V_ID in_count,
-                          const uint8_t* old_seg, const E_ID* push_row_ptr,
-                          const V_ID* push_col, const uint32_t* old_labels,
-                          const uint32_t* snapshot, uint32_t* new_labels,
-                          V_ID my_row_left, uint8_t* new_seg, V_ID capacity) {
-  hipStream_t s = (hipStream_t)stream;
-  if (in_count == 0) return;
-  int grid = grid_for(in_count);
-  auto launch = [&](auto kern) {
-    hipLaunchKernelGGL(kern, dim3(grid), dim3(BLOCK), 0, s, in_row_left,
-                       in_count, old_seg, push_row_ptr, push_col, old_labels,
-                       snapshot, new_labels, my_row_left, new_seg, capacity);
-  };
-  int key = (is_min ? 4 : 0) | (old_dense ? 2 : 0) | (new_dense ? 1 : 0);
-  switch (key) {
-    case 0: launch(push_scatter_kernel<false, false, false>); break;
-    case 1: launch(push_scatter_kernel<false, false, true>); break;
-    case 2: launch(push_scatter_kernel<false, true, false>); break;
-    case 3: launch(push_scatter_kernel<false, true, true>); break;
-    case 4: launch(push_scatter_kernel<true, false, false>); break;
-    case 5: launch(push_scatter_kernel<true, false, true>); break;
-    case 6: launch(push_scatter_kernel<true, true, false>); break;
-    case 7: launch(push_scatter_kernel<true, true, true>); break;
-  }
   LUX_POST_LAUNCH(stream);
 }
 

@@ -71,16 +71,6 @@ void lux_gpu_csr_scatter(uint64_t stream, uint64_t ep, const lux::V_ID* col,
                          const lux::E_ID* row_ptr_loc, lux::V_ID vp,
                          lux::V_ID row_left, unsigned long long* cursor,
                          lux::V_ID* push_col);
-void lux_gpu_push_scatter(uint64_t stream, int is_min, int old_dense,
-                          int new_dense, lux::V_ID in_row_left,
-                          lux::V_ID in_count, const uint8_t* old_seg,
-                          const lux::E_ID* push_row_ptr,
-                          const lux::V_ID* push_col,
-                          const uint32_t* old_labels,
-                          const uint32_t* snapshot, uint32_t* new_labels,
-                          lux::V_ID my_row_left, uint8_t* new_seg,
-                          lux::V_ID capacity);
-
 void lux_gpu_frontier_expand(uint64_t stream, int old_dense,
                              lux::V_ID in_row_left, lux::V_ID in_count,
                              const uint8_t* old_seg,
