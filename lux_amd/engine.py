@@ -252,7 +252,8 @@ class GraphPart:
         self.build_bins()
         if getattr(self, "blocks", None) is not None:
             return
-        shift = force_shift if force_shift is not None else LLC_BLOCK_SHIFT
+        shift = force_shift if force_shift is not None else int(
+            __import__("os").environ.get("LUX_BLOCK_SHIFT", LLC_BLOCK_SHIFT))
         bounds = self.pull_bounds(shift)
         if self.ep == 0 or len(bounds) <= 2:
             self.blocks = None
