@@ -23,9 +23,10 @@ def _stream():
     return torch.cuda.current_stream().cuda_stream
 
 
-LLC_BLOCK_SHIFT = 24  # 2^24 verts * 4 B = 64 MB gather window: measured best
-# on RMAT-27 (95.6 vs 86.4 GTEPS at 128 MB; the 256 MB LLC also carries the
-# col stream and newv partials)
+LLC_BLOCK_SHIFT = 23  # 2^23 verts * 4 B = 32 MB gather window: measured
+# best on RMAT-27 (112.8 / 95.6 / 86.4 / 83.0 GTEPS at 32/64/128/256 MB —
+# the 256 MB LLC also carries the col stream and newv partials; below 32 MB
+# the per-sweep row overhead wins)
 
 
 def _bins_for(row_ptr, vp, ep, device, compact=False):
@@ -257,6 +258,11 @@ class GraphPart:
         shift = force_shift if force_shift is not None else int(
             __import__("os").environ.get("LUX_BLOCK_SHIFT", LLC_BLOCK_SHIFT))
         bounds = self.pull_bounds(shift)
+        # the blocked-build scan size is sb*vp (u32): widen windows until it
+        # fits (at RMAT-27 shift 22 gives exactly 2^32 and silently wrapped)
+        while (len(bounds) - 1) * max(self.vp, 1) >= (1 << 32):
+            shift += 1
+            bounds = self.pull_bounds(shift)
         if self.ep == 0 or len(bounds) <= 2:
             self.blocks = None
             return
