@@ -178,3 +178,35 @@ def test_dist_pagerank_async_pipeline(world):
     g = Graph.rmat(scale, ne, seed=seed)
     want = cpu_ref.pagerank(g, iters)
     np.testing.assert_allclose(got, want, rtol=1e-6)
+
+
+def _dist_bytes_worker(rank, world, port, outq):
+    """Uneven uint8 segment exchange — the frontier (header+payload byte
+    segment) all-gather shape PushEngine uses (push_engine.py step)."""
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from lux_amd import dist as dx
+        seg_bytes = [48, 96]  # uneven, 16-aligned like _align16 segments
+        off = [0, 48]
+        total = sum(seg_bytes)
+        mine = torch.full((seg_bytes[rank],), 10 + rank, dtype=torch.uint8)
+        mine[0] = 7  # "header" byte
+        full = torch.zeros(total, dtype=torch.uint8)
+        dx.all_gather_slices(full, mine, seg_bytes, off)
+        if rank == 0:
+            outq.put(full.numpy().copy())
+    finally:
+        dist.destroy_process_group()
+
+
+def test_dist_byte_segments(world=2):
+    ctx = mp.get_context("spawn")
+    outq = ctx.SimpleQueue()
+    mp.spawn(_dist_bytes_worker, args=(world, _find_port(), outq),
+             nprocs=world, join=True)
+    got = outq.get()
+    assert got[0] == 7 and got[48] == 7
+    assert (got[1:48] == 10).all() and (got[49:] == 11).all()
