@@ -99,14 +99,16 @@ __global__ void cf_wave_kernel(uint32_t n, const V_ID* binlist, CFArgs a) {
 // Row stride 65 words makes both access patterns bank-conflict-free
 // ((e+k) mod 32 distinct within each 32-lane group).
 // ---------------------------------------------------------------------------
-constexpr int CF_TILE = 64;
-constexpr int CF_ROW = CF_TILE + 1;
-constexpr int CF_TB = 64;  // one wave per workgroup: LDS 17.2 KB -> 9 wg/CU
+constexpr int CF_ROW = 65;  // dim pitch: 64 dims + bank pad
+constexpr int CF_TB = 64;   // one wave per workgroup
 
-struct CFTileLds {
-  float S[CF_TILE * CF_ROW];
-  float dv[CF_TILE];
-  float err[CF_TILE];
+// TILE = edges staged per LDS tile. 64: 17.2 KB/wave (9 wg/CU); 32:
+// 8.8 KB/wave (16 wg/CU cap) — more waves to queue LLC gathers, but the
+// lane=edge dot phase runs half-idle. A/B via LUX_CF_TILE=32|64.
+template <int TILE> struct CFTileLds {
+  float S[TILE * CF_ROW];
+  float dv[64];
+  float err[TILE];
 };
 
 // Stage up to CF_TILE edges [t, t+rem) into lds->S and per-lane col/weight;
@@ -118,7 +120,8 @@ struct CFTileLds {
 // measured 35.9 ms/sweep on the NetFlix config, WORSE than the un-tiled v1).
 // Batching 16 loads into registers before the LDS writes keeps 16 reads in
 // flight per wave.
-__device__ __forceinline__ float cf_stage_tile(CFTileLds* lds, E_ID t,
+template <int TILE>
+__device__ __forceinline__ float cf_stage_tile(CFTileLds<TILE>* lds, E_ID t,
                                                int rem, const CFArgs& a,
                                                int lane) {
   uint32_t mycol = 0;
@@ -127,9 +130,9 @@ __device__ __forceinline__ float cf_stage_tile(CFTileLds* lds, E_ID t,
     mycol = a.col[t + lane];
     myw = (float)a.w[t + lane];
   }
-  if (rem == CF_TILE) {
+  if (rem == TILE) {
 #pragma unroll
-    for (int r0 = 0; r0 < CF_TILE; r0 += 16) {
+    for (int r0 = 0; r0 < TILE; r0 += 16) {
       float tmp[16];
 #pragma unroll
       for (int r = 0; r < 16; r++) {
@@ -161,24 +164,28 @@ __device__ __forceinline__ float cf_stage_tile(CFTileLds* lds, E_ID t,
   return myw;
 }
 
-// Full-tile (rem==64), K==64 pass: every loop statically unrolled.
-__device__ __forceinline__ float cf_tile_pass_fast(CFTileLds* lds, float myw,
-                                                   float acc, int lane) {
+// Full-tile (rem==TILE), K==64 pass: every loop statically unrolled.
+template <int TILE>
+__device__ __forceinline__ float cf_tile_pass_fast(CFTileLds<TILE>* lds,
+                                                   float myw, float acc,
+                                                   int lane) {
   wave_lds_sync();
-  const float* row = &lds->S[lane * CF_ROW];
-  float d0 = 0, d1 = 0, d2 = 0, d3 = 0;
+  if (lane < TILE) {  // lane = edge; rows only exist for lane < TILE
+    const float* row = &lds->S[lane * CF_ROW];
+    float d0 = 0, d1 = 0, d2 = 0, d3 = 0;
 #pragma unroll
-  for (int k = 0; k < CF_TILE; k += 4) {
-    d0 += row[k] * lds->dv[k];
-    d1 += row[k + 1] * lds->dv[k + 1];
-    d2 += row[k + 2] * lds->dv[k + 2];
-    d3 += row[k + 3] * lds->dv[k + 3];
+    for (int k = 0; k < 64; k += 4) {
+      d0 += row[k] * lds->dv[k];
+      d1 += row[k + 1] * lds->dv[k + 1];
+      d2 += row[k + 2] * lds->dv[k + 2];
+      d3 += row[k + 3] * lds->dv[k + 3];
+    }
+    lds->err[lane] = myw - ((d0 + d1) + (d2 + d3));
   }
-  lds->err[lane] = myw - ((d0 + d1) + (d2 + d3));
   wave_lds_sync();
   float a0 = 0, a1 = 0, a2 = 0, a3 = 0;
 #pragma unroll
-  for (int r = 0; r < CF_TILE; r += 4) {
+  for (int r = 0; r < TILE; r += 4) {
     a0 += lds->err[r] * lds->S[r * CF_ROW + lane];
     a1 += lds->err[r + 1] * lds->S[(r + 1) * CF_ROW + lane];
     a2 += lds->err[r + 2] * lds->S[(r + 2) * CF_ROW + lane];
@@ -189,22 +196,25 @@ __device__ __forceinline__ float cf_tile_pass_fast(CFTileLds* lds, float myw,
 }
 
 // dot+err+update over a staged tile; returns updated acc (lane=dim).
-__device__ __forceinline__ float cf_tile_pass(CFTileLds* lds, int rem,
+template <int TILE>
+__device__ __forceinline__ float cf_tile_pass(CFTileLds<TILE>* lds, int rem,
                                               float myw, int K, float acc,
                                               int lane) {
   wave_lds_sync();  // S rows visible
-  float d0 = 0, d1 = 0, d2 = 0, d3 = 0;
-  const float* row = &lds->S[lane * CF_ROW];
-  int k = 0;
-  for (; k + 3 < K; k += 4) {
-    d0 += row[k] * lds->dv[k];
-    d1 += row[k + 1] * lds->dv[k + 1];
-    d2 += row[k + 2] * lds->dv[k + 2];
-    d3 += row[k + 3] * lds->dv[k + 3];
+  if (lane < TILE) {  // lane = edge; rows only exist for lane < TILE
+    float d0 = 0, d1 = 0, d2 = 0, d3 = 0;
+    const float* row = &lds->S[lane * CF_ROW];
+    int k = 0;
+    for (; k + 3 < K; k += 4) {
+      d0 += row[k] * lds->dv[k];
+      d1 += row[k + 1] * lds->dv[k + 1];
+      d2 += row[k + 2] * lds->dv[k + 2];
+      d3 += row[k + 3] * lds->dv[k + 3];
+    }
+    for (; k < K; k++) d0 += row[k] * lds->dv[k];
+    float dot = (d0 + d1) + (d2 + d3);
+    lds->err[lane] = lane < rem ? myw - dot : 0.0f;
   }
-  for (; k < K; k++) d0 += row[k] * lds->dv[k];
-  float dot = (d0 + d1) + (d2 + d3);
-  lds->err[lane] = lane < rem ? myw - dot : 0.0f;
   wave_lds_sync();  // err visible
   float a0 = 0, a1 = 0, a2 = 0, a3 = 0;
   int r = 0;
@@ -220,29 +230,32 @@ __device__ __forceinline__ float cf_tile_pass(CFTileLds* lds, int rem,
 }
 
 // accumulate gradient over edge range [b, e); lane = dim
-__device__ __forceinline__ float cf_range_acc(CFTileLds* lds, E_ID b, E_ID e,
-                                              const CFArgs& a, int lane) {
+template <int TILE>
+__device__ __forceinline__ float cf_range_acc(CFTileLds<TILE>* lds, E_ID b,
+                                              E_ID e, const CFArgs& a,
+                                              int lane) {
   float acc = 0.0f;
   E_ID t = b;
-  if (a.K == CF_TILE) {
-    for (; t + CF_TILE <= e; t += CF_TILE) {
-      float myw = cf_stage_tile(lds, t, CF_TILE, a, lane);
-      acc = cf_tile_pass_fast(lds, myw, acc, lane);
+  if (a.K == 64) {
+    for (; t + TILE <= e; t += TILE) {
+      float myw = cf_stage_tile<TILE>(lds, t, TILE, a, lane);
+      acc = cf_tile_pass_fast<TILE>(lds, myw, acc, lane);
     }
   }
-  for (; t < e; t += CF_TILE) {
-    int rem = (int)(e - t < CF_TILE ? e - t : (E_ID)CF_TILE);
-    float myw = cf_stage_tile(lds, t, rem, a, lane);
-    acc = cf_tile_pass(lds, rem, myw, a.K, acc, lane);
+  for (; t < e; t += TILE) {
+    int rem = (int)(e - t < TILE ? e - t : (E_ID)TILE);
+    float myw = cf_stage_tile<TILE>(lds, t, rem, a, lane);
+    acc = cf_tile_pass<TILE>(lds, rem, myw, a.K, acc, lane);
   }
   return acc;
 }
 
 // one wave per dst vertex over a bin list (deg < T2)
+template <int TILE>
 __global__ __launch_bounds__(CF_TB) void cf_tile_kernel(uint32_t n,
                                                         const V_ID* binlist,
                                                         CFArgs a) {
-  __shared__ CFTileLds lds;
+  __shared__ CFTileLds<TILE> lds;
   int lane = threadIdx.x;
   uint64_t nwaves = gridDim.x;
   for (uint64_t i = blockIdx.x; i < n; i += nwaves) {
@@ -250,15 +263,16 @@ __global__ __launch_bounds__(CF_TB) void cf_tile_kernel(uint32_t n,
     E_ID b = a.row_ptr[v], e = a.row_ptr[v + 1];
     lds.dv[lane] =
         lane < a.K ? a.oldv[(uint64_t)(a.row_left + v) * a.K + lane] : 0.0f;
-    float acc = cf_range_acc(&lds, b, e, a, lane);
+    float acc = cf_range_acc<TILE>(&lds, b, e, a, lane);
     if (lane < a.K) a.newv[(uint64_t)v * a.K + lane] += CF_GAMMA * acc;
   }
 }
 
 // one wave per hub chunk (deg >= T2), atomic epilogue
+template <int TILE>
 __global__ __launch_bounds__(CF_TB) void cf_tile_chunk_kernel(
     uint32_t n2, const uint2* bin2, V_ID chunk_edges, CFArgs a) {
-  __shared__ CFTileLds lds;
+  __shared__ CFTileLds<TILE> lds;
   int lane = threadIdx.x;
   for (uint32_t i = blockIdx.x; i < n2; i += gridDim.x) {
     uint2 ent = bin2[i];
@@ -268,7 +282,7 @@ __global__ __launch_bounds__(CF_TB) void cf_tile_chunk_kernel(
     if (e > b + chunk_edges) e = b + chunk_edges;
     lds.dv[lane] =
         lane < a.K ? a.oldv[(uint64_t)(a.row_left + v) * a.K + lane] : 0.0f;
-    float acc = cf_range_acc(&lds, b, e, a, lane);
+    float acc = cf_range_acc<TILE>(&lds, b, e, a, lane);
     if (lane < a.K)
       atomicAdd(&a.newv[(uint64_t)v * a.K + lane], CF_GAMMA * acc);
   }
@@ -336,16 +350,30 @@ void lux_gpu_cf_iter(uint64_t stream, uint32_t n0, const V_ID* bin0,
                      V_ID row_left, int K) {
   hipStream_t s = (hipStream_t)stream;
   CFArgs a{row_ptr, col, w, oldv, newv, row_left, K};
-  if (K <= CF_TILE) {
+  if (K <= 64) {
     // LDS-tiled fast path (benchmark config K=64; reference K=20)
-    if (nbig)
-      hipLaunchKernelGGL(cf_tile_chunk_kernel,
-                         dim3(n2 > MAX_GRID ? MAX_GRID : n2), dim3(CF_TB), 0,
-                         s, n2, bin2, (V_ID)8192, a);
-    if (n1)
-      hipLaunchKernelGGL(cf_tile_kernel,
-                         dim3(n1 > MAX_GRID ? MAX_GRID : n1), dim3(CF_TB), 0,
-                         s, n1, bin1, a);
+    static int tile = [] {
+      const char* t = getenv("LUX_CF_TILE");
+      return t && atoi(t) == 32 ? 32 : 64;
+    }();
+    if (nbig) {
+      dim3 g(n2 > MAX_GRID ? MAX_GRID : n2);
+      if (tile == 32)
+        hipLaunchKernelGGL(cf_tile_chunk_kernel<32>, g, dim3(CF_TB), 0, s,
+                           n2, bin2, (V_ID)8192, a);
+      else
+        hipLaunchKernelGGL(cf_tile_chunk_kernel<64>, g, dim3(CF_TB), 0, s,
+                           n2, bin2, (V_ID)8192, a);
+    }
+    if (n1) {
+      dim3 g(n1 > MAX_GRID ? MAX_GRID : n1);
+      if (tile == 32)
+        hipLaunchKernelGGL(cf_tile_kernel<32>, g, dim3(CF_TB), 0, s, n1,
+                           bin1, a);
+      else
+        hipLaunchKernelGGL(cf_tile_kernel<64>, g, dim3(CF_TB), 0, s, n1,
+                           bin1, a);
+    }
     if (n0)
       hipLaunchKernelGGL(cf_wave_kernel, dim3(grid_for((uint64_t)n0 * WAVE)),
                          dim3(BLOCK), 0, s, n0, bin0, a);
