@@ -34,7 +34,9 @@ namespace lux {
 
 constexpr int ALS_K = 64;        // max latent rank of the MFMA path
 constexpr int ALS_ROW = 65;      // LDS row pitch (bank-conflict pad)
-constexpr int ALS_TILE = 64;     // edges staged per LDS tile
+constexpr int ALS_TILE = 32;     // edges staged per LDS tile (32: half the
+                                 // LDS per wave -> ~2x waves/CU; same
+                                 // occupancy lesson as cf.hip TILE=32)
 constexpr int ALS_TB = 64;       // one wave per workgroup
 
 using f32x4 = __attribute__((ext_vector_type(4))) float;
@@ -78,7 +80,7 @@ __device__ __forceinline__ void als_gram_range(const CFAlsArgs& a, E_ID b,
       mycol = a.col[t + lane];
       myw = (float)a.w[t + lane];
     }
-    lds->W[lane] = myw;
+    if (lane < ALS_TILE) lds->W[lane] = myw;
     // depth-16 batched loads: a dynamic-trip rolled loop would leave one
     // outstanding global read per edge (see cf.hip cf_stage_tile note)
     if (rem == ALS_TILE) {
