@@ -90,7 +90,7 @@ __global__ void cf_wave_kernel(uint32_t n, const V_ID* binlist, CFArgs a) {
 //
 // The generic wave kernel above serialises a ~500-cycle dependent chain per
 // edge (gather -> dot -> 6-step shuffle reduce -> broadcast -> axpy). Here a
-// wave stages CF_TILE=64 edges' src vectors in an LDS tile S[64][65] with a
+// wave stages TILE (32|64) edges' src vectors in an LDS tile S[TILE][65], a
 // coalesced lane=dim load, then switches lane mapping twice:
 //   dot phase    lane=edge: each lane walks its OWN row serially against the
 //                dst vector (LDS broadcast reads) -> 64 independent FMA

@@ -13,9 +13,9 @@
 //
 // MI355X mapping: the Gram matrix G = S^T S is the dense hot loop —
 // K^2 * deg MACs per vertex — and is exactly MFMA-shaped. One 64-lane wave
-// owns one dst vertex, stages 64-edge tiles of src vectors in LDS, and
-// accumulates the upper-triangular 16x16 tiles of G (K <= 64 -> 4x4 tile
-// grid, 10 upper tiles) with v_mfma_f32_16x16x4_f32: for each 4-edge group,
+// owns one dst vertex, stages ALS_TILE-edge tiles of src vectors in LDS,
+// and accumulates the upper-triangular 16x16 tiles of G (K <= 64 -> 4x4
+// tile grid, 10 upper tiles) with v_mfma_f32_16x16x4_f32: per 4-edge group,
 // fragment f[t] = S[e0 + (lane>>4)][t*16 + (lane&15)] serves as operand A of
 // tile-row t AND operand B of tile-col t (A[i][k]=S[e0+k][16ti+i],
 // B[k][j]=S[e0+k][16tj+j] per the gfx950 f32 MFMA lane maps), so 4 LDS reads
