@@ -7,11 +7,14 @@
 // multi-GPU path is `torchrun ... python -m lux_amd.apps.<app>` (one process
 // per GPU over RCCL). -ng > 1 here prints a pointer to that path.
 #pragma once
+#include <cstdint>
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
 #include <string>
+#include <vector>
 
+#include "../src/runtime/single_gpu.h"
 #include "lux/graph.h"
 
 struct AppArgs {
@@ -23,6 +26,8 @@ struct AppArgs {
   bool check = false;
   int k = 64;
   const char* synthetic = nullptr;
+  const char* dump = nullptr;  // write result vertex state (LUXS format,
+                               // lux_amd/checkpoint.py-compatible)
 };
 
 inline AppArgs parse_input_args(int argc, char** argv) {
@@ -38,6 +43,7 @@ inline AppArgs parse_input_args(int argc, char** argv) {
     else if (f == "-check" || f == "-c") a.check = true;
     else if (f == "-k") a.k = atoi(next());
     else if (f == "-synthetic") a.synthetic = next();
+    else if (f == "-dump") a.dump = next();
     else if (f.rfind("-ll:", 0) == 0 || f.rfind("-lg:", 0) == 0) {
       if (i + 1 < argc && argv[i + 1][0] != '-') i++;  // value-flag: skip
     } else {
@@ -45,6 +51,23 @@ inline AppArgs parse_input_args(int argc, char** argv) {
     }
   }
   return a;
+}
+
+// Vertex-state dump: u32 magic 'LUXS' | u32 dtype (0=f32,1=u32) | u32 K |
+// u32 nv | u64 iteration | payload (lux_amd/checkpoint.py format).
+inline void dump_state(const char* path, const void* dev_ptr, int dtype,
+                       uint32_t k, uint32_t nv, uint64_t iter) {
+  std::vector<char> host((size_t)nv * k * 4);
+  LUX_OK(hipMemcpy(host.data(), dev_ptr, host.size(),
+                   hipMemcpyDeviceToHost));
+  FILE* f = fopen(path, "wb");
+  if (!f) { perror(path); return; }
+  uint32_t hdr[4] = {0x5358554Cu, (uint32_t)dtype, k, nv};
+  fwrite(hdr, 4, 4, f);
+  fwrite(&iter, 8, 1, f);
+  fwrite(host.data(), 1, host.size(), f);
+  fclose(f);
+  printf("[lux] wrote %s (nv=%u k=%u)\n", path, nv, k);
 }
 
 inline bool load_graph(const AppArgs& a, lux::HostCSC* g, bool weighted) {

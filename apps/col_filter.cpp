@@ -37,6 +37,8 @@ int main(int argc, char** argv) {
   auto t1 = std::chrono::steady_clock::now();
   double secs = std::chrono::duration<double>(t1 - t0).count();
   printf("ELAPSED TIME = %7.7f s\n", secs);
+  if (a.dump)
+    dump_state(a.dump, engine.vectors(), 0, (uint32_t)a.k, g.nv, (uint64_t)a.num_iter);
   printf("[lux] %.3f GTEPS (%d sweeps, rank %d)\n",
          double(g.ne) * a.num_iter / secs / 1e9, a.num_iter, a.k);
   return 0;

@@ -36,6 +36,8 @@ int main(int argc, char** argv) {
   auto t1 = std::chrono::steady_clock::now();
   double secs = std::chrono::duration<double>(t1 - t0).count();
   printf("ELAPSED TIME = %7.7f s\n", secs);
+  if (a.dump)
+    dump_state(a.dump, engine.labels(), 1, 1, g.nv, (uint64_t)a.num_iter);
   printf("[lux] converged in %d iterations, %.3f GTEPS\n", iters,
          double(g.ne) / secs / 1e9);
   if (a.check) {

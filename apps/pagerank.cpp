@@ -37,6 +37,8 @@ int main(int argc, char** argv) {
   auto t1 = std::chrono::steady_clock::now();
   double secs = std::chrono::duration<double>(t1 - t0).count();
   printf("ELAPSED TIME = %7.7f s\n", secs);
+  if (a.dump)
+    dump_state(a.dump, engine.ranks(), 0, 1, g.nv, (uint64_t)a.num_iter);
   printf("[lux] %.3f GTEPS (%d iterations, %llu edges)\n",
          double(g.ne) * a.num_iter / secs / 1e9, a.num_iter,
          (unsigned long long)g.ne);

@@ -52,3 +52,39 @@ def test_native_lux_file_roundtrip(tmp_path):
           "100000", "-o", lux])
     out = _run([f"{BIN}/pagerank", "-file", lux, "-ni", "3", "-verbose"])
     assert "ELAPSED TIME" in out
+
+
+def test_native_pagerank_dump_matches_cpu(tmp_path):
+    """Cross-implementation parity: the native binary's -dump state equals
+    the CPU reference PageRank on the same .lux file (the native PR path
+    has no -check oracle; this is its numeric validation)."""
+    import numpy as np
+    from lux_amd import checkpoint as ck
+    from lux_amd import cpu_ref
+    from lux_amd.graph import Graph
+    lux = str(tmp_path / "g.lux")
+    out = str(tmp_path / "pr.luxs")
+    _run([f"{BIN}/rmat_gen", "-kind", "rmat", "-scale", "13", "-ne",
+          "120000", "-o", lux])
+    _run([f"{BIN}/pagerank", "-file", lux, "-ni", "5", "-dump", out])
+    got, it = ck.load_state(out)
+    assert it == 5
+    g = Graph.load(lux)
+    want = cpu_ref.pagerank(g, 5)
+    np.testing.assert_allclose(got, want, rtol=2e-4, atol=1e-9)
+
+
+def test_native_sssp_dump_matches_cpu(tmp_path):
+    import numpy as np
+    from lux_amd import checkpoint as ck
+    from lux_amd import cpu_ref
+    from lux_amd.graph import Graph
+    lux = str(tmp_path / "g.lux")
+    out = str(tmp_path / "l.luxs")
+    _run([f"{BIN}/rmat_gen", "-kind", "rmat", "-scale", "12", "-ne",
+          "80000", "-o", lux])
+    _run([f"{BIN}/sssp", "-file", lux, "-start", "0", "-dump", out])
+    got, _ = ck.load_state(out)
+    g = Graph.load(lux)
+    want, _ = cpu_ref.sssp(g, 0)
+    np.testing.assert_array_equal(got, want)
