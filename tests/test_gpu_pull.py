@@ -234,3 +234,36 @@ def test_pipelined_subset_schedule_matches_cpu():
     want = cpu_ref.pagerank(g, iters)
     np.testing.assert_allclose(old.cpu().numpy(), want, rtol=2e-4,
                                atol=1e-9)
+
+
+def test_empty_graph_engines():
+    """0-edge graph: engines must not crash; PR gives the uniform rank,
+    labels stay at init."""
+    nv = 1024
+    col_end = torch.zeros(nv, dtype=torch.int64, device="cuda")
+    src = torch.zeros(1, dtype=U32, device="cuda")
+    full = DeviceCSC(nv, 0, col_end, src)
+    part = GraphPart(full, 1, 0)
+    eng = PagerankEngine(part)
+    eng.step()
+    r = eng.ranks().cpu().numpy()
+    np.testing.assert_allclose(r, np.full(nv, 1.0 / nv, np.float32),
+                               rtol=1e-6)
+
+
+def test_self_loops_and_duplicates():
+    """Self-loops and duplicate edges flow through build + engines (the
+    reference neither filters nor documents them; we keep them as-is)."""
+    from lux_amd.graph import Graph
+    src = [0, 0, 1, 2, 2, 2, 3]
+    dst = [0, 1, 1, 2, 2, 0, 3]
+    g = Graph.from_edges(8, src, dst)
+    want = cpu_ref.pagerank(g, 3)
+    s = torch.tensor(src, dtype=U32, device="cuda")
+    d = torch.tensor(dst, dtype=U32, device="cuda")
+    full = DeviceCSC._from_device_edges(8, 7, s, d, None, "cuda")
+    eng = PagerankEngine(GraphPart(full, 1, 0))
+    for _ in range(3):
+        eng.step()
+    np.testing.assert_allclose(eng.ranks().cpu().numpy(), want, rtol=2e-4,
+                               atol=1e-9)
