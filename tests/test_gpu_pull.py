@@ -247,7 +247,8 @@ def test_empty_graph_engines():
     eng = PagerankEngine(part)
     eng.step()
     r = eng.ranks().cpu().numpy()
-    np.testing.assert_allclose(r, np.full(nv, 1.0 / nv, np.float32),
+    # zero in-sum everywhere: pr = (1-alpha)/nv (stored undivided, deg=0)
+    np.testing.assert_allclose(r, np.full(nv, 0.85 / nv, np.float32),
                                rtol=1e-6)
 
 
