@@ -177,12 +177,18 @@ def frontier_expand(stream, old_dense, in_row_left, in_count, old_seg,
 
 def push_chunk_scatter(stream, is_min, new_dense, items, counter, max_items,
                        push_row_ptr, push_col, old_labels, snapshot,
-                       new_labels, my_row_left, new_seg, capacity):
+                       new_labels, my_row_left, new_seg, capacity,
+                       visited_bits=None):
     lib().lux_gpu_push_chunk_scatter(
         _u64(stream), ctypes.c_int(is_min), ctypes.c_int(new_dense),
         dp(items), dp(counter), _u32(max_items), dp(push_row_ptr),
         dp(push_col), dp(old_labels), dp(snapshot), dp(new_labels),
-        _u32(my_row_left), dp(new_seg), _u32(capacity))
+        _u32(my_row_left), dp(new_seg), _u32(capacity), dp(visited_bits))
+
+
+def bits_from_labels(stream, vp, labels, bits):
+    lib().lux_gpu_bits_from_labels(_u64(stream), _u32(vp), dp(labels),
+                                   dp(bits))
 
 
 def uf_union_star(stream, nv, star, parent):

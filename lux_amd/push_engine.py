@@ -90,6 +90,13 @@ class PushEngine:
         self.items = torch.empty(self.max_items * 2, dtype=U32,
                                  device=device)
         self.item_counter = torch.zeros(2, dtype=U32, device=device)
+        # hop-SSSP (level-synchronous BFS) visited bitmap: push discovery
+        # is a test-and-set against vp/8 bytes (L2-resident) instead of an
+        # atomicMin against the 4*vp label array (push.hip BFS_BITS path);
+        # rebuilt from labels after pull iterations
+        self.visited = torch.empty((p.vp + 31) // 32, dtype=U32,
+                                   device=device) if self.is_min else None
+        self._bits_stale = True
 
         # ---- labels + frontier state ----
         self.labels = torch.empty(p.nv, dtype=U32, device=device)
@@ -141,6 +148,7 @@ class PushEngine:
         self.headers = list(self._fq_init[1])
         self.iterations = 0
         self.stats = []
+        self._bits_stale = True
 
     # -------- helpers --------
     def _my_seg_i32(self):
@@ -188,19 +196,32 @@ class PushEngine:
             # pull sweep (identical labels per iteration, src-blocked
             # LLC-resident gathers) is faster beyond ~ep/8 edges.
             n_edges = int(self.item_counter[1].cpu().item())
-            if n_edges > p.ep // 8:
+            # bitmap-BFS push touches vp/8 bytes of visited bits instead of
+            # the label array, so it stays cheaper than a dense pull sweep
+            # up to much larger frontiers
+            thresh = p.ep // 2 if self.visited is not None else p.ep // 8
+            if n_edges > thresh:
                 pull_fallback = True
         if pull_fallback:
             new_dense = True
             mode = ng.PULL_MIN if self.is_min else ng.PULL_MAX
             run_pull(p, mode, self.labels, self.labels_part, None, 0.0)
+            self._bits_stale = True  # pull writes labels directly
         else:
+            bits = None
+            if self.visited is not None and p.vp > 0:
+                if self._bits_stale:
+                    ng.bits_from_labels(s, p.vp, self.labels_part,
+                                        self.visited)
+                    self._bits_stale = False
+                bits = self.visited
             ng.push_chunk_scatter(s, int(self.is_min), int(new_dense),
                                   self.items, self.item_counter,
                                   self.max_items, self.push_row_ptr,
                                   self.push_col, self.labels, self.snapshot,
                                   self.labels_part, p.row_left,
-                                  self.new_seg, self.capacity)
+                                  self.new_seg, self.capacity,
+                                  visited_bits=bits)
 
         # ---- frontier format fix-ups (sssp_gpu.cu:462-491) ----
         if new_dense:

@@ -129,13 +129,23 @@ __global__ void frontier_expand_kernel(int old_dense, V_ID in_row_left,
   }
 }
 
-template <bool IS_MIN, bool NEW_DENSE>
+// BFS (IS_MIN) fast path: hop-SSSP is level-synchronous BFS — every vertex
+// in an iteration's frontier carries the same depth, and a finite label is
+// final (the settled-skip invariant in pull.hip). Discovery is therefore a
+// one-shot test-and-set against a per-partition VISITED BITMAP (vp/8 bytes
+// — ~2 MB at RMAT-27/8 ranks, 16.8 MB at 1 rank: L2/LLC-resident) instead
+// of an atomicMin against the 4*vp-byte label array; the label is written
+// once, on discovery. NOT valid for general min-plus relaxation (two
+// frontier labels could race to a non-min value) — the engine enables it
+// only for its single-source hop traversal; IS_MIN without a bitmap and
+// CC (IS_MAX) keep the label-atomic path.
+template <bool IS_MIN, bool NEW_DENSE, bool BFS_BITS>
 __global__ void push_chunk_scatter_kernel(
     const uint2* items, const uint32_t* counter, uint32_t max_items,
     const E_ID* push_row_ptr, const V_ID* push_col,
     const uint32_t* old_labels, const uint32_t* snapshot,
     uint32_t* new_labels, V_ID my_row_left, uint8_t* new_seg,
-    V_ID capacity) {
+    V_ID capacity, uint32_t* visited_bits) {
   using OP = LabOp<IS_MIN>;
   __shared__ uint32_t lds_scan[BLOCK / WAVE + 1];
   __shared__ uint32_t queue_base;
@@ -162,19 +172,35 @@ __global__ void push_chunk_scatter_kernel(
       V_ID dstv = 0;
       if (k < e) {
         V_ID v = push_col[k];
-        uint32_t* slot = &new_labels[v - my_row_left];
-        uint32_t cur = __hip_atomic_load(slot, __ATOMIC_RELAXED,
-                                         __HIP_MEMORY_SCOPE_AGENT);
-        if (OP::better(new_lab, cur)) {
-          if (NEW_DENSE) {
-            OP::atom(slot, new_lab);
-          } else {
-            // first-improvement enqueue (sssp_gpu.cu:63-82)
-            uint32_t last = snapshot[v - my_row_left];
-            uint32_t act = OP::atom(slot, new_lab);
-            if (act == last) {
-              flag = 1;
-              dstv = v;
+        V_ID lv = v - my_row_left;
+        if (BFS_BITS) {
+          uint32_t word = visited_bits[lv >> 5];
+          uint32_t bit = 1u << (lv & 31);
+          if (!(word & bit)) {
+            uint32_t old = atomicOr(&visited_bits[lv >> 5], bit);
+            if (!(old & bit)) {  // this thread discovered v
+              new_labels[lv] = new_lab;
+              if (!NEW_DENSE) {
+                flag = 1;
+                dstv = v;
+              }
+            }
+          }
+        } else {
+          uint32_t* slot = &new_labels[lv];
+          uint32_t cur = __hip_atomic_load(slot, __ATOMIC_RELAXED,
+                                           __HIP_MEMORY_SCOPE_AGENT);
+          if (OP::better(new_lab, cur)) {
+            if (NEW_DENSE) {
+              OP::atom(slot, new_lab);
+            } else {
+              // first-improvement enqueue (sssp_gpu.cu:63-82)
+              uint32_t last = snapshot[lv];
+              uint32_t act = OP::atom(slot, new_lab);
+              if (act == last) {
+                flag = 1;
+                dstv = v;
+              }
             }
           }
         }
@@ -194,6 +220,23 @@ __global__ void push_chunk_scatter_kernel(
         __syncthreads();
       }
     }
+  }
+}
+
+// visited_bits[w] bit i <=> labels[32w+i] != INF (rebuilt after pull
+// iterations, which write labels directly)
+__global__ void bits_from_labels_kernel(V_ID vp, const uint32_t* labels,
+                                        uint32_t* bits) {
+  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
+  V_ID nw = (vp + 31) / 32;
+  for (uint64_t w = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; w < nw;
+       w += stride) {
+    uint32_t word = 0;
+    V_ID base = (V_ID)w * 32;
+    int n = vp - base < 32 ? (int)(vp - base) : 32;
+    for (int i = 0; i < n; i++)
+      if (labels[base + i] != INF_LABEL) word |= 1u << i;
+    bits[w] = word;
   }
 }
 
@@ -313,36 +356,42 @@ void lux_gpu_push_chunk_scatter(uint64_t stream, int is_min, int new_dense,
                                 const uint32_t* old_labels,
                                 const uint32_t* snapshot,
                                 uint32_t* new_labels, V_ID my_row_left,
-                                uint8_t* new_seg, V_ID capacity) {
+                                uint8_t* new_seg, V_ID capacity,
+                                uint32_t* visited_bits /*nullable; IS_MIN
+                                level-synchronous BFS fast path*/) {
   hipStream_t s = (hipStream_t)stream;
   dim3 grid(MAX_GRID), blk(BLOCK);
   // grid-strided over the device-side item count: no host sync needed;
   // surplus blocks read the counter and exit
-  if (is_min) {
-    if (new_dense)
-      hipLaunchKernelGGL((push_chunk_scatter_kernel<true, true>), grid, blk,
-                         0, s, items, counter, max_items, push_row_ptr,
-                         push_col, old_labels, snapshot, new_labels,
-                         my_row_left, new_seg, capacity);
-    else
-      hipLaunchKernelGGL((push_chunk_scatter_kernel<true, false>), grid,
-                         blk, 0, s, items, counter, max_items, push_row_ptr,
-                         push_col, old_labels, snapshot, new_labels,
-                         my_row_left, new_seg, capacity);
+#define LUX_PCS(MIN_, DENSE_, BITS_)                                       \
+  hipLaunchKernelGGL((push_chunk_scatter_kernel<MIN_, DENSE_, BITS_>),     \
+                     grid, blk, 0, s, items, counter, max_items,           \
+                     push_row_ptr, push_col, old_labels, snapshot,         \
+                     new_labels, my_row_left, new_seg, capacity,           \
+                     visited_bits)
+  if (is_min && visited_bits) {
+    if (new_dense) LUX_PCS(true, true, true);
+    else LUX_PCS(true, false, true);
+  } else if (is_min) {
+    if (new_dense) LUX_PCS(true, true, false);
+    else LUX_PCS(true, false, false);
   } else {
-    if (new_dense)
-      hipLaunchKernelGGL((push_chunk_scatter_kernel<false, true>), grid,
-                         blk, 0, s, items, counter, max_items, push_row_ptr,
-                         push_col, old_labels, snapshot, new_labels,
-                         my_row_left, new_seg, capacity);
-    else
-      hipLaunchKernelGGL((push_chunk_scatter_kernel<false, false>), grid,
-                         blk, 0, s, items, counter, max_items, push_row_ptr,
-                         push_col, old_labels, snapshot, new_labels,
-                         my_row_left, new_seg, capacity);
+    if (new_dense) LUX_PCS(false, true, false);
+    else LUX_PCS(false, false, false);
   }
+#undef LUX_PCS
   LUX_POST_LAUNCH(stream);
 }
+
+void lux_gpu_bits_from_labels(uint64_t stream, V_ID vp,
+                              const uint32_t* labels, uint32_t* bits) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(bits_from_labels_kernel,
+                     dim3(grid_for((uint64_t)(vp + 31) / 32)), dim3(BLOCK),
+                     0, s, vp, labels, bits);
+  LUX_POST_LAUNCH(stream);
+}
+
 
 void lux_gpu_build_bitmap(uint64_t stream, V_ID vp, const uint32_t* snapshot,
                           const uint32_t* new_labels, uint8_t* seg) {

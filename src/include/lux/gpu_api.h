@@ -85,7 +85,10 @@ void lux_gpu_push_chunk_scatter(uint64_t stream, int is_min, int new_dense,
                                 const uint32_t* old_labels,
                                 const uint32_t* snapshot,
                                 uint32_t* new_labels, lux::V_ID my_row_left,
-                                uint8_t* new_seg, lux::V_ID capacity);
+                                uint8_t* new_seg, lux::V_ID capacity,
+                                uint32_t* visited_bits /*nullable*/);
+void lux_gpu_bits_from_labels(uint64_t stream, lux::V_ID vp,
+                              const uint32_t* labels, uint32_t* bits);
 
 void lux_gpu_build_bitmap(uint64_t stream, lux::V_ID vp,
                           const uint32_t* snapshot,

@@ -232,7 +232,7 @@ V_ID SingleGpuPush::step() {
                                new_dense ? 1 : 0, items_, item_counter_,
                                max_items_, push_row_ptr_, push_col_,
                                snapshot_, snapshot_, labels_, 0, new_fq_,
-                               capacity_);
+                               capacity_, nullptr);
   }
   FrontierHeader hh;
   if (new_dense) {
