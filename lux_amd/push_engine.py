@@ -202,6 +202,12 @@ class PushEngine:
             thresh = p.ep // 2 if self.visited is not None else p.ep // 8
             if n_edges > thresh:
                 pull_fallback = True
+            elif n_edges // 16 > self.capacity:
+                # expected discoveries cannot fit the sparse queue: choose
+                # the dense bitmap upfront instead of paying the sparse
+                # append machinery + guaranteed overflow rebuild (the
+                # reference always votes by input formats, sssp_gpu.cu:408)
+                new_dense = True
         if pull_fallback:
             new_dense = True
             mode = ng.PULL_MIN if self.is_min else ng.PULL_MAX
