@@ -7,7 +7,6 @@ expressed directly on one HIP stream per rank, with RCCL collectives over
 xGMI replacing the reference's zero-copy host staging
 (pull_app_task_impl, pagerank_gpu.cu:105-151).
 """
-import math
 
 import torch
 
