@@ -68,56 +68,79 @@ struct AlsLds {
 // ---- Gram accumulation over one vertex's edge range [b, e) ----
 // acc[10] are the upper tiles (ti<=tj) in order (0,0)(0,1)(0,2)(0,3)
 // (1,1)(1,2)(1,3)(2,2)(2,3)(3,3); rhs is lane=dim.
+//
+// Tile-level software pipeline: the NEXT tile's src-vector loads issue into
+// registers (tmp[ALS_TILE]) right before the CURRENT tile's MFMA burst
+// (~80 x 32-cycle issue at ALS_TILE=32), so their ~LLC latency hides under
+// the matrix work; one LDS buffer suffices (regs -> LDS at tile start).
+
+// Issue the whole tile's vector loads into registers (all outstanding).
+__device__ __forceinline__ void als_vec_loads(const CFAlsArgs& a,
+                                              uint32_t mycol, int rem,
+                                              int lane,
+                                              float tmp[ALS_TILE]) {
+#pragma unroll
+  for (int r = 0; r < ALS_TILE; r++) {
+    uint32_t src = __shfl(mycol, r, WAVE);
+    tmp[r] = (r < rem && lane < a.K)
+                 ? a.oldv[(uint64_t)src * a.K + lane]
+                 : 0.0f;
+  }
+}
+
 __device__ __forceinline__ void als_gram_range(const CFAlsArgs& a, E_ID b,
                                                E_ID e, int lane,
                                                AlsGramLds* lds,
                                                f32x4 acc[10], float* rhs) {
+  if (b >= e) return;
+  int rem = (int)(e - b < ALS_TILE ? e - b : (E_ID)ALS_TILE);
+  uint32_t mycol = 0;
+  float myw = 0.0f;
+  if (lane < rem) {
+    mycol = a.col[b + lane];
+    myw = (float)a.w[b + lane];
+  }
+  float tmp[ALS_TILE];
+  als_vec_loads(a, mycol, rem, lane, tmp);
   for (E_ID t = b; t < e; t += ALS_TILE) {
-    int rem = (int)(e - t < ALS_TILE ? e - t : (E_ID)ALS_TILE);
-    uint32_t mycol = 0;
-    float myw = 0.0f;
-    if (lane < rem) {
-      mycol = a.col[t + lane];
-      myw = (float)a.w[t + lane];
-    }
+    // stage the (prefetched) current tile
     if (lane < ALS_TILE) lds->W[lane] = myw;
-    // depth-16 batched loads: a dynamic-trip rolled loop would leave one
-    // outstanding global read per edge (see cf.hip cf_stage_tile note)
-    if (rem == ALS_TILE) {
 #pragma unroll
-      for (int r0 = 0; r0 < ALS_TILE; r0 += 16) {
-        float tmp[16];
-#pragma unroll
-        for (int r = 0; r < 16; r++) {
-          uint32_t src = __shfl(mycol, r0 + r, WAVE);
-          tmp[r] = lane < a.K ? a.oldv[(uint64_t)src * a.K + lane] : 0.0f;
-        }
-#pragma unroll
-        for (int r = 0; r < 16; r++)
-          lds->S[(r0 + r) * ALS_ROW + lane] = tmp[r];
-      }
-    } else {
-      for (int r = 0; r < rem; r++) {
-        uint32_t src = __shfl(mycol, r, WAVE);
-        lds->S[r * ALS_ROW + lane] =
-            lane < a.K ? a.oldv[(uint64_t)src * a.K + lane] : 0.0f;
+    for (int r = 0; r < ALS_TILE; r++)
+      lds->S[r * ALS_ROW + lane] = tmp[r];
+    int rem_cur = rem;
+    // prefetch next tile's col/weight now, vectors after the rhs pass
+    E_ID t2 = t + ALS_TILE;
+    bool more = t2 < e;
+    uint32_t ncol = 0;
+    float nw = 0.0f;
+    if (more) {
+      rem = (int)(e - t2 < ALS_TILE ? e - t2 : (E_ID)ALS_TILE);
+      if (lane < rem) {
+        ncol = a.col[t2 + lane];
+        nw = (float)a.w[t2 + lane];
       }
     }
-    int rem4 = (rem + 3) & ~3;
-    for (int r = rem; r < rem4; r++) lds->S[r * ALS_ROW + lane] = 0.0f;
+    int rem4 = (rem_cur + 3) & ~3;
     als_lds_sync();
     // rhs += sum_r w_r * S[r][lane] (4 independent LDS-read chains)
     float r0 = 0, r1 = 0, r2 = 0, r3 = 0;
     int rr = 0;
-    for (; rr + 4 <= rem; rr += 4) {
+    for (; rr + 4 <= rem_cur; rr += 4) {
       r0 += lds->W[rr] * lds->S[rr * ALS_ROW + lane];
       r1 += lds->W[rr + 1] * lds->S[(rr + 1) * ALS_ROW + lane];
       r2 += lds->W[rr + 2] * lds->S[(rr + 2) * ALS_ROW + lane];
       r3 += lds->W[rr + 3] * lds->S[(rr + 3) * ALS_ROW + lane];
     }
-    for (; rr < rem; rr++)
+    for (; rr < rem_cur; rr++)
       r0 += lds->W[rr] * lds->S[rr * ALS_ROW + lane];
     *rhs += ((r0 + r1) + (r2 + r3));
+    // issue the next tile's vector loads; they land under the MFMA burst
+    if (more) {
+      myw = nw;
+      als_vec_loads(a, ncol, rem, lane, tmp);
+      mycol = ncol;
+    }
     // MFMA over 4-edge groups
     int erow = lane >> 4, ecol = lane & 15;
     for (int kk = 0; kk < rem4; kk += 4) {
