@@ -295,10 +295,13 @@ class GraphPart:
         begin = 0
         for b in range(sb):
             end = int(ends[(b + 1) * vp - 1].item())
+            if end == begin:
+                continue  # empty src window (e.g. all-zero-degree tail
+                #           under the hot-source permutation): no sweeps
             row_ptr_b = torch.empty(vp + 1, dtype=U64, device=device)
             ng.local_row_ptr(s, vp, begin, ends.narrow(0, b * vp, vp),
                              row_ptr_b)
-            col_b = blk_col.narrow(0, begin, max(end - begin, 1))
+            col_b = blk_col.narrow(0, begin, end - begin)
             n0, n1, n2, nbig, b0, b1, b2, b2v = _bins_for(
                 row_ptr_b, vp, end - begin, device, compact=True)
             local = self.nparts > 1 and bounds[b] >= self.row_left \
