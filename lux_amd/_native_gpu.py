@@ -215,12 +215,3 @@ def uf_union_kth(stream, vp, row_ptr, col, row_left, parent, k):
 def cc_giant_bits(stream, nv, labels, giant, bits):
     lib().lux_gpu_cc_giant_bits(_u64(stream), _u32(nv), dp(labels),
                                 _u32(giant), dp(bits))
-
-
-def gather_f32(stream, n, vals, perm, out):
-    lib().lux_gpu_gather_f32(_u64(stream), _u32(n), dp(vals), dp(perm),
-                             dp(out))
-
-
-def remap_u32(stream, n, ids, mp):
-    lib().lux_gpu_remap_u32(_u64(stream), _u64(n), dp(ids), dp(mp))

@@ -331,20 +331,10 @@ class PagerankEngine:
         ng.hist_u32(s, part.ep, part.col, deg)
         dx.all_reduce_sum_(deg)
         self.deg = deg
-        # LUX_PR_PERM=1 (single-GPU experiment): degree-descending SOURCE
-        # permutation — the gather array old_p is rebuilt hot-first each
-        # iteration so the hottest sources compact into the first (L2-hot)
-        # part of the first 32 MB window; cols store permuted positions.
-        import os
-        self.perm = None
-        if os.environ.get("LUX_PR_PERM") == "1" and dx.world_size() == 1:
-            perm = torch.argsort(deg, descending=True,
-                                 stable=True).to(U32)
-            inv = torch.empty_like(perm)
-            inv[perm.long()] = torch.arange(part.nv, dtype=U32,
-                                            device=device)
-            ng.remap_u32(s, part.ep, part.col, inv)
-            self.perm = perm
+        # (a degree-descending source permutation was measured here and
+        # REGRESSED: 25.5 vs 19.2 ms/iter on RMAT-27 — sorting destroys the
+        # generator's natural hot-prefix locality and concentrates the
+        # per-sweep row folds into one window; see BENCHLOG r1)
         part.prepare_pull()
         # init: rank/deg (deg==0 -> rank), pagerank_gpu.cu:255-259
         rank0 = 1.0 / part.nv
@@ -352,8 +342,6 @@ class PagerankEngine:
         self.old = torch.where(deg == 0, torch.full_like(degf, rank0),
                                rank0 / degf.clamp(min=1.0))
         self.new_part = torch.empty(part.vp, dtype=F32, device=device)
-        self.old_p = torch.empty_like(self.old) if self.perm is not None \
-            else None
         self.init_rank = (1.0 - 0.15) / part.nv
         # pipelined state: cur_part = my slice's latest values; the
         # all-gather publishing them into `old` may still be in flight
@@ -373,13 +361,8 @@ class PagerankEngine:
 
     def _step_body(self):
         p = self.part
-        if self.perm is not None:
-            ng.gather_f32(_stream(), p.nv, self.old, self.perm, self.old_p)
-            run_pull(p, ng.PULL_PR, self.old_p, self.new_part, self.deg,
-                     self.init_rank)
-        else:
-            run_pull(p, ng.PULL_PR, self.old, self.new_part, self.deg,
-                     self.init_rank)
+        run_pull(p, ng.PULL_PR, self.old, self.new_part, self.deg,
+                 self.init_rank)
         self.old.narrow(0, p.row_left, p.vp).copy_(self.new_part)
 
     def step(self):

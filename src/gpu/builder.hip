@@ -208,26 +208,6 @@ __global__ void blocked_scatter_kernel(uint64_t ep, const V_ID* col,
   }
 }
 
-// ---------------- source-permutation helpers (PR hot-source layout) ------
-// out[i] = vals[perm[i]] — build the degree-sorted gather array each
-// iteration; and col[j] = inv[col[j]] — one-time remap of edge sources
-// into permuted positions (hot sources compact into the window prefix).
-
-__global__ void gather_f32_kernel(V_ID n, const float* vals,
-                                  const V_ID* perm, float* out) {
-  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
-  for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; i < n;
-       i += stride)
-    out[i] = vals[perm[i]];
-}
-
-__global__ void remap_u32_kernel(uint64_t n, V_ID* ids, const V_ID* map) {
-  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
-  for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; i < n;
-       i += stride)
-    ids[i] = map[ids[i]];
-}
-
 // ---------------- local row_ptr from global col_end slice ----------------
 
 __global__ void local_row_ptr_kernel(uint32_t vp, E_ID col_left,
@@ -331,22 +311,6 @@ void lux_gpu_blocked_scatter(uint64_t stream, uint64_t ep, const V_ID* col,
   hipLaunchKernelGGL(blocked_scatter_kernel, dim3(grid_for(ep)), dim3(BLOCK),
                      0, s, ep, col, row_ptr_loc, vp, bounds, nb, cursor,
                      out_col);
-  LUX_POST_LAUNCH(stream);
-}
-
-void lux_gpu_gather_f32(uint64_t stream, V_ID n, const float* vals,
-                        const V_ID* perm, float* out) {
-  hipStream_t s = (hipStream_t)stream;
-  hipLaunchKernelGGL(gather_f32_kernel, dim3(grid_for(n)), dim3(BLOCK), 0,
-                     s, n, vals, perm, out);
-  LUX_POST_LAUNCH(stream);
-}
-
-void lux_gpu_remap_u32(uint64_t stream, uint64_t n, V_ID* ids,
-                       const V_ID* map) {
-  hipStream_t s = (hipStream_t)stream;
-  hipLaunchKernelGGL(remap_u32_kernel, dim3(grid_for(n)), dim3(BLOCK), 0, s,
-                     n, ids, map);
   LUX_POST_LAUNCH(stream);
 }
 
