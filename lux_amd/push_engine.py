@@ -52,47 +52,21 @@ class PushEngine:
         # dense fallback
         part.prepare_pull()
 
-        # ---- push CSR(s): all nv sources -> my-partition dsts ----
-        # The scatter's random writes (visited bits / labels) land in a
-        # vp-sized target; when its bit array exceeds the 4 MB per-XCD L2
-        # (vp > 32M), the dst range is split into WINDOWS, one push CSR per
-        # window (its edges are a CONTIGUOUS slice of the local CSC), and
-        # the scatter runs window-by-window with the window's bits
-        # L2-resident. W==1 (vp <= 32M, e.g. every rank at N>=4) is the
-        # plain single-CSR layout.
-        WIN_VERTS = 32 * 1024 * 1024
-        nwin = max((p.vp + WIN_VERTS - 1) // WIN_VERTS, 1)
-        self.win = []
-        self.push_col = torch.empty(max(p.ep, 1), dtype=U32, device=device)
-        deg_src = torch.empty(p.nv, dtype=U32, device=device)
+        # ---- push CSR: all nv sources -> my-partition dsts ----
+        deg_src = torch.zeros(p.nv, dtype=U32, device=device)
+        ng.hist_u32(s, p.ep, p.col, deg_src)
         ends = torch.empty(p.nv, dtype=U64, device=device)
         partials = torch.empty(ng.scan_partials_size(p.nv), dtype=U64,
                                device=device)
-        col_base = 0
-        for w in range(nwin):
-            wl = w * WIN_VERTS
-            vp_w = min(WIN_VERTS, p.vp - wl)
-            e_lo = int(p.row_ptr[wl].item())
-            e_hi = int(p.row_ptr[wl + vp_w].item())
-            ep_w = e_hi - e_lo
-            col_w = p.col.narrow(0, e_lo, max(ep_w, 1))
-            row_ptr_w = p.row_ptr.narrow(0, wl, vp_w + 1) - e_lo
-            deg_src.zero_()
-            ng.hist_u32(s, ep_w, col_w, deg_src)
-            ng.scan_end_offsets(s, p.nv, deg_src, ends, partials)
-            prp = torch.empty(p.nv + 1, dtype=U64, device=device)
-            ng.local_row_ptr(s, p.nv, 0, ends, prp)
-            if col_base:
-                prp += col_base  # window region base inside push_col
-            cursor = prp[:p.nv].clone()
-            ng.csr_scatter(s, ep_w, col_w, row_ptr_w, vp_w,
-                           p.row_left + wl, cursor, self.push_col)
-            torch.cuda.synchronize()  # row_ptr_w/cursor temporaries
-            self.win.append(dict(row_ptr=prp, wl=wl, vp_w=vp_w))
-            col_base += ep_w
+        ng.scan_end_offsets(s, p.nv, deg_src, ends, partials)
+        self.push_row_ptr = torch.empty(p.nv + 1, dtype=U64, device=device)
+        ng.local_row_ptr(s, p.nv, 0, ends, self.push_row_ptr)
+        cursor = self.push_row_ptr[:p.nv].clone()
+        self.push_col = torch.empty(max(p.ep, 1), dtype=U32, device=device)
+        ng.csr_scatter(s, p.ep, p.col, p.row_ptr, p.vp, p.row_left, cursor,
+                       self.push_col)
         torch.cuda.synchronize()
-        self.push_row_ptr = self.win[0]["row_ptr"]  # W==1 fast reference
-        del deg_src, ends, partials
+        del deg_src, ends, partials, cursor
 
         # ---- frontier buffers ----
         self.seg_bytes = [_align16(frontier_bytes(v)) for v in p.verts_all]
@@ -110,12 +84,12 @@ class PushEngine:
         self.hdr_idx = torch.tensor(idx, dtype=torch.long, device=device)
 
         # ---- edge-balanced scatter work items (push.hip expand+chunk):
-        # push runs only while total frontier <= nv/16, so items per dst
-        # window are bounded by nv/16 active vertices + ep/8192 chunks ----
+        # push runs only while total frontier <= nv/16, so items are bounded
+        # by nv/16 active vertices + ep/8192 extra chunks ----
         self.max_items = p.nv // 16 + p.ep // 8192 + 1024
-        self.items = torch.empty(nwin * self.max_items * 2, dtype=U32,
+        self.items = torch.empty(self.max_items * 2, dtype=U32,
                                  device=device)
-        self.item_counter = torch.zeros(nwin * 2, dtype=U32, device=device)
+        self.item_counter = torch.zeros(2, dtype=U32, device=device)
         # hop-SSSP (level-synchronous BFS) visited bitmap: push discovery
         # is a test-and-set against vp/8 bytes (L2-resident) instead of an
         # atomicMin against the 4*vp label array (push.hip BFS_BITS path);
@@ -198,37 +172,30 @@ class PushEngine:
 
         pull_fallback = old_fq_size > p.nv // 16
         if not pull_fallback:
-            # expand all source segments into <=8192-edge work items, one
-            # item list per dst window (each window's scatter then writes
-            # an L2-resident slice of the visited bits / labels)
+            # expand all source segments into <=8192-edge work items
             self.item_counter.zero_()
-            for w, win in enumerate(self.win):
-                items_w = self.items.narrow(0, w * self.max_items * 2,
-                                            self.max_items * 2)
-                ctr_w = self.item_counter.narrow(0, w * 2, 2)
-                for q in range(nparts):
-                    typ, num = self.headers[q]
-                    if p.verts_all[q] == 0:
-                        continue
-                    seg = self.fq_all.narrow(0, int(self.seg_off[q]),
-                                             self.seg_bytes[q])
-                    if typ == DENSE_BITMAP:
-                        ng.frontier_expand(s, 1, p.row_left_all[q],
-                                           p.verts_all[q], seg,
-                                           win["row_ptr"], items_w, ctr_w,
-                                           self.max_items)
-                    elif num:
-                        ng.frontier_expand(s, 0, 0, num, seg,
-                                           win["row_ptr"], items_w, ctr_w,
-                                           self.max_items)
+            for q in range(nparts):
+                typ, num = self.headers[q]
+                if p.verts_all[q] == 0:
+                    continue
+                seg = self.fq_all.narrow(0, int(self.seg_off[q]),
+                                         self.seg_bytes[q])
+                if typ == DENSE_BITMAP:
+                    ng.frontier_expand(s, 1, p.row_left_all[q],
+                                       p.verts_all[q], seg,
+                                       self.push_row_ptr, self.items,
+                                       self.item_counter, self.max_items)
+                elif num:
+                    ng.frontier_expand(s, 0, 0, num, seg,
+                                       self.push_row_ptr, self.items,
+                                       self.item_counter, self.max_items)
             # second adaptivity axis (ours, not the reference's): the
             # vertex-count threshold misses RMAT's hub explosion — a 902K-
             # vertex frontier can cover ~half of all edges. The expand
             # kernel counts the frontier's out-edges (counter[1]); a dense
             # pull sweep (identical labels per iteration, src-blocked
             # LLC-resident gathers) is faster beyond ~ep/8 edges.
-            n_edges = int(self.item_counter.cpu().view(-1, 2)[:, 1]
-                          .sum().item())
+            n_edges = int(self.item_counter[1].cpu().item())
             # bitmap-BFS push touches vp/8 bytes of visited bits instead of
             # the label array, so it stays cheaper than a dense pull sweep
             # up to much larger frontiers
@@ -254,17 +221,13 @@ class PushEngine:
                                         self.visited)
                     self._bits_stale = False
                 bits = self.visited
-            for w, win in enumerate(self.win):
-                items_w = self.items.narrow(0, w * self.max_items * 2,
-                                            self.max_items * 2)
-                ctr_w = self.item_counter.narrow(0, w * 2, 2)
-                ng.push_chunk_scatter(s, int(self.is_min), int(new_dense),
-                                      items_w, ctr_w, self.max_items,
-                                      win["row_ptr"], self.push_col,
-                                      self.labels, self.snapshot,
-                                      self.labels_part, p.row_left,
-                                      self.new_seg, self.capacity,
-                                      visited_bits=bits)
+            ng.push_chunk_scatter(s, int(self.is_min), int(new_dense),
+                                  self.items, self.item_counter,
+                                  self.max_items, self.push_row_ptr,
+                                  self.push_col, self.labels, self.snapshot,
+                                  self.labels_part, p.row_left,
+                                  self.new_seg, self.capacity,
+                                  visited_bits=bits)
 
         # ---- frontier format fix-ups (sssp_gpu.cu:462-491) ----
         if new_dense:
