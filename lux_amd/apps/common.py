@@ -91,6 +91,25 @@ def print_memory_estimate(nv, ne, nparts, weighted=False, k=1):
           f"(of 294912 MB HBM3E)")
 
 
+def run_traced(engine_step, num_iter, ne=None):
+    """Reference `-verbose` parity: per-iteration wall timing (loadTime/
+    compTime analog, sssp_gpu.cu:516-518) via IterTrace; prints the CSV and
+    a GTEPS summary. Synchronises per iteration — for -verbose runs only."""
+    import torch
+
+    from ..trace import IterTrace
+    tr = IterTrace()
+    for i in range(num_iter):
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        engine_step()
+        torch.cuda.synchronize()
+        tr.record(iter=i, ms=round(1000 * (time.perf_counter() - t0), 4))
+    print(tr.to_csv(), end="")
+    print(f"[lux] trace summary: {tr.summary(ne=ne)}")
+    return tr
+
+
 class ElapsedTimer:
     """Reference-format timing: wall clock around the iteration loop only,
     printed as 'ELAPSED TIME = ... s' (pagerank.cc:118, sssp.cc:137)."""

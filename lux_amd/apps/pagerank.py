@@ -28,8 +28,12 @@ def main(argv=None):
     part = GraphPart(full, dx.world_size(), dx.rank())
     eng = PagerankEngine(part)
     with ElapsedTimer():
-        for _ in range(a.num_iter):
-            eng.step()
+        if a.verbose and dx.rank() == 0 and dx.world_size() == 1:
+            from .common import run_traced
+            run_traced(eng.step, a.num_iter, ne=full.ne)
+        else:
+            for _ in range(a.num_iter):
+                eng.step()
     if a.verbose and dx.rank() == 0:
         r = eng.ranks()[:5].cpu().tolist()
         print("[lux] first ranks (pr/out_degree):", r)

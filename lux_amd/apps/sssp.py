@@ -45,6 +45,12 @@ def main(argv=None):
         iters = eng.run()
     if dx.rank() == 0:
         print(f"[lux] converged in {iters} iterations")
+    if a.verbose and dx.rank() == 0 and hasattr(eng, "stats"):
+        from ..trace import IterTrace
+        tr = IterTrace()
+        for row in eng.stats:
+            tr.record(**row)
+        print(tr.to_csv(), end="")
     if a.check:
         mistakes = eng.check()
         tag = "PASS" if mistakes == 0 else "FAIL"
