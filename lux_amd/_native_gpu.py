@@ -130,12 +130,12 @@ PULL_MAX = 2
 
 
 def pull_iter(stream, mode, n0, bin0, n1, bin1, n2, bin2, nbig, bin2v,
-              row_ptr, col, oldv, newv, deg, row_left, init_rank, phase=3):
+              row_ptr, col, oldv, newv, deg, row_left, init_rank):
     lib().lux_gpu_pull_iter(_u64(stream), ctypes.c_int(mode), _u32(n0),
                             dp(bin0), _u32(n1), dp(bin1), _u32(n2), dp(bin2),
                             _u32(nbig), dp(bin2v), dp(row_ptr), dp(col),
                             dp(oldv), dp(newv), dp(deg), _u32(row_left),
-                            ctypes.c_float(init_rank), ctypes.c_int(phase))
+                            ctypes.c_float(init_rank))
 
 
 def blocked_count(stream, ep, col, row_ptr_loc, vp, bounds, nb, counts):

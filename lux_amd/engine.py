@@ -67,12 +67,12 @@ def run_pull_sweeps(part, mode, oldv, newv, deg, init_rank, subset=None):
             ng.pull_iter(s, mode, blk["n0"], blk["bin0"], blk["n1"],
                          blk["bin1"], blk["n2"], blk["bin2"], blk["nbig"],
                          blk["bin2v"], blk["row_ptr"], blk["col"], oldv,
-                         newv, deg, part.row_left, init_rank, 0)
+                         newv, deg, part.row_left, init_rank)
     else:
         assert subset is None, "subset sweeps need the blocked CSC"
         ng.pull_iter(s, mode, part.n0, part.bin0, part.n1, part.bin1,
                      part.n2, part.bin2, part.nbig, part.bin2v, part.row_ptr,
-                     part.col, oldv, newv, deg, part.row_left, init_rank, 0)
+                     part.col, oldv, newv, deg, part.row_left, init_rank)
 
 
 def seed_pull(part, mode, oldv, newv):

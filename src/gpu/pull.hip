@@ -141,9 +141,6 @@ struct PullArgs {
   const V_ID* deg;       // u32[nv] out-degrees (PR) or null
   V_ID row_left;
   float init_rank;
-  int phase;             // bit0 FIRST (init/seed), bit1 LAST (finish); 3 =
-                         // single-sweep. MID sweeps (0) fold the partial
-                         // already in newv (src-blocked CSC iteration).
 };
 
 // Result store for the non-atomic bins. The iteration contract: newv is
@@ -153,11 +150,7 @@ struct PullArgs {
 // so the fold needs no atomics.
 template <PullMode M>
 __device__ __forceinline__ void store_result(typename Val<M>::T* slot,
-                                             typename Val<M>::T acc,
-                                             typename Val<M>::T own,
-                                             float init_rank, V_ID deg,
-                                             int phase) {
-  (void)own; (void)init_rank; (void)deg; (void)phase;
+                                             typename Val<M>::T acc) {
   *slot = Val<M>::comb(acc, *slot);
 }
 
@@ -201,8 +194,7 @@ __global__ void pull_thread_kernel(uint32_t n0, const V_ID* bin0,
       continue;  // settled hop label: newv keeps the seed
     E_ID b = a.row_ptr[v], e = a.row_ptr[v + 1];
     T acc = gather_range<M>(oldv, a.col, b, e, 1);
-    store_result<M>(&newv[v], acc, oldv[a.row_left + v], a.init_rank,
-                    a.deg ? a.deg[a.row_left + v] : 0, a.phase);
+    store_result<M>(&newv[v], acc);
   }
 }
 
@@ -224,8 +216,7 @@ __global__ void pull_wave_kernel(uint32_t n1, const V_ID* bin1, PullArgs a) {
     T acc = gather_range<M>(oldv, a.col, b + lane, e, WAVE);
     acc = V::reduce_wave(acc);
     if (lane == 0)
-      store_result<M>(&newv[v], acc, oldv[a.row_left + v], a.init_rank,
-                      a.deg ? a.deg[a.row_left + v] : 0, a.phase);
+      store_result<M>(&newv[v], acc);
   }
 }
 
@@ -321,10 +312,9 @@ void lux_gpu_pull_iter(uint64_t stream, int mode, uint32_t n0,
                        uint32_t n2, const uint2* bin2, uint32_t nbig,
                        const V_ID* bin2v, const E_ID* row_ptr,
                        const V_ID* col, const void* oldv, void* newv,
-                       const V_ID* deg, V_ID row_left, float init_rank,
-                       int phase) {
+                       const V_ID* deg, V_ID row_left, float init_rank) {
   hipStream_t s = (hipStream_t)stream;
-  PullArgs a{row_ptr, col, oldv, newv, deg, row_left, init_rank, phase};
+  PullArgs a{row_ptr, col, oldv, newv, deg, row_left, init_rank};
   switch (mode) {
     case 0:
       pull_iter<PR_SUM>(s, n0, bin0, n1, bin1, n2, bin2, nbig, bin2v, a);

@@ -61,7 +61,7 @@ void lux_gpu_pull_iter(uint64_t stream, int mode, uint32_t n0,
                        const lux::V_ID* bin2v, const lux::E_ID* row_ptr,
                        const lux::V_ID* col, const void* oldv, void* newv,
                        const lux::V_ID* deg, lux::V_ID row_left,
-                       float init_rank, int phase);
+                       float init_rank);
 void lux_gpu_pull_finish_pr(uint64_t stream, lux::V_ID vp, float* newv,
                             const lux::V_ID* deg, lux::V_ID row_left,
                             float init_rank);
