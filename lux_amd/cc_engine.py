@@ -43,7 +43,8 @@ class CCUnionFindEngine:
         # Afforest-style: sample-hook 2 edges/vertex, find the giant
         # component by sampling, then sweep remaining edges skipping
         # both-endpoints-in-giant via an L2-resident bitmap (cc_uf.hip)
-        for k in range(2):
+        import os
+        for k in range(int(os.environ.get("LUX_CC_SAMPLE_ROUNDS", "2"))):
             ng.uf_union_kth(s, p.vp, p.row_ptr, p.col, p.row_left,
                             self.parent, k)
         ng.uf_flatten(s, p.nv, self.parent, self.labels_t)
