@@ -89,7 +89,7 @@ def _dist_sssp_worker(rank, world, port, scale, ne, seed, outq):
         dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("world", [2])
+@pytest.mark.parametrize("world", [2, 4])
 def test_dist_pagerank_matches_single(world):
     from lux_amd import cpu_ref
     from lux_amd.graph import Graph
@@ -164,7 +164,7 @@ def _dist_pagerank_pipelined_worker(rank, world, port, scale, ne, seed,
         dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("world", [2])
+@pytest.mark.parametrize("world", [2, 4])
 def test_dist_pagerank_async_pipeline(world):
     from lux_amd import cpu_ref
     from lux_amd.graph import Graph
