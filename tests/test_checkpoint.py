@@ -70,3 +70,13 @@ def test_engine_checkpoint_roundtrip_gpu(tmp_path):
     np.testing.assert_allclose(a.ranks().cpu().numpy(),
                                b.ranks().cpu().numpy(), rtol=1e-5,
                                atol=1e-12)
+
+
+def test_save_engine_rank_nonzero_noop(tmp_path, monkeypatch):
+    """Distributed contract: only rank 0 writes (save_engine no-ops
+    elsewhere)."""
+    from lux_amd import dist as dx
+    monkeypatch.setattr(dx, "rank", lambda: 1)
+    path = tmp_path / "x.luxs"
+    ckpt.save_engine(str(path), object())  # unknown engine wouldn't matter
+    assert not path.exists()
