@@ -28,6 +28,7 @@ struct AppArgs {
   const char* synthetic = nullptr;
   const char* dump = nullptr;  // write result vertex state (LUXS format,
                                // lux_amd/checkpoint.py-compatible)
+  bool als = false;            // col_filter: exact MFMA ALS optimizer
 };
 
 inline AppArgs parse_input_args(int argc, char** argv) {
@@ -44,6 +45,7 @@ inline AppArgs parse_input_args(int argc, char** argv) {
     else if (f == "-k") a.k = atoi(next());
     else if (f == "-synthetic") a.synthetic = next();
     else if (f == "-dump") a.dump = next();
+    else if (f == "-als") a.als = true;
     else if (f.rfind("-ll:", 0) == 0 || f.rfind("-lg:", 0) == 0) {
       if (i + 1 < argc && argv[i + 1][0] != '-') i++;  // value-flag: skip
     } else {

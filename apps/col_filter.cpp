@@ -23,14 +23,20 @@ int main(int argc, char** argv) {
 
   hipStream_t s;
   LUX_OK(hipStreamCreate(&s));
+  if (a.als && a.k > 64) {
+    fprintf(stderr, "[lux] -als covers K <= 64 (MFMA tile grid)\n");
+    return 1;
+  }
   size_t arena_bytes = 8ull * g.nv + 8ull * g.ne          // graph + weights
                        + 8ull * (g.nv + 1)
                        + 12ull * g.nv + (64ull << 20)
                        + 8ull * (uint64_t)g.nv * a.k      // old/new vectors
-                       + 8ull * (g.ne / 8192 + g.nv / 16);
+                       + 8ull * (g.ne / 8192 + g.nv / 16)
+                       + (a.als ? 4ull * g.nv +           // hub slot map
+                            (17ull << 20) * (g.ne / 2048 / 1024 + 1) : 0);
   DeviceArena arena(arena_bytes);
   DeviceGraph dg = DeviceGraph::upload(g, arena, s);
-  SingleGpuCF engine(dg, a.k, arena, s);
+  SingleGpuCF engine(dg, a.k, arena, s, a.als);
 
   auto t0 = std::chrono::steady_clock::now();
   engine.iterate(a.num_iter);

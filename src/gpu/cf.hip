@@ -334,11 +334,29 @@ __global__ void cf_chunk_kernel(uint32_t n2, const uint2* bin2,
 }
 
 
+// SGD sweep output seed: new = old * (1 - GAMMA*LAMBDA); the sweep kernels
+// then ADD GAMMA*acc (colfilter_gpu.cu:96-102 split, see cf_wave_kernel).
+__global__ void cf_seed_kernel(uint64_t n, const float* oldv, float* newv) {
+  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
+  const float f = 1.0f - CF_GAMMA * CF_LAMBDA;
+  for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    newv[i] = oldv[i] * f;
+}
+
 }  // namespace lux
 
 using namespace lux;
 
 extern "C" {
+
+void lux_gpu_cf_seed(uint64_t stream, uint64_t n, const float* oldv,
+                     float* newv) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(cf_seed_kernel, dim3(grid_for(n)), dim3(BLOCK), 0, s,
+                     n, oldv, newv);
+  LUX_POST_LAUNCH(stream);
+}
 
 // One CF sweep over my partition. Uses the pull.hip bin lists (bin0 and
 // bin1 both go to the wave kernel; bin2 chunked, chunk_edges = 8192).

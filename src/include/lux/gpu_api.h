@@ -101,6 +101,8 @@ void lux_gpu_check(uint64_t stream, int is_min, lux::V_ID vp,
                    unsigned long long* mistakes);
 
 // cf.hip
+void lux_gpu_cf_seed(uint64_t stream, uint64_t n, const float* oldv,
+                     float* newv);
 void lux_gpu_cf_iter(uint64_t stream, uint32_t n0, const lux::V_ID* bin0,
                      uint32_t n1, const lux::V_ID* bin1, uint32_t n2,
                      const lux_uint2* bin2, uint32_t nbig,
@@ -108,5 +110,15 @@ void lux_gpu_cf_iter(uint64_t stream, uint32_t n0, const lux::V_ID* bin0,
                      const lux::V_ID* col, const lux::WeightType* w,
                      const float* oldv, float* newv, lux::V_ID row_left,
                      int K);
+
+// cf_als.hip
+void lux_gpu_cf_als_iter(uint64_t stream, uint32_t n0, const lux::V_ID* bin0,
+                         uint32_t n1, const lux::V_ID* bin1, uint32_t n2,
+                         const lux_uint2* bin2, uint32_t nbig,
+                         const lux::V_ID* bin2v, const int* hubidx,
+                         float* gram_scratch, float* rhs_scratch,
+                         const lux::E_ID* row_ptr, const lux::V_ID* col,
+                         const lux::WeightType* w, const float* oldv,
+                         float* newv, lux::V_ID row_left, int K);
 
 }  // extern "C"

@@ -295,8 +295,8 @@ uint64_t SingleGpuPush::check() {
 // ---------------- CF ----------------
 
 SingleGpuCF::SingleGpuCF(const DeviceGraph& g, int K, DeviceArena& arena,
-                         hipStream_t s)
-    : g_(g), s_(s), K_(K) {
+                         hipStream_t s, bool als)
+    : g_(g), s_(s), K_(K), als_(als) {
   row_ptr_ = arena.alloc_n<E_ID>(g.nv + 1);
   lux_gpu_local_row_ptr((uint64_t)s, g.nv, 0, g.col_end, row_ptr_);
   bins_.build(row_ptr_, g.nv, g.ne, arena, s);
@@ -305,14 +305,49 @@ SingleGpuCF::SingleGpuCF(const DeviceGraph& g, int K, DeviceArena& arena,
   std::vector<float> h((size_t)g.nv * K, sqrtf(1.0f / K));
   LUX_OK(hipMemcpyAsync(old_, h.data(), sizeof(float) * h.size(),
                            hipMemcpyHostToDevice, s));
+  if (als_ && bins_.nbig) {
+    // hub scratch: slot map (local id -> [0, nbig)), Gram + rhs
+    hubidx_ = arena.alloc_n<int>(g.nv);
+    std::vector<V_ID> hubs(bins_.nbig);
+    LUX_OK(hipMemcpy(hubs.data(), bins_.bin2v,
+                     sizeof(V_ID) * bins_.nbig, hipMemcpyDeviceToHost));
+    std::vector<int> hidx(g.nv, -1);
+    for (uint32_t i = 0; i < bins_.nbig; i++) hidx[hubs[i]] = (int)i;
+    LUX_OK(hipMemcpyAsync(hubidx_, hidx.data(), sizeof(int) * g.nv,
+                          hipMemcpyHostToDevice, s));
+    gram_ = arena.alloc_n<float>((size_t)bins_.nbig * 64 * 64);
+    rhs_ = arena.alloc_n<float>((size_t)bins_.nbig * 64);
+  }
   LUX_OK(hipStreamSynchronize(s));
 }
 
 void SingleGpuCF::iterate(int iters) {
+  size_t n = (size_t)g_.nv * K_;
   for (int it = 0; it < iters; it++) {
-    lux_gpu_cf_iter((uint64_t)s_, bins_.n0, bins_.bin0, bins_.n1, bins_.bin1,
-                    bins_.n2, bins_.bin2, bins_.nbig, bins_.bin2v, row_ptr_,
-                    g_.src, g_.weight, old_, new_, 0, K_);
+    if (als_) {
+      // seed: solved rows are overwritten, deg-0 rows keep their vector
+      LUX_OK(hipMemcpyAsync(new_, old_, sizeof(float) * n,
+                            hipMemcpyDeviceToDevice, s_));
+      if (bins_.nbig) {
+        LUX_OK(hipMemsetAsync(gram_, 0,
+                              sizeof(float) * (size_t)bins_.nbig * 64 * 64,
+                              s_));
+        LUX_OK(hipMemsetAsync(rhs_, 0,
+                              sizeof(float) * (size_t)bins_.nbig * 64, s_));
+      }
+      lux_gpu_cf_als_iter((uint64_t)s_, bins_.n0, bins_.bin0, bins_.n1,
+                          bins_.bin1, bins_.n2, bins_.bin2, bins_.nbig,
+                          bins_.bin2v, hubidx_, gram_, rhs_, row_ptr_,
+                          g_.src, g_.weight, old_, new_, 0, K_);
+    } else {
+      // SGD contract (cf.hip): output pre-seeded old*(1-GAMMA*LAMBDA),
+      // sweeps add GAMMA*acc
+      lux_gpu_cf_seed((uint64_t)s_, n, old_, new_);
+      lux_gpu_cf_iter((uint64_t)s_, bins_.n0, bins_.bin0, bins_.n1,
+                      bins_.bin1, bins_.n2, bins_.bin2, bins_.nbig,
+                      bins_.bin2v, row_ptr_, g_.src, g_.weight, old_, new_,
+                      0, K_);
+    }
     std::swap(old_, new_);
   }
   LUX_OK(hipStreamSynchronize(s_));

@@ -113,8 +113,9 @@ class SingleGpuPush {
 
 class SingleGpuCF {
  public:
+  // als: exact MFMA ALS sweeps (cf_als.hip, K <= 64) instead of SGD
   SingleGpuCF(const DeviceGraph& g, int K, DeviceArena& arena,
-              hipStream_t s);
+              hipStream_t s, bool als = false);
   void iterate(int iters);
   const float* vectors() const { return old_; }
 
@@ -122,9 +123,13 @@ class SingleGpuCF {
   const DeviceGraph& g_;
   hipStream_t s_;
   int K_;
+  bool als_;
   E_ID* row_ptr_;
   Bins bins_;
   float *old_, *new_;
+  int* hubidx_ = nullptr;       // ALS hub scratch slot map
+  float* gram_ = nullptr;       // ALS: nbig x 64 x 64
+  float* rhs_ = nullptr;        // ALS: nbig x 64
 };
 
 }  // namespace lux
