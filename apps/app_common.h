@@ -29,6 +29,7 @@ struct AppArgs {
   const char* dump = nullptr;  // write result vertex state (LUXS format,
                                // lux_amd/checkpoint.py-compatible)
   bool als = false;            // col_filter: exact MFMA ALS optimizer
+  bool labelprop = false;      // components: reference-parity label prop
 };
 
 inline AppArgs parse_input_args(int argc, char** argv) {
@@ -46,6 +47,7 @@ inline AppArgs parse_input_args(int argc, char** argv) {
     else if (f == "-synthetic") a.synthetic = next();
     else if (f == "-dump") a.dump = next();
     else if (f == "-als") a.als = true;
+    else if (f == "-labelprop") a.labelprop = true;
     else if (f.rfind("-ll:", 0) == 0 || f.rfind("-lg:", 0) == 0) {
       if (i + 1 < argc && argv[i + 1][0] != '-') i++;  // value-flag: skip
     } else {

@@ -111,6 +111,22 @@ void lux_gpu_cf_iter(uint64_t stream, uint32_t n0, const lux::V_ID* bin0,
                      const float* oldv, float* newv, lux::V_ID row_left,
                      int K);
 
+// cc_uf.hip
+void lux_gpu_uf_union_binned(uint64_t stream, uint32_t n0,
+                             const lux::V_ID* bin0, uint32_t n1,
+                             const lux::V_ID* bin1, uint32_t n2,
+                             const lux_uint2* bin2, const lux::E_ID* row_ptr,
+                             const lux::V_ID* col, lux::V_ID row_left,
+                             lux::V_ID* parent, const uint32_t* gbits);
+void lux_gpu_uf_union_kth(uint64_t stream, lux::V_ID vp,
+                          const lux::E_ID* row_ptr, const lux::V_ID* col,
+                          lux::V_ID row_left, lux::V_ID* parent, uint32_t k);
+void lux_gpu_cc_giant_bits(uint64_t stream, lux::V_ID nv,
+                           const lux::V_ID* labels, lux::V_ID giant,
+                           uint32_t* bits);
+void lux_gpu_uf_flatten(uint64_t stream, lux::V_ID nv, lux::V_ID* parent,
+                        lux::V_ID* labels);
+
 // cf_als.hip
 void lux_gpu_cf_als_iter(uint64_t stream, uint32_t n0, const lux::V_ID* bin0,
                          uint32_t n1, const lux::V_ID* bin1, uint32_t n2,

@@ -292,6 +292,65 @@ uint64_t SingleGpuPush::check() {
   return h;
 }
 
+// ---------------- CC (union-find) ----------------
+
+SingleGpuCCUnionFind::SingleGpuCCUnionFind(const DeviceGraph& g,
+                                           DeviceArena& arena,
+                                           hipStream_t s)
+    : g_(g), s_(s) {
+  row_ptr_ = arena.alloc_n<E_ID>(g.nv + 1);
+  lux_gpu_local_row_ptr((uint64_t)s, g.nv, 0, g.col_end, row_ptr_);
+  bins_.build(row_ptr_, g.nv, g.ne, arena, s);
+  parent_ = arena.alloc_n<V_ID>(g.nv);
+  labels_ = arena.alloc_n<V_ID>(g.nv);
+  gbits_ = arena.alloc_n<uint32_t>((g.nv + 31) / 32);
+}
+
+void SingleGpuCCUnionFind::run() {
+  // parent = iota
+  std::vector<V_ID> h(g_.nv);
+  for (V_ID v = 0; v < g_.nv; v++) h[v] = v;
+  LUX_OK(hipMemcpyAsync(parent_, h.data(), sizeof(V_ID) * g_.nv,
+                        hipMemcpyHostToDevice, s_));
+  for (uint32_t k = 0; k < 2; k++)  // Afforest sample-hook rounds
+    lux_gpu_uf_union_kth((uint64_t)s_, g_.nv, row_ptr_, g_.src, 0, parent_,
+                         k);
+  lux_gpu_uf_flatten((uint64_t)s_, g_.nv, parent_, labels_);
+  // giant root by host sampling
+  V_ID nsamp = g_.nv < 4096 ? g_.nv : 4096;
+  V_ID stride = g_.nv / nsamp;
+  std::vector<V_ID> samp(nsamp);
+  LUX_OK(hipMemcpy2DAsync(samp.data(), sizeof(V_ID), labels_,
+                          (size_t)stride * sizeof(V_ID), sizeof(V_ID),
+                          nsamp, hipMemcpyDeviceToHost, s_));
+  LUX_OK(hipStreamSynchronize(s_));
+  std::sort(samp.begin(), samp.end());
+  V_ID giant = samp[0], best = 1, run = 1;
+  for (size_t i = 1; i < samp.size(); i++) {
+    run = samp[i] == samp[i - 1] ? run + 1 : 1;
+    if (run > best) { best = run; giant = samp[i]; }
+  }
+  lux_gpu_cc_giant_bits((uint64_t)s_, g_.nv, labels_, giant, gbits_);
+  lux_gpu_uf_union_binned((uint64_t)s_, bins_.n0, bins_.bin0, bins_.n1,
+                          bins_.bin1, bins_.n2, bins_.bin2, row_ptr_,
+                          g_.src, 0, parent_, gbits_);
+  lux_gpu_uf_flatten((uint64_t)s_, g_.nv, parent_, labels_);
+  LUX_OK(hipStreamSynchronize(s_));
+}
+
+uint64_t SingleGpuCCUnionFind::check() {
+  unsigned long long* mistakes;
+  LUX_OK(hipMalloc(&mistakes, 8));
+  LUX_OK(hipMemsetAsync(mistakes, 0, 8, s_));
+  lux_gpu_check((uint64_t)s_, 0, g_.nv, 0, row_ptr_, g_.src,
+                (const uint32_t*)labels_, mistakes);
+  unsigned long long h = 0;
+  LUX_OK(hipMemcpyAsync(&h, mistakes, 8, hipMemcpyDeviceToHost, s_));
+  LUX_OK(hipStreamSynchronize(s_));
+  hipFree(mistakes);
+  return h;
+}
+
 // ---------------- CF ----------------
 
 SingleGpuCF::SingleGpuCF(const DeviceGraph& g, int K, DeviceArena& arena,

@@ -111,6 +111,26 @@ class SingleGpuPush {
   int iters_ = 0;
 };
 
+// Union-find CC (cc_uf.hip): one edge pass + Afforest sampling; identical
+// labelling to converged max-label propagation on symmetric inputs
+// (components -labelprop selects the reference-parity SingleGpuPush).
+class SingleGpuCCUnionFind {
+ public:
+  SingleGpuCCUnionFind(const DeviceGraph& g, DeviceArena& arena,
+                       hipStream_t s);
+  void run();
+  uint64_t check();
+  const uint32_t* labels() const { return (const uint32_t*)labels_; }
+
+ private:
+  const DeviceGraph& g_;
+  hipStream_t s_;
+  E_ID* row_ptr_;
+  Bins bins_;
+  V_ID *parent_, *labels_;
+  uint32_t* gbits_;
+};
+
 class SingleGpuCF {
  public:
   // als: exact MFMA ALS sweeps (cf_als.hip, K <= 64) instead of SGD

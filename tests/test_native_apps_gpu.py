@@ -128,3 +128,38 @@ def test_native_col_filter_als(tmp_path):
     sgd, _ = ck.load_state(o1)
     als, _ = ck.load_state(o2)
     assert cpu_ref.cf_loss(g, 32, als) < cpu_ref.cf_loss(g, 32, sgd)
+
+
+def test_native_components_uf_dump_matches_cpu(tmp_path):
+    """Native union-find CC on an undirected graph: labels equal the CPU
+    reference's converged max-label propagation."""
+    import numpy as np
+    from lux_amd import checkpoint as ck
+    from lux_amd import cpu_ref
+    from lux_amd.graph import Graph
+    lux = str(tmp_path / "s.lux")
+    out = str(tmp_path / "l.luxs")
+    g = Graph.rmat(13, 60000, seed=3, sym=True)
+    g.save(lux)
+    o = _run([f"{BIN}/components", "-file", lux, "-dump", out, "-check"])
+    assert "PASS" in o
+    got, _ = ck.load_state(out)
+    want, _ = cpu_ref.cc(g)
+    np.testing.assert_array_equal(got, want)
+
+
+def test_native_components_labelprop_dump(tmp_path):
+    """-labelprop keeps the reference's directional propagation exactly
+    (directed input)."""
+    import numpy as np
+    from lux_amd import checkpoint as ck
+    from lux_amd import cpu_ref
+    from lux_amd.graph import Graph
+    lux = str(tmp_path / "d.lux")
+    out = str(tmp_path / "l.luxs")
+    g = Graph.rmat(12, 50000, seed=5)
+    g.save(lux)
+    _run([f"{BIN}/components", "-file", lux, "-labelprop", "-dump", out])
+    got, _ = ck.load_state(out)
+    want, _ = cpu_ref.cc(g)
+    np.testing.assert_array_equal(got, want)
