@@ -189,3 +189,18 @@ def test_cf_als_deg0_keeps_vector():
     want = np.full(K, v0, dtype=np.float32)
     for v in np.nonzero(deg == 0)[0][:5]:
         np.testing.assert_array_equal(out[v], want)
+
+
+def test_config1_pagerank_rmat16_cpu():
+    """BASELINE.json config 1 pinned: PageRank pull on the CPU reference
+    path over an RMAT-16 synthetic CSC (plumbing, no GPU). Checks the
+    stored-form invariant sum(pr_undivided) <= 1 (rank mass leaks only at
+    zero-out-degree sinks, the reference's convention) and stability."""
+    from lux_amd.graph import Graph
+    g = Graph.rmat(16, 1 << 20, seed=1)
+    pr = cpu_ref.pagerank(g, 10)
+    assert pr.shape == (1 << 16,)
+    assert np.isfinite(pr).all() and (pr > 0).all()
+    deg = g.out_degrees().astype(np.float64)
+    undiv = pr * np.maximum(deg, 1.0)
+    assert 0.2 < undiv.sum() <= 1.0 + 1e-6
