@@ -32,3 +32,21 @@ def test_converter_roundtrip(tmp_path):
     np.testing.assert_array_equal(g.col_end, [1, 3, 4])
     assert sorted(g.src[1:3].tolist()) == [0, 2]
     assert g.src[0] == 2 and g.src[3] == 1
+
+
+def test_rmat_gen_sym(tmp_path):
+    """-sym emits both directions of ne/2 pairs, matching Graph.rmat
+    sym=True (the components app's undirected input)."""
+    import numpy as np
+
+    from lux_amd.graph import Graph
+    out = str(tmp_path / "s.lux")
+    r = subprocess.run([_bin("rmat_gen"), "-kind", "rmat", "-scale", "9",
+                        "-ne", "3000", "-sym", "-o", out],
+                       capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stderr
+    g = Graph.load(out)
+    h = Graph.rmat(9, 3000, seed=1, sym=True)
+    assert (g.nv, g.ne) == (h.nv, h.ne)
+    np.testing.assert_array_equal(g.col_end, h.col_end)
+    np.testing.assert_array_equal(np.sort(g.src), np.sort(h.src))
