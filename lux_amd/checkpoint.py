@@ -78,6 +78,13 @@ def resume_engine(path, engine):
     if hasattr(engine, "ranks"):
         engine.old.copy_(torch.from_numpy(arr).to(engine.old.device))
         _refresh_slice(engine)
+    elif hasattr(engine, "labels_t"):  # CCUnionFindEngine
+        t = torch.from_numpy(arr.view(np.int32)).to(engine.labels_t.device)
+        engine.labels_t.copy_(t)
+        # a converged labelling (max vertex id per component) is itself a
+        # valid union-find forest: every root r has label[r] == r
+        engine.parent.copy_(t)
+        engine.iterations = iteration
     elif hasattr(engine, "labels"):
         lab = _labels_tensor(engine)
         t = torch.from_numpy(arr.view(np.int32)).to(lab.device)
@@ -87,6 +94,9 @@ def resume_engine(path, engine):
             engine.labels_part.copy_(engine.labels.narrow(0, p.row_left,
                                                           p.vp))
             engine.iterations = iteration
+            # restored labels can disagree with the BFS visited bitmap —
+            # force a rebuild before the next push step
+            engine._bits_stale = True
         else:  # LabelPullEngine
             engine.old.copy_(t)
             _refresh_slice(engine)

@@ -56,6 +56,52 @@ def barrier():
         dist.barrier()
 
 
+def _even_layout(full, verts, row_left):
+    """True when the slices are equal-sized and tile `full` exactly — the
+    all_gather_into_tensor fast path (one fused RCCL all-gather)."""
+    v = verts[0]
+    if v == 0 or any(x != v for x in verts):
+        return False
+    if any(row_left[p] != p * v for p in range(len(verts))):
+        return False
+    return full.numel() == v * len(verts)
+
+
+def _slice_ops(full, my_slice, verts, row_left, i):
+    """Batched p2p op list for an uneven all-gather(v): ship my slice to
+    every peer and receive each peer's slice straight into its window of
+    `full` — no staging copies (the r1 `repeat(ws)` sent P× the bytes
+    through a scratch buffer). Pairs are scheduled at ring offsets so at
+    offset k every rank talks to distinct peers: on the fully-connected
+    7-link xGMI mesh each round rides a different point-to-point link."""
+    ops = []
+    ws = len(verts)
+    for off in range(1, ws):
+        ps = (i + off) % ws
+        pr = (i - off) % ws
+        if my_slice.numel():
+            ops.append(dist.P2POp(dist.isend, my_slice, ps))
+        if verts[pr]:
+            ops.append(dist.P2POp(
+                dist.irecv, full.narrow(0, row_left[pr], verts[pr]), pr))
+    return ops
+
+
+class _DoneWork:
+    def wait(self):
+        return True
+
+
+class _BatchWork:
+    def __init__(self, reqs):
+        self.reqs = reqs
+
+    def wait(self):
+        for r in self.reqs:
+            r.wait()
+        return True
+
+
 def all_gather_slices(full, my_slice, verts, row_left, my_index=None):
     """All-gather(v) of per-rank slices into the replicated `full` tensor.
 
@@ -67,24 +113,7 @@ def all_gather_slices(full, my_slice, verts, row_left, my_index=None):
     my_index: which slice is mine (defaults to this process's rank; passed
               explicitly by single-process multi-partition simulations).
     """
-    ws = world_size()
-    if ws == 1:
-        i = 0 if my_index is None else my_index
-        dst = full.narrow(0, row_left[i], verts[i])
-        if dst.data_ptr() != my_slice.data_ptr():
-            dst.copy_(my_slice)
-        return
-    # all-gatherv as a single uneven all_to_all: each rank ships its slice
-    # to every peer (input tiled ws times), receives the concatenation.
-    inp = my_slice.repeat(ws)
-    in_splits = [my_slice.numel()] * ws
-    dist.all_to_all_single(full, inp, output_split_sizes=list(verts),
-                           input_split_sizes=in_splits)
-
-
-class _DoneWork:
-    def wait(self):
-        return True
+    all_gather_slices_async(full, my_slice, verts, row_left, my_index).wait()
 
 
 def all_gather_slices_async(full, my_slice, verts, row_left, my_index=None):
@@ -94,15 +123,43 @@ def all_gather_slices_async(full, my_slice, verts, row_left, my_index=None):
     launch this right after computing their slice and overlap the xGMI
     exchange with the next iteration's rank-local sweep."""
     ws = world_size()
+    i = rank() if my_index is None else my_index
     if ws == 1:
-        all_gather_slices(full, my_slice, verts, row_left, my_index)
+        dst = full.narrow(0, row_left[i], verts[i])
+        if dst.data_ptr() != my_slice.data_ptr():
+            dst.copy_(my_slice)
         return _DoneWork()
-    inp = my_slice.repeat(ws)
-    in_splits = [my_slice.numel()] * ws
-    return dist.all_to_all_single(full, inp,
-                                  output_split_sizes=list(verts),
-                                  input_split_sizes=in_splits,
-                                  async_op=True)
+    if _even_layout(full, verts, row_left) and dist.get_backend() == "nccl":
+        return dist.all_gather_into_tensor(full, my_slice, async_op=True)
+    dst = full.narrow(0, row_left[i], verts[i])
+    if verts[i] and dst.data_ptr() != my_slice.data_ptr():
+        dst.copy_(my_slice)
+    ops = _slice_ops(full, my_slice, verts, row_left, i)
+    if not ops:
+        return _DoneWork()
+    return _BatchWork(dist.batch_isend_irecv(ops))
+
+
+def exchange_multi_async(parts, my_index=None):
+    """Several all-gather(v)s fused into ONE batched p2p round: parts is a
+    list of (full, my_slice, verts, row_left) tuples. Each pairwise message
+    set rides the same ncclGroup, so a label slice and a frontier segment
+    to the same peer share one xGMI link pass. my_slice may be None to
+    skip sending a given part (peers must pass verts[i]==0 for it)."""
+    ws = world_size()
+    i = rank() if my_index is None else my_index
+    ops = []
+    for full, my_slice, verts, row_left in parts:
+        if ws == 1 or verts[i]:
+            dst = full.narrow(0, row_left[i], verts[i])
+            if my_slice is not None and \
+                    dst.data_ptr() != my_slice.data_ptr():
+                dst.copy_(my_slice)
+        if ws > 1 and my_slice is not None:
+            ops.extend(_slice_ops(full, my_slice, verts, row_left, i))
+    if not ops:
+        return _DoneWork()
+    return _BatchWork(dist.batch_isend_irecv(ops))
 
 
 def all_reduce_sum_(t):

@@ -72,6 +72,85 @@ def test_engine_checkpoint_roundtrip_gpu(tmp_path):
                                atol=1e-12)
 
 
+class _FakeUF:
+    """Shape of CCUnionFindEngine for branch-selection testing on CPU:
+    has a `labels` property AND labels_t/parent, no labels_part/old."""
+
+    def __init__(self, n):
+        import torch
+        self.labels_t = torch.zeros(n, dtype=torch.int32)
+        self.parent = torch.zeros(n, dtype=torch.int32)
+        self.iterations = 0
+
+    @property
+    def labels(self):
+        return self.labels_t
+
+
+def test_resume_uf_engine_branch(tmp_path):
+    """resume must hit the labels_t branch for the
+    union-find engine, not fall into the LabelPullEngine path."""
+    import torch
+    p = str(tmp_path / "uf.luxs")
+    n = 128
+    labs = np.arange(n, dtype=np.uint32)
+    labs[1:] = n - 1  # one giant component rooted at max id + a singleton
+    labs[0] = 0
+    ckpt.save_state(p, labs, iteration=4)
+    eng = _FakeUF(n)
+    it = ckpt.resume_engine(p, eng)
+    assert it == 4
+    want = torch.from_numpy(labs.view(np.int32))
+    assert torch.equal(eng.labels_t, want)
+    assert torch.equal(eng.parent, want)  # converged labelling = valid forest
+
+
+class _FakePush:
+    def __init__(self, n, vp):
+        import torch
+        self.labels = torch.zeros(n, dtype=torch.int32)
+        self.labels_part = torch.zeros(vp, dtype=torch.int32)
+        self.iterations = 0
+        self._bits_stale = False
+
+        class P:
+            row_left = 0
+        P.vp = vp
+        self.part = P
+
+
+def test_resume_push_marks_bits_stale(tmp_path):
+    """PushEngine resume must invalidate the BFS visited
+    bitmap."""
+    p = str(tmp_path / "push.luxs")
+    labs = np.arange(64, dtype=np.uint32)
+    ckpt.save_state(p, labs, iteration=2)
+    eng = _FakePush(64, 16)
+    ckpt.resume_engine(p, eng)
+    assert eng._bits_stale is True
+    assert eng.iterations == 2
+
+
+@pytest.mark.gpu
+def test_cc_uf_checkpoint_roundtrip_gpu(tmp_path):
+    """save+resume roundtrip through the real CCUnionFindEngine (the r1
+    resume crash case): converged labels survive and check() passes."""
+    import torch
+    from lux_amd import checkpoint as ck
+    from lux_amd.cc_engine import CCUnionFindEngine
+    from lux_amd.engine import DeviceCSC, GraphPart
+    path = str(tmp_path / "cc.luxs")
+    a = CCUnionFindEngine(GraphPart(
+        DeviceCSC.rmat(12, 60000, seed=5, sym=True), 1, 0))
+    a.run()
+    ck.save_engine(path, a)
+    b = CCUnionFindEngine(GraphPart(
+        DeviceCSC.rmat(12, 60000, seed=5, sym=True), 1, 0))
+    ck.resume_engine(path, b)
+    assert torch.equal(a.labels_t, b.labels_t)
+    assert b.check() == 0
+
+
 def test_save_engine_rank_nonzero_noop(tmp_path, monkeypatch):
     """Distributed contract: only rank 0 writes (save_engine no-ops
     elsewhere)."""
