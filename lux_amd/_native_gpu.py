@@ -168,11 +168,24 @@ def cf_als_iter(stream, n0, bin0, n1, bin1, n2, bin2, nbig, bin2v, hubidx,
 
 
 def frontier_expand(stream, old_dense, in_row_left, in_count, old_seg,
-                    push_row_ptr, items, counter, max_items):
+                    qlabels, labels_repair, push_row_ptr, items, counter,
+                    max_items):
     lib().lux_gpu_frontier_expand(
         _u64(stream), ctypes.c_int(old_dense), _u32(in_row_left),
-        _u32(in_count), dp(old_seg), dp(push_row_ptr), dp(items),
-        dp(counter), _u32(max_items))
+        _u32(in_count), dp(old_seg), dp(qlabels), dp(labels_repair),
+        dp(push_row_ptr), dp(items), dp(counter), _u32(max_items))
+
+
+def frontier_fixup(stream, vp, row_left, capacity, built_dense, snapshot,
+                   labels_part, deg_part, new_seg, annex, tmp_seg, meta,
+                   item_counter, max_items):
+    """Device-predicated conversion chain + meta record (push.hip
+    lux_gpu_frontier_fixup): zero host syncs."""
+    lib().lux_gpu_frontier_fixup(
+        _u64(stream), _u32(vp), _u32(row_left), _u32(capacity),
+        ctypes.c_int(built_dense), dp(snapshot), dp(labels_part),
+        dp(deg_part), dp(new_seg), dp(annex), dp(tmp_seg), dp(meta),
+        dp(item_counter), _u32(max_items))
 
 
 def push_chunk_scatter(stream, is_min, new_dense, items, counter, max_items,

@@ -79,7 +79,7 @@ def _slice_ops(full, my_slice, verts, row_left, i):
     for off in range(1, ws):
         ps = (i + off) % ws
         pr = (i - off) % ws
-        if my_slice.numel():
+        if my_slice is not None and my_slice.numel():
             ops.append(dist.P2POp(dist.isend, my_slice, ps))
         if verts[pr]:
             ops.append(dist.P2POp(
@@ -150,12 +150,13 @@ def exchange_multi_async(parts, my_index=None):
     i = rank() if my_index is None else my_index
     ops = []
     for full, my_slice, verts, row_left in parts:
-        if ws == 1 or verts[i]:
+        if verts[i] and my_slice is not None:
             dst = full.narrow(0, row_left[i], verts[i])
-            if my_slice is not None and \
-                    dst.data_ptr() != my_slice.data_ptr():
+            if dst.data_ptr() != my_slice.data_ptr():
                 dst.copy_(my_slice)
-        if ws > 1 and my_slice is not None:
+        if ws > 1:
+            # a rank that skips publishing a part still posts the recvs
+            # for its peers' slices of that part
             ops.extend(_slice_ops(full, my_slice, verts, row_left, i))
     if not ops:
         return _DoneWork()

@@ -7,14 +7,26 @@ The MI355X re-design of the reference's push pipeline
   - adaptive per-partition frontier segments (dense bitmap / sparse queue,
     FrontierHeader-compatible) with majority-vote output format
     (sssp_gpu.cu:395-408), dense->sparse conversion and sparse-overflow
-    fallback (sssp_gpu.cu:462-491);
-  - pull fallback when the global frontier exceeds nv/16 (sssp_gpu.cu:414),
-    served by the degree-binned pull kernels;
-  - exchange: RCCL all-gather(v) of label slices + frontier segments over
-    xGMI replaces the reference's zero-copy host staging; termination is
-    read off the exchanged headers (no extra collective, replacing the
-    FutureMap vote of sssp.cc:116-124).
+    fallback (sssp_gpu.cu:462-491) — ALL run as a device-predicated fixup
+    chain (push.hip frontier_fixup), so the conversions never block the
+    host;
+  - pull fallback when the global frontier exceeds nv/16 (sssp_gpu.cu:414)
+    or its out-edge volume dominates, served by the degree-binned pull
+    kernels;
+  - exchange: one batched RCCL p2p round per iteration over xGMI carrying
+    only the USED frontier bytes, a label annex for sparse queues, and —
+    only on dense iterations — the label slices. The reference instead
+    re-reads whole zero-copy regions each iteration.
+
+Latency hiding (reference: SLIDING_WINDOW=4 in-flight iterations,
+sssp.cc:111-129): here each iteration has exactly ONE blocking D2H read —
+the 32-byte-per-rank meta record produced on device and exchanged with the
+payloads. Frontier counts, format decisions, conversion outcomes,
+overflow flags and the push-vs-pull edge volume all ride that record, so
+the r1 engine's 3-4 blocking reads per iteration are gone.
 """
+import sys
+
 import numpy as np
 import torch
 
@@ -65,6 +77,12 @@ class PushEngine:
         self.push_col = torch.empty(max(p.ep, 1), dtype=U32, device=device)
         ng.csr_scatter(s, p.ep, p.col, p.row_ptr, p.vp, p.row_left, cursor,
                        self.push_col)
+        # global out-degrees (deg_src summed over ranks); keep only my
+        # partition's slice — it prices the NEW frontier's out-edge volume
+        # in the meta record (the exact push-vs-pull input)
+        dx.all_reduce_sum_(deg_src)
+        self.deg_part = deg_src.narrow(0, p.row_left, p.vp).clone() \
+            if p.vp else torch.zeros(1, dtype=U32, device=device)
         torch.cuda.synchronize()
         del deg_src, ends, partials, cursor
 
@@ -77,11 +95,16 @@ class PushEngine:
         self.new_seg = torch.zeros(self.seg_bytes[p.p], dtype=U8,
                                    device=device)
         self.tmp_seg = torch.zeros_like(self.new_seg)
-        # header byte index for the post-exchange D2H gather
-        idx = []
-        for q in range(p.nparts):
-            idx.extend(range(int(self.seg_off[q]), int(self.seg_off[q]) + 8))
-        self.hdr_idx = torch.tensor(idx, dtype=torch.long, device=device)
+        # label annex: sparse queues travel with their final labels so
+        # sparse iterations skip the O(nv) label slice exchange entirely
+        caps = [frontier_capacity(v) for v in p.verts_all]
+        self.annex_off = np.concatenate([[0], np.cumsum(caps)])
+        self.fq_annex_all = torch.zeros(int(self.annex_off[-1]), dtype=U32,
+                                        device=device)
+        self.new_annex = torch.zeros(self.capacity, dtype=U32, device=device)
+        # device meta record (u32[8]/rank: type,count,evol u64,overflow,..)
+        self.meta_mine = torch.zeros(8, dtype=U32, device=device)
+        self.meta_all = torch.zeros(8 * p.nparts, dtype=U32, device=device)
 
         # ---- edge-balanced scatter work items (push.hip expand+chunk):
         # push runs only while total frontier <= nv/16, so items are bounded
@@ -89,7 +112,7 @@ class PushEngine:
         self.max_items = p.nv // 16 + p.ep // 8192 + 1024
         self.items = torch.empty(self.max_items * 2, dtype=U32,
                                  device=device)
-        self.item_counter = torch.zeros(2, dtype=U32, device=device)
+        self.item_counter = torch.zeros(4, dtype=U32, device=device)
         # hop-SSSP (level-synchronous BFS) visited bitmap: push discovery
         # is a test-and-set against vp/8 bytes (L2-resident) instead of an
         # atomicMin against the 4*vp label array (push.hip BFS_BITS path);
@@ -120,10 +143,11 @@ class PushEngine:
             torch.arange(p.nv, dtype=U32, device=self.device,
                          out=self.labels)
         self.labels_part.copy_(self.labels.narrow(0, p.row_left, p.vp))
+        self.labels_current = True
 
         if self._fq_init is None:
             fq_host = np.zeros(int(self.seg_off[-1]), np.uint8)
-            headers = []
+            meta0 = np.zeros((p.nparts, 8), np.uint32)
             for q in range(p.nparts):
                 off = int(self.seg_off[q])
                 hv = fq_host[off:off + 8].view(np.uint32)
@@ -135,17 +159,25 @@ class PushEngine:
                     if owner:
                         hv[1] = 1
                         fq_host[off + 8:off + 12].view(np.uint32)[0] = source
-                    headers.append((SPARSE_QUEUE, int(hv[1])))
+                    meta0[q, 0] = SPARSE_QUEUE
+                    meta0[q, 1] = hv[1]
+                    # evol left 0: a 1-vertex seed frontier never flips the
+                    # push-vs-pull decision
                 else:
                     hv[0] = DENSE_BITMAP
                     hv[1] = p.verts_all[q]
                     nbytes = (p.verts_all[q] + 7) // 8
                     fq_host[off + 8:off + 8 + nbytes] = 0xFF
-                    headers.append((DENSE_BITMAP, p.verts_all[q]))
+                    meta0[q, 0] = DENSE_BITMAP
+                    meta0[q, 1] = p.verts_all[q]
             self._fq_init = (torch.from_numpy(fq_host).to(self.device),
-                             headers)
+                             meta0)
         self.fq_all.copy_(self._fq_init[0])
-        self.headers = list(self._fq_init[1])
+        self.fq_annex_all.zero_()  # seed labels are all 0 (SSSP source)
+        self.meta_host = self._fq_init[1].copy()
+        self.headers = [(int(self.meta_host[q, 0]),
+                         int(self.meta_host[q, 1]))
+                        for q in range(p.nparts)]
         self.iterations = 0
         self.stats = []
         self._bits_stale = True
@@ -154,55 +186,61 @@ class PushEngine:
     def _my_seg_i32(self):
         return self.new_seg.view(U32)
 
-    def _read_my_count(self):
-        return int(self.new_seg[4:8].view(U32).cpu().item())
+    def _sync_labels(self):
+        """Catch-up all-gather of label slices (needed before a pull
+        iteration or before reading the replicated labels) — slices go
+        stale while sparse iterations skip the label exchange."""
+        if self.labels_current:
+            return
+        p = self.part
+        dx.all_gather_slices(self.labels, self.labels_part, p.verts_all,
+                             p.row_left_all, my_index=p.p)
+        self.labels_current = True
+
+    def final_labels(self):
+        """Replicated labels, synchronised (use after run())."""
+        self._sync_labels()
+        return self.labels
 
     def step(self):
-        """One push iteration. Returns my partition's new-frontier count."""
+        """One push iteration. Returns the global new-frontier count
+        (from the exchanged meta — no extra sync)."""
         p = self.part
         s = _stream()
         nparts = p.nparts
-        old_fq_size = sum(h[1] for h in self.headers)
-        dense_votes = sum(1 for h in self.headers if h[0] == DENSE_BITMAP)
+        ws = dx.world_size()
+        mh = self.meta_host  # (nparts, 8) u32 describing CURRENT frontier
+        types = mh[:, 0]
+        counts = mh[:, 1]
+        evol = int((mh[:, 2].astype(np.uint64)
+                    | (mh[:, 3].astype(np.uint64) << np.uint64(32))).sum())
+        overflow = bool(mh[:, 4].any())
+        old_fq_size = int(counts.sum())
+        dense_votes = int((types == DENSE_BITMAP).sum())
         new_dense = dense_votes >= nparts - dense_votes
         self.snapshot.copy_(self.labels_part)
-        # zero my new header (type patched at the end)
+        # zero my new header (type patched by the device fixup chain)
         self._my_seg_i32()[0] = 0
         self._my_seg_i32()[1] = 0
 
-        pull_fallback = old_fq_size > p.nv // 16
+        pull_fallback = overflow or old_fq_size > p.nv // 16
+        if overflow:
+            print("[lux] frontier expand overflow: recovering with a "
+                  "forced pull iteration", file=sys.stderr)
         if not pull_fallback:
-            # expand all source segments into <=8192-edge work items
-            self.item_counter.zero_()
-            for q in range(nparts):
-                typ, num = self.headers[q]
-                if p.verts_all[q] == 0:
-                    continue
-                seg = self.fq_all.narrow(0, int(self.seg_off[q]),
-                                         self.seg_bytes[q])
-                if typ == DENSE_BITMAP:
-                    ng.frontier_expand(s, 1, p.row_left_all[q],
-                                       p.verts_all[q], seg,
-                                       self.push_row_ptr, self.items,
-                                       self.item_counter, self.max_items)
-                elif num:
-                    ng.frontier_expand(s, 0, 0, num, seg,
-                                       self.push_row_ptr, self.items,
-                                       self.item_counter, self.max_items)
             # second adaptivity axis (ours, not the reference's): the
             # vertex-count threshold misses RMAT's hub explosion — a 902K-
-            # vertex frontier can cover ~half of all edges. The expand
-            # kernel counts the frontier's out-edges (counter[1]); a dense
-            # pull sweep (identical labels per iteration, src-blocked
-            # LLC-resident gathers) is faster beyond ~ep/8 edges.
-            n_edges = int(self.item_counter[1].cpu().item())
-            # bitmap-BFS push touches vp/8 bytes of visited bits instead of
-            # the label array, so it stays cheaper than a dense pull sweep
-            # up to much larger frontiers
-            thresh = p.ep // 2 if self.visited is not None else p.ep // 8
-            if n_edges > thresh:
+            # vertex frontier can cover ~half of all edges. The meta
+            # record prices the frontier's out-edges exactly (global
+            # out-degrees, accumulated at frontier build time); a dense
+            # pull sweep (src-blocked LLC-resident gathers) is faster
+            # beyond ~ne/8 edges. bitmap-BFS push touches vp/8 bytes of
+            # visited bits instead of the label array, so it stays cheaper
+            # up to much larger frontiers (~ne/2).
+            thresh = p.ne // 2 if self.visited is not None else p.ne // 8
+            if evol > thresh:
                 pull_fallback = True
-            elif n_edges // 16 > self.capacity:
+            elif (evol * p.ep // max(p.ne, 1)) // 16 > self.capacity:
                 # expected discoveries cannot fit the sparse queue: choose
                 # the dense bitmap upfront instead of paying the sparse
                 # append machinery + guaranteed overflow rebuild (the
@@ -210,10 +248,38 @@ class PushEngine:
                 new_dense = True
         if pull_fallback:
             new_dense = True
+            self._sync_labels()  # pull reads every vertex's label
             mode = ng.PULL_MIN if self.is_min else ng.PULL_MAX
             run_pull(p, mode, self.labels, self.labels_part, None, 0.0)
             self._bits_stale = True  # pull writes labels directly
+            item_counter = None
         else:
+            # expand all source segments into <=8192-edge work items;
+            # sparse queues carry a label annex — expand repairs the
+            # (possibly stale) replicated labels from it before the
+            # scatter reads them
+            self.item_counter.zero_()
+            # repair even at ws==1: sparse iterations skip the label
+            # publish, so the replicated array is stale for exactly the
+            # queued vertices — the annex carries their fresh labels
+            repair = self.labels
+            for q in range(nparts):
+                typ, num = int(types[q]), int(counts[q])
+                if p.verts_all[q] == 0:
+                    continue
+                seg = self.fq_all.narrow(0, int(self.seg_off[q]),
+                                         self.seg_bytes[q])
+                if typ == DENSE_BITMAP:
+                    ng.frontier_expand(s, 1, p.row_left_all[q],
+                                       p.verts_all[q], seg, None, None,
+                                       self.push_row_ptr, self.items,
+                                       self.item_counter, self.max_items)
+                elif num:
+                    annex = self.fq_annex_all.narrow(
+                        0, int(self.annex_off[q]), num)
+                    ng.frontier_expand(s, 0, 0, num, seg, annex, repair,
+                                       self.push_row_ptr, self.items,
+                                       self.item_counter, self.max_items)
             bits = None
             if self.visited is not None and p.vp > 0:
                 if self._bits_stale:
@@ -228,60 +294,91 @@ class PushEngine:
                                   self.labels_part, p.row_left,
                                   self.new_seg, self.capacity,
                                   visited_bits=bits)
+            item_counter = self.item_counter
 
-        # ---- frontier format fix-ups (sssp_gpu.cu:462-491) ----
+        # ---- frontier fix-ups + meta, all device-side ----
         if new_dense:
             ng.build_bitmap(s, p.vp, self.snapshot, self.labels_part,
                             self.new_seg)
-            my_count = self._read_my_count()
-            if my_count < self.capacity:
-                # dense result fits the sparse capacity: convert
-                self.tmp_seg.copy_(self.new_seg)
-                self._my_seg_i32()[1] = 0
-                ng.d2s(s, p.vp, p.row_left, self.tmp_seg, self.new_seg)
-                new_dense = False
-        else:
-            my_count = self._read_my_count()
-            if my_count >= self.capacity:
-                # sparse overflow: rebuild as bitmap
-                new_dense = True
-                self._my_seg_i32()[1] = 0
-                ng.build_bitmap(s, p.vp, self.snapshot, self.labels_part,
-                                self.new_seg)
-                my_count = self._read_my_count()
-        self._my_seg_i32()[0] = DENSE_BITMAP if new_dense else SPARSE_QUEUE
+        ng.frontier_fixup(s, p.vp, p.row_left, self.capacity,
+                          int(new_dense), self.snapshot, self.labels_part,
+                          self.deg_part, self.new_seg, self.new_annex,
+                          self.tmp_seg, self.meta_mine, item_counter,
+                          self.max_items)
 
-        # ---- exchange: labels + frontier segments ----
-        dx.all_gather_slices(self.labels, self.labels_part, p.verts_all,
-                             p.row_left_all, my_index=p.p)
-        dx.all_gather_slices(self.fq_all, self.new_seg, self.seg_bytes,
-                             [int(o) for o in self.seg_off[:-1]],
+        # ---- exchange: meta first (the ONE host read), then payloads ----
+        dx.all_gather_slices(self.meta_all, self.meta_mine,
+                             [8] * nparts, [8 * q for q in range(nparts)],
                              my_index=p.p)
-        hdr = self.fq_all[self.hdr_idx].cpu().numpy().view(np.uint32)
-        self.headers = [(int(hdr[2 * q]), int(hdr[2 * q + 1]))
+        mh = self.meta_all.cpu().numpy().view(np.uint32).reshape(nparts, 8)
+        self.meta_host = mh
+        ntypes, ncounts = mh[:, 0], mh[:, 1]
+
+        # payload sizes from the fresh meta: only USED bytes travel
+        used = [0] * nparts
+        annex_n = [0] * nparts
+        lab_n = [0] * nparts
+        for q in range(nparts):
+            if p.verts_all[q] == 0:
+                continue
+            if ntypes[q] == DENSE_BITMAP:
+                used[q] = 8 + (p.verts_all[q] + 7) // 8
+                lab_n[q] = p.verts_all[q]  # dense ranks publish labels
+            else:
+                used[q] = 8 + 4 * int(ncounts[q])
+                annex_n[q] = int(ncounts[q])
+        me = p.p
+        parts = [(self.fq_all,
+                  self.new_seg.narrow(0, 0, used[me]) if used[me] else None,
+                  used, [int(o) for o in self.seg_off[:-1]]),
+                 (self.fq_annex_all,
+                  self.new_annex.narrow(0, 0, annex_n[me])
+                  if annex_n[me] else None,
+                  annex_n, [int(o) for o in self.annex_off[:-1]]),
+                 (self.labels,
+                  self.labels_part if lab_n[me] else None,
+                  lab_n, p.row_left_all)]
+        dx.exchange_multi_async(parts, my_index=me).wait()
+        # slice q is fresh iff q published it now, or it was fresh before
+        # and q changed nothing this iteration
+        if self.labels_current:
+            self.labels_current = all(
+                lab_n[q] > 0 or ncounts[q] == 0 or p.verts_all[q] == 0
+                for q in range(nparts))
+        else:
+            self.labels_current = all(
+                lab_n[q] > 0 or p.verts_all[q] == 0
+                for q in range(nparts))
+
+        self.headers = [(int(ntypes[q]), int(ncounts[q]))
                         for q in range(nparts)]
         self.iterations += 1
         # per-iteration trace row (reference -verbose parity,
         # sssp_gpu.cu:516-518: activeNodes + phase info per partition)
-        self.stats.append(dict(iter=self.iterations, old_frontier=old_fq_size,
+        my_new = int(ncounts[me])
+        self.stats.append(dict(iter=self.iterations,
+                               old_frontier=old_fq_size,
                                pull_fallback=bool(pull_fallback),
-                               out_dense=bool(new_dense), my_new=my_count))
-        return sum(h[1] for h in self.headers)
+                               out_dense=bool(ntypes[me] == DENSE_BITMAP),
+                               my_new=my_new))
+        return int(ncounts.sum())
 
     def run(self, max_iters=None):
         """Iterate to convergence (every partition reports an empty new
         frontier)."""
         while True:
             total = self.step()
-            if total == 0:
-                break
+            if total == 0 and not self.meta_host[:, 4].any():
+                break  # overflow never terminates: forced pull recovers
             if max_iters and self.iterations >= max_iters:
                 break
+        self._sync_labels()
         return self.iterations
 
     def check(self):
         """Device check oracle; returns global violation count."""
         p = self.part
+        self._sync_labels()
         mistakes = torch.zeros(1, dtype=U64, device=self.device)
         ng.check(_stream(), int(self.is_min), p.vp, p.row_left, p.row_ptr,
                  p.col, self.labels, mistakes)

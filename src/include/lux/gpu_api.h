@@ -74,9 +74,20 @@ void lux_gpu_csr_scatter(uint64_t stream, uint64_t ep, const lux::V_ID* col,
 void lux_gpu_frontier_expand(uint64_t stream, int old_dense,
                              lux::V_ID in_row_left, lux::V_ID in_count,
                              const uint8_t* old_seg,
+                             const uint32_t* qlabels /*nullable*/,
+                             uint32_t* labels_repair /*nullable*/,
                              const lux::E_ID* push_row_ptr,
                              lux_uint2* items, uint32_t* counter,
                              uint32_t max_items);
+void lux_gpu_frontier_fixup(uint64_t stream, lux::V_ID vp,
+                            lux::V_ID row_left, lux::V_ID capacity,
+                            int built_dense, const uint32_t* snapshot,
+                            const uint32_t* labels_part,
+                            const uint32_t* deg_part, uint8_t* new_seg,
+                            uint32_t* annex, uint8_t* tmp_seg,
+                            uint32_t* meta,
+                            const uint32_t* item_counter /*nullable*/,
+                            uint32_t max_items);
 void lux_gpu_push_chunk_scatter(uint64_t stream, int is_min, int new_dense,
                                 const lux_uint2* items,
                                 const uint32_t* counter, uint32_t max_items,

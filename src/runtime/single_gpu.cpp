@@ -182,7 +182,7 @@ SingleGpuPush::SingleGpuPush(const DeviceGraph& g, bool is_min, V_ID source,
   // the nv/16 push threshold + per-8192-edge chunk splits
   max_items_ = (uint32_t)(g.nv / SPARSE_THRESHOLD + g.ne / 8192 + 1024);
   items_ = arena.alloc_n<lux_uint2>(max_items_);
-  item_counter_ = arena.alloc_n<uint32_t>(2);
+  item_counter_ = arena.alloc_n<uint32_t>(4);
   capacity_ = frontier_capacity(g.nv);
   // seed labels + frontier (sssp_gpu.cu:733-744, components_gpu.cu:733-740)
   std::vector<uint32_t> hl(g.nv);
@@ -214,7 +214,8 @@ V_ID SingleGpuPush::step() {
                            hipMemcpyDeviceToDevice, s_));
   LUX_OK(hipMemsetAsync(new_fq_, 0, sizeof(FrontierHeader), s_));
   bool new_dense = fq_type_ == FrontierHeader::DENSE_BITMAP;
-  bool pull_fallback = fq_num_ > g_.nv / SPARSE_THRESHOLD;
+  bool pull_fallback = fq_num_ > g_.nv / SPARSE_THRESHOLD || force_pull_;
+  force_pull_ = false;
   if (pull_fallback) {
     new_dense = true;
     // dense pull iteration; labels_ serves as both old (all) and new (slice)
@@ -223,11 +224,11 @@ V_ID SingleGpuPush::step() {
                       bins_.bin2v, row_ptr_, g_.src, snapshot_, labels_,
                       nullptr, 0, 0.0f);
   } else {
-    LUX_OK(hipMemsetAsync(item_counter_, 0, 8, s_));
+    LUX_OK(hipMemsetAsync(item_counter_, 0, 16, s_));
     lux_gpu_frontier_expand(
         (uint64_t)s_, fq_type_ == FrontierHeader::DENSE_BITMAP ? 1 : 0, 0,
         fq_type_ == FrontierHeader::DENSE_BITMAP ? g_.nv : fq_num_, fq_,
-        push_row_ptr_, items_, item_counter_, max_items_);
+        nullptr, nullptr, push_row_ptr_, items_, item_counter_, max_items_);
     lux_gpu_push_chunk_scatter((uint64_t)s_, is_min_ ? 1 : 0,
                                new_dense ? 1 : 0, items_, item_counter_,
                                max_items_, push_row_ptr_, push_col_,
@@ -259,6 +260,20 @@ V_ID SingleGpuPush::step() {
       LUX_OK(hipStreamSynchronize(s_));
     }
   }
+  if (!pull_fallback) {
+    // expand overflow guard: truncated work items
+    // mean unrelaxed edges — recover loudly with a forced pull iteration
+    // (re-relaxes every edge; labels are monotone so this is safe)
+    uint32_t hc[4];
+    LUX_OK(hipMemcpyAsync(hc, item_counter_, 16, hipMemcpyDeviceToHost, s_));
+    LUX_OK(hipStreamSynchronize(s_));
+    if (hc[3] || hc[0] > max_items_) {
+      fprintf(stderr,
+              "[lux] frontier expand overflow (%u items > cap %u): forcing "
+              "a pull iteration to recover\n", hc[0], max_items_);
+      force_pull_ = true;
+    }
+  }
   fq_type_ = new_dense ? FrontierHeader::DENSE_BITMAP
                        : FrontierHeader::SPARSE_QUEUE;
   fq_num_ = hh.numNodes;
@@ -273,7 +288,7 @@ int SingleGpuPush::run(int max_iters) {
   while (true) {
     V_ID n = step();
     iters_++;
-    if (n == 0) break;
+    if (n == 0 && !force_pull_) break;  // overflow: don't terminate early
     if (max_iters && iters_ >= max_iters) break;
   }
   return iters_;

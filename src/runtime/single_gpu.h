@@ -108,6 +108,7 @@ class SingleGpuPush {
   uint32_t max_items_;
   V_ID capacity_;
   uint32_t fq_type_, fq_num_;
+  bool force_pull_ = false;  // set on expand overflow (loud recovery)
   int iters_ = 0;
 };
 

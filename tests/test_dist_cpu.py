@@ -180,6 +180,167 @@ def test_dist_pagerank_async_pipeline(world):
     np.testing.assert_allclose(got, want, rtol=1e-6)
 
 
+def _uf_find(parent, v):
+    while parent[v] != v:
+        parent[v] = parent[parent[v]]
+        v = parent[v]
+    return v
+
+
+def _uf_union(parent, a, b):
+    ra, rb = _uf_find(parent, a), _uf_find(parent, b)
+    if ra == rb:
+        return
+    if ra < rb:  # larger id wins: flattened labels = component max
+        parent[ra] = rb
+    else:
+        parent[rb] = ra
+
+
+def _dist_cc_star_worker(rank, world, port, scale, ne, seed, outq):
+    """The CCUnionFindEngine star-forest exchange (cc_engine.py run):
+    per-rank union of its edge partition, then all-gather label vectors
+    and union the peers' stars until fixpoint — with a CPU union-find."""
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from lux_amd import dist as dx
+        from lux_amd.graph import Graph
+        g = Graph.rmat(scale, ne, seed=seed, sym=True)
+        part = g.partition(world)
+        rl, rr, cl, ce, srcs, _w = g.slice(part, rank)
+        parent = np.arange(g.nv, dtype=np.int64)
+        counts = np.diff(np.concatenate([[cl], ce])).astype(np.int64)
+        dsts = np.repeat(np.arange(rl, rr + 1, dtype=np.int64), counts)
+        for s_, d_ in zip(srcs.astype(np.int64), dsts):
+            _uf_union(parent, s_, d_)
+        labels = np.array([_uf_find(parent, v) for v in range(g.nv)])
+        lab_t = torch.from_numpy(labels.astype(np.int32))
+        gathered = torch.empty(world * g.nv, dtype=torch.int32)
+        rounds = 0
+        while True:
+            dx.all_gather_slices(gathered, lab_t, [g.nv] * world,
+                                 [q * g.nv for q in range(world)])
+            gn = gathered.numpy()
+            for q in range(world):
+                if q != rank:
+                    star = gn[q * g.nv:(q + 1) * g.nv]
+                    for v in range(g.nv):
+                        _uf_union(parent, v, int(star[v]))
+            new = np.array([_uf_find(parent, v) for v in range(g.nv)])
+            changed = torch.tensor([int((new != labels).any())])
+            dx.all_reduce_sum_(changed)
+            labels = new
+            lab_t = torch.from_numpy(labels.astype(np.int32))
+            rounds += 1
+            if int(changed.item()) == 0:
+                break
+        if rank == 0:
+            outq.put((labels.copy(), rounds))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_dist_cc_star_forest(world=4):
+    from lux_amd.graph import Graph
+    scale, ne, seed = 9, 4000, 23
+    ctx = mp.get_context("spawn")
+    outq = ctx.SimpleQueue()
+    mp.spawn(_dist_cc_star_worker,
+             args=(world, _find_port(), scale, ne, seed, outq),
+             nprocs=world, join=True)
+    got, rounds = outq.get()
+    # expected: component max over the SAME undirected graph
+    g = Graph.rmat(scale, ne, seed=seed, sym=True)
+    parent = np.arange(g.nv, dtype=np.int64)
+    counts = np.diff(np.concatenate([[0], g.col_end])).astype(np.int64)
+    dsts = np.repeat(np.arange(g.nv, dtype=np.int64), counts)
+    for s_, d_ in zip(g.src.astype(np.int64), dsts):
+        _uf_union(parent, s_, d_)
+    want = np.array([_uf_find(parent, v) for v in range(g.nv)])
+    assert np.array_equal(got, want)
+    assert rounds <= 4  # monotone merges: O(log P) rounds
+
+
+def _dist_cf_slices_worker(rank, world, port, outq):
+    """CF latent-vector exchange shape (cf_engine.py step): uneven
+    K-scaled element slices of the [nv, K] factor matrix."""
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from lux_amd import dist as dx
+        nv, K = 10, 4
+        verts = [3, 7]
+        lefts = [0, 3]
+        old = torch.arange(nv * K, dtype=torch.float32)
+        my = old.narrow(0, lefts[rank] * K, verts[rank] * K).clone()
+        my *= (rank + 1) * 10.0
+        for _ in range(3):  # repeated sweeps reuse the same buffers
+            dx.all_gather_slices(old, my, [v * K for v in verts],
+                                 [off * K for off in lefts])
+        if rank == 0:
+            outq.put(old.numpy().copy())
+    finally:
+        dist.destroy_process_group()
+
+
+def test_dist_cf_vector_slices(world=2):
+    ctx = mp.get_context("spawn")
+    outq = ctx.SimpleQueue()
+    mp.spawn(_dist_cf_slices_worker, args=(world, _find_port(), outq),
+             nprocs=world, join=True)
+    got = outq.get()
+    want = np.arange(40, dtype=np.float32)
+    want[:12] *= 10.0
+    want[12:] *= 20.0
+    np.testing.assert_array_equal(got, want)
+
+
+def _dist_multi_worker(rank, world, port, outq):
+    """exchange_multi_async: one batched p2p round carrying a byte-segment
+    part AND a label part where rank 1 skips publishing (the push engine's
+    sparse-iteration label skip)."""
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from lux_amd import dist as dx
+        seg_used = [24, 40]
+        seg_off = [0, 48]
+        segs = torch.zeros(96, dtype=torch.uint8)
+        my_seg = torch.full((seg_used[rank],), 50 + rank, dtype=torch.uint8)
+        lab_n = [5, 0]  # rank 1 skips its label publish
+        lab_off = [0, 5]
+        labels = torch.full((9,), -1, dtype=torch.int32)
+        my_lab = torch.full((5,), 100 + rank, dtype=torch.int32) \
+            if rank == 0 else None
+        h = dx.exchange_multi_async([
+            (segs, my_seg, seg_used, seg_off),
+            (labels, my_lab, lab_n, lab_off)])
+        h.wait()
+        if rank == 0:
+            outq.put((segs.numpy().copy(), labels.numpy().copy()))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_dist_exchange_multi(world=2):
+    ctx = mp.get_context("spawn")
+    outq = ctx.SimpleQueue()
+    mp.spawn(_dist_multi_worker, args=(world, _find_port(), outq),
+             nprocs=world, join=True)
+    segs, labels = outq.get()
+    assert (segs[:24] == 50).all() and (segs[24:48] == 0).all()
+    assert (segs[48:88] == 51).all() and (segs[88:] == 0).all()
+    assert (labels[:5] == 100).all()
+    assert (labels[5:] == -1).all()  # skipped publisher left stale values
+
+
 def _dist_bytes_worker(rank, world, port, outq):
     """Uneven uint8 segment exchange — the frontier (header+payload byte
     segment) all-gather shape PushEngine uses (push_engine.py step)."""

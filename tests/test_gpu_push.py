@@ -80,11 +80,10 @@ def test_sssp_multipart_single_process():
     for it in range(4 * nv):
         for e in engines:
             e.step()
-        # manual exchange across the P single-rank engines
-        headers = []
-        for j, ej in enumerate(engines):
-            hdr = ej.new_seg[:8].view(U32).cpu().numpy().view(np.uint32)
-            headers.append((int(hdr[0]), int(hdr[1])))
+        # manual exchange across the P single-rank engines: segments,
+        # label annexes, label slices and the meta records
+        mh = np.stack([e.meta_mine.cpu().numpy().view(np.uint32)
+                       for e in engines])
         for ei in engines:
             for j, ej in enumerate(engines):
                 if parts[j].vp == 0:
@@ -93,8 +92,13 @@ def test_sssp_multipart_single_process():
                                  parts[j].vp).copy_(ej.labels_part)
                 ei.fq_all.narrow(0, int(ei.seg_off[j]),
                                  ei.seg_bytes[j]).copy_(ej.new_seg)
-            ei.headers = list(headers)
-        if sum(h[1] for h in headers) == 0:
+                cap_j = int(ei.annex_off[j + 1] - ei.annex_off[j])
+                ei.fq_annex_all.narrow(0, int(ei.annex_off[j]),
+                                       cap_j).copy_(ej.new_annex[:cap_j])
+            ei.meta_host = mh.copy()
+            ei.labels_current = True
+            ei.headers = [(int(mh[q, 0]), int(mh[q, 1])) for q in range(P)]
+        if int(mh[:, 1].sum()) == 0 and not mh[:, 4].any():
             break
     got = engines[0].labels.cpu().numpy().view(np.uint32)
     g = Graph.rmat(scale, ne, seed=seed)
