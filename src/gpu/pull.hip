@@ -134,7 +134,9 @@ __device__ __forceinline__ typename Val<M>::T finish(
 }
 
 struct PullArgs {
-  const E_ID* row_ptr;   // u64[vp+1], local 0-based
+  const void* row_ptr;   // RowT[vp+1], local 0-based: u64, or u32 for the
+                         // blocked path (block-local offsets < 2^32 — half
+                         // the per-row sweep traffic, NOTES_r2 item 1)
   const V_ID* col;       // u32[ep], local
   const void* oldv;      // full nv
   void* newv;            // vp
@@ -179,32 +181,34 @@ __device__ __forceinline__ typename Val<M>::T gather_range(
 }
 
 // ---- bin0: thread per vertex ----
-template <PullMode M>
+template <PullMode M, typename RowT>
 __global__ void pull_thread_kernel(uint32_t n0, const V_ID* bin0,
                                    PullArgs a) {
   using V = Val<M>;
   using T = typename V::T;
   const T* oldv = (const T*)a.oldv;
   T* newv = (T*)a.newv;
+  const RowT* row_ptr = (const RowT*)a.row_ptr;
   uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
   for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; i < n0;
        i += stride) {
     V_ID v = bin0[i];
     if (V::SKIP_SETTLED && oldv[a.row_left + v] != (T)INF_LABEL)
       continue;  // settled hop label: newv keeps the seed
-    E_ID b = a.row_ptr[v], e = a.row_ptr[v + 1];
+    E_ID b = row_ptr[v], e = row_ptr[v + 1];
     T acc = gather_range<M>(oldv, a.col, b, e, 1);
     store_result<M>(&newv[v], acc);
   }
 }
 
 // ---- bin1: wave per vertex ----
-template <PullMode M>
+template <PullMode M, typename RowT>
 __global__ void pull_wave_kernel(uint32_t n1, const V_ID* bin1, PullArgs a) {
   using V = Val<M>;
   using T = typename V::T;
   const T* oldv = (const T*)a.oldv;
   T* newv = (T*)a.newv;
+  const RowT* row_ptr = (const RowT*)a.row_ptr;
   int lane = threadIdx.x & (WAVE - 1);
   uint64_t wave_id = ((uint64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
   uint64_t nwaves = ((uint64_t)gridDim.x * blockDim.x) / WAVE;
@@ -212,7 +216,7 @@ __global__ void pull_wave_kernel(uint32_t n1, const V_ID* bin1, PullArgs a) {
     V_ID v = bin1[i];
     if (V::SKIP_SETTLED && oldv[a.row_left + v] != (T)INF_LABEL)
       continue;
-    E_ID b = a.row_ptr[v], e = a.row_ptr[v + 1];
+    E_ID b = row_ptr[v], e = row_ptr[v + 1];
     T acc = gather_range<M>(oldv, a.col, b + lane, e, WAVE);
     acc = V::reduce_wave(acc);
     if (lane == 0)
@@ -221,7 +225,7 @@ __global__ void pull_wave_kernel(uint32_t n1, const V_ID* bin1, PullArgs a) {
 }
 
 // ---- bin2: chunk accumulation (hubs) ----
-template <PullMode M>
+template <PullMode M, typename RowT>
 __global__ void pull_chunk_kernel(uint32_t n2, const uint2* bin2,
                                   PullArgs a) {
   using V = Val<M>;
@@ -229,13 +233,14 @@ __global__ void pull_chunk_kernel(uint32_t n2, const uint2* bin2,
   __shared__ T lds[BLOCK / WAVE];
   const T* oldv = (const T*)a.oldv;
   T* newv = (T*)a.newv;
+  const RowT* row_ptr = (const RowT*)a.row_ptr;
   for (uint32_t i = blockIdx.x; i < n2; i += gridDim.x) {
     uint2 ent = bin2[i];
     V_ID v = ent.x;
     if (V::SKIP_SETTLED && oldv[a.row_left + v] != (T)INF_LABEL)
       continue;
-    E_ID b = a.row_ptr[v] + (E_ID)ent.y * CHUNK_EDGES;
-    E_ID e = a.row_ptr[v + 1];
+    E_ID b = (E_ID)row_ptr[v] + (E_ID)ent.y * CHUNK_EDGES;
+    E_ID e = row_ptr[v + 1];
     if (e > b + CHUNK_EDGES) e = b + CHUNK_EDGES;
     T acc = gather_range<M>(oldv, a.col, b + threadIdx.x, e, blockDim.x);
     // block reduce (sum mode) or wave+lds fold (min/max)
@@ -255,23 +260,23 @@ __global__ void pull_chunk_kernel(uint32_t n2, const uint2* bin2,
 }
 
 
-template <PullMode M>
+template <PullMode M, typename RowT>
 static void pull_iter(hipStream_t s, uint32_t n0, const V_ID* bin0,
                       uint32_t n1, const V_ID* bin1, uint32_t n2,
                       const uint2* bin2, uint32_t nbig, const V_ID* bin2v,
                       const PullArgs& a) {
   if (nbig) {
-    hipLaunchKernelGGL(pull_chunk_kernel<M>,
+    hipLaunchKernelGGL((pull_chunk_kernel<M, RowT>),
                        dim3(n2 > MAX_GRID ? MAX_GRID : n2), dim3(BLOCK), 0, s,
                        n2, bin2, a);
   }
   if (n1)
-    hipLaunchKernelGGL(pull_wave_kernel<M>,
+    hipLaunchKernelGGL((pull_wave_kernel<M, RowT>),
                        dim3(grid_for((uint64_t)n1 * WAVE)), dim3(BLOCK), 0, s,
                        n1, bin1, a);
   if (n0)
-    hipLaunchKernelGGL(pull_thread_kernel<M>, dim3(grid_for(n0)), dim3(BLOCK),
-                       0, s, n0, bin0, a);
+    hipLaunchKernelGGL((pull_thread_kernel<M, RowT>), dim3(grid_for(n0)),
+                       dim3(BLOCK), 0, s, n0, bin0, a);
 }
 
 // PR epilogue over the whole partition: pr = (1-a)/nv + a*sum, stored
@@ -307,25 +312,22 @@ void lux_gpu_build_bins(uint64_t stream, uint32_t vp, const E_ID* row_ptr,
 }
 
 // mode: 0 = PageRank float-sum, 1 = u32 min (SSSP dense), 2 = u32 max (CC).
+// row_u32: row_ptr is u32[vp+1] (block-local offsets) instead of u64.
 void lux_gpu_pull_iter(uint64_t stream, int mode, uint32_t n0,
                        const V_ID* bin0, uint32_t n1, const V_ID* bin1,
                        uint32_t n2, const uint2* bin2, uint32_t nbig,
-                       const V_ID* bin2v, const E_ID* row_ptr,
+                       const V_ID* bin2v, const void* row_ptr, int row_u32,
                        const V_ID* col, const void* oldv, void* newv,
                        const V_ID* deg, V_ID row_left, float init_rank) {
   hipStream_t s = (hipStream_t)stream;
   PullArgs a{row_ptr, col, oldv, newv, deg, row_left, init_rank};
+#define LUX_PI(M_)                                                            do {                                                                          if (row_u32)                                                                  pull_iter<M_, uint32_t>(s, n0, bin0, n1, bin1, n2, bin2, nbig, bin2v,                               a);                                               else                                                                          pull_iter<M_, uint64_t>(s, n0, bin0, n1, bin1, n2, bin2, nbig, bin2v,                               a);                                             } while (0)
   switch (mode) {
-    case 0:
-      pull_iter<PR_SUM>(s, n0, bin0, n1, bin1, n2, bin2, nbig, bin2v, a);
-      break;
-    case 1:
-      pull_iter<LAB_MIN>(s, n0, bin0, n1, bin1, n2, bin2, nbig, bin2v, a);
-      break;
-    case 2:
-      pull_iter<LAB_MAX>(s, n0, bin0, n1, bin1, n2, bin2, nbig, bin2v, a);
-      break;
+    case 0: LUX_PI(PR_SUM); break;
+    case 1: LUX_PI(LAB_MIN); break;
+    case 2: LUX_PI(LAB_MAX); break;
   }
+#undef LUX_PI
   LUX_POST_LAUNCH(stream);
 }
 

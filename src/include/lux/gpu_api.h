@@ -27,10 +27,12 @@ void lux_gpu_bipartite_edges(uint64_t stream, uint64_t seed,
                              lux::WeightType* w);
 void lux_gpu_hist_u32(uint64_t stream, uint64_t n, const lux::V_ID* ids,
                       uint32_t* hist);
-uint32_t lux_gpu_scan_partials_size(uint32_t n);
-void lux_gpu_scan_end_offsets(uint64_t stream, uint32_t n,
+uint32_t lux_gpu_scan_partials_size(uint64_t n);
+void lux_gpu_scan_end_offsets(uint64_t stream, uint64_t n,
                               const uint32_t* in, lux::E_ID* out_end,
                               unsigned long long* partials);
+void lux_gpu_u64_to_u32(uint64_t stream, uint64_t n,
+                        const unsigned long long* in, uint32_t* out);
 void lux_gpu_edges_to_csc(uint64_t stream, uint32_t nv, uint64_t ne,
                           const lux::V_ID* src, const lux::V_ID* dst,
                           const lux::WeightType* w, lux::E_ID* col_end,
@@ -42,12 +44,12 @@ void lux_gpu_local_row_ptr(uint64_t stream, uint32_t vp, lux::E_ID col_left,
                            lux::E_ID* row_ptr_loc);
 void lux_gpu_blocked_count(uint64_t stream, uint64_t ep, const lux::V_ID* col,
                            const lux::E_ID* row_ptr_loc, lux::V_ID vp,
-                           int shift, uint32_t* counts);
+                           const lux::V_ID* bounds, int nb, uint32_t* counts);
 void lux_gpu_blocked_scatter(uint64_t stream, uint64_t ep,
                              const lux::V_ID* col,
                              const lux::E_ID* row_ptr_loc, lux::V_ID vp,
-                             int shift, unsigned long long* cursor,
-                             lux::V_ID* out_col);
+                             const lux::V_ID* bounds, int nb,
+                             unsigned long long* cursor, lux::V_ID* out_col);
 
 // pull.hip
 void lux_gpu_build_bins(uint64_t stream, uint32_t vp,
@@ -58,9 +60,9 @@ void lux_gpu_pull_iter(uint64_t stream, int mode, uint32_t n0,
                        const lux::V_ID* bin0, uint32_t n1,
                        const lux::V_ID* bin1, uint32_t n2,
                        const lux_uint2* bin2, uint32_t nbig,
-                       const lux::V_ID* bin2v, const lux::E_ID* row_ptr,
-                       const lux::V_ID* col, const void* oldv, void* newv,
-                       const lux::V_ID* deg, lux::V_ID row_left,
+                       const lux::V_ID* bin2v, const void* row_ptr,
+                       int row_u32, const lux::V_ID* col, const void* oldv,
+                       void* newv, const lux::V_ID* deg, lux::V_ID row_left,
                        float init_rank);
 void lux_gpu_pull_finish_pr(uint64_t stream, lux::V_ID vp, float* newv,
                             const lux::V_ID* deg, lux::V_ID row_left,

@@ -130,7 +130,7 @@ void SingleGpuPagerank::iterate(int iters) {
     LUX_OK(hipMemsetAsync(new_, 0, sizeof(float) * g_.nv, s_));
     lux_gpu_pull_iter((uint64_t)s_, 0, bins_.n0, bins_.bin0, bins_.n1,
                       bins_.bin1, bins_.n2, bins_.bin2, bins_.nbig,
-                      bins_.bin2v, row_ptr_, g_.src, old_, new_, deg_, 0,
+                      bins_.bin2v, row_ptr_, 0, g_.src, old_, new_, deg_, 0,
                       init_rank);
     lux_gpu_pull_finish_pr((uint64_t)s_, g_.nv, new_, deg_, 0, init_rank);
     std::swap(old_, new_);
@@ -221,7 +221,7 @@ V_ID SingleGpuPush::step() {
     // dense pull iteration; labels_ serves as both old (all) and new (slice)
     lux_gpu_pull_iter((uint64_t)s_, is_min_ ? 1 : 2, bins_.n0, bins_.bin0,
                       bins_.n1, bins_.bin1, bins_.n2, bins_.bin2, bins_.nbig,
-                      bins_.bin2v, row_ptr_, g_.src, snapshot_, labels_,
+                      bins_.bin2v, row_ptr_, 0, g_.src, snapshot_, labels_,
                       nullptr, 0, 0.0f);
   } else {
     LUX_OK(hipMemsetAsync(item_counter_, 0, 16, s_));
