@@ -136,6 +136,34 @@ def io_read(path, want_weights=False):
     return nv, ne, col_end, src, weight
 
 
+def io_read_col_end(path, nv):
+    col_end = np.empty(nv, np.uint64)
+    r = cpu.lux_io_read_col_end(path.encode(), ptr(col_end, u64p))
+    if r != 0:
+        raise IOError(f"lux_read_col_end failed: {path}")
+    return col_end
+
+
+def io_read_slice(path, row_left, row_right, ep, want_weights=False):
+    """Read ONLY [row_left, row_right]'s col_end + edge slice (ep edges,
+    known from a prior col_end read). Reference parity: the per-partition
+    fseeko load (core/push_model.inl:100-119)."""
+    nrows = row_right - row_left + 1
+    col_end = np.empty(nrows, np.uint64)
+    src = np.empty(max(ep, 1), np.uint32)
+    weight = np.empty(max(ep, 1), np.int32) if want_weights else None
+    cpu.lux_io_read_slice.restype = ctypes.c_int64
+    r = cpu.lux_io_read_slice(path.encode(), ctypes.c_uint32(row_left),
+                              ctypes.c_uint32(row_right),
+                              ptr(col_end, u64p), ptr(src, u32p),
+                              ptr(weight, i32p))
+    if r < 0:
+        raise IOError(f"lux_read_slice failed: {path}")
+    if r != ep:
+        raise IOError(f"lux_read_slice: {path}: got {r} edges, want {ep}")
+    return col_end, src[:ep], (weight[:ep] if weight is not None else None)
+
+
 def partition(nv, ne, col_end, nparts):
     rl = np.empty(nparts, np.uint32)
     rr = np.empty(nparts, np.uint32)

@@ -51,10 +51,35 @@ def rmat_edges(stream, seed, scale, ne, src, dst):
                              _u64(ne), dp(src), dp(dst))
 
 
+def rmat_edges_chunk(stream, seed, scale, e0, ne, src, dst):
+    lib().lux_gpu_rmat_edges_chunk(_u64(stream), _u64(seed),
+                                   ctypes.c_int(scale), _u64(e0), _u64(ne),
+                                   dp(src), dp(dst))
+
+
 def rmat_edges_folded(stream, seed, scale, nv, ne, src, dst):
     lib().lux_gpu_rmat_edges_folded(_u64(stream), _u64(seed),
                                     ctypes.c_int(scale), _u32(nv), _u64(ne),
                                     dp(src), dp(dst))
+
+
+def rmat_edges_folded_chunk(stream, seed, scale, nv, e0, ne, src, dst):
+    lib().lux_gpu_rmat_edges_folded_chunk(
+        _u64(stream), _u64(seed), ctypes.c_int(scale), _u32(nv), _u64(e0),
+        _u64(ne), dp(src), dp(dst))
+
+
+def bipartite_edges_chunk(stream, seed, n_users, n_items, e0, ne, src, dst,
+                          w):
+    lib().lux_gpu_bipartite_edges_chunk(
+        _u64(stream), _u64(seed), _u32(n_users), _u32(n_items), _u64(e0),
+        _u64(ne), dp(src), dp(dst), dp(w))
+
+
+def slice_scatter(stream, n, src, dst, w, rl, rr, cursor, out_col, out_w):
+    lib().lux_gpu_slice_scatter(_u64(stream), _u64(n), dp(src), dp(dst),
+                                dp(w), _u32(rl), _u32(rr), dp(cursor),
+                                dp(out_col), dp(out_w))
 
 
 def cf_iter(stream, n0, bin0, n1, bin1, n2, bin2, nbig, bin2v, row_ptr, col,

@@ -8,7 +8,7 @@ from .. import dist as dx
 from ..cc_engine import CCUnionFindEngine
 from ..engine import GraphPart
 from ..push_engine import PushEngine
-from .common import (ElapsedTimer, load_device_graph, parse_input_args,
+from .common import (ElapsedTimer, load_part, parse_input_args,
                      print_memory_estimate)
 
 
@@ -32,9 +32,14 @@ def build_cc_bench(args, device):
     # connected components is defined on undirected graphs, so the bench
     # graph stores both directions of ne/2 generated pairs (ne total edges)
     nv = getattr(args, "nv", None) or 41652230
-    full = DeviceCSC.rmat_folded(nv, args.edges, seed=args.seed,
-                                 device=device, sym=True)
-    part = GraphPart(full, dx.world_size(), dx.rank())
+    if dx.world_size() > 1:  # rank-sliced build: graph/P per rank
+        part = GraphPart.rmat_folded_sliced(nv, args.edges, dx.world_size(),
+                                            dx.rank(), seed=args.seed,
+                                            device=device, sym=True)
+    else:
+        full = DeviceCSC.rmat_folded(nv, args.edges, seed=args.seed,
+                                     device=device, sym=True)
+        part = GraphPart(full, 1, 0)
     return CCBench(part, labelprop=getattr(args, "labelprop", False)), part
 
 
@@ -45,10 +50,9 @@ def main(argv=None):
     local = dx.env_local_rank()
     torch.cuda.set_device(local)
     device = f"cuda:{local}"
-    full = load_device_graph(a, device)
+    part = load_part(a, device)
     if dx.rank() == 0:
-        print_memory_estimate(full.nv, full.ne, dx.world_size())
-    part = GraphPart(full, dx.world_size(), dx.rank())
+        print_memory_estimate(part.nv, part.ne, dx.world_size())
     eng = PushEngine(part, PushEngine.MODE_MAX) if a.labelprop \
         else CCUnionFindEngine(part)
     with ElapsedTimer():

@@ -5,7 +5,7 @@ import sys
 from .. import dist as dx
 from ..cf_engine import CFALSEngine, CFEngine
 from ..engine import GraphPart
-from .common import (ElapsedTimer, load_device_graph, parse_input_args,
+from .common import (ElapsedTimer, load_part, parse_input_args,
                      print_memory_estimate)
 
 # NetFlix-prize shape (reference README.md:86: 497,959 V / 200,961,014 E)
@@ -16,9 +16,14 @@ NETFLIX_ITEMS = 17770
 def build_cf_bench(args, device):
     from ..engine import DeviceCSC
     ne = args.edges if args.edges != (1 << 31) else 200961014
-    full = DeviceCSC.bipartite(NETFLIX_USERS, NETFLIX_ITEMS, ne,
-                               seed=args.seed, device=device)
-    part = GraphPart(full, dx.world_size(), dx.rank())
+    if dx.world_size() > 1:  # rank-sliced build: graph/P per rank
+        part = GraphPart.bipartite_sliced(NETFLIX_USERS, NETFLIX_ITEMS, ne,
+                                          dx.world_size(), dx.rank(),
+                                          seed=args.seed, device=device)
+    else:
+        full = DeviceCSC.bipartite(NETFLIX_USERS, NETFLIX_ITEMS, ne,
+                                   seed=args.seed, device=device)
+        part = GraphPart(full, 1, 0)
     cls = CFALSEngine if getattr(args, "als", False) else CFEngine
     return cls(part, K=64), part
 
@@ -30,11 +35,10 @@ def main(argv=None):
     local = dx.env_local_rank()
     torch.cuda.set_device(local)
     device = f"cuda:{local}"
-    full = load_device_graph(a, device, weighted=True)
+    part = load_part(a, device, weighted=True)
     if dx.rank() == 0:
-        print_memory_estimate(full.nv, full.ne, dx.world_size(),
+        print_memory_estimate(part.nv, part.ne, dx.world_size(),
                               weighted=True, k=a.k)
-    part = GraphPart(full, dx.world_size(), dx.rank())
     eng = (CFALSEngine if a.als else CFEngine)(part, K=a.k)
     with ElapsedTimer():
         for _ in range(a.num_iter):

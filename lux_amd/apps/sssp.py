@@ -5,7 +5,7 @@ import sys
 from .. import dist as dx
 from ..engine import GraphPart
 from ..push_engine import PushEngine
-from .common import (ElapsedTimer, load_device_graph, parse_input_args,
+from .common import (ElapsedTimer, load_part, parse_input_args,
                      print_memory_estimate)
 
 
@@ -23,9 +23,14 @@ class SSSPBench:
 
 def build_sssp_bench(args, device):
     from ..engine import DeviceCSC
-    full = DeviceCSC.rmat(args.scale, args.edges, seed=args.seed,
-                          device=device)
-    part = GraphPart(full, dx.world_size(), dx.rank())
+    if dx.world_size() > 1:  # rank-sliced build: graph/P per rank
+        part = GraphPart.rmat_sliced(args.scale, args.edges,
+                                     dx.world_size(), dx.rank(),
+                                     seed=args.seed, device=device)
+    else:
+        full = DeviceCSC.rmat(args.scale, args.edges, seed=args.seed,
+                              device=device)
+        part = GraphPart(full, 1, 0)
     return SSSPBench(part, 0), part
 
 
@@ -36,10 +41,9 @@ def main(argv=None):
     local = dx.env_local_rank()
     torch.cuda.set_device(local)
     device = f"cuda:{local}"
-    full = load_device_graph(a, device)
+    part = load_part(a, device)
     if dx.rank() == 0:
-        print_memory_estimate(full.nv, full.ne, dx.world_size())
-    part = GraphPart(full, dx.world_size(), dx.rank())
+        print_memory_estimate(part.nv, part.ne, dx.world_size())
     eng = PushEngine(part, PushEngine.MODE_MIN, source=a.start)
     with ElapsedTimer():
         iters = eng.run()

@@ -1,5 +1,6 @@
 // C ABI for the CPU core, consumed by lux_amd/_native.py via ctypes.
 // All pointers are caller-allocated (numpy / torch CPU tensors).
+#include <cstdio>
 #include <cstring>
 #include <vector>
 
@@ -105,6 +106,42 @@ int lux_io_read(const char* path, uint64_t* col_end, uint32_t* src,
   if (weight)
     std::memcpy(weight, g.weight.data(), sizeof(int32_t) * g.ne);
   return 0;
+}
+
+// Per-partition slice read (reference parity: the per-node fseeko load,
+// core/push_model.inl:100-119): reads ONLY [row_left, row_right]'s col_end
+// slice + edge slice. Caller sizes src/weight from a prior col_end read
+// (lux_io_read_col_end); returns the slice edge count or -1.
+int lux_io_read_col_end(const char* path, uint64_t* col_end) {
+  uint32_t nv;
+  uint64_t ne;
+  bool w;
+  if (!lux_read_header(path, &nv, &ne, &w)) return -1;
+  FILE* f = fopen(path, "rb");
+  if (!f) return -1;
+  if (fseeko(f, sizeof(uint32_t) + sizeof(uint64_t), SEEK_SET)) {
+    fclose(f);
+    return -1;
+  }
+  size_t got = fread(col_end, sizeof(uint64_t), nv, f);
+  fclose(f);
+  return got == nv ? 0 : -1;
+}
+
+int64_t lux_io_read_slice(const char* path, uint32_t row_left,
+                          uint32_t row_right, uint64_t* col_end_slice,
+                          uint32_t* src, int32_t* weight) {
+  std::vector<E_ID> ce;
+  std::vector<V_ID> s;
+  std::vector<WeightType> w;
+  if (!lux_read_slice(path, row_left, row_right, &ce, &s,
+                      weight ? &w : nullptr))
+    return -1;
+  if (weight && w.empty()) return -1;  // asked for weights, file has none
+  std::memcpy(col_end_slice, ce.data(), sizeof(uint64_t) * ce.size());
+  std::memcpy(src, s.data(), sizeof(uint32_t) * s.size());
+  if (weight) std::memcpy(weight, w.data(), sizeof(int32_t) * w.size());
+  return (int64_t)s.size();
 }
 
 // ---- partitioner ----

@@ -27,6 +27,22 @@ void lux_gpu_bipartite_edges(uint64_t stream, uint64_t seed,
                              lux::WeightType* w);
 void lux_gpu_hist_u32(uint64_t stream, uint64_t n, const lux::V_ID* ids,
                       uint32_t* hist);
+void lux_gpu_rmat_edges_chunk(uint64_t stream, uint64_t seed, int scale,
+                              uint64_t e0, uint64_t ne, lux::V_ID* src,
+                              lux::V_ID* dst);
+void lux_gpu_rmat_edges_folded_chunk(uint64_t stream, uint64_t seed,
+                                     int scale, lux::V_ID nv, uint64_t e0,
+                                     uint64_t ne, lux::V_ID* src,
+                                     lux::V_ID* dst);
+void lux_gpu_bipartite_edges_chunk(uint64_t stream, uint64_t seed,
+                                   lux::V_ID n_users, lux::V_ID n_items,
+                                   uint64_t e0, uint64_t ne, lux::V_ID* src,
+                                   lux::V_ID* dst, lux::WeightType* w);
+void lux_gpu_slice_scatter(uint64_t stream, uint64_t n, const lux::V_ID* src,
+                           const lux::V_ID* dst, const lux::WeightType* w,
+                           lux::V_ID rl, lux::V_ID rr,
+                           unsigned long long* cursor, lux::V_ID* out_col,
+                           lux::WeightType* out_w);
 uint32_t lux_gpu_scan_partials_size(uint64_t n);
 void lux_gpu_scan_end_offsets(uint64_t stream, uint64_t n,
                               const uint32_t* in, lux::E_ID* out_end,

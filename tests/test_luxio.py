@@ -90,3 +90,38 @@ def test_partition_slices_cover_graph():
         covered_e += part.edges(i)
     assert covered_v == g.nv
     assert covered_e == g.ne
+
+
+def test_load_sliced_matches_full_slices(tmp_path):
+    """GraphPart.load_sliced (per-partition fseeko read) == slicing the
+    fully-loaded graph, for every partition, weighted and not."""
+    import numpy as np
+
+    from lux_amd.engine import GraphPart
+    from lux_amd.graph import Graph
+    g = Graph.rmat(9, 4000, seed=31)
+    path = str(tmp_path / "g.lux")
+    g.save(path)
+    P = 3
+    part = g.partition(P)
+    for p in range(P):
+        gp = GraphPart.load_sliced(path, P, p, device="cpu")
+        rl, rr, cl, ce, srcs, _w = g.slice(part, p)
+        assert gp.row_left == rl and gp.row_right == rr
+        assert gp.col_left == cl
+        assert gp.ep == len(srcs)
+        rp = gp.row_ptr.numpy().view(np.uint64)
+        assert rp[0] == 0 and rp[-1] == gp.ep
+        np.testing.assert_array_equal(rp[1:], ce - np.uint64(cl))
+        np.testing.assert_array_equal(gp.col.numpy().view(np.uint32), srcs)
+
+    gw = Graph.bipartite(200, 32, 3000, seed=7)
+    wpath = str(tmp_path / "w.lux")
+    gw.save(wpath)
+    partw = gw.partition(2)
+    for p in range(2):
+        gp = GraphPart.load_sliced(wpath, 2, p, device="cpu",
+                                   want_weights=True)
+        rl, rr, cl, ce, srcs, w = gw.slice(partw, p)
+        np.testing.assert_array_equal(gp.col.numpy().view(np.uint32), srcs)
+        np.testing.assert_array_equal(gp.weight.numpy(), w)
