@@ -4,6 +4,7 @@
 #include <chrono>
 #include <cstdio>
 
+#include "../src/runtime/multi_gpu.h"
 #include "../src/runtime/single_gpu.h"
 #include "app_common.h"
 
@@ -12,10 +13,9 @@ using namespace lux;
 int main(int argc, char** argv) {
   AppArgs a = parse_input_args(argc, argv);
   if (a.num_gpu > 1) {
-    fprintf(stderr,
-            "[lux] multi-GPU runs use the RCCL engine: torchrun "
-            "--nproc-per-node %d -m lux_amd.apps.cf ...\n", a.num_gpu);
-    return 2;
+    // multi-GPU: exec the torchrun RCCL engine (one rank per GPU; same
+    // CLI, reference README.md:42-45 drop-in)
+    return exec_torchrun_app("lux_amd.apps.cf", a.num_gpu, argc, argv);
   }
   HostCSC g;
   if (!load_graph(a, &g, true)) return 1;

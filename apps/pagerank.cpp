@@ -3,6 +3,7 @@
 #include <chrono>
 #include <cstdio>
 
+#include "../src/runtime/multi_gpu.h"
 #include "../src/runtime/single_gpu.h"
 #include "app_common.h"
 
@@ -10,15 +11,16 @@ using namespace lux;
 
 int main(int argc, char** argv) {
   AppArgs a = parse_input_args(argc, argv);
-  if (a.num_gpu > 1) {
-    fprintf(stderr,
-            "[lux] multi-GPU runs use the RCCL engine: torchrun "
-            "--nproc-per-node %d -m lux_amd.apps.pagerank ...\n", a.num_gpu);
-    return 2;
-  }
   HostCSC g;
   if (!load_graph(a, &g, false)) return 1;
   print_memory_estimate(g.nv, g.ne, false, 1);
+  if (a.num_gpu > 1 || getenv("LUX_NATIVE_MULTI")) {
+    // native fork + RCCL engine, one child process per GPU (the
+    // reference's `pagerank -ll:gpu N` drop-in, README.md:42).
+    // LUX_NATIVE_MULTI=1 forces this path at -ll:gpu 1 so the fork+RCCL
+    // machinery is testable on a 1-GPU box.
+    return run_pagerank_multi(g, a.num_gpu, a.num_iter, a.verbose, a.dump);
+  }
 
   hipStream_t s;
   LUX_OK(hipStreamCreate(&s));

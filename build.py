@@ -96,8 +96,8 @@ def build_apps(force=False):
             if force or _newer(exe, [s, CPU_LIB, GPU_LIB] + runtime + hdrs):
                 libdir = os.path.dirname(CPU_LIB)
                 _run([HIPCC, *HIPFLAGS, "-Wno-unused-value", s, *runtime,
-                      f"-L{libdir}", "-llux_cpu", "-llux_gpu", "-o", exe,
-                      rpath])
+                      f"-L{libdir}", "-llux_cpu", "-llux_gpu",
+                      "-L/opt/rocm/lib", "-lrccl", "-o", exe, rpath])
             out.append(exe)
     return out
 

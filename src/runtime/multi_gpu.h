@@ -1,0 +1,22 @@
+// Native multi-GPU runtime: fork one child process per GPU, RCCL over
+// xGMI for the exchange (the BASELINE north star: C++/HIP + RCCL with no
+// Python in the hot path). `bin/pagerank -ll:gpu N` runs this engine
+// natively; the push/CF apps bridge to the torchrun engine (same RCCL
+// collectives, Python driver) via exec_torchrun_app.
+#pragma once
+#include "single_gpu.h"
+
+namespace lux {
+
+// Fork + RCCL distributed PageRank over an edge-balanced partition of g.
+// Parent must NOT have initialised the HIP runtime before this call
+// (children own their devices). Returns a process exit code.
+int run_pagerank_multi(const HostCSC& g, int ngpus, int iters, bool verbose,
+                       const char* dump);
+
+// Replace this process with `torchrun --nproc-per-node N -m <module>
+// <original args>` (one rank per GPU over the same RCCL exchange layer,
+// Python driver). Returns only on exec failure.
+int exec_torchrun_app(const char* module, int ngpus, int argc, char** argv);
+
+}  // namespace lux

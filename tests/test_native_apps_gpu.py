@@ -163,3 +163,28 @@ def test_native_components_labelprop_dump(tmp_path):
     got, _ = ck.load_state(out)
     want, _ = cpu_ref.cc(g)
     np.testing.assert_array_equal(got, want)
+
+
+def test_native_pagerank_multi_rccl_world1(tmp_path):
+    """The fork + RCCL native multi-GPU engine (run_pagerank_multi),
+    exercised at world 1 on this box (LUX_NATIVE_MULTI=1): same numeric
+    result as the CPU reference. The -ll:gpu N>1 path is this exact code
+    with more children."""
+    import numpy as np
+    from lux_amd import checkpoint as ck
+    from lux_amd import cpu_ref
+    from lux_amd.graph import Graph
+    lux = str(tmp_path / "g.lux")
+    out = str(tmp_path / "pr.luxs")
+    _run([f"{BIN}/rmat_gen", "-kind", "rmat", "-scale", "13", "-ne",
+          "120000", "-o", lux])
+    env = dict(os.environ, LUX_NATIVE_MULTI="1")
+    r = subprocess.run([f"{BIN}/pagerank", "-file", lux, "-ni", "5",
+                        "-dump", out], cwd=ROOT, capture_output=True,
+                       text=True, timeout=300, env=env)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "ELAPSED TIME" in r.stdout
+    got, _ = ck.load_state(out)
+    g = Graph.load(lux)
+    want = cpu_ref.pagerank(g, 5)
+    np.testing.assert_allclose(got, want, rtol=2e-4, atol=1e-9)
