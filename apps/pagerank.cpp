@@ -11,16 +11,24 @@ using namespace lux;
 
 int main(int argc, char** argv) {
   AppArgs a = parse_input_args(argc, argv);
-  HostCSC g;
-  if (!load_graph(a, &g, false)) return 1;
-  print_memory_estimate(g.nv, g.ne, false, 1);
+  if (const char* mr = getenv("LUX_MULTI_RANK")) {
+    // re-exec'd worker of the native multi-GPU launcher below
+    HostCSC g;
+    if (!load_graph(a, &g, false)) return 1;
+    return pagerank_multi_child(g, atoi(mr), atoi(getenv("LUX_MULTI_WORLD")),
+                                getenv("LUX_MULTI_IDFILE"), a.num_iter,
+                                a.verbose, a.dump);
+  }
   if (a.num_gpu > 1 || getenv("LUX_NATIVE_MULTI")) {
-    // native fork + RCCL engine, one child process per GPU (the
+    // native fork+exec + RCCL engine, one worker process per GPU (the
     // reference's `pagerank -ll:gpu N` drop-in, README.md:42).
     // LUX_NATIVE_MULTI=1 forces this path at -ll:gpu 1 so the fork+RCCL
     // machinery is testable on a 1-GPU box.
-    return run_pagerank_multi(g, a.num_gpu, a.num_iter, a.verbose, a.dump);
+    return run_pagerank_multi(a.num_gpu, argc, argv);
   }
+  HostCSC g;
+  if (!load_graph(a, &g, false)) return 1;
+  print_memory_estimate(g.nv, g.ne, false, 1);
 
   hipStream_t s;
   LUX_OK(hipStreamCreate(&s));

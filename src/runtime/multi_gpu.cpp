@@ -59,9 +59,15 @@ void dump_state_device(const char* path, const void* dev_ptr, int dtype,
   printf("[lux] wrote %s (nv=%u k=%u)\n", path, nv, k);
 }
 
-int child_pagerank(const HostCSC& g, int ngpus, int rank,
-                   const ncclUniqueId& id, int iters, bool verbose,
-                   const char* dump) {
+}  // namespace
+
+// Re-exec'd child entry (LUX_MULTI_RANK env): owns one device, joins the
+// RCCL communicator via the id file rank 0 writes (ncclGetUniqueId
+// initialises the HIP runtime, so it must never run in the launcher —
+// forked children of a HIP-initialised parent segfault in HSA).
+int pagerank_multi_child(const HostCSC& g, int rank, int ngpus,
+                         const char* idfile, int iters, bool verbose,
+                         const char* dump) {
   int ndev = 0;
   LUX_OK(hipGetDeviceCount(&ndev));
   if (rank >= ndev) {
@@ -73,6 +79,29 @@ int child_pagerank(const HostCSC& g, int ngpus, int rank,
   LUX_OK(hipSetDevice(rank));
   hipStream_t s;
   LUX_OK(hipStreamCreate(&s));
+  ncclUniqueId id;
+  if (rank == 0) {
+    LUX_NCCL(ncclGetUniqueId(&id));
+    std::string tmp = std::string(idfile) + ".tmp";
+    FILE* f = fopen(tmp.c_str(), "wb");
+    if (!f || fwrite(&id, sizeof(id), 1, f) != 1) {
+      perror(idfile);
+      return 4;
+    }
+    fclose(f);
+    rename(tmp.c_str(), idfile);
+  } else {
+    FILE* f = nullptr;
+    for (int tries = 0; tries < 1200 && !f; tries++) {  // <=120 s
+      f = fopen(idfile, "rb");
+      if (!f) usleep(100000);
+    }
+    if (!f || fread(&id, sizeof(id), 1, f) != 1) {
+      fprintf(stderr, "[lux] rank %d: no RCCL id file %s\n", rank, idfile);
+      return 4;
+    }
+    fclose(f);
+  }
   ncclComm_t comm;
   LUX_NCCL(ncclCommInitRank(&comm, ngpus, id, rank));
 
@@ -177,12 +206,19 @@ int child_pagerank(const HostCSC& g, int ngpus, int rank,
   return 0;
 }
 
-}  // namespace
-
-int run_pagerank_multi(const HostCSC& g, int ngpus, int iters, bool verbose,
-                       const char* dump) {
-  ncclUniqueId id;
-  LUX_NCCL(ncclGetUniqueId(&id));  // socket-only; no HIP init in parent
+// Launcher: fork + EXEC one child per GPU (a plain fork would inherit the
+// parent's process state; exec gives each rank a fresh runtime). The
+// parent itself never touches HIP or RCCL.
+int run_pagerank_multi(int ngpus, int argc, char** argv) {
+  char idfile[64];
+  snprintf(idfile, sizeof(idfile), "/tmp/lux_rccl_%d.id", (int)getpid());
+  unlink(idfile);
+  char exe[4096] = {0};
+  ssize_t n = readlink("/proc/self/exe", exe, sizeof(exe) - 1);
+  if (n <= 0) {
+    perror("readlink");
+    return 1;
+  }
   std::vector<pid_t> pids;
   for (int r = 0; r < ngpus; r++) {
     pid_t p = fork();
@@ -190,7 +226,20 @@ int run_pagerank_multi(const HostCSC& g, int ngpus, int iters, bool verbose,
       perror("fork");
       return 1;
     }
-    if (p == 0) _exit(child_pagerank(g, ngpus, r, id, iters, verbose, dump));
+    if (p == 0) {
+      char rs[16], ws[16];
+      snprintf(rs, sizeof(rs), "%d", r);
+      snprintf(ws, sizeof(ws), "%d", ngpus);
+      setenv("LUX_MULTI_RANK", rs, 1);
+      setenv("LUX_MULTI_WORLD", ws, 1);
+      setenv("LUX_MULTI_IDFILE", idfile, 1);
+      std::vector<char*> cargs;
+      for (int i = 0; i < argc; i++) cargs.push_back(argv[i]);
+      cargs.push_back(nullptr);
+      execv(exe, cargs.data());
+      perror("execv");
+      _exit(127);
+    }
     pids.push_back(p);
   }
   int rc = 0;
@@ -200,6 +249,7 @@ int run_pagerank_multi(const HostCSC& g, int ngpus, int iters, bool verbose,
     if (WIFEXITED(st) && WEXITSTATUS(st)) rc = WEXITSTATUS(st);
     if (WIFSIGNALED(st)) rc = 128 + WTERMSIG(st);
   }
+  unlink(idfile);
   return rc;
 }
 

@@ -8,11 +8,16 @@
 
 namespace lux {
 
-// Fork + RCCL distributed PageRank over an edge-balanced partition of g.
-// Parent must NOT have initialised the HIP runtime before this call
-// (children own their devices). Returns a process exit code.
-int run_pagerank_multi(const HostCSC& g, int ngpus, int iters, bool verbose,
-                       const char* dump);
+// Launcher: fork+exec one re-invocation of this binary per GPU (children
+// detected via LUX_MULTI_RANK env). The parent never initialises HIP.
+int run_pagerank_multi(int ngpus, int argc, char** argv);
+
+// Child entry: device `rank`, RCCL communicator joined via the id file
+// rank 0 writes; distributed pull PageRank over an edge-balanced
+// partition of g.
+int pagerank_multi_child(const HostCSC& g, int rank, int ngpus,
+                         const char* idfile, int iters, bool verbose,
+                         const char* dump);
 
 // Replace this process with `torchrun --nproc-per-node N -m <module>
 // <original args>` (one rank per GPU over the same RCCL exchange layer,
