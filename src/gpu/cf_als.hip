@@ -60,8 +60,30 @@ struct AlsGramLds {
   float W[ALS_TILE];            // staged edge weights
 };
 
+// bf16 Gram path (default): the fp32 MFMA formulation above runs at the
+// 155 TF f32 matrix rate and IS the sweep's bound (measured r2: the gram
+// chunk kernel ~14 ms/sweep ~= the f32-MFMA issue floor). bf16 inputs
+// with fp32 accumulation move the Gram onto v_mfma_f32_16x16x32_bf16
+// (2075 TF): one K=32 MFMA per 16x16 tile per 32-edge LDS tile — 10
+// MFMAs per tile instead of 80 — and the rhs folds from registers, so
+// the LDS image shrinks to a transposed bf16 [dim][edge] panel whose
+// fragments are single ds_read_b128s. LUX_ALS_F32=1 restores the exact
+// fp32 path.
+constexpr int ALS_BPITCH = 40;  // bf16 row pitch: 80 B = b128-aligned;
+                                // 16-lane b128 groups land on distinct
+                                // 4-bank runs (20 dwords/row, 20*16%64
+                                // spread) -> conflict-free reads
+
+struct alignas(16) AlsGramLdsBf {
+  __bf16 S[ALS_K * ALS_BPITCH];  // TRANSPOSED: row = dim, col = edge
+  float W[ALS_TILE];
+};
+
 struct AlsLds {
-  AlsGramLds g;                 // staging (S/W)
+  union {
+    AlsGramLds g;               // staging (S/W), fp32 path
+    AlsGramLdsBf gb;            // staging, bf16 path
+  };
   float G[ALS_K * ALS_ROW];     // Gram -> Cholesky factor (in place)
 };
 
@@ -161,6 +183,100 @@ __device__ __forceinline__ void als_gram_range(const CFAlsArgs& a, E_ID b,
   }
 }
 
+// ---- bf16 Gram accumulation (v_mfma_f32_16x16x32_bf16) ----
+// A/B lane map of the 16x16x32 family: lane holds 8 contiguous-k
+// elements, i (or j) = lane&15, k = (lane>>4)*8 + e (verified on
+// hardware by tests/test_gpu_cf.py bf16-vs-f32 equivalence). One
+// fragment per 16-dim block serves as A of tile-row t AND B of
+// tile-col t, exactly like the f32 path's shared fragment.
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+
+__device__ __forceinline__ void als_gram_range_bf16(const CFAlsArgs& a,
+                                                    E_ID b, E_ID e,
+                                                    int lane,
+                                                    AlsGramLdsBf* lds,
+                                                    f32x4 acc[10],
+                                                    float* rhs) {
+  if (b >= e) return;
+  int rem = (int)(e - b < ALS_TILE ? e - b : (E_ID)ALS_TILE);
+  uint32_t mycol = 0;
+  float myw = 0.0f;
+  if (lane < rem) {
+    mycol = a.col[b + lane];
+    myw = (float)a.w[b + lane];
+  }
+  float tmp[ALS_TILE];
+  als_vec_loads(a, mycol, rem, lane, tmp);
+  for (E_ID t = b; t < e; t += ALS_TILE) {
+    // rhs += sum_r w_r * S[r][lane] — straight from registers (lane=dim
+    // holds S[r][lane] in tmp[r]); exact fp32
+#pragma unroll
+    for (int r = 0; r < ALS_TILE; r++)
+      *rhs += __shfl(myw, r, WAVE) * tmp[r];
+    // stage TRANSPOSED bf16: my row = dim `lane`, cols = edges (pairs
+    // packed into dword stores)
+    uint32_t* row32 = (uint32_t*)&lds->S[(uint32_t)lane * ALS_BPITCH];
+#pragma unroll
+    for (int w = 0; w < ALS_TILE / 2; w++) {
+      union {
+        __bf16 h[2];
+        uint32_t u;
+      } pk;
+      pk.h[0] = (__bf16)tmp[2 * w];
+      pk.h[1] = (__bf16)tmp[2 * w + 1];
+      row32[w] = pk.u;
+    }
+    int rem_cur = rem;
+    (void)rem_cur;
+    // prefetch next tile's col/weight, then its vectors (they land
+    // under the MFMA burst)
+    E_ID t2 = t + ALS_TILE;
+    bool more = t2 < e;
+    uint32_t ncol = 0;
+    float nw = 0.0f;
+    if (more) {
+      rem = (int)(e - t2 < ALS_TILE ? e - t2 : (E_ID)ALS_TILE);
+      if (lane < rem) {
+        ncol = a.col[t2 + lane];
+        nw = (float)a.w[t2 + lane];
+      }
+    }
+    als_lds_sync();
+    bf16x8 f[4];
+    int m = lane & 15, g = lane >> 4;
+#pragma unroll
+    for (int ti = 0; ti < 4; ti++)
+      f[ti] = *(const bf16x8*)&lds->S[(uint32_t)(16 * ti + m) * ALS_BPITCH +
+                                      g * 8];
+    if (more) {
+      myw = nw;
+      als_vec_loads(a, ncol, rem, lane, tmp);
+      mycol = ncol;
+    }
+    acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[0], f[0], acc[0], 0,
+                                                     0, 0);
+    acc[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[0], f[1], acc[1], 0,
+                                                     0, 0);
+    acc[2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[0], f[2], acc[2], 0,
+                                                     0, 0);
+    acc[3] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[0], f[3], acc[3], 0,
+                                                     0, 0);
+    acc[4] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[1], f[1], acc[4], 0,
+                                                     0, 0);
+    acc[5] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[1], f[2], acc[5], 0,
+                                                     0, 0);
+    acc[6] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[1], f[3], acc[6], 0,
+                                                     0, 0);
+    acc[7] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[2], f[2], acc[7], 0,
+                                                     0, 0);
+    acc[8] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[2], f[3], acc[8], 0,
+                                                     0, 0);
+    acc[9] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[3], f[3], acc[9], 0,
+                                                     0, 0);
+    als_lds_sync();  // tile consumed before the next stage overwrites
+  }
+}
+
 // Scatter MFMA accumulators into the LDS Gram (mirroring the lower half)
 // using the 16x16x4 C/D lane map: element (row=(lane>>4)*4+reg, col=lane&15).
 __device__ __forceinline__ void als_dump_gram(AlsLds* lds, f32x4 acc[10],
@@ -217,6 +333,7 @@ __device__ __forceinline__ float wave_spd_solve64(const float* G, float r,
 }
 
 // ---- one wave per vertex: gram + solve fused (deg < T2 bin lists) ----
+template <bool BF16>
 __global__ __launch_bounds__(ALS_TB) void cf_als_solve_kernel(
     uint32_t n, const V_ID* binlist, CFAlsArgs a) {
   __shared__ AlsLds lds;
@@ -228,7 +345,10 @@ __global__ __launch_bounds__(ALS_TB) void cf_als_solve_kernel(
 #pragma unroll
     for (int t = 0; t < 10; t++) acc[t] = {0, 0, 0, 0};
     float rhs = 0.0f;
-    als_gram_range(a, b, e, lane, &lds.g, acc, &rhs);
+    if (BF16)
+      als_gram_range_bf16(a, b, e, lane, &lds.gb, acc, &rhs);
+    else
+      als_gram_range(a, b, e, lane, &lds.g, acc, &rhs);
     als_dump_gram(&lds, acc, lane, a.K);
     wave_cholesky64(lds.G, lane);
     float d = wave_spd_solve64(lds.G, rhs, lane);
@@ -242,11 +362,16 @@ __global__ __launch_bounds__(ALS_TB) void cf_als_solve_kernel(
 // rhs_scratch: f32[nbig * 64]; both pre-zeroed by the engine each sweep.
 constexpr int ALS_CHUNK_TB = 128;  // 2 independent waves per workgroup
 
+template <bool BF16>
 __global__ __launch_bounds__(ALS_CHUNK_TB) void cf_als_gram_chunk_kernel(
     uint32_t n2, const uint2* bin2, V_ID chunk_edges, const int* hubidx,
     float* gram_scratch, float* rhs_scratch, CFAlsArgs a) {
-  __shared__ AlsGramLds lds2[ALS_CHUNK_TB / WAVE];
-  AlsGramLds& lds = lds2[threadIdx.x >> 6];
+  union StageLds {
+    AlsGramLds g;
+    AlsGramLdsBf gb;
+  };
+  __shared__ StageLds lds2[ALS_CHUNK_TB / WAVE];
+  StageLds& lds = lds2[threadIdx.x >> 6];
   int lane = threadIdx.x & (WAVE - 1);
   uint32_t wave = (blockIdx.x * (ALS_CHUNK_TB / WAVE)) + (threadIdx.x >> 6);
   uint32_t nwaves = gridDim.x * (ALS_CHUNK_TB / WAVE);
@@ -260,7 +385,10 @@ __global__ __launch_bounds__(ALS_CHUNK_TB) void cf_als_gram_chunk_kernel(
 #pragma unroll
     for (int t = 0; t < 10; t++) acc[t] = {0, 0, 0, 0};
     float rhs = 0.0f;
-    als_gram_range(a, b, e, lane, &lds, acc, &rhs);
+    if (BF16)
+      als_gram_range_bf16(a, b, e, lane, &lds.gb, acc, &rhs);
+    else
+      als_gram_range(a, b, e, lane, &lds.g, acc, &rhs);
     int idx = hubidx[v];
     float* Gg = gram_scratch + (uint64_t)idx * ALS_K * ALS_K;
     atomicAdd(&rhs_scratch[(uint64_t)idx * ALS_K + lane], rhs);
@@ -326,24 +454,45 @@ void lux_gpu_cf_als_iter(uint64_t stream, uint32_t n0, const V_ID* bin0,
                          int K) {
   hipStream_t s = (hipStream_t)stream;
   CFAlsArgs a{row_ptr, col, w, oldv, newv, row_left, K};
+  // bf16 Gram (16x16x32 MFMA, fp32 accumulate) is the default; exact
+  // fp32 (16x16x4) via LUX_ALS_F32=1
+  bool bf16 = !getenv("LUX_ALS_F32");
   if (nbig) {
     uint32_t gw = (n2 + 1) / 2;
-    hipLaunchKernelGGL(cf_als_gram_chunk_kernel,
-                       dim3(gw > MAX_GRID ? MAX_GRID : gw),
-                       dim3(ALS_CHUNK_TB), 0, s, n2, bin2, (V_ID)8192,
-                       hubidx, gram_scratch, rhs_scratch, a);
+    if (bf16)
+      hipLaunchKernelGGL(cf_als_gram_chunk_kernel<true>,
+                         dim3(gw > MAX_GRID ? MAX_GRID : gw),
+                         dim3(ALS_CHUNK_TB), 0, s, n2, bin2, (V_ID)8192,
+                         hubidx, gram_scratch, rhs_scratch, a);
+    else
+      hipLaunchKernelGGL(cf_als_gram_chunk_kernel<false>,
+                         dim3(gw > MAX_GRID ? MAX_GRID : gw),
+                         dim3(ALS_CHUNK_TB), 0, s, n2, bin2, (V_ID)8192,
+                         hubidx, gram_scratch, rhs_scratch, a);
     hipLaunchKernelGGL(cf_als_hub_solve_kernel,
                        dim3(nbig > MAX_GRID ? MAX_GRID : nbig), dim3(ALS_TB),
                        0, s, nbig, bin2v, gram_scratch, rhs_scratch, a);
   }
-  if (n1)
-    hipLaunchKernelGGL(cf_als_solve_kernel,
-                       dim3(n1 > MAX_GRID ? MAX_GRID : n1), dim3(ALS_TB), 0,
-                       s, n1, bin1, a);
-  if (n0)
-    hipLaunchKernelGGL(cf_als_solve_kernel,
-                       dim3(n0 > MAX_GRID ? MAX_GRID : n0), dim3(ALS_TB), 0,
-                       s, n0, bin0, a);
+  if (n1) {
+    if (bf16)
+      hipLaunchKernelGGL(cf_als_solve_kernel<true>,
+                         dim3(n1 > MAX_GRID ? MAX_GRID : n1), dim3(ALS_TB),
+                         0, s, n1, bin1, a);
+    else
+      hipLaunchKernelGGL(cf_als_solve_kernel<false>,
+                         dim3(n1 > MAX_GRID ? MAX_GRID : n1), dim3(ALS_TB),
+                         0, s, n1, bin1, a);
+  }
+  if (n0) {
+    if (bf16)
+      hipLaunchKernelGGL(cf_als_solve_kernel<true>,
+                         dim3(n0 > MAX_GRID ? MAX_GRID : n0), dim3(ALS_TB),
+                         0, s, n0, bin0, a);
+    else
+      hipLaunchKernelGGL(cf_als_solve_kernel<false>,
+                         dim3(n0 > MAX_GRID ? MAX_GRID : n0), dim3(ALS_TB),
+                         0, s, n0, bin0, a);
+  }
   LUX_POST_LAUNCH(stream);
 }
 

@@ -79,9 +79,11 @@ def _rand_init(nv, K, seed):
 
 
 @pytest.mark.parametrize("K", [20, 64])
-def test_cf_als_vs_numpy(K):
+def test_cf_als_vs_numpy(K, monkeypatch):
     """MFMA ALS sweep (cf_als.hip) vs the float64 numpy normal-equations
-    reference from a common random init."""
+    reference from a common random init (exact-fp32 Gram path; the bf16
+    default is covered by test_cf_als_bf16_matches_f32 + the loss tests)."""
+    monkeypatch.setenv("LUX_ALS_F32", "1")
     import torch
     from lux_amd.cf_engine import CFALSEngine
     nu, ni, ne = 400, 100, 15000
@@ -105,8 +107,9 @@ def test_cf_als_vs_numpy(K):
     np.testing.assert_allclose(got / scale, want / scale, rtol=0, atol=2e-3)
 
 
-def test_cf_als_hub_path():
+def test_cf_als_hub_path(monkeypatch):
     """Extreme-degree items exercise the chunked Gram + hub-solve path."""
+    monkeypatch.setenv("LUX_ALS_F32", "1")
     import torch
     from lux_amd.cf_engine import CFALSEngine
     nu, ni, ne = 2000, 4, 60000
@@ -125,7 +128,8 @@ def test_cf_als_hub_path():
     np.testing.assert_allclose(got / scale, want / scale, rtol=0, atol=2e-3)
 
 
-def test_cf_als_normal_equation_residual():
+def test_cf_als_normal_equation_residual(monkeypatch):
+    monkeypatch.setenv("LUX_ALS_F32", "1")
     """From the reference's constant init (worst conditioning: sweep-1 Gram
     is rank-1 + lambda I), the fp32 solve must still satisfy its own normal
     equations to fp32 backward error — the condition-independent check."""
@@ -212,3 +216,31 @@ def test_cf_multipart_single_process(cls_name):
         scale = np.maximum(np.abs(want).max(axis=1, keepdims=True), 1.0)
         np.testing.assert_allclose(got / scale, want / scale, rtol=0,
                                    atol=2e-3)
+
+
+def test_cf_als_bf16_matches_f32(monkeypatch):
+    """The default bf16-Gram ALS (v_mfma_f32_16x16x32_bf16) vs the exact
+    fp32 path: same sweep within bf16 rounding. Also the hardware check of
+    the assumed 16x16x32 A/B fragment lane map — a wrong map produces a
+    wrong Gram, not a small error."""
+    import torch
+    from lux_amd.cf_engine import CFALSEngine
+    nu, ni, ne, K = 2000, 200, 60000, 64
+    init = _rand_init(nu + ni, K, seed=44)
+    outs = {}
+    for mode in ("bf16", "f32"):
+        if mode == "f32":
+            monkeypatch.setenv("LUX_ALS_F32", "1")
+        else:
+            monkeypatch.delenv("LUX_ALS_F32", raising=False)
+        full = DeviceCSC.bipartite(nu, ni, ne, seed=17)
+        part = GraphPart(full, 1, 0)
+        part.build_bins()
+        assert part.nbig > 0  # exercise the hub chunk path too
+        eng = CFALSEngine(part, K=K)
+        eng.old.copy_(torch.from_numpy(init.ravel()))
+        eng.step()
+        outs[mode] = eng.vectors().cpu().numpy().copy()
+    scale = np.maximum(np.abs(outs["f32"]).max(axis=1, keepdims=True), 1.0)
+    np.testing.assert_allclose(outs["bf16"] / scale, outs["f32"] / scale,
+                               rtol=0, atol=3e-2)
