@@ -237,7 +237,10 @@ class PushEngine:
             # beyond ~ne/8 edges. bitmap-BFS push touches vp/8 bytes of
             # visited bits instead of the label array, so it stays cheaper
             # up to much larger frontiers (~ne/2).
-            thresh = p.ne // 2 if self.visited is not None else p.ne // 8
+            import os
+            div = int(os.environ.get("LUX_PUSH_EVOL_DIV", "0") or 0)
+            thresh = p.ne // div if div else (
+                p.ne // 2 if self.visited is not None else p.ne // 8)
             if evol > thresh:
                 pull_fallback = True
             elif (evol * p.ep // max(p.ne, 1)) // 16 > self.capacity:
