@@ -186,9 +186,12 @@ def _sync_parts(engines, K):
 
 
 @pytest.mark.parametrize("cls_name", ["CFEngine", "CFALSEngine"])
-def test_cf_multipart_single_process(cls_name):
+def test_cf_multipart_single_process(cls_name, monkeypatch):
     """2 partitions stepped in one process match the whole-graph reference
-    (the N-GPU equivalence shape from SURVEY.md §4(d) for the CF family)."""
+    (the N-GPU equivalence shape from SURVEY.md §4(d) for the CF family).
+    Exact-fp32 ALS path: the comparison is against an f64 reference and
+    compounds over sweeps."""
+    monkeypatch.setenv("LUX_ALS_F32", "1")
     import lux_amd.cf_engine as cfe
     cls = getattr(cfe, cls_name)
     nu, ni, ne, K = 1000, 256, 40000, 32
