@@ -190,12 +190,12 @@ def pull_finish_pr(stream, vp, newv, deg, row_left, init_rank):
 
 def cf_als_iter(stream, n0, bin0, n1, bin1, n2, bin2, nbig, bin2v, hubidx,
                 gram_scratch, rhs_scratch, row_ptr, col, w, oldv, newv,
-                row_left, K):
+                row_left, K, oldv_bf=None):
     lib().lux_gpu_cf_als_iter(
         _u64(stream), _u32(n0), dp(bin0), _u32(n1), dp(bin1), _u32(n2),
         dp(bin2), _u32(nbig), dp(bin2v), dp(hubidx), dp(gram_scratch),
-        dp(rhs_scratch), dp(row_ptr), dp(col), dp(w), dp(oldv), dp(newv),
-        _u32(row_left), ctypes.c_int(K))
+        dp(rhs_scratch), dp(row_ptr), dp(col), dp(w), dp(oldv), dp(oldv_bf),
+        dp(newv), _u32(row_left), ctypes.c_int(K))
 
 
 def frontier_expand(stream, old_dense, in_row_left, in_count, old_seg,

@@ -78,6 +78,11 @@ class CFALSEngine:
                                      device=device)
         else:
             self.hubidx = self.gram = self.rhs_h = None
+        # bf16 gather replica for the bf16-Gram path: halves per-edge
+        # gather bytes and makes the hot src-vector table LLC-resident
+        import os
+        self.old_bf = None if os.environ.get("LUX_ALS_F32") else \
+            torch.empty(part.nv * K, dtype=torch.bfloat16, device=device)
 
     def step(self):
         p = self.part
@@ -87,10 +92,12 @@ class CFALSEngine:
         if p.nbig:
             self.gram.zero_()
             self.rhs_h.zero_()
+        if self.old_bf is not None:
+            self.old_bf.copy_(self.old)  # RNE cast, one fused torch kernel
         ng.cf_als_iter(_stream(), p.n0, p.bin0, p.n1, p.bin1, p.n2, p.bin2,
                        p.nbig, p.bin2v, self.hubidx, self.gram, self.rhs_h,
                        p.row_ptr, p.col, p.weight, self.old, self.new_part,
-                       p.row_left, self.K)
+                       p.row_left, self.K, oldv_bf=self.old_bf)
         dx.all_gather_slices(self.old, self.new_part, self.verts_elems,
                              self.left_elems, my_index=p.p)
 

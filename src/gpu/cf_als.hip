@@ -50,6 +50,9 @@ struct CFAlsArgs {
   const V_ID* col;       // u32[ep]
   const WeightType* w;   // i32[ep]
   const float* oldv;     // f32[nv*K]
+  const __bf16* oldv_bf; // bf16 gather replica (nullable): halves the
+                         // per-edge gather bytes AND the LLC footprint of
+                         // the hot src-vector table on the bf16 Gram path
   float* newv;           // f32[vp*K] (pre-seeded with old slice)
   V_ID row_left;
   int K;                 // <= 64
@@ -191,6 +194,19 @@ __device__ __forceinline__ void als_gram_range(const CFAlsArgs& a, E_ID b,
 // tile-col t, exactly like the f32 path's shared fragment.
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 
+__device__ __forceinline__ void als_vec_loads_bf(const CFAlsArgs& a,
+                                                 uint32_t mycol, int rem,
+                                                 int lane,
+                                                 float tmp[ALS_TILE]) {
+#pragma unroll
+  for (int r = 0; r < ALS_TILE; r++) {
+    uint32_t src = __shfl(mycol, r, WAVE);
+    tmp[r] = (r < rem && lane < a.K)
+                 ? (float)a.oldv_bf[(uint64_t)src * a.K + lane]
+                 : 0.0f;
+  }
+}
+
 __device__ __forceinline__ void als_gram_range_bf16(const CFAlsArgs& a,
                                                     E_ID b, E_ID e,
                                                     int lane,
@@ -206,7 +222,10 @@ __device__ __forceinline__ void als_gram_range_bf16(const CFAlsArgs& a,
     myw = (float)a.w[b + lane];
   }
   float tmp[ALS_TILE];
-  als_vec_loads(a, mycol, rem, lane, tmp);
+  if (a.oldv_bf)
+    als_vec_loads_bf(a, mycol, rem, lane, tmp);
+  else
+    als_vec_loads(a, mycol, rem, lane, tmp);
   for (E_ID t = b; t < e; t += ALS_TILE) {
     // rhs += sum_r w_r * S[r][lane] — straight from registers (lane=dim
     // holds S[r][lane] in tmp[r]); exact fp32
@@ -250,7 +269,10 @@ __device__ __forceinline__ void als_gram_range_bf16(const CFAlsArgs& a,
                                       g * 8];
     if (more) {
       myw = nw;
-      als_vec_loads(a, ncol, rem, lane, tmp);
+      if (a.oldv_bf)
+        als_vec_loads_bf(a, ncol, rem, lane, tmp);
+      else
+        als_vec_loads(a, ncol, rem, lane, tmp);
       mycol = ncol;
     }
     acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[0], f[0], acc[0], 0,
@@ -450,10 +472,12 @@ void lux_gpu_cf_als_iter(uint64_t stream, uint32_t n0, const V_ID* bin0,
                          const int* hubidx, float* gram_scratch,
                          float* rhs_scratch, const E_ID* row_ptr,
                          const V_ID* col, const WeightType* w,
-                         const float* oldv, float* newv, V_ID row_left,
-                         int K) {
+                         const float* oldv,
+                         const uint16_t* oldv_bf /*nullable*/, float* newv,
+                         V_ID row_left, int K) {
   hipStream_t s = (hipStream_t)stream;
-  CFAlsArgs a{row_ptr, col, w, oldv, newv, row_left, K};
+  CFAlsArgs a{row_ptr, col, w, oldv, (const __bf16*)oldv_bf, newv, row_left,
+              K};
   // bf16 Gram (16x16x32 MFMA, fp32 accumulate) is the default; exact
   // fp32 (16x16x4) via LUX_ALS_F32=1
   bool bf16 = !getenv("LUX_ALS_F32");

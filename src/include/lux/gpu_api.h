@@ -164,6 +164,7 @@ void lux_gpu_cf_als_iter(uint64_t stream, uint32_t n0, const lux::V_ID* bin0,
                          float* gram_scratch, float* rhs_scratch,
                          const lux::E_ID* row_ptr, const lux::V_ID* col,
                          const lux::WeightType* w, const float* oldv,
-                         float* newv, lux::V_ID row_left, int K);
+                         const uint16_t* oldv_bf /*nullable*/, float* newv,
+                         lux::V_ID row_left, int K);
 
 }  // extern "C"

@@ -412,7 +412,7 @@ void SingleGpuCF::iterate(int iters) {
       lux_gpu_cf_als_iter((uint64_t)s_, bins_.n0, bins_.bin0, bins_.n1,
                           bins_.bin1, bins_.n2, bins_.bin2, bins_.nbig,
                           bins_.bin2v, hubidx_, gram_, rhs_, row_ptr_,
-                          g_.src, g_.weight, old_, new_, 0, K_);
+                          g_.src, g_.weight, old_, nullptr, new_, 0, K_);
     } else {
       // SGD contract (cf.hip): output pre-seeded old*(1-GAMMA*LAMBDA),
       // sweeps add GAMMA*acc
