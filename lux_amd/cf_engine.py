@@ -78,11 +78,16 @@ class CFALSEngine:
                                      device=device)
         else:
             self.hubidx = self.gram = self.rhs_h = None
-        # bf16 gather replica for the bf16-Gram path: halves per-edge
-        # gather bytes and makes the hot src-vector table LLC-resident
+        # bf16 gather replica (LUX_ALS_BF_GATHER=1, off by default):
+        # halves per-edge gather bytes, but MEASURED 2.5x SLOWER on the
+        # NetFlix shape (10.6 -> 38 ms/sweep; sub-dword gathers are far
+        # below dword gather throughput regardless of load opcode —
+        # global_load_ushort and d16 variants both) — kept as an A/B knob
         import os
-        self.old_bf = None if os.environ.get("LUX_ALS_F32") else \
-            torch.empty(part.nv * K, dtype=torch.bfloat16, device=device)
+        self.old_bf = torch.empty(part.nv * K, dtype=torch.bfloat16,
+                                  device=device) \
+            if os.environ.get("LUX_ALS_BF_GATHER") == "1" \
+            and not os.environ.get("LUX_ALS_F32") else None
 
     def step(self):
         p = self.part

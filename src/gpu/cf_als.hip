@@ -198,13 +198,25 @@ __device__ __forceinline__ void als_vec_loads_bf(const CFAlsArgs& a,
                                                  uint32_t mycol, int rem,
                                                  int lane,
                                                  float tmp[ALS_TILE]) {
+  // raw u16 loads + integer shift (bf16 = high half of f32): typed __bf16
+  // scalar loads scalarise through conversion libcalls and serialise the
+  // 32-outstanding-load pipeline (measured 4x slower)
+  const uint16_t* src16 = (const uint16_t*)a.oldv_bf;
+  uint32_t raw[ALS_TILE];  // u32 destinations: zero-extending ushort
+                           // loads write FULL registers (u16 destinations
+                           // compile to d16 merge loads whose
+                           // read-modify-write false deps serialise the
+                           // 32-outstanding-load pipeline)
 #pragma unroll
   for (int r = 0; r < ALS_TILE; r++) {
     uint32_t src = __shfl(mycol, r, WAVE);
-    tmp[r] = (r < rem && lane < a.K)
-                 ? (float)a.oldv_bf[(uint64_t)src * a.K + lane]
-                 : 0.0f;
+    raw[r] = (r < rem && lane < a.K)
+                 ? (uint32_t)src16[(uint64_t)src * a.K + lane]
+                 : 0u;
   }
+#pragma unroll
+  for (int r = 0; r < ALS_TILE; r++)
+    tmp[r] = __uint_as_float(raw[r] << 16);
 }
 
 __device__ __forceinline__ void als_gram_range_bf16(const CFAlsArgs& a,
