@@ -47,6 +47,43 @@ def _align16(x):
     return (x + 15) & ~15
 
 
+def exchange_frontier_payloads(meta, verts_all, fq_all, new_seg, seg_off,
+                               annex_all, new_annex, annex_off, labels,
+                               labels_part, row_left_all, me):
+    """Post the per-iteration payload exchange from a fresh meta table:
+    frontier segments ship only their USED bytes, sparse queues ship a
+    label annex, and ONLY dense ranks publish their label slice (sparse
+    ranks' fresh labels ride the annex). One batched p2p round.
+
+    Returns (work_handle, lab_n) — lab_n[q] > 0 iff rank q published its
+    label slice. Factored out of PushEngine.step so the distributed
+    byte-accounting is CPU-testable under gloo (tests/test_dist_cpu.py).
+    """
+    nparts = len(verts_all)
+    used = [0] * nparts
+    annex_n = [0] * nparts
+    lab_n = [0] * nparts
+    for q in range(nparts):
+        if verts_all[q] == 0:
+            continue
+        if int(meta[q, 0]) == DENSE_BITMAP:
+            used[q] = 8 + (verts_all[q] + 7) // 8
+            lab_n[q] = verts_all[q]  # dense ranks publish labels
+        else:
+            used[q] = 8 + 4 * int(meta[q, 1])
+            annex_n[q] = int(meta[q, 1])
+    parts = [(fq_all,
+              new_seg.narrow(0, 0, used[me]) if used[me] else None,
+              used, seg_off),
+             (annex_all,
+              new_annex.narrow(0, 0, annex_n[me]) if annex_n[me] else None,
+              annex_n, annex_off),
+             (labels,
+              labels_part if lab_n[me] else None,
+              lab_n, row_left_all)]
+    return dx.exchange_multi_async(parts, my_index=me), lab_n
+
+
 class PushEngine:
     MODE_MIN = 1  # SSSP
     MODE_MAX = 2  # CC
@@ -317,31 +354,14 @@ class PushEngine:
         self.meta_host = mh
         ntypes, ncounts = mh[:, 0], mh[:, 1]
 
-        # payload sizes from the fresh meta: only USED bytes travel
-        used = [0] * nparts
-        annex_n = [0] * nparts
-        lab_n = [0] * nparts
-        for q in range(nparts):
-            if p.verts_all[q] == 0:
-                continue
-            if ntypes[q] == DENSE_BITMAP:
-                used[q] = 8 + (p.verts_all[q] + 7) // 8
-                lab_n[q] = p.verts_all[q]  # dense ranks publish labels
-            else:
-                used[q] = 8 + 4 * int(ncounts[q])
-                annex_n[q] = int(ncounts[q])
+        # payload exchange from the fresh meta: only USED bytes travel
         me = p.p
-        parts = [(self.fq_all,
-                  self.new_seg.narrow(0, 0, used[me]) if used[me] else None,
-                  used, [int(o) for o in self.seg_off[:-1]]),
-                 (self.fq_annex_all,
-                  self.new_annex.narrow(0, 0, annex_n[me])
-                  if annex_n[me] else None,
-                  annex_n, [int(o) for o in self.annex_off[:-1]]),
-                 (self.labels,
-                  self.labels_part if lab_n[me] else None,
-                  lab_n, p.row_left_all)]
-        dx.exchange_multi_async(parts, my_index=me).wait()
+        h, lab_n = exchange_frontier_payloads(
+            mh, p.verts_all, self.fq_all, self.new_seg,
+            [int(o) for o in self.seg_off[:-1]], self.fq_annex_all,
+            self.new_annex, [int(o) for o in self.annex_off[:-1]],
+            self.labels, self.labels_part, p.row_left_all, me)
+        h.wait()
         # slice q is fresh iff q published it now, or it was fresh before
         # and q changed nothing this iteration
         if self.labels_current:

@@ -371,3 +371,76 @@ def test_dist_byte_segments(world=2):
     got = outq.get()
     assert got[0] == 7 and got[48] == 7
     assert (got[1:48] == 10).all() and (got[49:] == 11).all()
+
+
+def _dist_push_payload_worker(rank, world, port, outq):
+    """The push engine's per-iteration payload exchange
+    (push_engine.exchange_frontier_payloads) with synthetic meta/segments:
+    rank 0 dense (publishes bitmap bytes + label slice), rank 1 sparse
+    (publishes used queue bytes + label annex, label slice SKIPPED)."""
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        import torch
+
+        from lux_amd.push_engine import exchange_frontier_payloads
+        from lux_amd.types import DENSE_BITMAP, SPARSE_QUEUE
+        verts_all = [20, 12]
+        row_left_all = [0, 20]
+        seg_bytes = [64, 64]
+        seg_off = [0, 64]
+        annex_off = [0, 16]
+        meta = np.zeros((2, 8), np.uint32)
+        meta[0, 0], meta[0, 1] = DENSE_BITMAP, 5
+        meta[1, 0], meta[1, 1] = SPARSE_QUEUE, 3
+        fq_all = torch.zeros(128, dtype=torch.uint8)
+        annex_all = torch.zeros(32, dtype=torch.int32)
+        labels = torch.full((32,), -1, dtype=torch.int32)
+        labels_part = torch.full((verts_all[rank],), 70 + rank,
+                                 dtype=torch.int32)
+        if rank == 0:  # dense seg: 8B hdr + ceil(20/8)=3 bitmap bytes
+            new_seg = torch.full((64,), 0xA0, dtype=torch.uint8)
+            new_seg[:8].view(torch.int32)[0] = DENSE_BITMAP
+            new_seg[:8].view(torch.int32)[1] = 5
+            new_annex = torch.zeros(16, dtype=torch.int32)
+        else:  # sparse: 8B hdr + 3*4 queue ids; annex carries 3 labels
+            new_seg = torch.full((64,), 0xB1, dtype=torch.uint8)
+            new_seg[:8].view(torch.int32)[0] = SPARSE_QUEUE
+            new_seg[:8].view(torch.int32)[1] = 3
+            new_annex = torch.arange(100, 116, dtype=torch.int32)
+        h, lab_n = exchange_frontier_payloads(
+            meta, verts_all, fq_all, new_seg, seg_off, annex_all, new_annex,
+            annex_off, labels, labels_part, row_left_all, rank)
+        h.wait()
+        assert lab_n == [20, 0]
+        if rank == 0:
+            outq.put((fq_all.numpy().copy(), annex_all.numpy().copy(),
+                      labels.numpy().copy()))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_dist_push_payload_exchange(world=2):
+    from lux_amd.types import DENSE_BITMAP, SPARSE_QUEUE
+    ctx = mp.get_context("spawn")
+    outq = ctx.SimpleQueue()
+    mp.spawn(_dist_push_payload_worker, args=(world, _find_port(), outq),
+             nprocs=world, join=True)
+    fq, annex, labels = outq.get()
+    # rank 0's dense seg: used = 8 + 3 = 11 bytes of 0xA0 payload
+    assert fq[:4].view(np.uint32)[0] == DENSE_BITMAP
+    assert (fq[8:11] == 0xA0).all()
+    assert (fq[11:64] == 0).all()  # beyond used bytes: never shipped
+    # rank 1's sparse seg at offset 64: used = 8 + 12
+    assert fq[64:68].view(np.uint32)[0] == SPARSE_QUEUE
+    assert (fq[72:84] == 0xB1).all()
+    assert (fq[84:] == 0).all()
+    # annex: only rank 1's 3 labels (at its annex offset 16)
+    assert (annex[:16] == 0).all()
+    assert list(annex[16:19]) == [100, 101, 102]
+    assert (annex[19:] == 0).all()
+    # labels: rank 0 (dense) published its slice; rank 1 (sparse) did not
+    assert (labels[:20] == 70).all()
+    assert (labels[20:] == -1).all()
