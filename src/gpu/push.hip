@@ -176,6 +176,53 @@ __global__ void push_chunk_scatter_kernel(
     if (e > b + PUSH_CHUNK) e = b + PUSH_CHUNK;
     uint32_t nstripes =
         (uint32_t)((e - b + blockDim.x - 1) / blockDim.x);
+    // dense-output path: no block-wide scan per stripe, so 4 stripes are
+    // processed per pass — 4 col loads then 4 dependent visited/label
+    // line reads in flight (same MLP lever as pull.hip gather_range; the
+    // hub-explosion iteration is LLC-random-line-bound)
+    if (NEW_DENSE) {
+      uint32_t s = 0;
+      for (; s + 4 <= nstripes; s += 4) {
+        E_ID k0 = b + (E_ID)s * blockDim.x + threadIdx.x;
+        V_ID v0 = 0, v1 = 0, v2 = 0, v3 = 0;
+        bool p0 = k0 < e, p1 = k0 + blockDim.x < e,
+             p2 = k0 + 2 * blockDim.x < e, p3 = k0 + 3 * blockDim.x < e;
+        if (p0) v0 = push_col[k0];
+        if (p1) v1 = push_col[k0 + blockDim.x];
+        if (p2) v2 = push_col[k0 + 2 * blockDim.x];
+        if (p3) v3 = push_col[k0 + 3 * blockDim.x];
+#define LUX_PD_EDGE(P_, V_)                                                 \
+        if (P_) {                                                           \
+          V_ID lv = (V_) - my_row_left;                                     \
+          if (BFS_BITS) {                                                   \
+            uint32_t word = visited_bits[lv >> 5];                          \
+            uint32_t bit = 1u << (lv & 31);                                 \
+            if (!(word & bit)) {                                            \
+              uint32_t old = atomicOr(&visited_bits[lv >> 5], bit);         \
+              if (!(old & bit)) new_labels[lv] = new_lab;                   \
+            }                                                               \
+          } else {                                                          \
+            uint32_t* slot = &new_labels[lv];                               \
+            uint32_t cur = __hip_atomic_load(slot, __ATOMIC_RELAXED,        \
+                                             __HIP_MEMORY_SCOPE_AGENT);     \
+            if (OP::better(new_lab, cur)) OP::atom(slot, new_lab);          \
+          }                                                                 \
+        }
+        LUX_PD_EDGE(p0, v0)
+        LUX_PD_EDGE(p1, v1)
+        LUX_PD_EDGE(p2, v2)
+        LUX_PD_EDGE(p3, v3)
+      }
+      for (; s < nstripes; s++) {
+        E_ID k = b + (E_ID)s * blockDim.x + threadIdx.x;
+        if (k < e) {
+          V_ID v = push_col[k];
+          LUX_PD_EDGE(true, v)
+        }
+      }
+#undef LUX_PD_EDGE
+      continue;
+    }
     for (uint32_t s = 0; s < nstripes; s++) {
       E_ID k = b + (E_ID)s * blockDim.x + threadIdx.x;
       uint32_t flag = 0;
