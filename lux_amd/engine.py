@@ -475,6 +475,42 @@ class GraphPart:
         del ends, cursor
 
 
+def _maybe_halo(part):
+    """Build the in_vtxs halo exchange when it beats the full slice
+    all-gather (halo.py; reference parity: pagerank_gpu.cu:229-241).
+    RMAT-shaped partitions have halo ~ nv and stay on the full gather;
+    locality-structured graphs (meshes, roads) ship only their halo."""
+    if dx.world_size() <= 1 or part.vp == 0:
+        return None
+    from .halo import HaloExchange
+    h = HaloExchange(part.nv, part.row_left_all, part.verts_all, part.p,
+                     part.col)
+    return h if h.worth_it() else None
+
+
+def _publish_async(engine, my_slice):
+    p = engine.part
+    if engine.halo is not None:
+        return engine.halo.publish_async(engine.old, my_slice)
+    return dx.all_gather_slices_async(engine.old, my_slice, p.verts_all,
+                                      p.row_left_all, my_index=p.p)
+
+
+def _finalize_full(engine):
+    """Collective: one full slice all-gather so the replicated array is
+    current EVERYWHERE (halo iterations keep it current only at read
+    positions). Every rank must call (apps do, after the timed loop)."""
+    if engine._handle is not None:
+        engine._handle.wait()
+        engine._handle = None
+    if engine.halo is None:
+        return
+    p = engine.part
+    dx.all_gather_slices(engine.old,
+                         engine.old.narrow(0, p.row_left, p.vp),
+                         p.verts_all, p.row_left_all, my_index=p.p)
+
+
 class PagerankEngine:
     """Distributed pull PageRank. State: old ranks replicated f32[nv]
     (pre-divided by out-degree, the reference's stored form), new slice
@@ -508,6 +544,7 @@ class PagerankEngine:
         self.cur_part = self.old.narrow(0, part.row_left,
                                         part.vp).clone()
         self._handle = None
+        self.halo = _maybe_halo(part)
         # single-GPU iteration is a fixed ~50-launch sequence (seed + 16
         # blocked sweeps x 3 bins + epilogue + publish): captured into a
         # hipGraph on the 2nd step and replayed (one graph launch per
@@ -532,8 +569,7 @@ class PagerankEngine:
             if dx.world_size() > 1:  # degenerate rank (vp==0 / unblocked)
                 run_pull(p, ng.PULL_PR, self.old, self.new_part, self.deg,
                          self.init_rank)
-                dx.all_gather_slices(self.old, self.new_part, p.verts_all,
-                                     p.row_left_all, my_index=p.p)
+                _publish_async(self, self.new_part).wait()
                 return
             self._steps += 1
             if self._graph is not None:
@@ -569,12 +605,17 @@ class PagerankEngine:
         ng.pull_finish_pr(_stream(), p.vp, self.new_part, self.deg,
                           p.row_left, self.init_rank)
         self.cur_part, self.new_part = self.new_part, self.cur_part
-        self._handle = dx.all_gather_slices_async(
-            self.old, self.cur_part, p.verts_all, p.row_left_all,
-            my_index=p.p)
+        self._handle = _publish_async(self, self.cur_part)
+
+    def finalize(self):
+        """Collective (all ranks): complete the replicated rank vector
+        after halo-exchange iterations. No-op without halo."""
+        _finalize_full(self)
 
     def ranks(self):
-        """Replicated stored ranks (pr/out_degree) as a torch tensor."""
+        """Replicated stored ranks (pr/out_degree) as a torch tensor.
+        With the halo exchange active, call finalize() (collectively)
+        first if you need non-halo positions of peers' slices."""
         if self._handle is not None:
             self._handle.wait()
             self._handle = None
@@ -595,6 +636,7 @@ class LabelPullEngine:
         self.cur_part = init_labels.narrow(0, part.row_left,
                                            part.vp).clone()
         self._handle = None
+        self.halo = _maybe_halo(part)
 
     def _pipelined(self):
         p = self.part
@@ -606,8 +648,7 @@ class LabelPullEngine:
             run_pull(p, self.mode, self.old, self.new_part, None, 0.0)
             changed = (self.new_part
                        != self.old.narrow(0, p.row_left, p.vp)).sum()
-            dx.all_gather_slices(self.old, self.new_part, p.verts_all,
-                                 p.row_left_all, my_index=p.p)
+            _publish_async(self, self.new_part).wait()
             return changed
         # pipelined (see PagerankEngine.step): local block vs cur_part
         # while the previous publish is in flight
@@ -622,15 +663,13 @@ class LabelPullEngine:
                         subset="remote")
         changed = (self.new_part != self.cur_part).sum()
         self.cur_part, self.new_part = self.new_part, self.cur_part
-        self._handle = dx.all_gather_slices_async(
-            self.old, self.cur_part, p.verts_all, p.row_left_all,
-            my_index=p.p)
+        self._handle = _publish_async(self, self.cur_part)
         return changed
 
     def labels(self):
-        if self._handle is not None:
-            self._handle.wait()
-            self._handle = None
+        """Replicated labels; completes the halo-skipped positions with
+        one full all-gather (collective when halo is active)."""
+        _finalize_full(self)
         return self.old
 
     def run_to_fixpoint(self, max_iters=None):

@@ -444,3 +444,105 @@ def test_dist_push_payload_exchange(world=2):
     # labels: rank 0 (dense) published its slice; rank 1 (sparse) did not
     assert (labels[:20] == 70).all()
     assert (labels[20:] == -1).all()
+
+
+def _dist_halo_pagerank_worker(rank, world, port, iters, outq):
+    """PageRank over a ring-of-cliques graph exchanged through the
+    in_vtxs halo machinery (halo.py, LUX_HALO=1): only boundary vertices
+    travel. Verifies the packed/scattered values reproduce the full
+    all-gather result exactly."""
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["LUX_HALO"] = "1"
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from lux_amd import _native as nat
+        from lux_amd import dist as dx
+        from lux_amd.graph import Graph
+        from lux_amd.halo import HaloExchange
+        # ring graph + a few local chords: halo per partition ~ 1 vertex
+        nv = 4096
+        src = list(range(nv)) + [v for v in range(0, nv, 4)]
+        dst = [(v + 1) % nv for v in range(nv)] + \
+            [(v + 2) % nv for v in range(0, nv, 4)]
+        g = Graph.from_edges(nv, src, dst)
+        part = g.partition(world)
+        rl, rr, cl, ce, srcs, _w = g.slice(part, rank)
+        vp = part.verts(rank)
+        verts = [part.verts(p) for p in range(world)]
+        lefts = [int(part.row_left[p]) for p in range(world)]
+        col_t = torch.from_numpy(srcs.astype(np.int32))
+        halo = HaloExchange(g.nv, lefts, verts, rank, col_t)
+        assert halo.worth_it()
+        assert halo.total_halo() < nv // world  # boundary-sized, not nv
+        deg_t = torch.zeros(g.nv, dtype=torch.int32)
+        deg_t += torch.from_numpy(
+            np.bincount(srcs, minlength=g.nv).astype(np.int32))
+        dx.all_reduce_sum_(deg_t)
+        deg = deg_t.numpy().view(np.uint32)
+        old_t = torch.from_numpy(nat.pagerank_init(g.nv, deg))
+        new_t = torch.empty(vp, dtype=torch.float32)
+        for _ in range(iters):
+            nat.pagerank_iter_part(g.nv, rl, rr, cl, ce, srcs, deg,
+                                   old_t.numpy(), new_t.numpy())
+            halo.publish(old_t, new_t)
+        # finalize: one full gather so every position is current
+        dx.all_gather_slices(old_t, old_t.narrow(0, lefts[rank], vp),
+                             verts, lefts)
+        if rank == 0:
+            outq.put(old_t.numpy().copy())
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_dist_halo_pagerank(world):
+    from lux_amd import cpu_ref
+    from lux_amd.graph import Graph
+    iters = 5
+    ctx = mp.get_context("spawn")
+    outq = ctx.SimpleQueue()
+    mp.spawn(_dist_halo_pagerank_worker,
+             args=(world, _find_port(), iters, outq), nprocs=world,
+             join=True)
+    got = outq.get()
+    nv = 4096
+    src = list(range(nv)) + [v for v in range(0, nv, 4)]
+    dst = [(v + 1) % nv for v in range(nv)] + \
+        [(v + 2) % nv for v in range(0, nv, 4)]
+    g = Graph.from_edges(nv, src, dst)
+    want = cpu_ref.pagerank(g, iters)
+    np.testing.assert_allclose(got, want, rtol=1e-6)
+
+
+def _dist_halo_autodisable_worker(rank, world, port, outq):
+    """On an RMAT graph the halo covers most of nv — worth_it() must say
+    no (the engines then keep the plain slice all-gather)."""
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ.pop("LUX_HALO", None)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from lux_amd.graph import Graph
+        from lux_amd.halo import HaloExchange
+        g = Graph.rmat(10, 30000, seed=3)
+        part = g.partition(world)
+        _rl, _rr, _cl, _ce, srcs, _w = g.slice(part, rank)
+        verts = [part.verts(p) for p in range(world)]
+        lefts = [int(part.row_left[p]) for p in range(world)]
+        halo = HaloExchange(g.nv, lefts, verts, rank,
+                            torch.from_numpy(srcs.astype(np.int32)))
+        if rank == 0:
+            outq.put(bool(halo.worth_it()))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_dist_halo_autodisable_on_rmat(world=2):
+    ctx = mp.get_context("spawn")
+    outq = ctx.SimpleQueue()
+    mp.spawn(_dist_halo_autodisable_worker, args=(world, _find_port(), outq),
+             nprocs=world, join=True)
+    assert outq.get() is False
