@@ -72,42 +72,56 @@ class HaloExchange:
                                          device=device)] * ws
             self.send_counts = [0] * ws
             self.recv_counts = recv_counts
+            self._cnt = None
             self._bufs_ready = False
             return
 
-        # exchange counts (all-gather of the ws x ws count matrix row;
-        # tensors live on `device` so the collective works on NCCL too)
+        # exchange counts only (the cheap part: a ws*ws int matrix;
+        # tensors live on `device` so the collective works on NCCL too).
+        # The index lists themselves are exchanged LAZILY — only after
+        # worth_it() said yes on every rank (the decision is computed
+        # from the full matrix, so it is globally identical; per-rank
+        # decisions could deadlock the list exchange).
         cnt = torch.zeros(ws * ws, dtype=torch.int64, device=device)
         mine = torch.tensor(recv_counts, dtype=torch.int64, device=device)
         dx.all_gather_slices(cnt, mine, [ws] * ws,
                              [q * ws for q in range(ws)], my_index=my_part)
-        cnt = cnt.view(ws, ws).cpu()  # cnt[p][q] = what p wants from q
-        self.send_counts = [int(cnt[q, my_part]) for q in range(ws)]
+        self._cnt = cnt.view(ws, ws).cpu()  # cnt[p][q] = p wants from q
+        self.send_counts = [int(self._cnt[q, my_part]) for q in range(ws)]
         self.recv_counts = recv_counts
-        # exchange the index lists themselves (one batched p2p round;
-        # indices are local to MY range on the send side)
+        self.send_idx = None  # exchanged by _exchange_lists()
+        self._bufs_ready = False
+
+    def _exchange_lists(self):
+        """Collective: ship my per-peer index lists; receive the lists
+        peers will be served from. Call only after worth_it() (globally
+        consistent) returned True on every rank."""
+        if self.send_idx is not None:
+            return
         import torch.distributed as dist
+        ws = len(self.verts_all)
+        my_part = self.p
+        device = self.recv_idx[0].device if self.recv_idx else "cpu"
         ops = []
-        self.send_idx = []
+        pend = []
         for off in range(1, ws):
             ps = (my_part + off) % ws
             pr = (my_part - off) % ws
-            if recv_counts[ps]:
+            if self.recv_counts[ps]:
                 ops.append(dist.P2POp(dist.isend,
                                       self.recv_idx[ps].contiguous(), ps))
             if self.send_counts[pr]:
                 buf = torch.empty(self.send_counts[pr], dtype=torch.long,
                                   device=device)
-                self.send_idx.append((pr, buf))
+                pend.append((pr, buf))
                 ops.append(dist.P2POp(dist.irecv, buf, pr))
         if ops:
             for r in dist.batch_isend_irecv(ops):
                 r.wait()
         sidx = [torch.zeros(0, dtype=torch.long, device=device)] * ws
-        for pr, buf in self.send_idx:
+        for pr, buf in pend:
             sidx[pr] = buf
         self.send_idx = sidx
-        self._bufs_ready = False
 
     def total_halo(self):
         """Vertices received per iteration (the wire cost)."""
@@ -123,10 +137,15 @@ class HaloExchange:
         f = os.environ.get("LUX_HALO")
         if f is not None:
             return f == "1"
+        if getattr(self, "_cnt", None) is None:
+            return False  # single process: nothing to exchange
+        # GLOBAL decision (identical on every rank): total halo elements
+        # over all rank pairs vs the total full-slice exchange
         ws = len(self.verts_all)
-        vp = self.verts_all[self.p]
-        full_cost = (sum(self.verts_all) - vp) + vp * (ws - 1)
-        halo_cost = self.total_halo() + sum(self.send_counts)
+        halo_cost = int(self._cnt.sum()) * 2  # each element recv'd + sent
+        full_cost = sum(
+            (sum(self.verts_all) - self.verts_all[p]) * 2
+            for p in range(ws))
         return halo_cost * 100 < full_cost * 35
 
     def _ensure_bufs(self, dtype, device):
@@ -157,6 +176,7 @@ class HaloExchange:
                 dst.copy_(my_slice)
         if not dx.initialized() or dx.world_size() == 1:
             return dx._DoneWork()
+        self._exchange_lists()  # lazy, first publish (collective)
         self._ensure_bufs(full.dtype, full.device)
         K = self.K
         mv = my_slice.view(self.verts_all[p], K) if K > 1 else my_slice
