@@ -244,6 +244,15 @@ def test_cf_als_bf16_matches_f32(monkeypatch):
         eng.old.copy_(torch.from_numpy(init.ravel()))
         eng.step()
         outs[mode] = eng.vectors().cpu().numpy().copy()
-    scale = np.maximum(np.abs(outs["f32"]).max(axis=1, keepdims=True), 1.0)
-    np.testing.assert_allclose(outs["bf16"] / scale, outs["f32"] / scale,
-                               rtol=0, atol=3e-2)
+        if mode == "bf16":
+            indeg = np.diff(part.row_ptr.cpu().numpy().view(np.uint64))
+    # compare WELL-CONDITIONED rows only (indeg >= 2K): for deg < K the
+    # Gram is rank-deficient and lambda=1e-3 leaves cond ~ 1e5+, where
+    # both the bf16 rounding AND the fp32 atomic-order nondeterminism
+    # blow up along near-null directions (the f32 residual test covers
+    # those rows' correctness contract instead)
+    well = indeg >= 2 * K
+    assert well.sum() > 50  # the lane-map check needs real coverage
+    a, b = outs["bf16"][well], outs["f32"][well]
+    scale = np.maximum(np.abs(b).max(axis=1, keepdims=True), 1.0)
+    np.testing.assert_allclose(a / scale, b / scale, rtol=0, atol=3e-2)
