@@ -162,6 +162,11 @@ class PushEngine:
         self.labels = torch.empty(p.nv, dtype=U32, device=device)
         self.labels_part = torch.empty(p.vp, dtype=U32, device=device)
         self.snapshot = torch.empty_like(self.labels_part)
+        # hipGraph for the pull-fallback body (~50 fixed-address ctypes
+        # launches: seed + blocked sweeps + bitmap + fixup); captured on
+        # 2nd use, single-rank only (multi-rank syncs labels first)
+        self._pull_graph = None
+        self._pull_graph_uses = 0
         self._fq_init = None
         self.reset(source)
 
@@ -239,6 +244,44 @@ class PushEngine:
         self._sync_labels()
         return self.labels
 
+    def _pull_body_eager(self):
+        p = self.part
+        s = _stream()
+        mode = ng.PULL_MIN if self.is_min else ng.PULL_MAX
+        run_pull(p, mode, self.labels, self.labels_part, None, 0.0)
+        ng.build_bitmap(s, p.vp, self.snapshot, self.labels_part,
+                        self.new_seg)
+        ng.frontier_fixup(s, p.vp, p.row_left, self.capacity, 1,
+                          self.snapshot, self.labels_part, self.deg_part,
+                          self.new_seg, self.new_annex, self.tmp_seg,
+                          self.meta_mine, None, self.max_items)
+
+    def _run_pull_body(self):
+        """Pull fallback iteration body (sweep + bitmap + device fixups).
+        At world 1 it is a fixed ~50-launch sequence over fixed addresses:
+        captured into a hipGraph on its 2nd use and replayed (one launch
+        instead of ~50 ctypes calls — the same lever as PagerankEngine's
+        single-GPU capture)."""
+        import os
+        if self._pull_graph is not None:
+            self._pull_graph.replay()
+            return
+        self._pull_graph_uses += 1
+        if (dx.world_size() == 1 and self._pull_graph_uses >= 2
+                and os.environ.get("LUX_HIPGRAPH", "1") == "1"
+                and not os.environ.get("LUX_SYNC_CHECK")):
+            try:
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self._pull_body_eager()
+                self._pull_graph = g
+                self._pull_graph.replay()  # capture records, doesn't run
+                return
+            except Exception as e:  # capture unsupported: stay eager
+                print(f"[lux] push pull-body hipGraph disabled: {e}")
+                os.environ["LUX_HIPGRAPH"] = "0"
+        self._pull_body_eager()
+
     def step(self):
         """One push iteration. Returns the global new-frontier count
         (from the exchanged meta — no extra sync)."""
@@ -289,8 +332,7 @@ class PushEngine:
         if pull_fallback:
             new_dense = True
             self._sync_labels()  # pull reads every vertex's label
-            mode = ng.PULL_MIN if self.is_min else ng.PULL_MAX
-            run_pull(p, mode, self.labels, self.labels_part, None, 0.0)
+            self._run_pull_body()
             self._bits_stale = True  # pull writes labels directly
             item_counter = None
         else:
@@ -337,14 +379,15 @@ class PushEngine:
             item_counter = self.item_counter
 
         # ---- frontier fix-ups + meta, all device-side ----
-        if new_dense:
-            ng.build_bitmap(s, p.vp, self.snapshot, self.labels_part,
-                            self.new_seg)
-        ng.frontier_fixup(s, p.vp, p.row_left, self.capacity,
-                          int(new_dense), self.snapshot, self.labels_part,
-                          self.deg_part, self.new_seg, self.new_annex,
-                          self.tmp_seg, self.meta_mine, item_counter,
-                          self.max_items)
+        if item_counter is not None:  # push path (pull body fused above)
+            if new_dense:
+                ng.build_bitmap(s, p.vp, self.snapshot, self.labels_part,
+                                self.new_seg)
+            ng.frontier_fixup(s, p.vp, p.row_left, self.capacity,
+                              int(new_dense), self.snapshot,
+                              self.labels_part, self.deg_part, self.new_seg,
+                              self.new_annex, self.tmp_seg, self.meta_mine,
+                              item_counter, self.max_items)
 
         # ---- exchange: meta first (the ONE host read), then payloads ----
         dx.all_gather_slices(self.meta_all, self.meta_mine,
