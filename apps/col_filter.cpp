@@ -12,10 +12,22 @@ using namespace lux;
 
 int main(int argc, char** argv) {
   AppArgs a = parse_input_args(argc, argv);
-  if (a.num_gpu > 1) {
-    // multi-GPU: exec the torchrun RCCL engine (one rank per GPU; same
-    // CLI, reference README.md:42-45 drop-in)
-    return exec_torchrun_app("lux_amd.apps.cf", a.num_gpu, argc, argv);
+  if (const char* mr = getenv("LUX_MULTI_RANK")) {
+    // re-exec'd native multi-GPU worker (SGD or MFMA ALS sweeps)
+    HostCSC g;
+    if (!load_graph(a, &g, true)) return 1;
+    if (a.als && a.k > 64) {
+      fprintf(stderr, "[lux] -als covers K <= 64 (MFMA tile grid)\n");
+      return 1;
+    }
+    return col_filter_multi_child(g, atoi(mr),
+                                  atoi(getenv("LUX_MULTI_WORLD")),
+                                  getenv("LUX_MULTI_IDFILE"), a.k,
+                                  a.num_iter, a.als, a.dump);
+  }
+  if (a.num_gpu > 1 || getenv("LUX_NATIVE_MULTI")) {
+    // native fork+exec + RCCL engine, one worker per GPU
+    return run_multi_workers(a.num_gpu, argc, argv);
   }
   HostCSC g;
   if (!load_graph(a, &g, true)) return 1;

@@ -192,6 +192,19 @@ __global__ void uf_flatten_kernel(V_ID nv, V_ID* parent, V_ID* labels) {
   }
 }
 
+__global__ void count_diff_kernel(uint64_t n, const uint32_t* a,
+                                  const uint32_t* b,
+                                  unsigned long long* out) {
+  __shared__ unsigned long long lds[BLOCK / WAVE];
+  unsigned long long c = 0;
+  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
+  for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    if (a[i] != b[i]) c++;
+  c = block_reduce_sum(c, lds);
+  if (threadIdx.x == 0 && c) atomicAdd(out, c);
+}
+
 }  // namespace lux
 
 using namespace lux;
@@ -267,6 +280,16 @@ void lux_gpu_uf_flatten(uint64_t stream, V_ID nv, V_ID* parent,
   hipStream_t s = (hipStream_t)stream;
   hipLaunchKernelGGL(uf_flatten_kernel, dim3(grid_for(nv)), dim3(BLOCK), 0,
                      s, nv, parent, labels);
+  LUX_POST_LAUNCH(stream);
+}
+
+// element diff count (the native multi-GPU star-exchange convergence test)
+void lux_gpu_count_diff(uint64_t stream, uint64_t n, const uint32_t* a,
+                        const uint32_t* b,
+                        unsigned long long* out /*pre-zeroed*/) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(count_diff_kernel, dim3(grid_for(n)), dim3(BLOCK), 0, s,
+                     n, a, b, out);
   LUX_POST_LAUNCH(stream);
 }
 

@@ -141,6 +141,8 @@ void lux_gpu_cf_iter(uint64_t stream, uint32_t n0, const lux::V_ID* bin0,
                      int K);
 
 // cc_uf.hip
+void lux_gpu_count_diff(uint64_t stream, uint64_t n, const uint32_t* a,
+                        const uint32_t* b, unsigned long long* out);
 void lux_gpu_uf_union_binned(uint64_t stream, uint32_t n0,
                              const lux::V_ID* bin0, uint32_t n1,
                              const lux::V_ID* bin1, uint32_t n2,
@@ -153,6 +155,8 @@ void lux_gpu_uf_union_kth(uint64_t stream, lux::V_ID vp,
 void lux_gpu_cc_giant_bits(uint64_t stream, lux::V_ID nv,
                            const lux::V_ID* labels, lux::V_ID giant,
                            uint32_t* bits);
+void lux_gpu_uf_union_star(uint64_t stream, lux::V_ID nv,
+                           const lux::V_ID* star, lux::V_ID* parent);
 void lux_gpu_uf_flatten(uint64_t stream, lux::V_ID nv, lux::V_ID* parent,
                         lux::V_ID* labels);
 

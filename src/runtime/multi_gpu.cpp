@@ -16,6 +16,7 @@
 #include <cstdio>
 #include <cstring>
 #include <string>
+#include <unordered_map>
 #include <vector>
 
 #include "multi_gpu.h"
@@ -206,10 +207,309 @@ int pagerank_multi_child(const HostCSC& g, int rank, int ngpus,
   return 0;
 }
 
+// ---- shared worker bootstrap (device + RCCL + slice upload) ----
+namespace {
+
+struct MultiCtx {
+  hipStream_t s;
+  ncclComm_t comm;
+  Partition part;
+  V_ID rl = 0, rr = 0, vp = 0;
+  E_ID cl = 0, ep = 0;
+  std::vector<V_ID> verts;
+  E_ID* row_ptr = nullptr;  // local, 0-based
+  V_ID* col = nullptr;
+  WeightType* w = nullptr;
+  Bins bins;
+};
+
+// device + communicator + partition (NO allocations — the caller builds
+// its DeviceArena AFTER this call so it lands on the worker's device)
+int multi_join(const HostCSC& g, int ngpus, int rank, const char* idfile,
+               MultiCtx* c) {
+  int ndev = 0;
+  LUX_OK(hipGetDeviceCount(&ndev));
+  if (rank >= ndev) {
+    if (rank == ndev)
+      fprintf(stderr, "[lux] -ll:gpu %d but only %d visible GPU(s)\n",
+              ngpus, ndev);
+    return 3;
+  }
+  LUX_OK(hipSetDevice(rank));
+  LUX_OK(hipStreamCreate(&c->s));
+  ncclUniqueId id;
+  if (rank == 0) {
+    LUX_NCCL(ncclGetUniqueId(&id));
+    std::string tmp = std::string(idfile) + ".tmp";
+    FILE* f = fopen(tmp.c_str(), "wb");
+    if (!f || fwrite(&id, sizeof(id), 1, f) != 1) {
+      perror(idfile);
+      return 4;
+    }
+    fclose(f);
+    rename(tmp.c_str(), idfile);
+  } else {
+    FILE* f = nullptr;
+    for (int tries = 0; tries < 1200 && !f; tries++) {
+      f = fopen(idfile, "rb");
+      if (!f) usleep(100000);
+    }
+    if (!f || fread(&id, sizeof(id), 1, f) != 1) {
+      fprintf(stderr, "[lux] rank %d: no RCCL id file %s\n", rank, idfile);
+      return 4;
+    }
+    fclose(f);
+  }
+  LUX_NCCL(ncclCommInitRank(&c->comm, ngpus, id, rank));
+  c->part = partition_edge_balanced(g.nv, g.ne, g.col_end.data(), ngpus);
+  c->rl = c->part.row_left[rank];
+  c->rr = c->part.row_right[rank];
+  c->vp = c->rr >= c->rl ? c->rr - c->rl + 1 : 0;
+  c->cl = c->part.col_left[rank];
+  c->ep = c->part.col_right[rank] - c->cl;
+  c->verts.resize(ngpus);
+  for (int r = 0; r < ngpus; r++)
+    c->verts[r] = c->part.row_right[r] >= c->part.row_left[r]
+                      ? c->part.row_right[r] - c->part.row_left[r] + 1
+                      : 0;
+  return 0;
+}
+
+// upload ONLY my slice (call after the arena exists on my device)
+void multi_upload(const HostCSC& g, bool weighted, DeviceArena& arena,
+                  MultiCtx* c) {
+  E_ID* col_end_sl = arena.alloc_n<E_ID>(c->vp ? c->vp : 1);
+  c->col = arena.alloc_n<V_ID>(c->ep ? c->ep : 1);
+  if (c->vp)
+    LUX_OK(hipMemcpyAsync(col_end_sl, g.col_end.data() + c->rl,
+                          sizeof(E_ID) * c->vp, hipMemcpyHostToDevice,
+                          c->s));
+  if (c->ep)
+    LUX_OK(hipMemcpyAsync(c->col, g.src.data() + c->cl,
+                          sizeof(V_ID) * c->ep, hipMemcpyHostToDevice,
+                          c->s));
+  if (weighted && c->ep) {
+    c->w = arena.alloc_n<WeightType>(c->ep);
+    LUX_OK(hipMemcpyAsync(c->w, g.weight.data() + c->cl,
+                          sizeof(WeightType) * c->ep, hipMemcpyHostToDevice,
+                          c->s));
+  }
+  c->row_ptr = arena.alloc_n<E_ID>(c->vp + 1);
+  lux_gpu_local_row_ptr((uint64_t)c->s, c->vp, c->cl, col_end_sl,
+                        c->row_ptr);
+  c->bins.build(c->row_ptr, c->vp, c->ep, arena, c->s);
+}
+
+void multi_barrier(MultiCtx& c, float* bar) {
+  LUX_NCCL(ncclAllReduce(bar, bar, 1, ncclFloat, ncclSum, c.comm, c.s));
+  LUX_OK(hipStreamSynchronize(c.s));
+}
+
+}  // namespace
+
+int components_multi_child(const HostCSC& g, int rank, int ngpus,
+                           const char* idfile, bool check, const char* dump,
+                           bool verbose) {
+  MultiCtx c;
+  int rc = multi_join(g, ngpus, rank, idfile, &c);
+  if (rc) return rc;
+  DeviceArena arena(16ull * g.nv                     // parent + labels
+                    + (uint64_t)ngpus * 4ull * g.nv  // gathered stars
+                    + 8ull * g.nv                    // col_end slice
+                    + 8ull * (g.ne / (ngpus ? ngpus : 1) + 1)
+                    + 12ull * g.nv + (96ull << 20));
+  multi_upload(g, false, arena, &c);
+  V_ID* parent = arena.alloc_n<V_ID>(g.nv);
+  V_ID* labels = arena.alloc_n<V_ID>(g.nv);
+  V_ID* gathered = arena.alloc_n<V_ID>((uint64_t)ngpus * g.nv);
+  uint32_t* gbits = arena.alloc_n<uint32_t>((g.nv + 31) / 32);
+  unsigned long long* diff = arena.alloc_n<unsigned long long>(1);
+  float* bar = arena.alloc_n<float>(1);
+  {
+    std::vector<V_ID> h(g.nv);
+    for (V_ID v = 0; v < g.nv; v++) h[v] = v;
+    LUX_OK(hipMemcpyAsync(parent, h.data(), 4ull * g.nv,
+                          hipMemcpyHostToDevice, c.s));
+  }
+  multi_barrier(c, bar);
+  double t0 = now_seconds();
+  // Afforest sample-hook + giant-skip sweep over MY edge slice
+  // (cc_engine.py run, per rank)
+  for (uint32_t k = 0; k < 2; k++)
+    lux_gpu_uf_union_kth((uint64_t)c.s, c.vp, c.row_ptr, c.col, c.rl,
+                         parent, k);
+  lux_gpu_uf_flatten((uint64_t)c.s, g.nv, parent, labels);
+  V_ID giant;
+  {
+    std::vector<V_ID> hl(g.nv);
+    LUX_OK(hipMemcpyAsync(hl.data(), labels, 4ull * g.nv,
+                          hipMemcpyDeviceToHost, c.s));
+    LUX_OK(hipStreamSynchronize(c.s));
+    V_ID stride = g.nv / 65536 ? g.nv / 65536 : 1;
+    std::unordered_map<V_ID, uint32_t> freq;
+    V_ID best = hl[0];
+    uint32_t bestc = 0;
+    for (V_ID v = 0; v < g.nv; v += stride) {
+      uint32_t n = ++freq[hl[v]];
+      if (n > bestc) {
+        bestc = n;
+        best = hl[v];
+      }
+    }
+    giant = best;
+  }
+  lux_gpu_cc_giant_bits((uint64_t)c.s, g.nv, labels, giant, gbits);
+  lux_gpu_uf_union_binned((uint64_t)c.s, c.bins.n0, c.bins.bin0, c.bins.n1,
+                          c.bins.bin1, c.bins.n2, c.bins.bin2, c.row_ptr,
+                          c.col, c.rl, parent, gbits);
+  lux_gpu_uf_flatten((uint64_t)c.s, g.nv, parent, labels);
+  // star-forest exchange: allgather full label vectors, union peers'
+  // stars, repeat until no label moves (monotone merges -> O(log P))
+  int rounds = 1;
+  while (ngpus > 1) {
+    LUX_NCCL(ncclAllGather(labels, gathered, g.nv, ncclUint32, c.comm,
+                           c.s));
+    for (int q = 0; q < ngpus; q++)
+      if (q != rank)
+        lux_gpu_uf_union_star((uint64_t)c.s, g.nv,
+                              gathered + (uint64_t)q * g.nv, parent);
+    lux_gpu_uf_flatten((uint64_t)c.s, g.nv, parent, labels);
+    LUX_OK(hipMemsetAsync(diff, 0, 8, c.s));
+    // prev labels live in my gathered copy
+    lux_gpu_count_diff((uint64_t)c.s, g.nv,
+                       gathered + (uint64_t)rank * g.nv, labels, diff);
+    LUX_NCCL(ncclAllReduce(diff, diff, 1, ncclUint64, ncclSum, c.comm,
+                           c.s));
+    unsigned long long hd;
+    LUX_OK(hipMemcpyAsync(&hd, diff, 8, hipMemcpyDeviceToHost, c.s));
+    LUX_OK(hipStreamSynchronize(c.s));
+    rounds++;
+    if (hd == 0) break;
+  }
+  multi_barrier(c, bar);
+  double secs = now_seconds() - t0;
+  uint64_t mistakes = 0;
+  if (check) {
+    unsigned long long* m = arena.alloc_n<unsigned long long>(1);
+    LUX_OK(hipMemsetAsync(m, 0, 8, c.s));
+    lux_gpu_check((uint64_t)c.s, 0, c.vp, c.rl, c.row_ptr, c.col, labels,
+                  m);
+    LUX_NCCL(ncclAllReduce(m, m, 1, ncclUint64, ncclSum, c.comm, c.s));
+    unsigned long long hm;
+    LUX_OK(hipMemcpyAsync(&hm, m, 8, hipMemcpyDeviceToHost, c.s));
+    LUX_OK(hipStreamSynchronize(c.s));
+    mistakes = hm;
+  }
+  if (rank == 0) {
+    printf("ELAPSED TIME = %7.7f s\n", secs);
+    if (dump) dump_state_device(dump, labels, 1, 1, g.nv, (uint64_t)rounds);
+    printf("[lux] converged in %d exchange rounds, %.3f GTEPS (%d GPUs)\n",
+           rounds, double(g.ne) / secs / 1e9, ngpus);
+    if (check)
+      printf("[%s] %llu mistakes\n", mistakes == 0 ? "PASS" : "FAIL",
+             (unsigned long long)mistakes);
+  }
+  (void)verbose;
+  ncclCommDestroy(c.comm);
+  return check && mistakes ? 1 : 0;
+}
+
+int col_filter_multi_child(const HostCSC& g, int rank, int ngpus,
+                           const char* idfile, int K, int iters, bool als,
+                           const char* dump) {
+  MultiCtx c;
+  int rc = multi_join(g, ngpus, rank, idfile, &c);
+  if (rc) return rc;
+  DeviceArena arena(
+      8ull * g.nv + 8ull * (g.ne / (ngpus ? ngpus : 1) + 1) * 2
+      + 8ull * (uint64_t)g.nv * K  // old + new slack
+      + 12ull * g.nv + (96ull << 20)
+      + (als ? 4ull * g.nv + (17ull << 20) * (g.ne / 2048 / 1024 + 1) : 0));
+  multi_upload(g, true, arena, &c);
+  float* old_ = arena.alloc_n<float>((uint64_t)g.nv * K);
+  float* new_ = arena.alloc_n<float>((uint64_t)(c.vp ? c.vp : 1) * K);
+  {
+    std::vector<float> h((uint64_t)g.nv * K, sqrtf(1.0f / K));
+    LUX_OK(hipMemcpyAsync(old_, h.data(), 4ull * g.nv * K,
+                          hipMemcpyHostToDevice, c.s));
+  }
+  int* hubidx = nullptr;
+  float *gram = nullptr, *rhs = nullptr;
+  if (als && c.bins.nbig) {
+    hubidx = arena.alloc_n<int>(c.vp ? c.vp : 1);
+    gram = arena.alloc_n<float>((uint64_t)c.bins.nbig * 64 * 64);
+    rhs = arena.alloc_n<float>((uint64_t)c.bins.nbig * 64);
+    std::vector<V_ID> hv(c.bins.nbig);
+    LUX_OK(hipMemcpyAsync(hv.data(), c.bins.bin2v, 4ull * c.bins.nbig,
+                          hipMemcpyDeviceToHost, c.s));
+    LUX_OK(hipStreamSynchronize(c.s));
+    std::vector<int> hidx(c.vp, -1);
+    for (uint32_t i = 0; i < c.bins.nbig; i++) hidx[hv[i]] = (int)i;
+    LUX_OK(hipMemcpyAsync(hubidx, hidx.data(), 4ull * c.vp,
+                          hipMemcpyHostToDevice, c.s));
+  }
+  float* bar = arena.alloc_n<float>(1);
+  multi_barrier(c, bar);
+  double t0 = now_seconds();
+  for (int it = 0; it < iters; it++) {
+    if (c.vp) {
+      if (als) {
+        LUX_OK(hipMemcpyAsync(new_, old_ + (uint64_t)c.rl * K,
+                              4ull * c.vp * K, hipMemcpyDeviceToDevice,
+                              c.s));
+        if (c.bins.nbig) {
+          LUX_OK(hipMemsetAsync(gram, 0, 4ull * c.bins.nbig * 64 * 64,
+                                c.s));
+          LUX_OK(hipMemsetAsync(rhs, 0, 4ull * c.bins.nbig * 64, c.s));
+        }
+        lux_gpu_cf_als_iter((uint64_t)c.s, c.bins.n0, c.bins.bin0,
+                            c.bins.n1, c.bins.bin1, c.bins.n2, c.bins.bin2,
+                            c.bins.nbig, c.bins.bin2v, hubidx, gram, rhs,
+                            c.row_ptr, c.col, c.w, old_, nullptr, new_,
+                            c.rl, K);
+      } else {
+        lux_gpu_cf_seed((uint64_t)c.s, (uint64_t)c.vp * K,
+                        old_ + (uint64_t)c.rl * K, new_);
+        lux_gpu_cf_iter((uint64_t)c.s, c.bins.n0, c.bins.bin0, c.bins.n1,
+                        c.bins.bin1, c.bins.n2, c.bins.bin2, c.bins.nbig,
+                        c.bins.bin2v, c.row_ptr, c.col, c.w, old_, new_,
+                        c.rl, K);
+      }
+    }
+    // all-gather(v) of K-dim vector slices (direct pairwise over xGMI)
+    LUX_NCCL(ncclGroupStart());
+    for (int r = 0; r < ngpus; r++) {
+      if (r == rank) continue;
+      if (c.vp)
+        LUX_NCCL(ncclSend(new_, (uint64_t)c.vp * K, ncclFloat, r, c.comm,
+                          c.s));
+      if (c.verts[r])
+        LUX_NCCL(ncclRecv(old_ + (uint64_t)c.part.row_left[r] * K,
+                          (uint64_t)c.verts[r] * K, ncclFloat, r, c.comm,
+                          c.s));
+    }
+    LUX_NCCL(ncclGroupEnd());
+    if (c.vp)
+      LUX_OK(hipMemcpyAsync(old_ + (uint64_t)c.rl * K, new_,
+                            4ull * c.vp * K, hipMemcpyDeviceToDevice, c.s));
+  }
+  multi_barrier(c, bar);
+  double secs = now_seconds() - t0;
+  if (rank == 0) {
+    printf("ELAPSED TIME = %7.7f s\n", secs);
+    if (dump)
+      dump_state_device(dump, old_, 0, (uint32_t)K, g.nv, (uint64_t)iters);
+    printf("[lux] %.3f GTEPS (%d sweeps, rank %d, %d GPUs)\n",
+           double(g.ne) * iters / secs / 1e9, iters, K, ngpus);
+  }
+  ncclCommDestroy(c.comm);
+  return 0;
+}
+
 // Launcher: fork + EXEC one child per GPU (a plain fork would inherit the
 // parent's process state; exec gives each rank a fresh runtime). The
 // parent itself never touches HIP or RCCL.
-int run_pagerank_multi(int ngpus, int argc, char** argv) {
+int run_multi_workers(int ngpus, int argc, char** argv) {
   char idfile[64];
   snprintf(idfile, sizeof(idfile), "/tmp/lux_rccl_%d.id", (int)getpid());
   unlink(idfile);
@@ -251,6 +551,10 @@ int run_pagerank_multi(int ngpus, int argc, char** argv) {
   }
   unlink(idfile);
   return rc;
+}
+
+int run_pagerank_multi(int ngpus, int argc, char** argv) {
+  return run_multi_workers(ngpus, argc, argv);
 }
 
 int exec_torchrun_app(const char* module, int ngpus, int argc, char** argv) {

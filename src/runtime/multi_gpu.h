@@ -19,6 +19,18 @@ int pagerank_multi_child(const HostCSC& g, int rank, int ngpus,
                          const char* idfile, int iters, bool verbose,
                          const char* dump);
 
+// Native multi-GPU workers for the other apps (same launcher: the app
+// main re-invokes these under LUX_MULTI_RANK).
+int components_multi_child(const HostCSC& g, int rank, int ngpus,
+                           const char* idfile, bool check, const char* dump,
+                           bool verbose);
+int col_filter_multi_child(const HostCSC& g, int rank, int ngpus,
+                           const char* idfile, int K, int iters, bool als,
+                           const char* dump);
+// Generic launcher (same as run_pagerank_multi's body): fork+exec one
+// re-invocation per GPU.
+int run_multi_workers(int ngpus, int argc, char** argv);
+
 // Replace this process with `torchrun --nproc-per-node N -m <module>
 // <original args>` (one rank per GPU over the same RCCL exchange layer,
 // Python driver). Returns only on exec failure.

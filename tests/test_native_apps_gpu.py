@@ -188,3 +188,56 @@ def test_native_pagerank_multi_rccl_world1(tmp_path):
     g = Graph.load(lux)
     want = cpu_ref.pagerank(g, 5)
     np.testing.assert_allclose(got, want, rtol=2e-4, atol=1e-9)
+
+
+def test_native_components_multi_rccl_world1(tmp_path):
+    """Native multi-GPU union-find worker (components_multi_child) at
+    world 1: same labelling as the CPU reference, and the star-exchange
+    convergence loop terminates."""
+    import numpy as np
+    from lux_amd import checkpoint as ck
+    from lux_amd import cpu_ref
+    from lux_amd.graph import Graph
+    lux = str(tmp_path / "s.lux")
+    out = str(tmp_path / "l.luxs")
+    g = Graph.rmat(13, 60000, seed=3, sym=True)
+    g.save(lux)
+    env = dict(os.environ, LUX_NATIVE_MULTI="1")
+    r = subprocess.run([f"{BIN}/components", "-file", lux, "-dump", out,
+                        "-check"], cwd=ROOT, capture_output=True, text=True,
+                       timeout=300, env=env)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "PASS" in r.stdout
+    got, _ = ck.load_state(out)
+    want, _ = cpu_ref.cc(g)
+    np.testing.assert_array_equal(got, want)
+
+
+def test_native_col_filter_multi_rccl_world1(tmp_path):
+    """Native multi-GPU CF worker (col_filter_multi_child) at world 1:
+    SGD sweeps match the CPU reference; -als runs and beats SGD loss."""
+    import numpy as np
+    from lux_amd import checkpoint as ck
+    from lux_amd import cpu_ref
+    from lux_amd.graph import Graph
+    lux = str(tmp_path / "b.lux")
+    o1 = str(tmp_path / "v.luxs")
+    o2 = str(tmp_path / "a.luxs")
+    _run([f"{BIN}/rmat_gen", "-kind", "bipartite", "-users", "800",
+          "-items", "120", "-ne", "20000", "-o", lux])
+    env = dict(os.environ, LUX_NATIVE_MULTI="1")
+    r = subprocess.run([f"{BIN}/col_filter", "-file", lux, "-ni", "3",
+                        "-k", "20", "-dump", o1], cwd=ROOT,
+                       capture_output=True, text=True, timeout=300, env=env)
+    assert r.returncode == 0, r.stdout + r.stderr
+    got, _ = ck.load_state(o1)
+    g = Graph.load(lux, want_weights=True)
+    want = cpu_ref.cf(g, 20, 3)
+    np.testing.assert_allclose(got, want, rtol=2e-4, atol=1e-7)
+    r = subprocess.run([f"{BIN}/col_filter", "-file", lux, "-ni", "3",
+                        "-k", "32", "-als", "-dump", o2], cwd=ROOT,
+                       capture_output=True, text=True, timeout=300, env=env)
+    assert r.returncode == 0, r.stdout + r.stderr
+    als, _ = ck.load_state(o2)
+    sgd = cpu_ref.cf(g, 32, 3)
+    assert cpu_ref.cf_loss(g, 32, als) < cpu_ref.cf_loss(g, 32, sgd)
