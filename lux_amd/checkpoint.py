@@ -109,8 +109,13 @@ def resume_engine(path, engine):
 def _refresh_slice(engine):
     """Pipelined pull engines mirror their slice in cur_part and may hold an
     in-flight publish handle — resync after overwriting `old`."""
-    if getattr(engine, "_handle", None) is not None:
-        engine._handle.wait()
+    h = getattr(engine, "_handle", None)
+    if h is not None:
+        if isinstance(h, list):  # per-peer publish handles
+            for _q, w in h:
+                w.wait()
+        else:
+            h.wait()
         engine._handle = None
     if hasattr(engine, "cur_part"):
         p = engine.part

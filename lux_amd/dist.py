@@ -140,6 +140,37 @@ def all_gather_slices_async(full, my_slice, verts, row_left, my_index=None):
     return _BatchWork(dist.batch_isend_irecv(ops))
 
 
+def all_gather_slices_per_peer(full, my_slice, verts, row_left,
+                               my_index=None):
+    """Like all_gather_slices_async, but one batched p2p group PER PEER:
+    returns [(peer, work), ...] in ring-offset order so the caller can
+    start consuming peer q's slice as soon as ITS group lands instead of
+    after the whole exchange (the pipelined pull engines sweep each
+    peer's src blocks behind its own wait). All groups are posted
+    up-front (host never blocks between groups), so mismatched
+    completion order cannot deadlock."""
+    ws = world_size()
+    i = rank() if my_index is None else my_index
+    dst = full.narrow(0, row_left[i], verts[i])
+    if verts[i] and dst.data_ptr() != my_slice.data_ptr():
+        dst.copy_(my_slice)
+    out = []
+    if ws == 1:
+        return out
+    for off in range(1, ws):
+        ps = (i + off) % ws
+        pr = (i - off) % ws
+        ops = []
+        if my_slice.numel():
+            ops.append(dist.P2POp(dist.isend, my_slice, ps))
+        if verts[pr]:
+            ops.append(dist.P2POp(
+                dist.irecv, full.narrow(0, row_left[pr], verts[pr]), pr))
+        if ops:
+            out.append((pr, _BatchWork(dist.batch_isend_irecv(ops))))
+    return out
+
+
 def exchange_multi_async(parts, my_index=None):
     """Several all-gather(v)s fused into ONE batched p2p round: parts is a
     list of (full, my_slice, verts, row_left) tuples. Each pairwise message

@@ -64,6 +64,9 @@ def run_pull_sweeps(part, mode, oldv, newv, deg, init_rank, subset=None):
                 continue
             if subset == "remote" and blk["local"]:
                 continue
+            if isinstance(subset, tuple) and subset[0] == "peer" \
+                    and blk.get("owner") != subset[1]:
+                continue
             ng.pull_iter(s, mode, blk["n0"], blk["bin0"], blk["n1"],
                          blk["bin1"], blk["n2"], blk["bin2"], blk["nbig"],
                          blk["bin2v"], blk["row_ptr"], blk["col"], oldv,
@@ -395,7 +398,13 @@ class GraphPart:
         rank-local block(s) while the all-gather of remote slices flies."""
         bounds = set(range(0, self.nv, 1 << shift)) | {self.nv}
         if self.nparts > 1 and self.vp > 0:
-            bounds |= {self.row_left, self.row_right + 1}
+            # EVERY rank's boundaries: each block then lies inside exactly
+            # one rank's range, so remote sweeps can run per-peer as that
+            # peer's slice arrives (all_gather_slices_per_peer)
+            for q in range(self.nparts):
+                if self.verts_all[q]:
+                    bounds |= {self.row_left_all[q],
+                               self.row_right_all[q] + 1}
         return sorted(bounds)
 
     def prepare_pull(self, force_shift=None):
@@ -464,12 +473,19 @@ class GraphPart:
             row32 = torch.empty(vp + 1, dtype=U32, device=device)
             ng.u64_to_u32(s, vp + 1, row_ptr_b, row32)
             del row_ptr_b
-            local = self.nparts > 1 and bounds[b] >= self.row_left \
-                and bounds[b + 1] <= self.row_right + 1
+            owner = -1
+            if self.nparts > 1:
+                for q in range(self.nparts):
+                    if self.verts_all[q] \
+                            and bounds[b] >= self.row_left_all[q] \
+                            and bounds[b + 1] <= self.row_right_all[q] + 1:
+                        owner = q
+                        break
+            local = self.nparts > 1 and owner == self.p
             self.blocks.append(dict(row_ptr=row32, row_u32=1, col=col_b,
                                     n0=n0, n1=n1, n2=n2, nbig=nbig, bin0=b0,
                                     bin1=b1, bin2=b2, bin2v=b2v,
-                                    local=local))
+                                    local=local, owner=owner))
             begin = end
         self._blk_col = blk_col  # keep the narrow()s' base alive
         del ends, cursor
@@ -496,13 +512,23 @@ def _publish_async(engine, my_slice):
                                       p.row_left_all, my_index=p.p)
 
 
+def _wait_handle(engine):
+    h = engine._handle
+    engine._handle = None
+    if h is None:
+        return
+    if isinstance(h, list):
+        for _q, w in h:
+            w.wait()
+    else:
+        h.wait()
+
+
 def _finalize_full(engine):
     """Collective: one full slice all-gather so the replicated array is
     current EVERYWHERE (halo iterations keep it current only at read
     positions). Every rank must call (apps do, after the timed loop)."""
-    if engine._handle is not None:
-        engine._handle.wait()
-        engine._handle = None
+    _wait_handle(engine)
     if engine.halo is None:
         return
     p = engine.part
@@ -592,20 +618,35 @@ class PagerankEngine:
         # overlap: sweep the rank-local src block against cur_part (offset
         # base so global src ids index it) while the gather of remote
         # slices from the previous step is still in flight (SURVEY.md §7
-        # M2: xGMI exchange hidden under local compute)
+        # M2: xGMI exchange hidden under local compute); with per-peer
+        # publishes, each peer's remote blocks are swept right after ITS
+        # slice lands instead of after the whole exchange
         self.new_part.zero_()
         local_base = self.cur_part.data_ptr() - p.row_left * 4
         run_pull_sweeps(p, ng.PULL_PR, local_base, self.new_part, self.deg,
                         self.init_rank, subset="local")
-        if self._handle is not None:
-            self._handle.wait()
+        if isinstance(self._handle, list):
+            for q, w in self._handle:
+                w.wait()
+                run_pull_sweeps(p, ng.PULL_PR, self.old, self.new_part,
+                                self.deg, self.init_rank,
+                                subset=("peer", q))
             self._handle = None
-        run_pull_sweeps(p, ng.PULL_PR, self.old, self.new_part, self.deg,
-                        self.init_rank, subset="remote")
+        else:
+            if self._handle is not None:
+                self._handle.wait()
+                self._handle = None
+            run_pull_sweeps(p, ng.PULL_PR, self.old, self.new_part,
+                            self.deg, self.init_rank, subset="remote")
         ng.pull_finish_pr(_stream(), p.vp, self.new_part, self.deg,
                           p.row_left, self.init_rank)
         self.cur_part, self.new_part = self.new_part, self.cur_part
-        self._handle = _publish_async(self, self.cur_part)
+        if self.halo is None:
+            self._handle = dx.all_gather_slices_per_peer(
+                self.old, self.cur_part, p.verts_all, p.row_left_all,
+                my_index=p.p)
+        else:
+            self._handle = _publish_async(self, self.cur_part)
 
     def finalize(self):
         """Collective (all ranks): complete the replicated rank vector
@@ -616,9 +657,7 @@ class PagerankEngine:
         """Replicated stored ranks (pr/out_degree) as a torch tensor.
         With the halo exchange active, call finalize() (collectively)
         first if you need non-halo positions of peers' slices."""
-        if self._handle is not None:
-            self._handle.wait()
-            self._handle = None
+        _wait_handle(self)
         return self.old
 
 
@@ -656,14 +695,26 @@ class LabelPullEngine:
         local_base = self.cur_part.data_ptr() - p.row_left * 4
         run_pull_sweeps(p, self.mode, local_base, self.new_part, None, 0.0,
                         subset="local")
-        if self._handle is not None:
-            self._handle.wait()
+        if isinstance(self._handle, list):
+            for q, w in self._handle:
+                w.wait()
+                run_pull_sweeps(p, self.mode, self.old, self.new_part,
+                                None, 0.0, subset=("peer", q))
             self._handle = None
-        run_pull_sweeps(p, self.mode, self.old, self.new_part, None, 0.0,
-                        subset="remote")
+        else:
+            if self._handle is not None:
+                self._handle.wait()
+                self._handle = None
+            run_pull_sweeps(p, self.mode, self.old, self.new_part, None,
+                            0.0, subset="remote")
         changed = (self.new_part != self.cur_part).sum()
         self.cur_part, self.new_part = self.new_part, self.cur_part
-        self._handle = _publish_async(self, self.cur_part)
+        if self.halo is None:
+            self._handle = dx.all_gather_slices_per_peer(
+                self.old, self.cur_part, p.verts_all, p.row_left_all,
+                my_index=p.p)
+        else:
+            self._handle = _publish_async(self, self.cur_part)
         return changed
 
     def labels(self):
