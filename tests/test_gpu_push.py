@@ -119,3 +119,40 @@ def test_sssp_unreachable_stays_inf():
     want, _ = cpu_ref.sssp(g, 0)
     assert np.array_equal(got, want)
     assert got[4] == 0xFFFFFFFF
+
+
+def test_cc_labelprop_multipart_single_process():
+    """MODE_MAX (CC label propagation) across 3 partitions with the manual
+    meta/annex exchange: unlike BFS, queued labels here are arbitrary
+    values, so this exercises the annex-fill-after-fixup path (labels
+    must be iteration-final, not enqueue-time)."""
+    scale, ne, seed, P = 10, 30000, 29, 3
+    nv = 1 << scale
+    full = DeviceCSC.rmat(scale, ne, seed=seed)
+    parts = [GraphPart(full, P, p, keep_full=True) for p in range(P)]
+    engines = [PushEngine(parts[p], PushEngine.MODE_MAX) for p in range(P)]
+    for it in range(4 * nv):
+        for e in engines:
+            e.step()
+        mh = np.stack([e.meta_mine.cpu().numpy().view(np.uint32)
+                       for e in engines])
+        for ei in engines:
+            for j, ej in enumerate(engines):
+                if parts[j].vp == 0:
+                    continue
+                ei.labels.narrow(0, parts[j].row_left,
+                                 parts[j].vp).copy_(ej.labels_part)
+                ei.fq_all.narrow(0, int(ei.seg_off[j]),
+                                 ei.seg_bytes[j]).copy_(ej.new_seg)
+                cap_j = int(ei.annex_off[j + 1] - ei.annex_off[j])
+                ei.fq_annex_all.narrow(0, int(ei.annex_off[j]),
+                                       cap_j).copy_(ej.new_annex[:cap_j])
+            ei.meta_host = mh.copy()
+            ei.labels_current = True
+            ei.headers = [(int(mh[q, 0]), int(mh[q, 1])) for q in range(P)]
+        if int(mh[:, 1].sum()) == 0 and not mh[:, 4].any():
+            break
+    got = engines[0].labels.cpu().numpy().view(np.uint32)
+    g = Graph.rmat(scale, ne, seed=seed)
+    want, _ = cpu_ref.cc(g)
+    assert np.array_equal(got, want)
