@@ -25,6 +25,9 @@ int main(int argc, char** argv) {
                                   getenv("LUX_MULTI_IDFILE"), a.k,
                                   a.num_iter, a.als, a.dump);
   }
+  if (a.num_gpu > 1 && getenv("LUX_TORCHRUN"))
+    // escape hatch: the torchrun RCCL engine (same CLI, Python driver)
+    return exec_torchrun_app("lux_amd.apps.cf", a.num_gpu, argc, argv);
   if (a.num_gpu > 1 || getenv("LUX_NATIVE_MULTI")) {
     // native fork+exec + RCCL engine, one worker per GPU
     return run_multi_workers(a.num_gpu, argc, argv);

@@ -14,19 +14,25 @@ using namespace lux;
 int main(int argc, char** argv) {
   AppArgs a = parse_input_args(argc, argv);
   if (const char* mr = getenv("LUX_MULTI_RANK")) {
-    // re-exec'd native multi-GPU worker (union-find + star exchange)
+    // re-exec'd native multi-GPU worker: union-find + star exchange, or
+    // the reference-parity push label propagation under -labelprop
     HostCSC g;
     if (!load_graph(a, &g, false)) return 1;
+    if (a.labelprop)
+      return push_multi_child(g, atoi(mr), atoi(getenv("LUX_MULTI_WORLD")),
+                              /*is_min=*/false, 0,
+                              getenv("LUX_MULTI_IDFILE"), a.check, a.dump,
+                              a.verbose);
     return components_multi_child(g, atoi(mr),
                                   atoi(getenv("LUX_MULTI_WORLD")),
                                   getenv("LUX_MULTI_IDFILE"), a.check,
                                   a.dump, a.verbose);
   }
+  if (a.num_gpu > 1 && getenv("LUX_TORCHRUN"))
+    // escape hatch: the torchrun RCCL engine (same CLI, Python driver)
+    return exec_torchrun_app("lux_amd.apps.cc", a.num_gpu, argc, argv);
   if (a.num_gpu > 1 || getenv("LUX_NATIVE_MULTI")) {
-    if (a.labelprop)  // multi-GPU label propagation lives in the
-                      // torchrun push engine (same CLI)
-      return exec_torchrun_app("lux_amd.apps.cc", a.num_gpu, argc, argv);
-    // native fork+exec + RCCL union-find engine, one worker per GPU
+    // native fork+exec + RCCL engine, one worker per GPU
     return run_multi_workers(a.num_gpu, argc, argv);
   }
   HostCSC g;

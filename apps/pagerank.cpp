@@ -19,6 +19,9 @@ int main(int argc, char** argv) {
                                 getenv("LUX_MULTI_IDFILE"), a.num_iter,
                                 a.verbose, a.dump);
   }
+  if (a.num_gpu > 1 && getenv("LUX_TORCHRUN"))
+    // escape hatch: the torchrun RCCL engine (same CLI, Python driver)
+    return exec_torchrun_app("lux_amd.apps.pagerank", a.num_gpu, argc, argv);
   if (a.num_gpu > 1 || getenv("LUX_NATIVE_MULTI")) {
     // native fork+exec + RCCL engine, one worker process per GPU (the
     // reference's `pagerank -ll:gpu N` drop-in, README.md:42).

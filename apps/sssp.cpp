@@ -11,10 +11,21 @@ using namespace lux;
 
 int main(int argc, char** argv) {
   AppArgs a = parse_input_args(argc, argv);
-  if (a.num_gpu > 1) {
-    // multi-GPU: exec the torchrun RCCL engine (one rank per GPU; same
-    // CLI, reference README.md:42-45 drop-in)
+  if (const char* mr = getenv("LUX_MULTI_RANK")) {
+    // re-exec'd native multi-GPU push worker (meta-record exchange)
+    HostCSC g;
+    if (!load_graph(a, &g, false)) return 1;
+    return push_multi_child(g, atoi(mr), atoi(getenv("LUX_MULTI_WORLD")),
+                            /*is_min=*/true, a.start,
+                            getenv("LUX_MULTI_IDFILE"), a.check, a.dump,
+                            a.verbose);
+  }
+  if (a.num_gpu > 1 && getenv("LUX_TORCHRUN"))
+    // escape hatch: the torchrun RCCL engine (same CLI, Python driver)
     return exec_torchrun_app("lux_amd.apps.sssp", a.num_gpu, argc, argv);
+  if (a.num_gpu > 1 || getenv("LUX_NATIVE_MULTI")) {
+    // native fork+exec + RCCL push engine, one worker per GPU
+    return run_multi_workers(a.num_gpu, argc, argv);
   }
   HostCSC g;
   if (!load_graph(a, &g, false)) return 1;

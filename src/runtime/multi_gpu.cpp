@@ -506,6 +506,338 @@ int col_filter_multi_child(const HostCSC& g, int rank, int ngpus,
   return 0;
 }
 
+// ---- native multi-GPU push engine (SSSP hop / CC label prop) ----
+// The C++/RCCL twin of lux_amd/push_engine.py: per iteration exactly ONE
+// blocking host read (the all-gathered 32 B/rank meta record); dense/
+// sparse conversions run in the device fixup chain; payloads ship only
+// USED bytes; label slices travel only on dense iterations (sparse
+// queues carry the label annex, repaired into the replicated array by
+// frontier_expand).
+int push_multi_child(const HostCSC& g, int rank, int ngpus, bool is_min,
+                     V_ID source, const char* idfile, bool check,
+                     const char* dump, bool verbose) {
+  MultiCtx c;
+  int rc = multi_join(g, ngpus, rank, idfile, &c);
+  if (rc) return rc;
+  // seg/annex layout for all ranks
+  std::vector<uint64_t> seg_bytes(ngpus), seg_off(ngpus + 1, 0);
+  std::vector<uint64_t> annex_cap(ngpus), annex_off(ngpus + 1, 0);
+  for (int q = 0; q < ngpus; q++) {
+    seg_bytes[q] = (frontier_bytes(c.verts[q]) + 15) & ~15ull;
+    seg_off[q + 1] = seg_off[q] + seg_bytes[q];
+    annex_cap[q] = frontier_capacity(c.verts[q]);
+    annex_off[q + 1] = annex_off[q] + annex_cap[q];
+  }
+  V_ID capacity = frontier_capacity(c.vp);
+  uint64_t max_items = (uint64_t)g.nv / SPARSE_THRESHOLD +
+                       (c.ep ? c.ep : 1) / 8192 + 1024;
+  DeviceArena arena(
+      4ull * g.nv * 3                  // labels + deg + slack
+      + 8ull * (g.nv + 2)              // push row_ptr
+      + 4ull * (c.ep + 1) * 3          // pull col + push col + cursor slack
+      + 8ull * (g.nv + 2)              // col_end slice + cursor
+      + 12ull * g.nv + (96ull << 20)   // bins + slack
+      + seg_off[ngpus] + 3 * seg_bytes[rank] + 4 * annex_off[ngpus]
+      + 16ull * max_items + (16ull << 20));
+  multi_upload(g, false, arena, &c);
+  // push CSR: all nv sources -> my-partition dsts
+  E_ID* push_row_ptr = arena.alloc_n<E_ID>(g.nv + 1);
+  V_ID* push_col = arena.alloc_n<V_ID>(c.ep ? c.ep : 1);
+  V_ID* deg = arena.alloc_n<V_ID>(g.nv);  // global out-degrees after
+                                          // allreduce; slice at +rl
+  {
+    E_ID* ends;
+    unsigned long long *cursor, *partials;
+    LUX_OK(hipMalloc(&ends, sizeof(E_ID) * g.nv));
+    LUX_OK(hipMalloc(&cursor, sizeof(uint64_t) * g.nv));
+    LUX_OK(hipMalloc(&partials,
+                     sizeof(uint64_t) * lux_gpu_scan_partials_size(g.nv)));
+    LUX_OK(hipMemsetAsync(deg, 0, sizeof(V_ID) * g.nv, c.s));
+    lux_gpu_hist_u32((uint64_t)c.s, c.ep, c.col, deg);
+    lux_gpu_scan_end_offsets((uint64_t)c.s, g.nv, deg, ends, partials);
+    lux_gpu_local_row_ptr((uint64_t)c.s, g.nv, 0, ends, push_row_ptr);
+    LUX_OK(hipMemcpyAsync(cursor, push_row_ptr, sizeof(E_ID) * g.nv,
+                          hipMemcpyDeviceToDevice, c.s));
+    lux_gpu_csr_scatter((uint64_t)c.s, c.ep, c.col, c.row_ptr, c.vp, c.rl,
+                        cursor, push_col);
+    // global out-degrees (the meta edge-volume pricing input)
+    LUX_NCCL(ncclAllReduce(deg, deg, g.nv, ncclUint32, ncclSum, c.comm,
+                           c.s));
+    LUX_OK(hipStreamSynchronize(c.s));
+    hipFree(ends);
+    hipFree(cursor);
+    hipFree(partials);
+  }
+  uint32_t* labels = arena.alloc_n<uint32_t>(g.nv);
+  uint32_t* labels_part = arena.alloc_n<uint32_t>(c.vp ? c.vp : 1);
+  uint32_t* snapshot = arena.alloc_n<uint32_t>(c.vp ? c.vp : 1);
+  uint8_t* fq_all = arena.alloc_n<uint8_t>(seg_off[ngpus]);
+  uint8_t* new_seg = arena.alloc_n<uint8_t>(seg_bytes[rank]);
+  uint8_t* tmp_seg = arena.alloc_n<uint8_t>(seg_bytes[rank]);
+  uint32_t* annex_all = arena.alloc_n<uint32_t>(annex_off[ngpus]);
+  uint32_t* new_annex = arena.alloc_n<uint32_t>(capacity ? capacity : 1);
+  lux_uint2* items = arena.alloc_n<lux_uint2>(max_items);
+  uint32_t* item_counter = arena.alloc_n<uint32_t>(4);
+  uint32_t* meta_mine = arena.alloc_n<uint32_t>(8);
+  uint32_t* meta_all = arena.alloc_n<uint32_t>(8ull * ngpus);
+  uint32_t* bits = nullptr;
+  if (is_min) bits = arena.alloc_n<uint32_t>((c.vp + 31) / 32 + 1);
+  bool bits_stale = true;
+  float* bar = arena.alloc_n<float>(1);
+
+  // seed labels + frontier + meta (sssp_gpu.cu:733-744 / components)
+  std::vector<uint32_t> meta_h(8ull * ngpus, 0);
+  {
+    std::vector<uint32_t> hl(g.nv);
+    std::vector<uint8_t> hfq(seg_off[ngpus], 0);
+    for (int q = 0; q < ngpus; q++) {
+      FrontierHeader* h = (FrontierHeader*)(hfq.data() + seg_off[q]);
+      if (is_min) {
+        h->type = FrontierHeader::SPARSE_QUEUE;
+        bool owner = c.verts[q] && c.part.row_left[q] <= source &&
+                     source <= c.part.row_right[q];
+        h->numNodes = owner ? 1 : 0;
+        if (owner)
+          *(V_ID*)(hfq.data() + seg_off[q] + sizeof(FrontierHeader)) =
+              source;
+      } else {
+        h->type = FrontierHeader::DENSE_BITMAP;
+        h->numNodes = c.verts[q];
+        memset(hfq.data() + seg_off[q] + sizeof(FrontierHeader), 0xFF,
+               (c.verts[q] + 7) / 8);
+      }
+      meta_h[8ull * q + 0] = h->type;
+      meta_h[8ull * q + 1] = h->numNodes;
+    }
+    if (is_min) {
+      for (V_ID v = 0; v < g.nv; v++) hl[v] = INF_LABEL;
+      hl[source] = 0;
+    } else {
+      for (V_ID v = 0; v < g.nv; v++) hl[v] = v;
+    }
+    LUX_OK(hipMemcpyAsync(labels, hl.data(), 4ull * g.nv,
+                          hipMemcpyHostToDevice, c.s));
+    if (c.vp)
+      LUX_OK(hipMemcpyAsync(labels_part, hl.data() + c.rl, 4ull * c.vp,
+                            hipMemcpyHostToDevice, c.s));
+    LUX_OK(hipMemcpyAsync(fq_all, hfq.data(), seg_off[ngpus],
+                          hipMemcpyHostToDevice, c.s));
+    LUX_OK(hipMemsetAsync(annex_all, 0, 4ull * annex_off[ngpus], c.s));
+  }
+  bool labels_current = true;
+  auto sync_labels = [&]() {
+    if (labels_current) return;
+    LUX_NCCL(ncclGroupStart());
+    for (int r = 0; r < ngpus; r++) {
+      if (r == rank) continue;
+      if (c.vp)
+        LUX_NCCL(ncclSend(labels_part, c.vp, ncclUint32, r, c.comm, c.s));
+      if (c.verts[r])
+        LUX_NCCL(ncclRecv(labels + c.part.row_left[r], c.verts[r],
+                          ncclUint32, r, c.comm, c.s));
+    }
+    LUX_NCCL(ncclGroupEnd());
+    if (c.vp)
+      LUX_OK(hipMemcpyAsync(labels + c.rl, labels_part, 4ull * c.vp,
+                            hipMemcpyDeviceToDevice, c.s));
+    labels_current = true;
+  };
+
+  multi_barrier(c, bar);
+  double t0 = now_seconds();
+  int iters = 0;
+  while (true) {
+    // decisions from the CURRENT frontier's meta
+    uint64_t old_fq = 0, evol = 0;
+    int dense_votes = 0;
+    bool overflow = false;
+    for (int q = 0; q < ngpus; q++) {
+      old_fq += meta_h[8ull * q + 1];
+      evol |= 0;  // (accumulated below)
+      evol += ((uint64_t)meta_h[8ull * q + 3] << 32) | meta_h[8ull * q + 2];
+      if (meta_h[8ull * q + 0] == FrontierHeader::DENSE_BITMAP)
+        dense_votes++;
+      if (meta_h[8ull * q + 4]) overflow = true;
+    }
+    bool new_dense = dense_votes >= ngpus - dense_votes;
+    if (c.vp) {
+      LUX_OK(hipMemcpyAsync(snapshot, labels_part, 4ull * c.vp,
+                            hipMemcpyDeviceToDevice, c.s));
+      LUX_OK(hipMemsetAsync(new_seg, 0, sizeof(FrontierHeader), c.s));
+    }
+    bool pull_fallback =
+        overflow || old_fq > (uint64_t)g.nv / SPARSE_THRESHOLD;
+    if (overflow && rank == 0)
+      fprintf(stderr, "[lux] frontier expand overflow: forced pull\n");
+    bool did_push = false;
+    if (!pull_fallback) {
+      uint64_t thresh = is_min ? g.ne / 2 : g.ne / 8;
+      if (evol > thresh) pull_fallback = true;
+      else if ((evol * (c.ep ? c.ep : 1) / g.ne) / 16 > capacity)
+        new_dense = true;
+    }
+    if (pull_fallback) {
+      new_dense = true;
+      sync_labels();
+      if (c.vp) {
+        LUX_OK(hipMemcpyAsync(labels_part, labels + c.rl, 4ull * c.vp,
+                              hipMemcpyDeviceToDevice, c.s));
+        lux_gpu_pull_iter((uint64_t)c.s, is_min ? 1 : 2, c.bins.n0,
+                          c.bins.bin0, c.bins.n1, c.bins.bin1, c.bins.n2,
+                          c.bins.bin2, c.bins.nbig, c.bins.bin2v,
+                          c.row_ptr, 0, c.col, labels, labels_part, nullptr,
+                          c.rl, 0.0f);
+      }
+      bits_stale = true;
+    } else {
+      did_push = true;
+      LUX_OK(hipMemsetAsync(item_counter, 0, 16, c.s));
+      for (int q = 0; q < ngpus; q++) {
+        uint32_t typ = meta_h[8ull * q + 0], num = meta_h[8ull * q + 1];
+        if (!c.verts[q]) continue;
+        if (typ == FrontierHeader::DENSE_BITMAP) {
+          lux_gpu_frontier_expand((uint64_t)c.s, 1, c.part.row_left[q],
+                                  c.verts[q], fq_all + seg_off[q], nullptr,
+                                  nullptr, push_row_ptr, items,
+                                  item_counter, (uint32_t)max_items);
+        } else if (num) {
+          lux_gpu_frontier_expand((uint64_t)c.s, 0, 0, num,
+                                  fq_all + seg_off[q],
+                                  annex_all + annex_off[q], labels,
+                                  push_row_ptr, items, item_counter,
+                                  (uint32_t)max_items);
+        }
+      }
+      uint32_t* b = nullptr;
+      if (is_min && c.vp) {
+        if (bits_stale) {
+          lux_gpu_bits_from_labels((uint64_t)c.s, c.vp, labels_part, bits);
+          bits_stale = false;
+        }
+        b = bits;
+      }
+      lux_gpu_push_chunk_scatter((uint64_t)c.s, is_min ? 1 : 0,
+                                 new_dense ? 1 : 0, items, item_counter,
+                                 (uint32_t)max_items, push_row_ptr,
+                                 push_col, labels, snapshot, labels_part,
+                                 c.rl, new_seg, capacity, b);
+    }
+    if (c.vp && new_dense)
+      lux_gpu_build_bitmap((uint64_t)c.s, c.vp, snapshot, labels_part,
+                           new_seg);
+    lux_gpu_frontier_fixup((uint64_t)c.s, c.vp, c.rl, capacity,
+                           new_dense ? 1 : 0, snapshot, labels_part,
+                           deg + c.rl, new_seg, new_annex, tmp_seg,
+                           meta_mine, did_push ? item_counter : nullptr,
+                           (uint32_t)max_items);
+    // meta first (the ONE host read), then payloads sized from it
+    LUX_NCCL(ncclAllGather(meta_mine, meta_all, 8, ncclUint32, c.comm,
+                           c.s));
+    LUX_OK(hipMemcpyAsync(meta_h.data(), meta_all, 32ull * ngpus,
+                          hipMemcpyDeviceToHost, c.s));
+    LUX_OK(hipStreamSynchronize(c.s));
+    // payload exchange: used bytes + annex + (dense ranks') label slices
+    uint64_t used_me = 0, annex_me = 0;
+    bool dense_me = meta_h[8ull * rank + 0] == FrontierHeader::DENSE_BITMAP;
+    if (c.vp) {
+      if (dense_me) used_me = 8 + ((uint64_t)c.vp + 7) / 8;
+      else {
+        used_me = 8 + 4ull * meta_h[8ull * rank + 1];
+        annex_me = meta_h[8ull * rank + 1];
+      }
+    }
+    LUX_NCCL(ncclGroupStart());
+    for (int r = 0; r < ngpus; r++) {
+      if (r == rank) continue;
+      if (used_me)
+        LUX_NCCL(ncclSend(new_seg, used_me, ncclUint8, r, c.comm, c.s));
+      if (annex_me)
+        LUX_NCCL(ncclSend(new_annex, annex_me, ncclUint32, r, c.comm,
+                          c.s));
+      if (dense_me && c.vp)
+        LUX_NCCL(ncclSend(labels_part, c.vp, ncclUint32, r, c.comm, c.s));
+      if (c.verts[r]) {
+        bool dense_r =
+            meta_h[8ull * r + 0] == FrontierHeader::DENSE_BITMAP;
+        uint64_t used_r = dense_r ? 8 + ((uint64_t)c.verts[r] + 7) / 8
+                                  : 8 + 4ull * meta_h[8ull * r + 1];
+        LUX_NCCL(ncclRecv(fq_all + seg_off[r], used_r, ncclUint8, r,
+                          c.comm, c.s));
+        if (!dense_r && meta_h[8ull * r + 1])
+          LUX_NCCL(ncclRecv(annex_all + annex_off[r],
+                            meta_h[8ull * r + 1], ncclUint32, r, c.comm,
+                            c.s));
+        if (dense_r)
+          LUX_NCCL(ncclRecv(labels + c.part.row_left[r], c.verts[r],
+                            ncclUint32, r, c.comm, c.s));
+      }
+    }
+    LUX_NCCL(ncclGroupEnd());
+    // my own copies
+    if (c.vp) {
+      LUX_OK(hipMemcpyAsync(fq_all + seg_off[rank], new_seg, used_me,
+                            hipMemcpyDeviceToDevice, c.s));
+      if (annex_me)
+        LUX_OK(hipMemcpyAsync(annex_all + annex_off[rank], new_annex,
+                              4ull * annex_me, hipMemcpyDeviceToDevice,
+                              c.s));
+      if (dense_me)
+        LUX_OK(hipMemcpyAsync(labels + c.rl, labels_part, 4ull * c.vp,
+                              hipMemcpyDeviceToDevice, c.s));
+    }
+    // slice q fresh iff published now, or fresh before and unchanged
+    bool all_ok = true;
+    for (int q = 0; q < ngpus; q++) {
+      if (!c.verts[q]) continue;
+      bool pub = meta_h[8ull * q + 0] == FrontierHeader::DENSE_BITMAP;
+      if (labels_current) {
+        if (!pub && meta_h[8ull * q + 1]) all_ok = false;
+      } else if (!pub) {
+        all_ok = false;
+      }
+    }
+    labels_current = all_ok;
+    iters++;
+    uint64_t total = 0;
+    bool ovf = false;
+    for (int q = 0; q < ngpus; q++) {
+      total += meta_h[8ull * q + 1];
+      if (meta_h[8ull * q + 4]) ovf = true;
+    }
+    if (verbose && rank == 0)
+      printf("iter %d: activeNodes(%llu)\n", iters,
+             (unsigned long long)total);
+    if (total == 0 && !ovf) break;
+    if (iters > (int)g.nv * 4) break;  // safety
+  }
+  sync_labels();
+  multi_barrier(c, bar);
+  double secs = now_seconds() - t0;
+  uint64_t mistakes = 0;
+  if (check) {
+    unsigned long long* m = arena.alloc_n<unsigned long long>(1);
+    LUX_OK(hipMemsetAsync(m, 0, 8, c.s));
+    lux_gpu_check((uint64_t)c.s, is_min ? 1 : 0, c.vp, c.rl, c.row_ptr,
+                  c.col, labels, m);
+    LUX_NCCL(ncclAllReduce(m, m, 1, ncclUint64, ncclSum, c.comm, c.s));
+    unsigned long long hm;
+    LUX_OK(hipMemcpyAsync(&hm, m, 8, hipMemcpyDeviceToHost, c.s));
+    LUX_OK(hipStreamSynchronize(c.s));
+    mistakes = hm;
+  }
+  if (rank == 0) {
+    printf("ELAPSED TIME = %7.7f s\n", secs);
+    if (dump) dump_state_device(dump, labels, 1, 1, g.nv, (uint64_t)iters);
+    printf("[lux] converged in %d iterations, %.3f GTEPS (%d GPUs)\n",
+           iters, double(g.ne) / secs / 1e9, ngpus);
+    if (check)
+      printf("[%s] %llu mistakes\n", mistakes == 0 ? "PASS" : "FAIL",
+             (unsigned long long)mistakes);
+  }
+  ncclCommDestroy(c.comm);
+  return check && mistakes ? 1 : 0;
+}
+
 // Launcher: fork + EXEC one child per GPU (a plain fork would inherit the
 // parent's process state; exec gives each rank a fresh runtime). The
 // parent itself never touches HIP or RCCL.

@@ -27,6 +27,9 @@ int components_multi_child(const HostCSC& g, int rank, int ngpus,
 int col_filter_multi_child(const HostCSC& g, int rank, int ngpus,
                            const char* idfile, int K, int iters, bool als,
                            const char* dump);
+int push_multi_child(const HostCSC& g, int rank, int ngpus, bool is_min,
+                     lux::V_ID source, const char* idfile, bool check,
+                     const char* dump, bool verbose);
 // Generic launcher (same as run_pagerank_multi's body): fork+exec one
 // re-invocation per GPU.
 int run_multi_workers(int ngpus, int argc, char** argv);
