@@ -241,3 +241,46 @@ def test_native_col_filter_multi_rccl_world1(tmp_path):
     als, _ = ck.load_state(o2)
     sgd = cpu_ref.cf(g, 32, 3)
     assert cpu_ref.cf_loss(g, 32, als) < cpu_ref.cf_loss(g, 32, sgd)
+
+
+def test_native_sssp_multi_rccl_world1(tmp_path):
+    """Native multi-GPU push worker (push_multi_child) at world 1: exact
+    SSSP labels + passing oracle through the meta-record exchange path."""
+    import numpy as np
+    from lux_amd import checkpoint as ck
+    from lux_amd import cpu_ref
+    from lux_amd.graph import Graph
+    lux = str(tmp_path / "g.lux")
+    out = str(tmp_path / "l.luxs")
+    _run([f"{BIN}/rmat_gen", "-kind", "rmat", "-scale", "13", "-ne",
+          "100000", "-o", lux])
+    env = dict(os.environ, LUX_NATIVE_MULTI="1")
+    r = subprocess.run([f"{BIN}/sssp", "-file", lux, "-start", "0",
+                        "-check", "-dump", out], cwd=ROOT,
+                       capture_output=True, text=True, timeout=300, env=env)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "PASS" in r.stdout
+    got, _ = ck.load_state(out)
+    g = Graph.load(lux)
+    want, _ = cpu_ref.sssp(g, 0)
+    np.testing.assert_array_equal(got, want)
+
+
+def test_native_components_labelprop_multi_rccl_world1(tmp_path):
+    """Native multi-GPU push worker in MODE_MAX (-labelprop) at world 1."""
+    import numpy as np
+    from lux_amd import checkpoint as ck
+    from lux_amd import cpu_ref
+    from lux_amd.graph import Graph
+    lux = str(tmp_path / "d.lux")
+    out = str(tmp_path / "l.luxs")
+    g = Graph.rmat(12, 50000, seed=5)
+    g.save(lux)
+    env = dict(os.environ, LUX_NATIVE_MULTI="1")
+    r = subprocess.run([f"{BIN}/components", "-file", lux, "-labelprop",
+                        "-dump", out], cwd=ROOT, capture_output=True,
+                       text=True, timeout=300, env=env)
+    assert r.returncode == 0, r.stdout + r.stderr
+    got, _ = ck.load_state(out)
+    want, _ = cpu_ref.cc(g)
+    np.testing.assert_array_equal(got, want)
