@@ -78,16 +78,16 @@ class CFALSEngine:
                                      device=device)
         else:
             self.hubidx = self.gram = self.rhs_h = None
-        # bf16 gather replica (LUX_ALS_BF_GATHER=1, off by default):
-        # halves per-edge gather bytes, but MEASURED 2.5x SLOWER on the
-        # NetFlix shape (10.6 -> 38 ms/sweep; sub-dword gathers are far
-        # below dword gather throughput regardless of load opcode —
-        # global_load_ushort and d16 variants both) — kept as an A/B knob
+        # bf16 gather replica, loaded as DWORD pairs (two dims per load):
+        # halves gather bytes at full dword load rate. Per-lane SUB-dword
+        # gathers of the same table measured 2.5x SLOWER (BENCHLOG r2.4)
+        # — the dword-pair staging is what makes the replica pay.
+        # LUX_ALS_BF_GATHER=0 forces fp32 gathers; needs even K.
         import os
         self.old_bf = torch.empty(part.nv * K, dtype=torch.bfloat16,
                                   device=device) \
-            if os.environ.get("LUX_ALS_BF_GATHER") == "1" \
-            and not os.environ.get("LUX_ALS_F32") else None
+            if os.environ.get("LUX_ALS_BF_GATHER", "1") == "1" \
+            and K % 2 == 0 and not os.environ.get("LUX_ALS_F32") else None
 
     def step(self):
         p = self.part

@@ -219,6 +219,117 @@ __device__ __forceinline__ void als_vec_loads_bf(const CFAlsArgs& a,
     tmp[r] = __uint_as_float(raw[r] << 16);
 }
 
+// bf16-replica gathers, DWORD-shaped: each lane loads one dword = TWO
+// adjacent dims of one staged edge row (sub-dword gathers measured 2.5x
+// below dword gather throughput — BENCHLOG r2.4; this keeps the halved
+// bytes AND full-rate loads). Lanes 0..31 cover row r's 32 dim-pairs,
+// lanes 32..63 row r+1's; 16 passes stage a 32-edge tile. The pair is
+// split into the transposed [dim][edge] bf16 panel with two b16 writes;
+// rhs folds from the staged panel. Requires even K.
+__device__ __forceinline__ void als_stage_tile_bf2(const CFAlsArgs& a,
+                                                   uint32_t mycol, int rem,
+                                                   int lane, float myw,
+                                                   AlsGramLdsBf* lds,
+                                                   uint32_t pair[16]) {
+  const uint32_t* tab = (const uint32_t*)a.oldv_bf;
+  int half = lane >> 5;        // which of the pass's two rows
+  int m = lane & 31;           // dim pair index (dims 2m, 2m+1)
+  int kp = a.K >> 1;
+#pragma unroll
+  for (int pass = 0; pass < 16; pass++) {
+    int r = pass * 2 + half;
+    uint32_t src = __shfl(mycol, r, WAVE);
+    pair[pass] = (r < rem && m < kp)
+                     ? tab[(uint64_t)src * kp + m]
+                     : 0u;
+  }
+  if (lane < ALS_TILE) lds->W[lane] = myw;
+#pragma unroll
+  for (int pass = 0; pass < 16; pass++) {
+    int r = pass * 2 + half;
+    lds->S[(uint32_t)(2 * m) * ALS_BPITCH + r] =
+        ((const __bf16*)&pair[pass])[0];
+    lds->S[(uint32_t)(2 * m + 1) * ALS_BPITCH + r] =
+        ((const __bf16*)&pair[pass])[1];
+  }
+}
+
+__device__ __forceinline__ void als_gram_range_bf2(const CFAlsArgs& a,
+                                                   E_ID b, E_ID e,
+                                                   int lane,
+                                                   AlsGramLdsBf* lds,
+                                                   f32x4 acc[10],
+                                                   float* rhs) {
+  if (b >= e) return;
+  int rem = (int)(e - b < ALS_TILE ? e - b : (E_ID)ALS_TILE);
+  uint32_t mycol = 0;
+  float myw = 0.0f;
+  if (lane < rem) {
+    mycol = a.col[b + lane];
+    myw = (float)a.w[b + lane];
+  }
+  for (E_ID t = b; t < e; t += ALS_TILE) {
+    uint32_t pair[16];
+    als_stage_tile_bf2(a, mycol, rem, lane, myw, lds, pair);
+    int rem_cur = rem;
+    // prefetch next tile's col/weight
+    E_ID t2 = t + ALS_TILE;
+    bool more = t2 < e;
+    uint32_t ncol = 0;
+    float nw = 0.0f;
+    if (more) {
+      rem = (int)(e - t2 < ALS_TILE ? e - t2 : (E_ID)ALS_TILE);
+      if (lane < rem) {
+        ncol = a.col[t2 + lane];
+        nw = (float)a.w[t2 + lane];
+      }
+    }
+    als_lds_sync();
+    // rhs += sum_r w_r * S[r][lane], from the staged bf16 panel
+    if (lane < a.K) {
+      const __bf16* row = &lds->S[(uint32_t)lane * ALS_BPITCH];
+      float r0 = 0, r1 = 0;
+      for (int r = 0; r + 2 <= rem_cur; r += 2) {
+        r0 += lds->W[r] * (float)row[r];
+        r1 += lds->W[r + 1] * (float)row[r + 1];
+      }
+      if (rem_cur & 1) r0 += lds->W[rem_cur - 1] * (float)row[rem_cur - 1];
+      *rhs += r0 + r1;
+    }
+    bf16x8 f[4];
+    int m = lane & 15, g = lane >> 4;
+#pragma unroll
+    for (int ti = 0; ti < 4; ti++)
+      f[ti] = *(const bf16x8*)&lds->S[(uint32_t)(16 * ti + m) * ALS_BPITCH +
+                                      g * 8];
+    if (more) {  // next tile's gathers land under the MFMA burst... the
+      myw = nw;  // bf2 path stages lazily per tile, so just roll state
+      mycol = ncol;
+    }
+    acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[0], f[0], acc[0], 0,
+                                                     0, 0);
+    acc[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[0], f[1], acc[1], 0,
+                                                     0, 0);
+    acc[2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[0], f[2], acc[2], 0,
+                                                     0, 0);
+    acc[3] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[0], f[3], acc[3], 0,
+                                                     0, 0);
+    acc[4] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[1], f[1], acc[4], 0,
+                                                     0, 0);
+    acc[5] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[1], f[2], acc[5], 0,
+                                                     0, 0);
+    acc[6] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[1], f[3], acc[6], 0,
+                                                     0, 0);
+    acc[7] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[2], f[2], acc[7], 0,
+                                                     0, 0);
+    acc[8] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[2], f[3], acc[8], 0,
+                                                     0, 0);
+    acc[9] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[3], f[3], acc[9], 0,
+                                                     0, 0);
+    als_lds_sync();
+  }
+}
+
 __device__ __forceinline__ void als_gram_range_bf16(const CFAlsArgs& a,
                                                     E_ID b, E_ID e,
                                                     int lane,
@@ -234,10 +345,7 @@ __device__ __forceinline__ void als_gram_range_bf16(const CFAlsArgs& a,
     myw = (float)a.w[b + lane];
   }
   float tmp[ALS_TILE];
-  if (a.oldv_bf)
-    als_vec_loads_bf(a, mycol, rem, lane, tmp);
-  else
-    als_vec_loads(a, mycol, rem, lane, tmp);
+  als_vec_loads(a, mycol, rem, lane, tmp);
   for (E_ID t = b; t < e; t += ALS_TILE) {
     // rhs += sum_r w_r * S[r][lane] — straight from registers (lane=dim
     // holds S[r][lane] in tmp[r]); exact fp32
@@ -281,10 +389,7 @@ __device__ __forceinline__ void als_gram_range_bf16(const CFAlsArgs& a,
                                       g * 8];
     if (more) {
       myw = nw;
-      if (a.oldv_bf)
-        als_vec_loads_bf(a, ncol, rem, lane, tmp);
-      else
-        als_vec_loads(a, ncol, rem, lane, tmp);
+      als_vec_loads(a, ncol, rem, lane, tmp);
       mycol = ncol;
     }
     acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(f[0], f[0], acc[0], 0,
@@ -379,10 +484,14 @@ __global__ __launch_bounds__(ALS_TB) void cf_als_solve_kernel(
 #pragma unroll
     for (int t = 0; t < 10; t++) acc[t] = {0, 0, 0, 0};
     float rhs = 0.0f;
-    if (BF16)
-      als_gram_range_bf16(a, b, e, lane, &lds.gb, acc, &rhs);
-    else
+    if (BF16) {
+      if (a.oldv_bf)
+        als_gram_range_bf2(a, b, e, lane, &lds.gb, acc, &rhs);
+      else
+        als_gram_range_bf16(a, b, e, lane, &lds.gb, acc, &rhs);
+    } else {
       als_gram_range(a, b, e, lane, &lds.g, acc, &rhs);
+    }
     als_dump_gram(&lds, acc, lane, a.K);
     wave_cholesky64(lds.G, lane);
     float d = wave_spd_solve64(lds.G, rhs, lane);
@@ -419,10 +528,14 @@ __global__ __launch_bounds__(ALS_CHUNK_TB) void cf_als_gram_chunk_kernel(
 #pragma unroll
     for (int t = 0; t < 10; t++) acc[t] = {0, 0, 0, 0};
     float rhs = 0.0f;
-    if (BF16)
-      als_gram_range_bf16(a, b, e, lane, &lds.gb, acc, &rhs);
-    else
+    if (BF16) {
+      if (a.oldv_bf)
+        als_gram_range_bf2(a, b, e, lane, &lds.gb, acc, &rhs);
+      else
+        als_gram_range_bf16(a, b, e, lane, &lds.gb, acc, &rhs);
+    } else {
       als_gram_range(a, b, e, lane, &lds.g, acc, &rhs);
+    }
     int idx = hubidx[v];
     float* Gg = gram_scratch + (uint64_t)idx * ALS_K * ALS_K;
     atomicAdd(&rhs_scratch[(uint64_t)idx * ALS_K + lane], rhs);
