@@ -78,15 +78,16 @@ class CFALSEngine:
                                      device=device)
         else:
             self.hubidx = self.gram = self.rhs_h = None
-        # bf16 gather replica, loaded as DWORD pairs (two dims per load):
-        # halves gather bytes at full dword load rate. Per-lane SUB-dword
-        # gathers of the same table measured 2.5x SLOWER (BENCHLOG r2.4)
-        # — the dword-pair staging is what makes the replica pay.
-        # LUX_ALS_BF_GATHER=0 forces fp32 gathers; needs even K.
+        # bf16 gather replica: MEASURED SLOWER both ways and off by
+        # default (BENCHLOG r2.4/r2.5) — per-lane sub-dword gathers run
+        # 2.5x below dword rate, and the dword-pair staging variant loses
+        # the register-prefetch pipeline (20.4 vs 10.5 ms/sweep). The
+        # fp32-gather + bf16-LDS-stage split is the winning shape.
+        # LUX_ALS_BF_GATHER=1 re-enables the experiment (needs even K).
         import os
         self.old_bf = torch.empty(part.nv * K, dtype=torch.bfloat16,
                                   device=device) \
-            if os.environ.get("LUX_ALS_BF_GATHER", "1") == "1" \
+            if os.environ.get("LUX_ALS_BF_GATHER") == "1" \
             and K % 2 == 0 and not os.environ.get("LUX_ALS_F32") else None
 
     def step(self):
