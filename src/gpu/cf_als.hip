@@ -194,31 +194,6 @@ __device__ __forceinline__ void als_gram_range(const CFAlsArgs& a, E_ID b,
 // tile-col t, exactly like the f32 path's shared fragment.
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 
-__device__ __forceinline__ void als_vec_loads_bf(const CFAlsArgs& a,
-                                                 uint32_t mycol, int rem,
-                                                 int lane,
-                                                 float tmp[ALS_TILE]) {
-  // raw u16 loads + integer shift (bf16 = high half of f32): typed __bf16
-  // scalar loads scalarise through conversion libcalls and serialise the
-  // 32-outstanding-load pipeline (measured 4x slower)
-  const uint16_t* src16 = (const uint16_t*)a.oldv_bf;
-  uint32_t raw[ALS_TILE];  // u32 destinations: zero-extending ushort
-                           // loads write FULL registers (u16 destinations
-                           // compile to d16 merge loads whose
-                           // read-modify-write false deps serialise the
-                           // 32-outstanding-load pipeline)
-#pragma unroll
-  for (int r = 0; r < ALS_TILE; r++) {
-    uint32_t src = __shfl(mycol, r, WAVE);
-    raw[r] = (r < rem && lane < a.K)
-                 ? (uint32_t)src16[(uint64_t)src * a.K + lane]
-                 : 0u;
-  }
-#pragma unroll
-  for (int r = 0; r < ALS_TILE; r++)
-    tmp[r] = __uint_as_float(raw[r] << 16);
-}
-
 // bf16-replica gathers, DWORD-shaped: each lane loads one dword = TWO
 // adjacent dims of one staged edge row (sub-dword gathers measured 2.5x
 // below dword gather throughput — BENCHLOG r2.4; this keeps the halved
@@ -365,8 +340,6 @@ __device__ __forceinline__ void als_gram_range_bf16(const CFAlsArgs& a,
       pk.h[1] = (__bf16)tmp[2 * w + 1];
       row32[w] = pk.u;
     }
-    int rem_cur = rem;
-    (void)rem_cur;
     // prefetch next tile's col/weight, then its vectors (they land
     // under the MFMA burst)
     E_ID t2 = t + ALS_TILE;
