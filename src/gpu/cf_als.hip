@@ -445,7 +445,12 @@ __device__ __forceinline__ float wave_spd_solve64(const float* G, float r,
 }
 
 // ---- one wave per vertex: gram + solve fused (deg < T2 bin lists) ----
-template <bool BF16>
+// MODE selects the Gram path at compile time (0 = exact fp32, 1 = bf16
+// MFMA with fp32 gathers, 2 = bf16 MFMA with dword-pair bf16 gathers):
+// a runtime branch would co-inline every path and the register
+// allocator sizes for their union (measured: +40 VGPRs, occupancy 2->1,
+// 10.5 -> 18 ms/sweep with the branch never taken).
+template <int MODE>
 __global__ __launch_bounds__(ALS_TB) void cf_als_solve_kernel(
     uint32_t n, const V_ID* binlist, CFAlsArgs a) {
   __shared__ AlsLds lds;
@@ -457,14 +462,12 @@ __global__ __launch_bounds__(ALS_TB) void cf_als_solve_kernel(
 #pragma unroll
     for (int t = 0; t < 10; t++) acc[t] = {0, 0, 0, 0};
     float rhs = 0.0f;
-    if (BF16) {
-      if (a.oldv_bf)
-        als_gram_range_bf2(a, b, e, lane, &lds.gb, acc, &rhs);
-      else
-        als_gram_range_bf16(a, b, e, lane, &lds.gb, acc, &rhs);
-    } else {
+    if (MODE == 2)
+      als_gram_range_bf2(a, b, e, lane, &lds.gb, acc, &rhs);
+    else if (MODE == 1)
+      als_gram_range_bf16(a, b, e, lane, &lds.gb, acc, &rhs);
+    else
       als_gram_range(a, b, e, lane, &lds.g, acc, &rhs);
-    }
     als_dump_gram(&lds, acc, lane, a.K);
     wave_cholesky64(lds.G, lane);
     float d = wave_spd_solve64(lds.G, rhs, lane);
@@ -478,7 +481,7 @@ __global__ __launch_bounds__(ALS_TB) void cf_als_solve_kernel(
 // rhs_scratch: f32[nbig * 64]; both pre-zeroed by the engine each sweep.
 constexpr int ALS_CHUNK_TB = 128;  // 2 independent waves per workgroup
 
-template <bool BF16>
+template <int MODE>
 __global__ __launch_bounds__(ALS_CHUNK_TB) void cf_als_gram_chunk_kernel(
     uint32_t n2, const uint2* bin2, V_ID chunk_edges, const int* hubidx,
     float* gram_scratch, float* rhs_scratch, CFAlsArgs a) {
@@ -501,14 +504,12 @@ __global__ __launch_bounds__(ALS_CHUNK_TB) void cf_als_gram_chunk_kernel(
 #pragma unroll
     for (int t = 0; t < 10; t++) acc[t] = {0, 0, 0, 0};
     float rhs = 0.0f;
-    if (BF16) {
-      if (a.oldv_bf)
-        als_gram_range_bf2(a, b, e, lane, &lds.gb, acc, &rhs);
-      else
-        als_gram_range_bf16(a, b, e, lane, &lds.gb, acc, &rhs);
-    } else {
+    if (MODE == 2)
+      als_gram_range_bf2(a, b, e, lane, &lds.gb, acc, &rhs);
+    else if (MODE == 1)
+      als_gram_range_bf16(a, b, e, lane, &lds.gb, acc, &rhs);
+    else
       als_gram_range(a, b, e, lane, &lds.g, acc, &rhs);
-    }
     int idx = hubidx[v];
     float* Gg = gram_scratch + (uint64_t)idx * ALS_K * ALS_K;
     atomicAdd(&rhs_scratch[(uint64_t)idx * ALS_K + lane], rhs);
@@ -577,44 +578,36 @@ void lux_gpu_cf_als_iter(uint64_t stream, uint32_t n0, const V_ID* bin0,
   CFAlsArgs a{row_ptr, col, w, oldv, (const __bf16*)oldv_bf, newv, row_left,
               K};
   // bf16 Gram (16x16x32 MFMA, fp32 accumulate) is the default; exact
-  // fp32 (16x16x4) via LUX_ALS_F32=1
-  bool bf16 = !getenv("LUX_ALS_F32");
+  // fp32 (16x16x4) via LUX_ALS_F32=1; mode 2 = dword-pair bf16 gathers
+  // (only when the engine passed a bf16 replica)
+  int mode = getenv("LUX_ALS_F32") ? 0 : (oldv_bf ? 2 : 1);
+#define LUX_ALS_LAUNCH(K_, GRID_, N_, LIST_)                                  do {                                                                          if (mode == 2)                                                                hipLaunchKernelGGL(K_<2>, GRID_, dim3(ALS_TB), 0, s, N_, LIST_, a);       else if (mode == 1)                                                           hipLaunchKernelGGL(K_<1>, GRID_, dim3(ALS_TB), 0, s, N_, LIST_, a);       else                                                                          hipLaunchKernelGGL(K_<0>, GRID_, dim3(ALS_TB), 0, s, N_, LIST_, a);     } while (0)
   if (nbig) {
     uint32_t gw = (n2 + 1) / 2;
-    if (bf16)
-      hipLaunchKernelGGL(cf_als_gram_chunk_kernel<true>,
-                         dim3(gw > MAX_GRID ? MAX_GRID : gw),
+    dim3 grid(gw > MAX_GRID ? MAX_GRID : gw);
+    if (mode == 2)
+      hipLaunchKernelGGL(cf_als_gram_chunk_kernel<2>, grid,
+                         dim3(ALS_CHUNK_TB), 0, s, n2, bin2, (V_ID)8192,
+                         hubidx, gram_scratch, rhs_scratch, a);
+    else if (mode == 1)
+      hipLaunchKernelGGL(cf_als_gram_chunk_kernel<1>, grid,
                          dim3(ALS_CHUNK_TB), 0, s, n2, bin2, (V_ID)8192,
                          hubidx, gram_scratch, rhs_scratch, a);
     else
-      hipLaunchKernelGGL(cf_als_gram_chunk_kernel<false>,
-                         dim3(gw > MAX_GRID ? MAX_GRID : gw),
+      hipLaunchKernelGGL(cf_als_gram_chunk_kernel<0>, grid,
                          dim3(ALS_CHUNK_TB), 0, s, n2, bin2, (V_ID)8192,
                          hubidx, gram_scratch, rhs_scratch, a);
     hipLaunchKernelGGL(cf_als_hub_solve_kernel,
                        dim3(nbig > MAX_GRID ? MAX_GRID : nbig), dim3(ALS_TB),
                        0, s, nbig, bin2v, gram_scratch, rhs_scratch, a);
   }
-  if (n1) {
-    if (bf16)
-      hipLaunchKernelGGL(cf_als_solve_kernel<true>,
-                         dim3(n1 > MAX_GRID ? MAX_GRID : n1), dim3(ALS_TB),
-                         0, s, n1, bin1, a);
-    else
-      hipLaunchKernelGGL(cf_als_solve_kernel<false>,
-                         dim3(n1 > MAX_GRID ? MAX_GRID : n1), dim3(ALS_TB),
-                         0, s, n1, bin1, a);
-  }
-  if (n0) {
-    if (bf16)
-      hipLaunchKernelGGL(cf_als_solve_kernel<true>,
-                         dim3(n0 > MAX_GRID ? MAX_GRID : n0), dim3(ALS_TB),
-                         0, s, n0, bin0, a);
-    else
-      hipLaunchKernelGGL(cf_als_solve_kernel<false>,
-                         dim3(n0 > MAX_GRID ? MAX_GRID : n0), dim3(ALS_TB),
-                         0, s, n0, bin0, a);
-  }
+  if (n1)
+    LUX_ALS_LAUNCH(cf_als_solve_kernel, dim3(n1 > MAX_GRID ? MAX_GRID : n1),
+                   n1, bin1);
+  if (n0)
+    LUX_ALS_LAUNCH(cf_als_solve_kernel, dim3(n0 > MAX_GRID ? MAX_GRID : n0),
+                   n0, bin0);
+#undef LUX_ALS_LAUNCH
   LUX_POST_LAUNCH(stream);
 }
 
