@@ -45,7 +45,9 @@ int main(int argc, char** argv) {
                        + 16ull * (g.nv + 1) + 4ull * g.ne
                        + 12ull * g.nv + (64ull << 20)
                        + 8ull * g.nv + 3ull * frontier_bytes(g.nv)
-                       + 8ull * (g.ne / 8192 + g.nv / 16);
+                       + 8ull * (g.ne / 8192 + g.nv / 16)
+                       + BlockedPull::arena_bytes(g.nv, g.nv, g.ne)
+                       + g.nv / 2;  // BFS bits
   DeviceArena arena(arena_bytes);
   DeviceGraph dg = DeviceGraph::upload(g, arena, s);
 

@@ -40,7 +40,8 @@ int main(int argc, char** argv) {
                        + 12ull * g.nv + (64ull << 20)   // bins + slack
                        + 4ull * g.nv                    // degrees
                        + 8ull * g.nv                    // old/new
-                       + 8ull * (g.ne / 8192 + g.nv / 16);
+                       + 8ull * (g.ne / 8192 + g.nv / 16)
+                       + BlockedPull::arena_bytes(g.nv, g.nv, g.ne);
   DeviceArena arena(arena_bytes);
   DeviceGraph dg = DeviceGraph::upload(g, arena, s);
   SingleGpuPagerank engine(dg, arena, s);

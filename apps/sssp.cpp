@@ -38,7 +38,9 @@ int main(int argc, char** argv) {
                        + 12ull * g.nv + (64ull << 20)     // bins + slack
                        + 8ull * g.nv                      // labels+snapshot
                        + 3ull * frontier_bytes(g.nv)
-                       + 8ull * (g.ne / 8192 + g.nv / 16);
+                       + 8ull * (g.ne / 8192 + g.nv / 16)
+                       + BlockedPull::arena_bytes(g.nv, g.nv, g.ne)
+                       + g.nv / 2;  // BFS bits
   DeviceArena arena(arena_bytes);
   DeviceGraph dg = DeviceGraph::upload(g, arena, s);
   SingleGpuPush engine(dg, /*is_min=*/true, a.start, arena, s, a.verbose);

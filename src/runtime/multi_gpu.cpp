@@ -122,6 +122,7 @@ int pagerank_multi_child(const HostCSC& g, int rank, int ngpus,
                        + 12ull * vp + (64ull << 20)             // bins
                        + 4ull * g.nv                            // degrees
                        + 4ull * g.nv + 4ull * vp                // old/new
+                       + BlockedPull::arena_bytes(g.nv, vp, ep)
                        + (8ull << 20);
   DeviceArena arena(arena_bytes);
   // upload ONLY my slice (col_end slice + col slice); the replicated
@@ -139,6 +140,8 @@ int pagerank_multi_child(const HostCSC& g, int rank, int ngpus,
   lux_gpu_local_row_ptr((uint64_t)s, vp, cl, col_end_sl, row_ptr);
   Bins bins;
   bins.build(row_ptr, vp, ep, arena, s);
+  BlockedPull blocked;
+  blocked.build(row_ptr, col, vp, ep, g.nv, arena, s);
   // global out-degrees: slice histogram + RCCL all-reduce
   V_ID* deg = arena.alloc_n<V_ID>(g.nv);
   LUX_OK(hipMemsetAsync(deg, 0, sizeof(V_ID) * g.nv, s));
@@ -168,9 +171,13 @@ int pagerank_multi_child(const HostCSC& g, int rank, int ngpus,
   for (int it = 0; it < iters; it++) {
     if (vp) {
       LUX_OK(hipMemsetAsync(new_, 0, sizeof(float) * vp, s));
-      lux_gpu_pull_iter((uint64_t)s, 0, bins.n0, bins.bin0, bins.n1,
-                        bins.bin1, bins.n2, bins.bin2, bins.nbig, bins.bin2v,
-                        row_ptr, 0, col, old_, new_, deg, rl, init_rank);
+      if (blocked.active())
+        blocked.sweep(0, old_, new_, deg, rl, init_rank, s);
+      else
+        lux_gpu_pull_iter((uint64_t)s, 0, bins.n0, bins.bin0, bins.n1,
+                          bins.bin1, bins.n2, bins.bin2, bins.nbig,
+                          bins.bin2v, row_ptr, 0, col, old_, new_, deg, rl,
+                          init_rank);
       lux_gpu_pull_finish_pr((uint64_t)s, vp, new_, deg, rl, init_rank);
     }
     // all-gather(v) of slices: direct pairwise sends on the xGMI mesh
@@ -221,6 +228,7 @@ struct MultiCtx {
   V_ID* col = nullptr;
   WeightType* w = nullptr;
   Bins bins;
+  BlockedPull blocked;
 };
 
 // device + communicator + partition (NO allocations — the caller builds
@@ -298,6 +306,7 @@ void multi_upload(const HostCSC& g, bool weighted, DeviceArena& arena,
   lux_gpu_local_row_ptr((uint64_t)c->s, c->vp, c->cl, col_end_sl,
                         c->row_ptr);
   c->bins.build(c->row_ptr, c->vp, c->ep, arena, c->s);
+  c->blocked.build(c->row_ptr, c->col, c->vp, c->ep, g.nv, arena, c->s);
 }
 
 void multi_barrier(MultiCtx& c, float* bar) {
@@ -538,7 +547,8 @@ int push_multi_child(const HostCSC& g, int rank, int ngpus, bool is_min,
       + 8ull * (g.nv + 2)              // col_end slice + cursor
       + 12ull * g.nv + (96ull << 20)   // bins + slack
       + seg_off[ngpus] + 3 * seg_bytes[rank] + 4 * annex_off[ngpus]
-      + 16ull * max_items + (16ull << 20));
+      + 16ull * max_items
+      + BlockedPull::arena_bytes(g.nv, c.vp, c.ep) + (16ull << 20));
   multi_upload(g, false, arena, &c);
   // push CSR: all nv sources -> my-partition dsts
   E_ID* push_row_ptr = arena.alloc_n<E_ID>(g.nv + 1);
@@ -682,11 +692,15 @@ int push_multi_child(const HostCSC& g, int rank, int ngpus, bool is_min,
       if (c.vp) {
         LUX_OK(hipMemcpyAsync(labels_part, labels + c.rl, 4ull * c.vp,
                               hipMemcpyDeviceToDevice, c.s));
-        lux_gpu_pull_iter((uint64_t)c.s, is_min ? 1 : 2, c.bins.n0,
-                          c.bins.bin0, c.bins.n1, c.bins.bin1, c.bins.n2,
-                          c.bins.bin2, c.bins.nbig, c.bins.bin2v,
-                          c.row_ptr, 0, c.col, labels, labels_part, nullptr,
-                          c.rl, 0.0f);
+        if (c.blocked.active())
+          c.blocked.sweep(is_min ? 1 : 2, labels, labels_part, nullptr,
+                          c.rl, 0.0f, c.s);
+        else
+          lux_gpu_pull_iter((uint64_t)c.s, is_min ? 1 : 2, c.bins.n0,
+                            c.bins.bin0, c.bins.n1, c.bins.bin1, c.bins.n2,
+                            c.bins.bin2, c.bins.nbig, c.bins.bin2v,
+                            c.row_ptr, 0, c.col, labels, labels_part,
+                            nullptr, c.rl, 0.0f);
       }
       bits_stale = true;
     } else {

@@ -4,6 +4,7 @@
 #include <cassert>
 #include <cmath>
 #include <cstdio>
+#include <cstdlib>
 #include <cstring>
 
 namespace lux {
@@ -97,6 +98,135 @@ void Bins::build(const E_ID* row_ptr_loc, V_ID vp, E_ID ep,
   nbig = c[3];
 }
 
+// ---------------- src-blocked pull (BlockedPull) ----------------
+
+static int blocked_shift(V_ID nv) {
+  // LUX_NATIVE_BLOCK_SHIFT forces blocking at any nv (tests); default:
+  // 32 MB windows, only when the gather window exceeds the 256 MiB LLC
+  if (const char* e = getenv("LUX_NATIVE_BLOCK_SHIFT")) return atoi(e);
+  return (uint64_t)nv * 4 > (256ull << 20) ? 23 : 0;
+}
+
+size_t BlockedPull::arena_bytes(V_ID nv, V_ID vp, E_ID ep) {
+  int shift = blocked_shift(nv);
+  if (!shift || ep == 0) return 0;
+  uint64_t sb = ((uint64_t)nv + (1u << shift) - 1) >> shift;
+  // blk_col + per-block u32 row tables + compacted bins (each (block,row)
+  // pair with edges lands in exactly one bin list => <= ep entries) +
+  // bin2 chunk lists + slack
+  return 4ull * ep + sb * 4ull * (vp + 1) + 5ull * ep + (64ull << 20);
+}
+
+void BlockedPull::build(const E_ID* row_ptr_loc, const V_ID* col, V_ID vp_,
+                        E_ID ep, V_ID nv, DeviceArena& arena,
+                        hipStream_t s) {
+  vp = vp_;
+  int shift = blocked_shift(nv);
+  if (!shift || ep == 0 || vp == 0) return;
+  std::vector<V_ID> hb;
+  for (uint64_t b = 0; b < nv; b += (1u << shift)) hb.push_back((V_ID)b);
+  hb.push_back(nv);
+  int sb = (int)hb.size() - 1;
+  if (sb <= 1) return;
+  V_ID* bounds;
+  LUX_OK(hipMalloc(&bounds, sizeof(V_ID) * hb.size()));
+  LUX_OK(hipMemcpyAsync(bounds, hb.data(), sizeof(V_ID) * hb.size(),
+                        hipMemcpyHostToDevice, s));
+  uint64_t n = (uint64_t)sb * vp;
+  uint32_t* counts;
+  unsigned long long *cursor, *partials;
+  LUX_OK(hipMalloc(&counts, 4ull * n));
+  LUX_OK(hipMalloc(&cursor, 8ull * (n + 1)));
+  LUX_OK(hipMalloc(&partials,
+                   8ull * lux_gpu_scan_partials_size(n)));
+  LUX_OK(hipMemsetAsync(counts, 0, 4ull * n, s));
+  LUX_OK(hipMemsetAsync(cursor, 0, 8, s));
+  lux_gpu_blocked_count((uint64_t)s, ep, col, row_ptr_loc, vp, bounds,
+                        sb + 1, counts);
+  lux_gpu_scan_end_offsets((uint64_t)s, n, counts, (E_ID*)cursor + 1,
+                           partials);
+  V_ID* blk_col = arena.alloc_n<V_ID>(ep);
+  lux_gpu_blocked_scatter((uint64_t)s, ep, col, row_ptr_loc, vp, bounds,
+                          sb + 1, cursor, blk_col);
+  // post-scatter, cursor[i] == end offset of slot i
+  E_ID* row64;
+  LUX_OK(hipMalloc(&row64, 8ull * (vp + 1)));
+  // temp (uncompacted) bin buffers reused across blocks
+  V_ID *t0, *t1, *t2v;
+  lux_uint2* t2;
+  uint32_t* counters;
+  uint32_t nbig_max = (uint32_t)std::min<uint64_t>(vp, ep / 2048 + 1);
+  uint32_t n2_max = (uint32_t)(ep / 8192 + nbig_max + 1);
+  LUX_OK(hipMalloc(&t0, 4ull * vp));
+  LUX_OK(hipMalloc(&t1, 4ull * vp));
+  LUX_OK(hipMalloc(&t2, 8ull * n2_max));
+  LUX_OK(hipMalloc(&t2v, 4ull * nbig_max));
+  LUX_OK(hipMalloc(&counters, 16));
+  uint64_t begin = 0;
+  for (int b = 0; b < sb; b++) {
+    // post-scatter, cursor[i] (i < n) holds slot i's END offset
+    unsigned long long end;
+    LUX_OK(hipMemcpyAsync(&end, cursor + ((uint64_t)(b + 1) * vp - 1), 8,
+                          hipMemcpyDeviceToHost, s));
+    LUX_OK(hipStreamSynchronize(s));
+    if (end == begin) continue;
+    lux_gpu_local_row_ptr((uint64_t)s, vp, begin,
+                          (E_ID*)(cursor + (uint64_t)b * vp), row64);
+    LUX_OK(hipMemsetAsync(counters, 0, 16, s));
+    lux_gpu_build_bins((uint64_t)s, vp, row64, t0, t1, t2, t2v, counters);
+    uint32_t hc[4];
+    LUX_OK(hipMemcpyAsync(hc, counters, 16, hipMemcpyDeviceToHost, s));
+    LUX_OK(hipStreamSynchronize(s));
+    Blk blk;
+    blk.n0 = hc[0];
+    blk.n1 = hc[1];
+    blk.n2 = hc[2];
+    blk.nbig = hc[3];
+    blk.bin0 = arena.alloc_n<V_ID>(hc[0] ? hc[0] : 1);
+    blk.bin1 = arena.alloc_n<V_ID>(hc[1] ? hc[1] : 1);
+    blk.bin2 = arena.alloc_n<lux_uint2>(hc[2] ? hc[2] : 1);
+    blk.bin2v = arena.alloc_n<V_ID>(hc[3] ? hc[3] : 1);
+    if (hc[0])
+      LUX_OK(hipMemcpyAsync(blk.bin0, t0, 4ull * hc[0],
+                            hipMemcpyDeviceToDevice, s));
+    if (hc[1])
+      LUX_OK(hipMemcpyAsync(blk.bin1, t1, 4ull * hc[1],
+                            hipMemcpyDeviceToDevice, s));
+    if (hc[2])
+      LUX_OK(hipMemcpyAsync(blk.bin2, t2, 8ull * hc[2],
+                            hipMemcpyDeviceToDevice, s));
+    if (hc[3])
+      LUX_OK(hipMemcpyAsync(blk.bin2v, t2v, 4ull * hc[3],
+                            hipMemcpyDeviceToDevice, s));
+    blk.row32 = arena.alloc_n<uint32_t>(vp + 1);
+    lux_gpu_u64_to_u32((uint64_t)s, (uint64_t)vp + 1,
+                       (const unsigned long long*)row64, blk.row32);
+    blk.col = blk_col + begin;
+    blocks.push_back(blk);
+    begin = end;
+  }
+  LUX_OK(hipStreamSynchronize(s));
+  hipFree(bounds);
+  hipFree(counts);
+  hipFree(cursor);
+  hipFree(partials);
+  hipFree(row64);
+  hipFree(t0);
+  hipFree(t1);
+  hipFree(t2);
+  hipFree(t2v);
+  hipFree(counters);
+}
+
+void BlockedPull::sweep(int mode, const void* oldv, void* newv,
+                        const V_ID* deg, V_ID row_left, float init_rank,
+                        hipStream_t s) const {
+  for (const Blk& b : blocks)
+    lux_gpu_pull_iter((uint64_t)s, mode, b.n0, b.bin0, b.n1, b.bin1, b.n2,
+                      b.bin2, b.nbig, b.bin2v, b.row32, 1, b.col, oldv,
+                      newv, deg, row_left, init_rank);
+}
+
 // ---------------- PageRank ----------------
 
 SingleGpuPagerank::SingleGpuPagerank(const DeviceGraph& g, DeviceArena& arena,
@@ -105,6 +235,7 @@ SingleGpuPagerank::SingleGpuPagerank(const DeviceGraph& g, DeviceArena& arena,
   row_ptr_ = arena.alloc_n<E_ID>(g.nv + 1);
   lux_gpu_local_row_ptr((uint64_t)s, g.nv, 0, g.col_end, row_ptr_);
   bins_.build(row_ptr_, g.nv, g.ne, arena, s);
+  blocked_.build(row_ptr_, g.src, g.nv, g.ne, g.nv, arena, s);
   deg_ = arena.alloc_n<V_ID>(g.nv);
   LUX_OK(hipMemsetAsync(deg_, 0, sizeof(V_ID) * g.nv, s));
   lux_gpu_hist_u32((uint64_t)s, g.ne, g.src, deg_);
@@ -128,10 +259,13 @@ void SingleGpuPagerank::iterate(int iters) {
   float init_rank = (1.0f - PR_ALPHA) / g_.nv;
   for (int it = 0; it < iters; it++) {
     LUX_OK(hipMemsetAsync(new_, 0, sizeof(float) * g_.nv, s_));
-    lux_gpu_pull_iter((uint64_t)s_, 0, bins_.n0, bins_.bin0, bins_.n1,
-                      bins_.bin1, bins_.n2, bins_.bin2, bins_.nbig,
-                      bins_.bin2v, row_ptr_, 0, g_.src, old_, new_, deg_, 0,
-                      init_rank);
+    if (blocked_.active())
+      blocked_.sweep(0, old_, new_, deg_, 0, init_rank, s_);
+    else
+      lux_gpu_pull_iter((uint64_t)s_, 0, bins_.n0, bins_.bin0, bins_.n1,
+                        bins_.bin1, bins_.n2, bins_.bin2, bins_.nbig,
+                        bins_.bin2v, row_ptr_, 0, g_.src, old_, new_, deg_,
+                        0, init_rank);
     lux_gpu_pull_finish_pr((uint64_t)s_, g_.nv, new_, deg_, 0, init_rank);
     std::swap(old_, new_);
   }
@@ -146,6 +280,8 @@ SingleGpuPush::SingleGpuPush(const DeviceGraph& g, bool is_min, V_ID source,
   row_ptr_ = arena.alloc_n<E_ID>(g.nv + 1);
   lux_gpu_local_row_ptr((uint64_t)s, g.nv, 0, g.col_end, row_ptr_);
   bins_.build(row_ptr_, g.nv, g.ne, arena, s);
+  blocked_.build(row_ptr_, g.src, g.nv, g.ne, g.nv, arena, s);
+  if (is_min) bits_ = arena.alloc_n<uint32_t>((g.nv + 31) / 32 + 1);
   // push CSR (single partition: transpose over all nv)
   push_row_ptr_ = arena.alloc_n<E_ID>(g.nv + 1);
   push_col_ = arena.alloc_n<V_ID>(g.ne);
@@ -216,24 +352,56 @@ V_ID SingleGpuPush::step() {
   bool new_dense = fq_type_ == FrontierHeader::DENSE_BITMAP;
   bool pull_fallback = fq_num_ > g_.nv / SPARSE_THRESHOLD || force_pull_;
   force_pull_ = false;
+  auto run_pull = [&]() {
+    // dense pull iteration; labels_ serves as both old (all) and new
+    // (slice); src-blocked sweeps when the gather window exceeds the LLC
+    if (blocked_.active())
+      blocked_.sweep(is_min_ ? 1 : 2, snapshot_, labels_, nullptr, 0, 0.0f,
+                     s_);
+    else
+      lux_gpu_pull_iter((uint64_t)s_, is_min_ ? 1 : 2, bins_.n0, bins_.bin0,
+                        bins_.n1, bins_.bin1, bins_.n2, bins_.bin2,
+                        bins_.nbig, bins_.bin2v, row_ptr_, 0, g_.src,
+                        snapshot_, labels_, nullptr, 0, 0.0f);
+    bits_stale_ = true;
+  };
   if (pull_fallback) {
     new_dense = true;
-    // dense pull iteration; labels_ serves as both old (all) and new (slice)
-    lux_gpu_pull_iter((uint64_t)s_, is_min_ ? 1 : 2, bins_.n0, bins_.bin0,
-                      bins_.n1, bins_.bin1, bins_.n2, bins_.bin2, bins_.nbig,
-                      bins_.bin2v, row_ptr_, 0, g_.src, snapshot_, labels_,
-                      nullptr, 0, 0.0f);
+    run_pull();
   } else {
     LUX_OK(hipMemsetAsync(item_counter_, 0, 16, s_));
     lux_gpu_frontier_expand(
         (uint64_t)s_, fq_type_ == FrontierHeader::DENSE_BITMAP ? 1 : 0, 0,
         fq_type_ == FrontierHeader::DENSE_BITMAP ? g_.nv : fq_num_, fq_,
         nullptr, nullptr, push_row_ptr_, items_, item_counter_, max_items_);
-    lux_gpu_push_chunk_scatter((uint64_t)s_, is_min_ ? 1 : 0,
-                               new_dense ? 1 : 0, items_, item_counter_,
-                               max_items_, push_row_ptr_, push_col_,
-                               snapshot_, snapshot_, labels_, 0, new_fq_,
-                               capacity_, nullptr);
+    // second adaptivity axis (python engine parity): the frontier's
+    // out-edge volume, counted by the expand kernel (u32 — exact for
+    // ne < 2^32), decides push vs a dense pull sweep
+    uint32_t hc[4];
+    LUX_OK(hipMemcpyAsync(hc, item_counter_, 16, hipMemcpyDeviceToHost,
+                          s_));
+    LUX_OK(hipStreamSynchronize(s_));
+    uint64_t thresh = (is_min_ && bits_) ? g_.ne / 2 : g_.ne / 8;
+    if (g_.ne < (1ull << 32) && hc[1] > thresh) {
+      pull_fallback = true;
+      new_dense = true;
+      run_pull();
+    } else {
+      if ((uint64_t)hc[1] / 16 > capacity_) new_dense = true;
+      uint32_t* bits = nullptr;
+      if (is_min_ && bits_) {
+        if (bits_stale_) {
+          lux_gpu_bits_from_labels((uint64_t)s_, g_.nv, labels_, bits_);
+          bits_stale_ = false;
+        }
+        bits = bits_;
+      }
+      lux_gpu_push_chunk_scatter((uint64_t)s_, is_min_ ? 1 : 0,
+                                 new_dense ? 1 : 0, items_, item_counter_,
+                                 max_items_, push_row_ptr_, push_col_,
+                                 snapshot_, snapshot_, labels_, 0, new_fq_,
+                                 capacity_, bits);
+    }
   }
   FrontierHeader hh;
   if (new_dense) {

@@ -69,6 +69,31 @@ struct Bins {
              hipStream_t s);
 };
 
+// src-blocked CSC + per-block compacted bins — the native twin of
+// engine.py build_blocked: regroups edges by 32 MB src window so the
+// random old-property gather stays Infinity-Cache-resident, with
+// u32 block-local row offsets (half the per-row sweep traffic).
+// Built only when the gather window exceeds the LLC (nv*4 B > 256 MiB).
+struct BlockedPull {
+  struct Blk {
+    uint32_t* row32;
+    V_ID* col;
+    uint32_t n0, n1, n2, nbig;
+    V_ID *bin0, *bin1, *bin2v;
+    lux_uint2* bin2;
+  };
+  std::vector<Blk> blocks;
+  V_ID vp = 0;
+  bool active() const { return !blocks.empty(); }
+  static size_t arena_bytes(V_ID nv, V_ID vp, E_ID ep);
+  void build(const E_ID* row_ptr_loc, const V_ID* col, V_ID vp, E_ID ep,
+             V_ID nv, DeviceArena& arena, hipStream_t s);
+  // one fold-sweep over every block (same iteration contract as the
+  // unblocked pull: newv pre-seeded, PR epilogue applied by the caller)
+  void sweep(int mode, const void* oldv, void* newv, const V_ID* deg,
+             V_ID row_left, float init_rank, hipStream_t s) const;
+};
+
 class SingleGpuPagerank {
  public:
   SingleGpuPagerank(const DeviceGraph& g, DeviceArena& arena, hipStream_t s);
@@ -80,6 +105,7 @@ class SingleGpuPagerank {
   hipStream_t s_;
   E_ID* row_ptr_;
   Bins bins_;
+  BlockedPull blocked_;
   V_ID* deg_;
   float *old_, *new_;
 };
@@ -101,6 +127,9 @@ class SingleGpuPush {
   E_ID *row_ptr_, *push_row_ptr_;
   V_ID* push_col_;
   Bins bins_;
+  BlockedPull blocked_;
+  uint32_t* bits_ = nullptr;  // BFS visited bitmap (is_min fast path)
+  bool bits_stale_ = true;
   uint32_t *labels_, *snapshot_;
   uint8_t *fq_, *new_fq_, *tmp_fq_;
   lux_uint2* items_;
