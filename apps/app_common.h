@@ -74,6 +74,36 @@ inline void dump_state(const char* path, const void* dev_ptr, int dtype,
   printf("[lux] wrote %s (nv=%u k=%u)\n", path, nv, k);
 }
 
+// Parse -synthetic without building: returns 0 (not synthetic / unknown),
+// 1 (rmat:S:NE), 2 (bipartite:NU:NI:NE); fills the shape outputs.
+inline int parse_synthetic(const char* spec, int* scale, long long* nu,
+                           long long* ni, long long* ne) {
+  if (!spec) return 0;
+  if (sscanf(spec, "rmat:%d:%lld", scale, ne) == 2) return 1;
+  if (sscanf(spec, "bipartite:%lld:%lld:%lld", nu, ni, ne) == 3) return 2;
+  return 0;
+}
+
+// Device-side synthetic build for the single-GPU drivers (the CPU
+// generator at RMAT-27 takes minutes; the GPU one milliseconds). Returns
+// true and fills *dg when -synthetic was used.
+inline bool build_synthetic_device(const AppArgs& a, lux::DeviceArena& arena,
+                                   hipStream_t s, lux::DeviceGraph* dg) {
+  int scale = 0;
+  long long nu = 0, ni = 0, ne = 0;
+  int kind = parse_synthetic(a.synthetic, &scale, &nu, &ni, &ne);
+  if (kind == 1) {
+    *dg = lux::DeviceGraph::rmat(scale, (lux::E_ID)ne, 1, arena, s);
+    return true;
+  }
+  if (kind == 2) {
+    *dg = lux::DeviceGraph::bipartite((lux::V_ID)nu, (lux::V_ID)ni,
+                                      (lux::E_ID)ne, 1, arena, s);
+    return true;
+  }
+  return false;
+}
+
 inline bool load_graph(const AppArgs& a, lux::HostCSC* g, bool weighted) {
   if (a.file) return lux::lux_read(a.file, g, weighted);
   if (a.synthetic) {

@@ -32,9 +32,25 @@ int main(int argc, char** argv) {
     // native fork+exec + RCCL engine, one worker per GPU
     return run_multi_workers(a.num_gpu, argc, argv);
   }
+  int sscale = 0;
+  long long snu = 0, sni = 0, sne = 0;
+  int skind = parse_synthetic(a.synthetic, &sscale, &snu, &sni, &sne);
+  if (skind == 1) {
+    fprintf(stderr, "[lux] col_filter needs a weighted graph: use "
+                    "-synthetic bipartite:NU:NI:NE\n");
+    return 1;
+  }
   HostCSC g;
-  if (!load_graph(a, &g, true)) return 1;
-  print_memory_estimate(g.nv, g.ne, true, a.k);
+  uint64_t NV = 0, NE = 0;
+  if (skind == 2) {
+    NV = (uint64_t)(snu + sni);
+    NE = (uint64_t)sne;
+  } else {
+    if (!load_graph(a, &g, true)) return 1;
+    NV = g.nv;
+    NE = g.ne;
+  }
+  print_memory_estimate((V_ID)NV, NE, true, a.k);
 
   hipStream_t s;
   LUX_OK(hipStreamCreate(&s));
@@ -42,15 +58,20 @@ int main(int argc, char** argv) {
     fprintf(stderr, "[lux] -als covers K <= 64 (MFMA tile grid)\n");
     return 1;
   }
-  size_t arena_bytes = 8ull * g.nv + 8ull * g.ne          // graph + weights
-                       + 8ull * (g.nv + 1)
-                       + 12ull * g.nv + (64ull << 20)
-                       + 8ull * (uint64_t)g.nv * a.k      // old/new vectors
-                       + 8ull * (g.ne / 8192 + g.nv / 16)
-                       + (a.als ? 4ull * g.nv +           // hub slot map
-                            (17ull << 20) * (g.ne / 2048 / 1024 + 1) : 0);
+  size_t arena_bytes = 8ull * NV + 8ull * NE              // graph + weights
+                       + 8ull * (NV + 1)
+                       + 12ull * NV + (64ull << 20)
+                       + 8ull * NV * a.k                  // old/new vectors
+                       + 8ull * (NE / 8192 + NV / 16)
+                       + (a.als ? 4ull * NV +             // hub slot map
+                            (17ull << 20) * (NE / 2048 / 1024 + 1) : 0);
   DeviceArena arena(arena_bytes);
-  DeviceGraph dg = DeviceGraph::upload(g, arena, s);
+  DeviceGraph dg;
+  if (skind) {
+    build_synthetic_device(a, arena, s, &dg);
+  } else {
+    dg = DeviceGraph::upload(g, arena, s);
+  }
   SingleGpuCF engine(dg, a.k, arena, s, a.als);
 
   auto t0 = std::chrono::steady_clock::now();
@@ -59,8 +80,8 @@ int main(int argc, char** argv) {
   double secs = std::chrono::duration<double>(t1 - t0).count();
   printf("ELAPSED TIME = %7.7f s\n", secs);
   if (a.dump)
-    dump_state(a.dump, engine.vectors(), 0, (uint32_t)a.k, g.nv, (uint64_t)a.num_iter);
+    dump_state(a.dump, engine.vectors(), 0, (uint32_t)a.k, dg.nv, (uint64_t)a.num_iter);
   printf("[lux] %.3f GTEPS (%d sweeps, rank %d)\n",
-         double(g.ne) * a.num_iter / secs / 1e9, a.num_iter, a.k);
+         double(dg.ne) * a.num_iter / secs / 1e9, a.num_iter, a.k);
   return 0;
 }

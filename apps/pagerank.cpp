@@ -29,21 +29,41 @@ int main(int argc, char** argv) {
     // machinery is testable on a 1-GPU box.
     return run_pagerank_multi(a.num_gpu, argc, argv);
   }
+  int sscale = 0;
+  long long snu = 0, sni = 0, sne = 0;
+  int skind = parse_synthetic(a.synthetic, &sscale, &snu, &sni, &sne);
   HostCSC g;
-  if (!load_graph(a, &g, false)) return 1;
-  print_memory_estimate(g.nv, g.ne, false, 1);
+  uint64_t NV = 0, NE = 0;
+  if (skind == 1) {
+    NV = 1ull << sscale;
+    NE = (uint64_t)sne;
+  } else if (skind == 2) {
+    NV = (uint64_t)(snu + sni);
+    NE = (uint64_t)sne;
+  } else {
+    if (!load_graph(a, &g, false)) return 1;
+    NV = g.nv;
+    NE = g.ne;
+  }
+  print_memory_estimate((V_ID)NV, NE, false, 1);
 
   hipStream_t s;
   LUX_OK(hipStreamCreate(&s));
-  size_t arena_bytes = 8ull * g.nv + 4ull * g.ne        // graph
-                       + 8ull * (g.nv + 1)              // row_ptr
-                       + 12ull * g.nv + (64ull << 20)   // bins + slack
-                       + 4ull * g.nv                    // degrees
-                       + 8ull * g.nv                    // old/new
-                       + 8ull * (g.ne / 8192 + g.nv / 16)
-                       + BlockedPull::arena_bytes(g.nv, g.nv, g.ne);
+  size_t arena_bytes = 8ull * NV + 4ull * NE            // graph
+                       + 8ull * (NV + 1)                // row_ptr
+                       + 12ull * NV + (64ull << 20)     // bins + slack
+                       + 4ull * NV                      // degrees
+                       + 8ull * NV                      // old/new
+                       + 8ull * (NE / 8192 + NV / 16)
+                       + BlockedPull::arena_bytes((V_ID)NV, (V_ID)NV, NE);
   DeviceArena arena(arena_bytes);
-  DeviceGraph dg = DeviceGraph::upload(g, arena, s);
+  DeviceGraph dg;
+  if (skind) {
+    // synthetic graphs build on-device (CPU gen at RMAT-27 takes minutes)
+    build_synthetic_device(a, arena, s, &dg);
+  } else {
+    dg = DeviceGraph::upload(g, arena, s);
+  }
   SingleGpuPagerank engine(dg, arena, s);
 
   auto t0 = std::chrono::steady_clock::now();
@@ -52,10 +72,10 @@ int main(int argc, char** argv) {
   double secs = std::chrono::duration<double>(t1 - t0).count();
   printf("ELAPSED TIME = %7.7f s\n", secs);
   if (a.dump)
-    dump_state(a.dump, engine.ranks(), 0, 1, g.nv, (uint64_t)a.num_iter);
+    dump_state(a.dump, engine.ranks(), 0, 1, dg.nv, (uint64_t)a.num_iter);
   printf("[lux] %.3f GTEPS (%d iterations, %llu edges)\n",
-         double(g.ne) * a.num_iter / secs / 1e9, a.num_iter,
-         (unsigned long long)g.ne);
+         double(dg.ne) * a.num_iter / secs / 1e9, a.num_iter,
+         (unsigned long long)dg.ne);
   if (a.verbose) {
     float first[5];
     LUX_OK(hipMemcpy(first, engine.ranks(), sizeof(first),

@@ -27,22 +27,45 @@ int main(int argc, char** argv) {
     // native fork+exec + RCCL push engine, one worker per GPU
     return run_multi_workers(a.num_gpu, argc, argv);
   }
+  int sscale = 0;
+  long long snu = 0, sni = 0, sne = 0;
+  int skind = parse_synthetic(a.synthetic, &sscale, &snu, &sni, &sne);
   HostCSC g;
-  if (!load_graph(a, &g, false)) return 1;
-  print_memory_estimate(g.nv, g.ne, false, 1);
+  uint64_t NV, NE;
+  if (skind == 1) {
+    NV = 1ull << sscale;
+    NE = (uint64_t)sne;
+  } else if (skind == 2) {
+    NV = (uint64_t)(snu + sni);
+    NE = (uint64_t)sne;
+  } else {
+    skind = 0;
+  }
+  if (!skind) {
+    if (!load_graph(a, &g, false)) return 1;
+    NV = g.nv;
+    NE = g.ne;
+  }
+  print_memory_estimate((V_ID)NV, NE, false, 1);
 
   hipStream_t s;
   LUX_OK(hipStreamCreate(&s));
-  size_t arena_bytes = 8ull * g.nv + 4ull * g.ne          // graph
-                       + 16ull * (g.nv + 1) + 4ull * g.ne // csrs
-                       + 12ull * g.nv + (64ull << 20)     // bins + slack
-                       + 8ull * g.nv                      // labels+snapshot
-                       + 3ull * frontier_bytes(g.nv)
-                       + 8ull * (g.ne / 8192 + g.nv / 16)
-                       + BlockedPull::arena_bytes(g.nv, g.nv, g.ne)
-                       + g.nv / 2;  // BFS bits
+  size_t arena_bytes = 8ull * NV + 4ull * NE              // graph
+                       + 16ull * (NV + 1) + 4ull * NE     // csrs
+                       + 12ull * NV + (64ull << 20)       // bins + slack
+                       + 8ull * NV                        // labels+snapshot
+                       + 3ull * frontier_bytes((V_ID)NV)
+                       + 8ull * (NE / 8192 + NV / 16)
+                       + BlockedPull::arena_bytes((V_ID)NV, (V_ID)NV, NE)
+                       + NV / 2;  // BFS bits
   DeviceArena arena(arena_bytes);
-  DeviceGraph dg = DeviceGraph::upload(g, arena, s);
+  DeviceGraph dg;
+  if (skind) {
+    // synthetic graphs build on-device (CPU gen at RMAT-27 takes minutes)
+    build_synthetic_device(a, arena, s, &dg);
+  } else {
+    dg = DeviceGraph::upload(g, arena, s);
+  }
   SingleGpuPush engine(dg, /*is_min=*/true, a.start, arena, s, a.verbose);
 
   auto t0 = std::chrono::steady_clock::now();
@@ -51,9 +74,9 @@ int main(int argc, char** argv) {
   double secs = std::chrono::duration<double>(t1 - t0).count();
   printf("ELAPSED TIME = %7.7f s\n", secs);
   if (a.dump)
-    dump_state(a.dump, engine.labels(), 1, 1, g.nv, (uint64_t)a.num_iter);
+    dump_state(a.dump, engine.labels(), 1, 1, dg.nv, (uint64_t)a.num_iter);
   printf("[lux] converged in %d iterations, %.3f GTEPS\n", iters,
-         double(g.ne) / secs / 1e9);
+         double(dg.ne) / secs / 1e9);
   if (a.check) {
     uint64_t mistakes = engine.check();
     printf("[%s] %llu mistakes\n", mistakes == 0 ? "PASS" : "FAIL",

@@ -75,6 +75,41 @@ DeviceGraph DeviceGraph::rmat(int scale, E_ID ne, uint64_t seed,
   return d;
 }
 
+DeviceGraph DeviceGraph::bipartite(V_ID n_users, V_ID n_items, E_ID ne,
+                                   uint64_t seed, DeviceArena& arena,
+                                   hipStream_t s) {
+  DeviceGraph d;
+  d.nv = n_users + n_items;
+  d.ne = ne;
+  d.col_end = arena.alloc_n<E_ID>(d.nv);
+  d.src = arena.alloc_n<V_ID>(ne);
+  d.weight = arena.alloc_n<WeightType>(ne);
+  V_ID *esrc, *edst;
+  WeightType* ew;
+  uint32_t* hist;
+  unsigned long long *cursor, *partials;
+  LUX_OK(hipMalloc(&esrc, sizeof(V_ID) * ne));
+  LUX_OK(hipMalloc(&edst, sizeof(V_ID) * ne));
+  LUX_OK(hipMalloc(&ew, sizeof(WeightType) * ne));
+  LUX_OK(hipMalloc(&hist, sizeof(uint32_t) * d.nv));
+  LUX_OK(hipMalloc(&cursor, sizeof(uint64_t) * d.nv));
+  LUX_OK(hipMalloc(&partials,
+                   sizeof(uint64_t) * lux_gpu_scan_partials_size(d.nv)));
+  LUX_OK(hipMemsetAsync(hist, 0, sizeof(uint32_t) * d.nv, s));
+  lux_gpu_bipartite_edges((uint64_t)s, seed, n_users, n_items, ne, esrc,
+                          edst, ew);
+  lux_gpu_edges_to_csc((uint64_t)s, d.nv, ne, esrc, edst, ew, d.col_end,
+                       d.src, d.weight, hist, cursor, partials);
+  LUX_OK(hipStreamSynchronize(s));
+  hipFree(esrc);
+  hipFree(edst);
+  hipFree(ew);
+  hipFree(hist);
+  hipFree(cursor);
+  hipFree(partials);
+  return d;
+}
+
 void Bins::build(const E_ID* row_ptr_loc, V_ID vp, E_ID ep,
                  DeviceArena& arena, hipStream_t s) {
   bin0 = arena.alloc_n<V_ID>(vp);

@@ -58,6 +58,9 @@ struct DeviceGraph {
   // Synthetic generation fully on-device.
   static DeviceGraph rmat(int scale, E_ID ne, uint64_t seed,
                           DeviceArena& arena, hipStream_t s);
+  static DeviceGraph bipartite(V_ID n_users, V_ID n_items, E_ID ne,
+                               uint64_t seed, DeviceArena& arena,
+                               hipStream_t s);
 };
 
 // Degree bins shared by the pull/CF engines (built once; see pull.hip).
