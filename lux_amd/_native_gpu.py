@@ -207,6 +207,15 @@ def frontier_expand(stream, old_dense, in_row_left, in_count, old_seg,
         dp(push_row_ptr), dp(items), dp(counter), _u32(max_items))
 
 
+def frontier_expand_auto(stream, verts, in_row_left, seg, qlabels,
+                         labels_repair, push_row_ptr, items, counter,
+                         max_items):
+    lib().lux_gpu_frontier_expand_auto(
+        _u64(stream), _u32(verts), _u32(in_row_left), dp(seg), dp(qlabels),
+        dp(labels_repair), dp(push_row_ptr), dp(items), dp(counter),
+        _u32(max_items))
+
+
 def frontier_fixup(stream, vp, row_left, capacity, built_dense, snapshot,
                    labels_part, deg_part, new_seg, annex, tmp_seg, meta,
                    item_counter, max_items):

@@ -162,11 +162,15 @@ class PushEngine:
         self.labels = torch.empty(p.nv, dtype=U32, device=device)
         self.labels_part = torch.empty(p.vp, dtype=U32, device=device)
         self.snapshot = torch.empty_like(self.labels_part)
-        # hipGraph for the pull-fallback body (~50 fixed-address ctypes
-        # launches: seed + blocked sweeps + bitmap + fixup); captured on
-        # 2nd use, single-rank only (multi-rank syncs labels first)
-        self._pull_graph = None
-        self._pull_graph_uses = 0
+        # hipGraphs for the single-rank iteration bodies (pull / push
+        # sparse-out / push dense-out): at world 1 every launch dimension
+        # is static (frontier_expand_auto sizes itself from the segment
+        # header on device), so a whole iteration replays as ONE graph
+        # launch + one 32-byte meta read — removing the ~12-ctypes-call
+        # host gap per iteration that made the python engine ~10% slower
+        # than the native C++ runtime
+        self._graphs = {}
+        self._graph_uses = {}
         self._fq_init = None
         self.reset(source)
 
@@ -244,43 +248,111 @@ class PushEngine:
         self._sync_labels()
         return self.labels
 
-    def _pull_body_eager(self):
+    def _single_body(self, pull, new_dense):
+        """One complete world-1 iteration body: compute + fixups + the
+        device-side "exchange" (full-size self copies + meta). Every op
+        has static shapes and fixed addresses -> hipGraph-capturable."""
         p = self.part
         s = _stream()
-        mode = ng.PULL_MIN if self.is_min else ng.PULL_MAX
-        run_pull(p, mode, self.labels, self.labels_part, None, 0.0)
-        ng.build_bitmap(s, p.vp, self.snapshot, self.labels_part,
-                        self.new_seg)
-        ng.frontier_fixup(s, p.vp, p.row_left, self.capacity, 1,
-                          self.snapshot, self.labels_part, self.deg_part,
-                          self.new_seg, self.new_annex, self.tmp_seg,
-                          self.meta_mine, None, self.max_items)
+        self.snapshot.copy_(self.labels_part)
+        self._my_seg_i32()[0] = 0
+        self._my_seg_i32()[1] = 0
+        if pull:
+            mode = ng.PULL_MIN if self.is_min else ng.PULL_MAX
+            run_pull(p, mode, self.labels, self.labels_part, None, 0.0)
+            counter = None
+        else:
+            self.item_counter.zero_()
+            ng.frontier_expand_auto(s, p.vp, p.row_left, self.fq_all, None,
+                                    None, self.push_row_ptr, self.items,
+                                    self.item_counter, self.max_items)
+            bits = self.visited if self.visited is not None and p.vp > 0 \
+                else None
+            ng.push_chunk_scatter(s, int(self.is_min), int(new_dense),
+                                  self.items, self.item_counter,
+                                  self.max_items, self.push_row_ptr,
+                                  self.push_col, self.labels, self.snapshot,
+                                  self.labels_part, p.row_left,
+                                  self.new_seg, self.capacity,
+                                  visited_bits=bits)
+            counter = self.item_counter
+        if pull or new_dense:
+            ng.build_bitmap(s, p.vp, self.snapshot, self.labels_part,
+                            self.new_seg)
+        ng.frontier_fixup(s, p.vp, p.row_left, self.capacity,
+                          int(pull or new_dense), self.snapshot,
+                          self.labels_part, self.deg_part, self.new_seg,
+                          self.new_annex, self.tmp_seg, self.meta_mine,
+                          counter, self.max_items)
+        # world-1 "exchange": fixed-size self copies keep fq_all and the
+        # replicated labels current every iteration (labels never stale)
+        self.fq_all.copy_(self.new_seg)
+        self.labels.narrow(0, p.row_left, p.vp).copy_(self.labels_part)
 
-    def _run_pull_body(self):
-        """Pull fallback iteration body (sweep + bitmap + device fixups).
-        At world 1 it is a fixed ~50-launch sequence over fixed addresses:
-        captured into a hipGraph on its 2nd use and replayed (one launch
-        instead of ~50 ctypes calls — the same lever as PagerankEngine's
-        single-GPU capture)."""
+    def _run_single_body(self, pull, new_dense):
         import os
-        if self._pull_graph is not None:
-            self._pull_graph.replay()
+        key = ("pull",) if pull else ("push", bool(new_dense))
+        g = self._graphs.get(key)
+        if g is not None:
+            g.replay()
             return
-        self._pull_graph_uses += 1
-        if (dx.world_size() == 1 and self._pull_graph_uses >= 2
-                and os.environ.get("LUX_HIPGRAPH", "1") == "1"
+        uses = self._graph_uses.get(key, 0) + 1
+        self._graph_uses[key] = uses
+        if (uses >= 2 and os.environ.get("LUX_HIPGRAPH", "1") == "1"
                 and not os.environ.get("LUX_SYNC_CHECK")):
             try:
-                g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g):
-                    self._pull_body_eager()
-                self._pull_graph = g
-                self._pull_graph.replay()  # capture records, doesn't run
+                cg = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(cg):
+                    self._single_body(pull, new_dense)
+                self._graphs[key] = cg
+                cg.replay()  # capture records, doesn't run
                 return
             except Exception as e:  # capture unsupported: stay eager
-                print(f"[lux] push pull-body hipGraph disabled: {e}")
+                print(f"[lux] push hipGraph capture disabled: {e}")
                 os.environ["LUX_HIPGRAPH"] = "0"
-        self._pull_body_eager()
+        self._single_body(pull, new_dense)
+
+    def _step_single(self):
+        """World-1 fast path: replay the captured iteration graph, then
+        ONE 32-byte meta read."""
+        p = self.part
+        mh = self.meta_host
+        evol = int((mh[:, 2].astype(np.uint64)
+                    | (mh[:, 3].astype(np.uint64) << np.uint64(32))).sum())
+        overflow = bool(mh[:, 4].any())
+        old_fq_size = int(mh[:, 1].sum())
+        new_dense = int(mh[0, 0]) == DENSE_BITMAP
+        pull = overflow or old_fq_size > p.nv // 16
+        if overflow:
+            print("[lux] frontier expand overflow: recovering with a "
+                  "forced pull iteration", file=sys.stderr)
+        if not pull:
+            import os
+            div = int(os.environ.get("LUX_PUSH_EVOL_DIV", "0") or 0)
+            thresh = p.ne // div if div else (
+                p.ne // 2 if self.visited is not None else p.ne // 8)
+            if evol > thresh:
+                pull = True
+            elif (evol * p.ep // max(p.ne, 1)) // 16 > self.capacity:
+                new_dense = True
+        if not pull and self.visited is not None and p.vp > 0 \
+                and self._bits_stale:
+            ng.bits_from_labels(_stream(), p.vp, self.labels_part,
+                                self.visited)
+            self._bits_stale = False
+        self._run_single_body(pull, new_dense)
+        if pull:
+            self._bits_stale = True
+        mh = self.meta_mine.cpu().numpy().view(np.uint32).reshape(1, 8)
+        self.meta_host = mh
+        self.headers = [(int(mh[0, 0]), int(mh[0, 1]))]
+        self.iterations += 1
+        self.stats.append(dict(iter=self.iterations,
+                               old_frontier=old_fq_size,
+                               pull_fallback=bool(pull),
+                               out_dense=bool(mh[0, 0] == DENSE_BITMAP),
+                               my_new=int(mh[0, 1])))
+        return int(mh[0, 1])
 
     def step(self):
         """One push iteration. Returns the global new-frontier count
@@ -289,6 +361,8 @@ class PushEngine:
         s = _stream()
         nparts = p.nparts
         ws = dx.world_size()
+        if ws == 1 and nparts == 1:
+            return self._step_single()
         mh = self.meta_host  # (nparts, 8) u32 describing CURRENT frontier
         types = mh[:, 0]
         counts = mh[:, 1]
@@ -332,7 +406,8 @@ class PushEngine:
         if pull_fallback:
             new_dense = True
             self._sync_labels()  # pull reads every vertex's label
-            self._run_pull_body()
+            mode = ng.PULL_MIN if self.is_min else ng.PULL_MAX
+            run_pull(p, mode, self.labels, self.labels_part, None, 0.0)
             self._bits_stale = True  # pull writes labels directly
             item_counter = None
         else:
@@ -379,15 +454,14 @@ class PushEngine:
             item_counter = self.item_counter
 
         # ---- frontier fix-ups + meta, all device-side ----
-        if item_counter is not None:  # push path (pull body fused above)
-            if new_dense:
-                ng.build_bitmap(s, p.vp, self.snapshot, self.labels_part,
-                                self.new_seg)
-            ng.frontier_fixup(s, p.vp, p.row_left, self.capacity,
-                              int(new_dense), self.snapshot,
-                              self.labels_part, self.deg_part, self.new_seg,
-                              self.new_annex, self.tmp_seg, self.meta_mine,
-                              item_counter, self.max_items)
+        if new_dense:
+            ng.build_bitmap(s, p.vp, self.snapshot, self.labels_part,
+                            self.new_seg)
+        ng.frontier_fixup(s, p.vp, p.row_left, self.capacity,
+                          int(new_dense), self.snapshot, self.labels_part,
+                          self.deg_part, self.new_seg, self.new_annex,
+                          self.tmp_seg, self.meta_mine, item_counter,
+                          self.max_items)
 
         # ---- exchange: meta first (the ONE host read), then payloads ----
         dx.all_gather_slices(self.meta_all, self.meta_mine,

@@ -139,6 +139,66 @@ __global__ void frontier_expand_kernel(int old_dense, V_ID in_row_left,
   }
 }
 
+// Device-sized expand: reads the segment's OWN header (type + count) so
+// every launch dimension is static — the single-GPU engine captures the
+// whole push iteration into a hipGraph and replays it (one launch + one
+// 32 B meta read per iteration instead of ~12 ctypes calls; the same
+// lever as the pull-body capture). Grid is fixed; the loop strides over
+// whatever count the header holds.
+__global__ void frontier_expand_auto_kernel(V_ID verts, V_ID in_row_left,
+                                            const uint8_t* seg,
+                                            const uint32_t* qlabels,
+                                            uint32_t* labels_repair,
+                                            const E_ID* push_row_ptr,
+                                            uint2* items, uint32_t* counter,
+                                            uint32_t max_items) {
+  __shared__ uint32_t lds_scan[BLOCK / WAVE + 1];
+  __shared__ unsigned long long lds_red[BLOCK / WAVE];
+  __shared__ uint32_t blk_base;
+  const FrontierHeader* h = (const FrontierHeader*)seg;
+  int dense = h->type == FrontierHeader::DENSE_BITMAP;
+  V_ID in_count = dense ? verts : h->numNodes;
+  const uint8_t* bitmap = seg + sizeof(FrontierHeader);
+  const V_ID* queue = (const V_ID*)(seg + sizeof(FrontierHeader));
+  for (V_ID blk = blockIdx.x * blockDim.x; blk < in_count;
+       blk += blockDim.x * gridDim.x) {
+    V_ID idx = blk + threadIdx.x;
+    V_ID u = 0;
+    uint32_t nch = 0;
+    E_ID deg = 0;
+    if (idx < in_count) {
+      bool active;
+      if (dense) {
+        u = in_row_left + idx;
+        active = (bitmap[idx >> 3] >> (idx & 7)) & 1;
+      } else {
+        u = queue[idx];
+        active = true;
+        if (labels_repair && qlabels) labels_repair[u] = qlabels[idx];
+      }
+      if (active) {
+        deg = push_row_ptr[u + 1] - push_row_ptr[u];
+        nch = (uint32_t)((deg + PUSH_CHUNK - 1) / PUSH_CHUNK);
+      }
+    }
+    unsigned long long esum =
+        block_reduce_sum((unsigned long long)deg, lds_red);
+    uint32_t total;
+    uint32_t ex = block_exscan<uint32_t, BLOCK>(nch, lds_scan, &total);
+    if (threadIdx.x == 0) {
+      blk_base = total ? atomicAdd(counter, total) : 0;
+      if (esum) atomicAdd(&counter[1], (uint32_t)esum);
+      if (total && blk_base + total > max_items) counter[3] = 1;
+    }
+    __syncthreads();
+    for (uint32_t c = 0; c < nch; c++) {
+      uint32_t pos = blk_base + ex + c;
+      if (pos < max_items) items[pos] = make_uint2(u, c);
+    }
+    __syncthreads();
+  }
+}
+
 // BFS (IS_MIN) fast path: hop-SSSP is level-synchronous BFS — every vertex
 // in an iteration's frontier carries the same depth, and a finite label is
 // final (the settled-skip invariant in pull.hip). Discovery is therefore a
@@ -524,6 +584,20 @@ void lux_gpu_frontier_expand(uint64_t stream, int old_dense,
                      dim3(BLOCK), 0, s, old_dense, in_row_left, in_count,
                      old_seg, qlabels, labels_repair, push_row_ptr, items,
                      counter, max_items);
+  LUX_POST_LAUNCH(stream);
+}
+
+void lux_gpu_frontier_expand_auto(uint64_t stream, V_ID verts,
+                                  V_ID in_row_left, const uint8_t* seg,
+                                  const uint32_t* qlabels,
+                                  uint32_t* labels_repair,
+                                  const E_ID* push_row_ptr, uint2* items,
+                                  uint32_t* counter, uint32_t max_items) {
+  hipStream_t s = (hipStream_t)stream;
+  if (verts == 0) return;
+  hipLaunchKernelGGL(frontier_expand_auto_kernel, dim3(grid_for(verts)),
+                     dim3(BLOCK), 0, s, verts, in_row_left, seg, qlabels,
+                     labels_repair, push_row_ptr, items, counter, max_items);
   LUX_POST_LAUNCH(stream);
 }
 

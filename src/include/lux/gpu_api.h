@@ -97,6 +97,13 @@ void lux_gpu_frontier_expand(uint64_t stream, int old_dense,
                              const lux::E_ID* push_row_ptr,
                              lux_uint2* items, uint32_t* counter,
                              uint32_t max_items);
+void lux_gpu_frontier_expand_auto(uint64_t stream, lux::V_ID verts,
+                                  lux::V_ID in_row_left, const uint8_t* seg,
+                                  const uint32_t* qlabels,
+                                  uint32_t* labels_repair,
+                                  const lux::E_ID* push_row_ptr,
+                                  lux_uint2* items, uint32_t* counter,
+                                  uint32_t max_items);
 void lux_gpu_frontier_fixup(uint64_t stream, lux::V_ID vp,
                             lux::V_ID row_left, lux::V_ID capacity,
                             int built_dense, const uint32_t* snapshot,
