@@ -139,6 +139,10 @@ class PushEngine:
         self.fq_annex_all = torch.zeros(int(self.annex_off[-1]), dtype=U32,
                                         device=device)
         self.new_annex = torch.zeros(self.capacity, dtype=U32, device=device)
+        # persistent zero header: zeroing via D2D copy (a python scalar
+        # write does a pageable H2D that ABORTS hipGraph capture — and an
+        # aborted capture can poison the stream)
+        self._hdr_zero = torch.zeros(8, dtype=U8, device=device)
         # device meta record (u32[8]/rank: type,count,evol u64,overflow,..)
         self.meta_mine = torch.zeros(8, dtype=U32, device=device)
         self.meta_all = torch.zeros(8 * p.nparts, dtype=U32, device=device)
@@ -255,8 +259,7 @@ class PushEngine:
         p = self.part
         s = _stream()
         self.snapshot.copy_(self.labels_part)
-        self._my_seg_i32()[0] = 0
-        self._my_seg_i32()[1] = 0
+        self.new_seg[:8].copy_(self._hdr_zero)
         if pull:
             mode = ng.PULL_MIN if self.is_min else ng.PULL_MAX
             run_pull(p, mode, self.labels, self.labels_part, None, 0.0)
@@ -301,6 +304,7 @@ class PushEngine:
         if (uses >= 2 and os.environ.get("LUX_HIPGRAPH", "1") == "1"
                 and not os.environ.get("LUX_SYNC_CHECK")):
             try:
+                torch.cuda.synchronize()
                 cg = torch.cuda.CUDAGraph()
                 with torch.cuda.graph(cg):
                     self._single_body(pull, new_dense)
@@ -310,6 +314,7 @@ class PushEngine:
             except Exception as e:  # capture unsupported: stay eager
                 print(f"[lux] push hipGraph capture disabled: {e}")
                 os.environ["LUX_HIPGRAPH"] = "0"
+                torch.cuda.synchronize()
         self._single_body(pull, new_dense)
 
     def _step_single(self):
@@ -374,8 +379,7 @@ class PushEngine:
         new_dense = dense_votes >= nparts - dense_votes
         self.snapshot.copy_(self.labels_part)
         # zero my new header (type patched by the device fixup chain)
-        self._my_seg_i32()[0] = 0
-        self._my_seg_i32()[1] = 0
+        self.new_seg[:8].copy_(self._hdr_zero)
 
         pull_fallback = overflow or old_fq_size > p.nv // 16
         if overflow:
