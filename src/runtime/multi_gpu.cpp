@@ -663,7 +663,6 @@ int push_multi_child(const HostCSC& g, int rank, int ngpus, bool is_min,
     bool overflow = false;
     for (int q = 0; q < ngpus; q++) {
       old_fq += meta_h[8ull * q + 1];
-      evol |= 0;  // (accumulated below)
       evol += ((uint64_t)meta_h[8ull * q + 3] << 32) | meta_h[8ull * q + 2];
       if (meta_h[8ull * q + 0] == FrontierHeader::DENSE_BITMAP)
         dense_votes++;
@@ -822,7 +821,7 @@ int push_multi_child(const HostCSC& g, int rank, int ngpus, bool is_min,
       printf("iter %d: activeNodes(%llu)\n", iters,
              (unsigned long long)total);
     if (total == 0 && !ovf) break;
-    if (iters > (int)g.nv * 4) break;  // safety
+    if ((uint64_t)iters > 4ull * g.nv) break;  // safety
   }
   sync_labels();
   multi_barrier(c, bar);
