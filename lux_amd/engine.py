@@ -631,6 +631,10 @@ class PagerankEngine:
                 run_pull_sweeps(p, ng.PULL_PR, self.old, self.new_part,
                                 self.deg, self.init_rank,
                                 subset=("peer", q))
+            # safety: any block without a resolved owner (cannot happen
+            # when every rank boundary is block-aligned) sweeps last
+            run_pull_sweeps(p, ng.PULL_PR, self.old, self.new_part,
+                            self.deg, self.init_rank, subset=("peer", -1))
             self._handle = None
         else:
             if self._handle is not None:
@@ -700,6 +704,8 @@ class LabelPullEngine:
                 w.wait()
                 run_pull_sweeps(p, self.mode, self.old, self.new_part,
                                 None, 0.0, subset=("peer", q))
+            run_pull_sweeps(p, self.mode, self.old, self.new_part, None,
+                            0.0, subset=("peer", -1))
             self._handle = None
         else:
             if self._handle is not None:
