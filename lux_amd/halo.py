@@ -20,7 +20,6 @@ its own slice). Positions outside the halo union may be stale — callers
 that hand the replicated array to users do one full all-gather first
 (LabelPullEngine.labels, PushEngine.final_labels).
 """
-import numpy as np
 import torch
 
 from . import dist as dx
