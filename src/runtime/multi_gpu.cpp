@@ -283,9 +283,11 @@ int multi_join(const HostCSC& g, int ngpus, int rank, const char* idfile,
   return 0;
 }
 
-// upload ONLY my slice (call after the arena exists on my device)
+// upload ONLY my slice (call after the arena exists on my device).
+// want_blocked: build the src-blocked CSC (only the push worker's pull
+// fallback sweeps use it; CC/CF workers skip the memory)
 void multi_upload(const HostCSC& g, bool weighted, DeviceArena& arena,
-                  MultiCtx* c) {
+                  MultiCtx* c, bool want_blocked = false) {
   E_ID* col_end_sl = arena.alloc_n<E_ID>(c->vp ? c->vp : 1);
   c->col = arena.alloc_n<V_ID>(c->ep ? c->ep : 1);
   if (c->vp)
@@ -306,7 +308,8 @@ void multi_upload(const HostCSC& g, bool weighted, DeviceArena& arena,
   lux_gpu_local_row_ptr((uint64_t)c->s, c->vp, c->cl, col_end_sl,
                         c->row_ptr);
   c->bins.build(c->row_ptr, c->vp, c->ep, arena, c->s);
-  c->blocked.build(c->row_ptr, c->col, c->vp, c->ep, g.nv, arena, c->s);
+  if (want_blocked)
+    c->blocked.build(c->row_ptr, c->col, c->vp, c->ep, g.nv, arena, c->s);
 }
 
 void multi_barrier(MultiCtx& c, float* bar) {
@@ -549,7 +552,7 @@ int push_multi_child(const HostCSC& g, int rank, int ngpus, bool is_min,
       + seg_off[ngpus] + 3 * seg_bytes[rank] + 4 * annex_off[ngpus]
       + 16ull * max_items
       + BlockedPull::arena_bytes(g.nv, c.vp, c.ep) + (16ull << 20));
-  multi_upload(g, false, arena, &c);
+  multi_upload(g, false, arena, &c, /*want_blocked=*/true);
   // push CSR: all nv sources -> my-partition dsts
   E_ID* push_row_ptr = arena.alloc_n<E_ID>(g.nv + 1);
   V_ID* push_col = arena.alloc_n<V_ID>(c.ep ? c.ep : 1);
