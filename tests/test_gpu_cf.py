@@ -221,14 +221,15 @@ def test_cf_multipart_single_process(cls_name, monkeypatch):
                                    atol=2e-3)
 
 
-def test_cf_als_bf16_matches_f32(monkeypatch):
+@pytest.mark.parametrize("K", [20, 32, 64])
+def test_cf_als_bf16_matches_f32(monkeypatch, K):
     """The default bf16-Gram ALS (v_mfma_f32_16x16x32_bf16) vs the exact
     fp32 path: same sweep within bf16 rounding. Also the hardware check of
     the assumed 16x16x32 A/B fragment lane map — a wrong map produces a
     wrong Gram, not a small error."""
     import torch
     from lux_amd.cf_engine import CFALSEngine
-    nu, ni, ne, K = 2000, 200, 60000, 64
+    nu, ni, ne = 2000, 200, 60000
     init = _rand_init(nu + ni, K, seed=44)
     outs = {}
     for mode in ("bf16", "f32"):

@@ -87,3 +87,18 @@ def test_pagerank_on_sliced_part_end_to_end():
     got = eng.ranks().cpu().numpy()
     want = cpu_ref.pagerank(Graph.rmat(scale, ne, seed=seed), iters)
     np.testing.assert_allclose(got, want, rtol=1e-5, atol=1e-12)
+
+
+def test_rmat_sliced_more_parts_than_busy():
+    """More partitions than the edge distribution can fill: trailing
+    partitions may be empty/thin; the sliced build must agree with the
+    full-build slices anyway."""
+    scale, ne, P, seed = 8, 400, 6, 3  # sparse graph, 256 verts
+    full = DeviceCSC.rmat(scale, ne, seed=seed)
+    for p in range(P):
+        fp = GraphPart(full, P, p, keep_full=True)
+        sp = GraphPart.rmat_sliced(scale, ne, P, p, seed=seed)
+        if fp.vp == 0:
+            assert sp.vp == 0
+            continue
+        _assert_same_partition(sp, fp)
