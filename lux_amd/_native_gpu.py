@@ -239,6 +239,11 @@ def push_chunk_scatter(stream, is_min, new_dense, items, counter, max_items,
         _u32(my_row_left), dp(new_seg), _u32(capacity), dp(visited_bits))
 
 
+def publish_labels_guarded(stream, vp, meta, labels_part, labels_slice):
+    lib().lux_gpu_publish_labels_guarded(_u64(stream), _u32(vp), dp(meta),
+                                         dp(labels_part), dp(labels_slice))
+
+
 def bits_from_labels(stream, vp, labels, bits):
     lib().lux_gpu_bits_from_labels(_u64(stream), _u32(vp), dp(labels),
                                    dp(bits))

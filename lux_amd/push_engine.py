@@ -266,8 +266,9 @@ class PushEngine:
             counter = None
         else:
             self.item_counter.zero_()
-            ng.frontier_expand_auto(s, p.vp, p.row_left, self.fq_all, None,
-                                    None, self.push_row_ptr, self.items,
+            ng.frontier_expand_auto(s, p.vp, p.row_left, self.fq_all,
+                                    self.fq_annex_all, self.labels,
+                                    self.push_row_ptr, self.items,
                                     self.item_counter, self.max_items)
             bits = self.visited if self.visited is not None and p.vp > 0 \
                 else None
@@ -287,10 +288,15 @@ class PushEngine:
                           self.labels_part, self.deg_part, self.new_seg,
                           self.new_annex, self.tmp_seg, self.meta_mine,
                           counter, self.max_items)
-        # world-1 "exchange": fixed-size self copies keep fq_all and the
-        # replicated labels current every iteration (labels never stale)
+        # world-1 "exchange": fixed-size self copies of the segment +
+        # annex; the O(nv) label copy is device-predicated on the FINAL
+        # type (sparse iterations skip it — the annex carries the queued
+        # labels, exactly like the distributed label-skip)
         self.fq_all.copy_(self.new_seg)
-        self.labels.narrow(0, p.row_left, p.vp).copy_(self.labels_part)
+        self.fq_annex_all[:self.capacity].copy_(self.new_annex)
+        ng.publish_labels_guarded(s, p.vp, self.meta_mine,
+                                  self.labels_part,
+                                  self.labels.narrow(0, p.row_left, p.vp))
 
     def _run_single_body(self, pull, new_dense):
         import os
@@ -340,6 +346,11 @@ class PushEngine:
                 pull = True
             elif (evol * p.ep // max(p.ne, 1)) // 16 > self.capacity:
                 new_dense = True
+        if pull and not self.labels_current:
+            # catch up the replicated labels before the pull sweep (the
+            # guarded publish skipped them on sparse iterations)
+            self.labels.narrow(0, p.row_left, p.vp).copy_(self.labels_part)
+            self.labels_current = True
         if not pull and self.visited is not None and p.vp > 0 \
                 and self._bits_stale:
             ng.bits_from_labels(_stream(), p.vp, self.labels_part,
@@ -350,6 +361,7 @@ class PushEngine:
             self._bits_stale = True
         mh = self.meta_mine.cpu().numpy().view(np.uint32).reshape(1, 8)
         self.meta_host = mh
+        self.labels_current = int(mh[0, 0]) == DENSE_BITMAP
         self.headers = [(int(mh[0, 0]), int(mh[0, 1]))]
         self.iterations += 1
         self.stats.append(dict(iter=self.iterations,

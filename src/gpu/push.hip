@@ -528,6 +528,21 @@ __global__ void seg_meta_kernel(V_ID vp, V_ID row_left, const uint8_t* seg,
     atomicAdd((unsigned long long*)(meta + 2), acc);
 }
 
+// World-1 publish: labels slice copy predicated on the FINAL segment
+// type (sparse iterations skip the O(nv) copy — queued labels ride the
+// annex and expand repairs them; the distributed engine makes the same
+// decision from the exchanged meta). Runs inside the captured iteration
+// graph, where the host cannot branch on the fixup outcome.
+__global__ void publish_labels_guarded_kernel(V_ID vp, const uint32_t* meta,
+                                              const uint32_t* labels_part,
+                                              uint32_t* labels_slice) {
+  if (meta[0] != FrontierHeader::DENSE_BITMAP) return;
+  uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
+  for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; i < vp;
+       i += stride)
+    labels_slice[i] = labels_part[i];
+}
+
 // Check oracles on device (check_kernel, sssp_gpu.cu:773-798 /
 // components_gpu.cu:767-791): count violations over my pull-CSC partition.
 template <bool IS_MIN>
@@ -699,6 +714,16 @@ void lux_gpu_d2s(uint64_t stream, V_ID vp, V_ID row_left,
   hipLaunchKernelGGL(d2s_kernel, dim3(grid_for(vp)), dim3(BLOCK), 0, s, vp,
                      row_left, dense_seg, sparse_seg,
                      (const uint32_t*)nullptr, 0u);
+  LUX_POST_LAUNCH(stream);
+}
+
+void lux_gpu_publish_labels_guarded(uint64_t stream, V_ID vp,
+                                    const uint32_t* meta,
+                                    const uint32_t* labels_part,
+                                    uint32_t* labels_slice) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(publish_labels_guarded_kernel, dim3(grid_for(vp)),
+                     dim3(BLOCK), 0, s, vp, meta, labels_part, labels_slice);
   LUX_POST_LAUNCH(stream);
 }
 

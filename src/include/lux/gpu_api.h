@@ -131,6 +131,10 @@ void lux_gpu_build_bitmap(uint64_t stream, lux::V_ID vp,
                           const uint32_t* new_labels, uint8_t* seg);
 void lux_gpu_d2s(uint64_t stream, lux::V_ID vp, lux::V_ID row_left,
                  const uint8_t* dense_seg, uint8_t* sparse_seg);
+void lux_gpu_publish_labels_guarded(uint64_t stream, lux::V_ID vp,
+                                    const uint32_t* meta,
+                                    const uint32_t* labels_part,
+                                    uint32_t* labels_slice);
 void lux_gpu_check(uint64_t stream, int is_min, lux::V_ID vp,
                    lux::V_ID row_left, const lux::E_ID* row_ptr_loc,
                    const lux::V_ID* col, const uint32_t* labels,
