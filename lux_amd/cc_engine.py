@@ -48,8 +48,13 @@ class CCUnionFindEngine:
             ng.uf_union_kth(s, p.vp, p.row_ptr, p.col, p.row_left,
                             self.parent, k)
         ng.uf_flatten(s, p.nv, self.parent, self.labels_t)
-        stride = max(1, p.nv // 65536)
-        giant = int(self.labels_t[::stride].mode().values.item())
+        # majority label of a 16K sample (torch.mode() sorted a 65K
+        # sample at ~2.8 ms — a fifth of the whole run; unique+argmax on
+        # a smaller sample detects the giant just as reliably)
+        stride = max(1, p.nv // 16384)
+        sample = self.labels_t[::stride]
+        vals, counts = torch.unique(sample, return_counts=True)
+        giant = int(vals[counts.argmax()].item())
         nwords = (p.nv + 31) // 32
         if not hasattr(self, "_gbits") or self._gbits.numel() < nwords:
             self._gbits = torch.empty(nwords, dtype=U32, device=self.device)
