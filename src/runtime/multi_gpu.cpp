@@ -160,6 +160,7 @@ int pagerank_multi_child(const HostCSC& g, int rank, int ngpus,
       hpr[v] = hdeg[v] == 0 ? r0 : r0 / hdeg[v];
     LUX_OK(hipMemcpyAsync(old_, hpr.data(), sizeof(float) * g.nv,
                           hipMemcpyHostToDevice, s));
+    LUX_OK(hipStreamSynchronize(s));  // hpr leaves scope
   }
   float init_rank = (1.0f - PR_ALPHA) / g.nv;
 
@@ -342,6 +343,7 @@ int components_multi_child(const HostCSC& g, int rank, int ngpus,
     for (V_ID v = 0; v < g.nv; v++) h[v] = v;
     LUX_OK(hipMemcpyAsync(parent, h.data(), 4ull * g.nv,
                           hipMemcpyHostToDevice, c.s));
+    LUX_OK(hipStreamSynchronize(c.s));  // h leaves scope
   }
   multi_barrier(c, bar);
   double t0 = now_seconds();
@@ -444,6 +446,7 @@ int col_filter_multi_child(const HostCSC& g, int rank, int ngpus,
     std::vector<float> h((uint64_t)g.nv * K, sqrtf(1.0f / K));
     LUX_OK(hipMemcpyAsync(old_, h.data(), 4ull * g.nv * K,
                           hipMemcpyHostToDevice, c.s));
+    LUX_OK(hipStreamSynchronize(c.s));  // h leaves scope
   }
   int* hubidx = nullptr;
   float *gram = nullptr, *rhs = nullptr;
@@ -636,6 +639,8 @@ int push_multi_child(const HostCSC& g, int rank, int ngpus, bool is_min,
     LUX_OK(hipMemcpyAsync(fq_all, hfq.data(), seg_off[ngpus],
                           hipMemcpyHostToDevice, c.s));
     LUX_OK(hipMemsetAsync(annex_all, 0, 4ull * annex_off[ngpus], c.s));
+    // drain before hl/hfq leave scope (async copy from pageable host)
+    LUX_OK(hipStreamSynchronize(c.s));
   }
   bool labels_current = true;
   auto sync_labels = [&]() {
