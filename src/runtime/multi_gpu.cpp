@@ -462,6 +462,7 @@ int col_filter_multi_child(const HostCSC& g, int rank, int ngpus,
     for (uint32_t i = 0; i < c.bins.nbig; i++) hidx[hv[i]] = (int)i;
     LUX_OK(hipMemcpyAsync(hubidx, hidx.data(), 4ull * c.vp,
                           hipMemcpyHostToDevice, c.s));
+    LUX_OK(hipStreamSynchronize(c.s));  // hidx leaves scope
   }
   float* bar = arena.alloc_n<float>(1);
   multi_barrier(c, bar);
