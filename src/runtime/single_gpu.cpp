@@ -592,6 +592,7 @@ SingleGpuCF::SingleGpuCF(const DeviceGraph& g, int K, DeviceArena& arena,
     for (uint32_t i = 0; i < bins_.nbig; i++) hidx[hubs[i]] = (int)i;
     LUX_OK(hipMemcpyAsync(hubidx_, hidx.data(), sizeof(int) * g.nv,
                           hipMemcpyHostToDevice, s));
+    LUX_OK(hipStreamSynchronize(s));  // hidx leaves scope
     gram_ = arena.alloc_n<float>((size_t)bins_.nbig * 64 * 64);
     rhs_ = arena.alloc_n<float>((size_t)bins_.nbig * 64);
   }
