@@ -591,7 +591,7 @@ def _dist_per_peer_worker(rank, world, port, scale, ne, seed, iters, outq):
         dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("world", [2, 4])
+@pytest.mark.parametrize("world", [2, 3, 4])
 def test_dist_per_peer_exchange(world):
     from lux_amd import cpu_ref
     from lux_amd.graph import Graph
