@@ -415,9 +415,17 @@ __device__ __forceinline__ void als_dump_gram(AlsLds* lds, f32x4 acc[10],
 }
 
 // In-place lower Cholesky of the 64x64 LDS matrix (one wave; lane = row).
+// Pivot floor: a (near-)rank-deficient Gram — e.g. the parity init makes
+// sweep 1's Gram rank-1 with entries ~deg/K, whose elimination cancels
+// later pivots to rounding noise (negative under bf16 inputs -> sqrt ->
+// NaN at NetFlix-scale hub degrees). Flooring the pivot at lambda keeps
+// the factor finite; the affected directions get a bounded (regularised)
+// step and subsequent sweeps decorrelate the factors.
 __device__ __forceinline__ void wave_cholesky64(float* G, int lane) {
   for (int k = 0; k < ALS_K; k++) {
-    float dkk = sqrtf(G[k * ALS_ROW + k]);
+    float piv = G[k * ALS_ROW + k];
+    if (!(piv > (float)CF_LAMBDA * 0.5f)) piv = (float)CF_LAMBDA;
+    float dkk = sqrtf(piv);
     float lik = lane > k ? G[lane * ALS_ROW + k] / dkk : 0.0f;
     if (lane == k) G[k * ALS_ROW + k] = dkk;
     if (lane > k) G[lane * ALS_ROW + k] = lik;

@@ -257,3 +257,24 @@ def test_cf_als_bf16_matches_f32(monkeypatch, K):
     a, b = outs["bf16"][well], outs["f32"][well]
     scale = np.maximum(np.abs(b).max(axis=1, keepdims=True), 1.0)
     np.testing.assert_allclose(a / scale, b / scale, rtol=0, atol=3e-2)
+
+
+def test_cf_als_parity_init_hub_scale_finite():
+    """From the reference's constant init the sweep-1 Gram is rank-1 with
+    entries ~deg/K; at hub-scale degrees its Cholesky pivots cancel
+    negative and produced NaN before the pivot floor. ALS must stay
+    finite and beat SGD from the parity init on a hub-heavy shape."""
+    from lux_amd.cf_engine import CFALSEngine
+    nu, ni, ne, K = 200000, 50, 2000000, 64  # item degree ~40K
+    full = DeviceCSC.bipartite(nu, ni, ne, seed=21)
+    g = Graph.bipartite(nu, ni, ne, seed=21)
+    sgd = CFEngine(GraphPart(full, 1, 0, keep_full=True), K=K)
+    als = CFALSEngine(GraphPart(full, 1, 0), K=K)
+    for _ in range(3):
+        sgd.step()
+        als.step()
+    va = als.vectors().cpu().numpy()
+    assert np.isfinite(va).all()
+    l_als = cpu_ref.cf_loss(g, K, va)
+    l_sgd = cpu_ref.cf_loss(g, K, sgd.vectors().cpu().numpy())
+    assert np.isfinite(l_als) and l_als < l_sgd
