@@ -415,19 +415,22 @@ __device__ __forceinline__ void als_dump_gram(AlsLds* lds, f32x4 acc[10],
 }
 
 // In-place lower Cholesky of the 64x64 LDS matrix (one wave; lane = row).
-// Pivot floor: a (near-)rank-deficient Gram — e.g. the parity init makes
-// sweep 1's Gram rank-1 with entries ~deg/K, whose elimination cancels
-// later pivots to rounding noise (negative under bf16 inputs -> sqrt ->
-// NaN at NetFlix-scale hub degrees). Flooring the pivot at lambda keeps
-// the factor finite; the affected directions get a bounded (regularised)
-// step and subsequent sweeps decorrelate the factors.
+// Semi-definite handling: a (near-)rank-deficient Gram — e.g. the parity
+// init makes sweep 1's Gram rank-1 with entries ~deg/K, whose
+// elimination cancels later pivots to rounding noise (negative under
+// bf16 inputs at NetFlix-scale hub degrees; sqrt -> NaN, and a naive
+// lambda floor instead explodes the degenerate directions' step, which
+// overflowed the NEXT sweep's Gram to inf/NaN). Deficient pivots are
+// marked with a -1 sentinel, their column is zeroed, and the solve
+// takes NO step along those directions — the exact-arithmetic
+// pseudoinverse behaviour on the deficient subspace.
 __device__ __forceinline__ void wave_cholesky64(float* G, int lane) {
   for (int k = 0; k < ALS_K; k++) {
     float piv = G[k * ALS_ROW + k];
-    if (!(piv > (float)CF_LAMBDA * 0.5f)) piv = (float)CF_LAMBDA;
-    float dkk = sqrtf(piv);
-    float lik = lane > k ? G[lane * ALS_ROW + k] / dkk : 0.0f;
-    if (lane == k) G[k * ALS_ROW + k] = dkk;
+    bool ok = piv > (float)CF_LAMBDA * 0.5f;
+    float dkk = ok ? sqrtf(piv) : 1.0f;
+    float lik = (lane > k && ok) ? G[lane * ALS_ROW + k] / dkk : 0.0f;
+    if (lane == k) G[k * ALS_ROW + k] = ok ? dkk : -1.0f;
     if (lane > k) G[lane * ALS_ROW + k] = lik;
     als_lds_sync();
     for (int j = k + 1; j <= lane; j++)
@@ -437,15 +440,18 @@ __device__ __forceinline__ void wave_cholesky64(float* G, int lane) {
 }
 
 // Solve L L^T d = rhs; rhs/result live lane=dim in a register.
+// Sentinel (-1) pivots contribute zero (see wave_cholesky64).
 __device__ __forceinline__ float wave_spd_solve64(const float* G, float r,
                                                   int lane) {
   for (int k = 0; k < ALS_K; k++) {  // forward: L y = r
-    float yk = __shfl(r, k, WAVE) / G[k * ALS_ROW + k];
+    float lkk = G[k * ALS_ROW + k];
+    float yk = lkk > 0.0f ? __shfl(r, k, WAVE) / lkk : 0.0f;
     if (lane == k) r = yk;
     else if (lane > k) r -= G[lane * ALS_ROW + k] * yk;
   }
   for (int k = ALS_K - 1; k >= 0; k--) {  // backward: L^T d = y
-    float dk = __shfl(r, k, WAVE) / G[k * ALS_ROW + k];
+    float lkk = G[k * ALS_ROW + k];
+    float dk = lkk > 0.0f ? __shfl(r, k, WAVE) / lkk : 0.0f;
     if (lane == k) r = dk;
     else if (lane < k) r -= G[k * ALS_ROW + lane] * dk;
   }
