@@ -169,18 +169,20 @@ def pull_iter(stream, mode, n0, bin0, n1, bin1, n2, bin2, nbig, bin2v,
                             ctypes.c_float(init_rank))
 
 
-def blocked_count(stream, ep, col, row_ptr_loc, vp, bounds, nb, counts):
+def blocked_count(stream, ep, col, row_ptr_loc, vp, bounds, nb, counts,
+                  lo=0, hi=0xFFFFFFFF):
     lib().lux_gpu_blocked_count(_u64(stream), _u64(ep), dp(col),
                                 dp(row_ptr_loc), _u32(vp), dp(bounds),
-                                ctypes.c_int(nb), dp(counts))
+                                ctypes.c_int(nb), _u32(lo), _u32(hi),
+                                dp(counts))
 
 
 def blocked_scatter(stream, ep, col, row_ptr_loc, vp, bounds, nb, cursor,
-                    out_col):
+                    out_col, lo=0, hi=0xFFFFFFFF):
     lib().lux_gpu_blocked_scatter(_u64(stream), _u64(ep), dp(col),
                                   dp(row_ptr_loc), _u32(vp), dp(bounds),
-                                  ctypes.c_int(nb), dp(cursor),
-                                  dp(out_col))
+                                  ctypes.c_int(nb), _u32(lo), _u32(hi),
+                                  dp(cursor), dp(out_col))
 
 
 def pull_finish_pr(stream, vp, newv, deg, row_left, init_rank):

@@ -445,64 +445,87 @@ class GraphPart:
         self.build_blocked(bounds)
 
     def build_blocked(self, bounds):
+        """Src-blocked CSC build, processed in window GROUPS sized so the
+        transient (block, row) slot table (counts 4 B + cursor 8 B per
+        slot) fits free HBM — 32 MB windows therefore hold at every
+        scale (the full table at RMAT-29 x 1 GPU would be 384 GB). Each
+        extra group costs one more filtered pass over the edge list.
+        LUX_BLOCK_GROUP_SLOTS caps the slot budget (tests)."""
+        import os
         device = self.device
         s = _stream()
         vp, ep = self.vp, self.ep
         sb = len(bounds) - 1
-        bounds_t = torch.tensor(bounds, dtype=U32, device=device)
-        n = sb * vp
-        counts = torch.zeros(n, dtype=U32, device=device)
-        ng.blocked_count(s, ep, self.col, self.row_ptr, vp, bounds_t, sb + 1,
-                         counts)
-        # scan straight into cursor[1:]: the scatter advances every slot by
-        # exactly its count, so the mutated cursor ends up equal to the end
-        # offsets — no separate `ends` array (a peak-memory lever: at
-        # RMAT-28 / 32 MB windows the slot table alone is 2^33 entries)
-        cursor = torch.zeros(n + 1, dtype=U64, device=device)
-        partials = torch.empty(ng.scan_partials_size(n), dtype=U64,
-                               device=device)
-        ng.scan_end_offsets(s, n, counts, cursor.narrow(0, 1, n), partials)
-        torch.cuda.synchronize()
-        del counts, partials
-        torch.cuda.empty_cache()
+        cap = int(os.environ.get("LUX_BLOCK_GROUP_SLOTS", "0") or 0)
+        if not cap and torch.cuda.is_available():
+            free, _total = torch.cuda.mem_get_info(device)
+            cap = int(max(free - 4 * ep - (2 << 30), 1 << 30) * 0.85) // 12
+        bpg = sb if not cap else max(1, min(sb, cap // max(vp, 1)))
         blk_col = torch.empty(max(ep, 1), dtype=U32, device=device)
-        ng.blocked_scatter(s, ep, self.col, self.row_ptr, vp, bounds_t,
-                           sb + 1, cursor.narrow(0, 0, n), blk_col)
-        ends = cursor.narrow(0, 0, n)  # == end offsets after the scatter
         self.blocks = []
-        begin = 0
-        for b in range(sb):
-            end = int(ends[(b + 1) * vp - 1].item())
-            if end == begin:
-                continue  # empty src window (e.g. all-zero-degree tail
-                #           under the hot-source permutation): no sweeps
-            row_ptr_b = torch.empty(vp + 1, dtype=U64, device=device)
-            ng.local_row_ptr(s, vp, begin, ends.narrow(0, b * vp, vp),
-                             row_ptr_b)
-            col_b = blk_col.narrow(0, begin, end - begin)
-            n0, n1, n2, nbig, b0, b1, b2, b2v = _bins_for(
-                row_ptr_b, vp, end - begin, device, compact=True)
-            # block-local offsets fit u32 (block edge counts < 2^32): half
-            # the per-row sweep traffic vs u64 rows (NOTES_r2 item 1)
-            row32 = torch.empty(vp + 1, dtype=U32, device=device)
-            ng.u64_to_u32(s, vp + 1, row_ptr_b, row32)
-            del row_ptr_b
-            owner = -1
-            if self.nparts > 1:
-                for q in range(self.nparts):
-                    if self.verts_all[q] \
-                            and bounds[b] >= self.row_left_all[q] \
-                            and bounds[b + 1] <= self.row_right_all[q] + 1:
-                        owner = q
-                        break
-            local = self.nparts > 1 and owner == self.p
-            self.blocks.append(dict(row_ptr=row32, row_u32=1, col=col_b,
-                                    n0=n0, n1=n1, n2=n2, nbig=nbig, bin0=b0,
-                                    bin1=b1, bin2=b2, bin2v=b2v,
-                                    local=local, owner=owner))
-            begin = end
+        global_begin = 0
+        for g0 in range(0, sb, bpg):
+            g1 = min(sb, g0 + bpg)
+            nb = g1 - g0
+            slots = nb * vp
+            bounds_g = torch.tensor(bounds[g0:g1 + 1], dtype=U32,
+                                    device=device)
+            counts = torch.zeros(slots, dtype=U32, device=device)
+            ng.blocked_count(s, ep, self.col, self.row_ptr, vp, bounds_g,
+                             nb + 1, counts, lo=bounds[g0], hi=bounds[g1])
+            # scan straight into cursor[1:]: the scatter advances every
+            # slot by exactly its count, so the mutated cursor ends up
+            # equal to the end offsets — no separate `ends` array
+            cursor = torch.zeros(slots + 1, dtype=U64, device=device)
+            partials = torch.empty(ng.scan_partials_size(slots), dtype=U64,
+                                   device=device)
+            ng.scan_end_offsets(s, slots, counts,
+                                cursor.narrow(0, 1, slots), partials)
+            torch.cuda.synchronize()
+            del counts, partials
+            torch.cuda.empty_cache()
+            if global_begin:
+                cursor += global_begin  # offsets into the shared blk_col
+            ng.blocked_scatter(s, ep, self.col, self.row_ptr, vp, bounds_g,
+                               nb + 1, cursor.narrow(0, 0, slots), blk_col,
+                               lo=bounds[g0], hi=bounds[g1])
+            ends = cursor.narrow(0, 0, slots)  # post-scatter = end offsets
+            begin = global_begin
+            for b in range(nb):
+                end = int(ends[(b + 1) * vp - 1].item())
+                if end == begin:
+                    continue  # empty src window: no sweeps
+                row_ptr_b = torch.empty(vp + 1, dtype=U64, device=device)
+                ng.local_row_ptr(s, vp, begin, ends.narrow(0, b * vp, vp),
+                                 row_ptr_b)
+                col_b = blk_col.narrow(0, begin, end - begin)
+                n0, n1, n2, nbig, b0, b1, b2, b2v = _bins_for(
+                    row_ptr_b, vp, end - begin, device, compact=True)
+                # block-local offsets fit u32 (block edge counts < 2^32):
+                # half the per-row sweep traffic vs u64 rows
+                row32 = torch.empty(vp + 1, dtype=U32, device=device)
+                ng.u64_to_u32(s, vp + 1, row_ptr_b, row32)
+                del row_ptr_b
+                gb = g0 + b
+                owner = -1
+                if self.nparts > 1:
+                    for q in range(self.nparts):
+                        if self.verts_all[q] \
+                                and bounds[gb] >= self.row_left_all[q] \
+                                and bounds[gb + 1] \
+                                <= self.row_right_all[q] + 1:
+                            owner = q
+                            break
+                local = self.nparts > 1 and owner == self.p
+                self.blocks.append(dict(row_ptr=row32, row_u32=1, col=col_b,
+                                        n0=n0, n1=n1, n2=n2, nbig=nbig,
+                                        bin0=b0, bin1=b1, bin2=b2,
+                                        bin2v=b2v, local=local,
+                                        owner=owner))
+                begin = end
+            global_begin = int(ends[slots - 1].item())
+            del ends, cursor
         self._blk_col = blk_col  # keep the narrow()s' base alive
-        del ends, cursor
 
 
 def _maybe_halo(part):

@@ -205,15 +205,22 @@ __device__ __forceinline__ uint32_t block_of(V_ID src, const V_ID* bounds,
   return (uint32_t)lo;
 }
 
+// lo/hi: src-range filter for the GROUPED build (the (block,row) slot
+// table at 32 MB windows is sb*vp entries — 2^35 at RMAT-29 on one GPU;
+// processing groups of windows keeps the transient table inside free
+// HBM at the cost of one extra edge pass per group). bounds here is the
+// GROUP's boundary slice; block indices are group-local.
 __global__ void blocked_count_kernel(uint64_t ep, const V_ID* col,
                                      const E_ID* row_ptr_loc, V_ID vp,
                                      const V_ID* bounds, int nb,
-                                     uint32_t* counts) {
+                                     V_ID lo, V_ID hi, uint32_t* counts) {
   uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
   for (uint64_t j = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; j < ep;
        j += stride) {
+    V_ID c = col[j];
+    if (c < lo || c >= hi) continue;
     V_ID v = row_of_edge(row_ptr_loc, vp, j);
-    uint32_t b = block_of(col[j], bounds, nb);
+    uint32_t b = block_of(c, bounds, nb);
     atomicAdd(&counts[(uint64_t)b * vp + v], 1u);
   }
 }
@@ -221,15 +228,18 @@ __global__ void blocked_count_kernel(uint64_t ep, const V_ID* col,
 __global__ void blocked_scatter_kernel(uint64_t ep, const V_ID* col,
                                        const E_ID* row_ptr_loc, V_ID vp,
                                        const V_ID* bounds, int nb,
+                                       V_ID lo, V_ID hi,
                                        unsigned long long* cursor,
                                        V_ID* out_col) {
   uint64_t stride = (uint64_t)blockDim.x * gridDim.x;
   for (uint64_t j = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; j < ep;
        j += stride) {
+    V_ID c = col[j];
+    if (c < lo || c >= hi) continue;
     V_ID v = row_of_edge(row_ptr_loc, vp, j);
-    uint32_t b = block_of(col[j], bounds, nb);
+    uint32_t b = block_of(c, bounds, nb);
     unsigned long long pos = atomicAdd(&cursor[(uint64_t)b * vp + v], 1ull);
-    out_col[pos] = col[j];
+    out_col[pos] = c;
   }
 }
 
@@ -372,21 +382,23 @@ void lux_gpu_edges_to_csc(uint64_t stream, uint32_t nv, uint64_t ne,
 void lux_gpu_blocked_count(uint64_t stream, uint64_t ep, const V_ID* col,
                            const E_ID* row_ptr_loc, V_ID vp,
                            const V_ID* bounds /*device u32[nb+1]*/, int nb,
+                           V_ID lo, V_ID hi,
                            uint32_t* counts /*pre-zeroed u32[nb*vp]*/) {
   hipStream_t s = (hipStream_t)stream;
   hipLaunchKernelGGL(blocked_count_kernel, dim3(grid_for(ep)), dim3(BLOCK),
-                     0, s, ep, col, row_ptr_loc, vp, bounds, nb, counts);
+                     0, s, ep, col, row_ptr_loc, vp, bounds, nb, lo, hi,
+                     counts);
   LUX_POST_LAUNCH(stream);
 }
 
 void lux_gpu_blocked_scatter(uint64_t stream, uint64_t ep, const V_ID* col,
                              const E_ID* row_ptr_loc, V_ID vp,
-                             const V_ID* bounds, int nb,
+                             const V_ID* bounds, int nb, V_ID lo, V_ID hi,
                              unsigned long long* cursor, V_ID* out_col) {
   hipStream_t s = (hipStream_t)stream;
   hipLaunchKernelGGL(blocked_scatter_kernel, dim3(grid_for(ep)), dim3(BLOCK),
-                     0, s, ep, col, row_ptr_loc, vp, bounds, nb, cursor,
-                     out_col);
+                     0, s, ep, col, row_ptr_loc, vp, bounds, nb, lo, hi,
+                     cursor, out_col);
   LUX_POST_LAUNCH(stream);
 }
 

@@ -60,12 +60,14 @@ void lux_gpu_local_row_ptr(uint64_t stream, uint32_t vp, lux::E_ID col_left,
                            lux::E_ID* row_ptr_loc);
 void lux_gpu_blocked_count(uint64_t stream, uint64_t ep, const lux::V_ID* col,
                            const lux::E_ID* row_ptr_loc, lux::V_ID vp,
-                           const lux::V_ID* bounds, int nb, uint32_t* counts);
+                           const lux::V_ID* bounds, int nb, lux::V_ID lo,
+                           lux::V_ID hi, uint32_t* counts);
 void lux_gpu_blocked_scatter(uint64_t stream, uint64_t ep,
                              const lux::V_ID* col,
                              const lux::E_ID* row_ptr_loc, lux::V_ID vp,
-                             const lux::V_ID* bounds, int nb,
-                             unsigned long long* cursor, lux::V_ID* out_col);
+                             const lux::V_ID* bounds, int nb, lux::V_ID lo,
+                             lux::V_ID hi, unsigned long long* cursor,
+                             lux::V_ID* out_col);
 
 // pull.hip
 void lux_gpu_build_bins(uint64_t stream, uint32_t vp,

@@ -177,12 +177,12 @@ void BlockedPull::build(const E_ID* row_ptr_loc, const V_ID* col, V_ID vp_,
   LUX_OK(hipMemsetAsync(counts, 0, 4ull * n, s));
   LUX_OK(hipMemsetAsync(cursor, 0, 8, s));
   lux_gpu_blocked_count((uint64_t)s, ep, col, row_ptr_loc, vp, bounds,
-                        sb + 1, counts);
+                        sb + 1, 0, nv, counts);
   lux_gpu_scan_end_offsets((uint64_t)s, n, counts, (E_ID*)cursor + 1,
                            partials);
   V_ID* blk_col = arena.alloc_n<V_ID>(ep);
   lux_gpu_blocked_scatter((uint64_t)s, ep, col, row_ptr_loc, vp, bounds,
-                          sb + 1, cursor, blk_col);
+                          sb + 1, 0, nv, cursor, blk_col);
   // post-scatter, cursor[i] == end offset of slot i
   E_ID* row64;
   LUX_OK(hipMalloc(&row64, 8ull * (vp + 1)));
