@@ -68,7 +68,7 @@ void lux_bipartite_edges(uint64_t seed, uint32_t n_users, uint32_t n_items,
   while (((uint32_t)1 << item_scale) < n_items) item_scale++;
   for (uint64_t e = 0; e < ne; e++) {
     bipartite_edge(seed, e, n_users, n_items, item_scale, &src[e], &dst[e]);
-    w[e] = rmat_weight(seed, e);
+    w[e] = rmat_weight(seed, e >> 1);  // both directions share the rating
   }
 }
 void lux_edges_to_csc(uint32_t nv, uint64_t ne, const uint32_t* src,

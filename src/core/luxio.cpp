@@ -164,7 +164,7 @@ HostCSC bipartite_csc_cpu(V_ID n_users, V_ID n_items, E_ID ne, uint64_t seed) {
   std::vector<WeightType> w(ne);
   for (E_ID e = 0; e < ne; e++) {
     bipartite_edge(seed, e, n_users, n_items, item_scale, &s[e], &d[e]);
-    w[e] = rmat_weight(seed, e);
+    w[e] = rmat_weight(seed, e >> 1);  // both directions share the rating
   }
   return edges_to_csc(n_users + n_items, s, d, &w);
 }

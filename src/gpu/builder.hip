@@ -45,7 +45,7 @@ __global__ void bipartite_edges_kernel(uint64_t seed, V_ID n_users,
        e += stride) {
     bipartite_edge(seed, e0 + e, n_users, n_items, item_scale, &src[e],
                    &dst[e]);
-    w[e] = rmat_weight(seed, e0 + e);
+    w[e] = rmat_weight(seed, (e0 + e) >> 1);  // both directions share the rating
   }
 }
 

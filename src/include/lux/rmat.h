@@ -82,17 +82,29 @@ LUX_HD WeightType rmat_weight(uint64_t seed, uint64_t e) {
 }
 
 // Bipartite synthetic generator for CF (NetFlix-shaped: users x items).
-// Edge e connects src = user id in [0, n_users) to dst = n_users + item id
-// in [0, n_items), with an RMAT-skewed item (popularity) and a hash-uniform
-// user. dst-skew makes the CSC degree distribution realistic.
+// Edge pair 2r/2r+1 stores rating r in BOTH directions (user <-> item),
+// matching the reference's NetFlix file whose 200.9M edges are 100.5M
+// ratings x 2 — the pull-model CF sweeps then update BOTH sides (items
+// gather user vectors and vice versa; a one-directional store would
+// freeze the user factors forever). Items are RMAT-popularity-skewed,
+// users hash-uniform; both directions of a rating share the weight
+// (rmat_weight(seed, e >> 1)).
 LUX_HD void bipartite_edge(uint64_t seed, uint64_t e, V_ID n_users,
                            V_ID n_items, int item_scale, V_ID* src,
                            V_ID* dst) {
+  uint64_t r = e >> 1;
   V_ID iu, iv;
-  rmat_edge(seed, e, item_scale, &iu, &iv);
-  uint64_t s = splitmix64(seed ^ 0x5EEDF00Dull ^ (e * 0x9E3779B97F4A7C15ull));
-  *src = fold_id((V_ID)s, n_users);
-  *dst = n_users + fold_id(iv, n_items);
+  rmat_edge(seed, r, item_scale, &iu, &iv);
+  uint64_t s = splitmix64(seed ^ 0x5EEDF00Dull ^ (r * 0x9E3779B97F4A7C15ull));
+  V_ID user = fold_id((V_ID)s, n_users);
+  V_ID item = n_users + fold_id(iv, n_items);
+  if (e & 1) {
+    *src = item;
+    *dst = user;
+  } else {
+    *src = user;
+    *dst = item;
+  }
 }
 
 }  // namespace lux

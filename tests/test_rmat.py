@@ -42,5 +42,19 @@ def test_bipartite_weighted():
     assert g.nv == 1256
     assert g.weight is not None
     assert g.weight.min() >= 1 and g.weight.max() <= 5
-    # users (first 1000 ids) must have no in-edges: all edges point to items
-    assert g.row_end(999) == 0
+    # each rating is stored in BOTH directions (reference NetFlix parity:
+    # the .nf edge file holds user->item and item->user copies), so edges
+    # split evenly and both sides have in-edges
+    assert g.row_end(999) == g.ne // 2      # user in-edges = item->user half
+    assert g.row_end(1255) == g.ne          # item in-edges = user->item half
+
+
+def test_bipartite_pair_symmetry():
+    """Even/odd generator indices form one rating: reversed endpoints,
+    identical weight (rmat.h bipartite_edge pair emission)."""
+    src, dst, w = nat.bipartite_edges(7, 500, 128, 10000)
+    assert np.array_equal(src[0::2], dst[1::2])
+    assert np.array_equal(dst[0::2], src[1::2])
+    assert np.array_equal(w[0::2], w[1::2])
+    # direction check: even edges go user->item, odd item->user
+    assert src[0::2].max() < 500 and dst[0::2].min() >= 500
