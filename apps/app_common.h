@@ -30,6 +30,9 @@ struct AppArgs {
                                // lux_amd/checkpoint.py-compatible)
   bool als = false;            // col_filter: exact MFMA ALS optimizer
   bool labelprop = false;      // components: reference-parity label prop
+  lux::V_ID users = 0;         // col_filter -als: bipartite user/item
+                               // boundary for Gauss-Seidel alternation
+                               // (auto-set from -synthetic bipartite:...)
 };
 
 inline AppArgs parse_input_args(int argc, char** argv) {
@@ -47,12 +50,18 @@ inline AppArgs parse_input_args(int argc, char** argv) {
     else if (f == "-synthetic") a.synthetic = next();
     else if (f == "-dump") a.dump = next();
     else if (f == "-als") a.als = true;
+    else if (f == "-users") a.users = (lux::V_ID)atoll(next());
     else if (f == "-labelprop") a.labelprop = true;
     else if (f.rfind("-ll:", 0) == 0 || f.rfind("-lg:", 0) == 0) {
       if (i + 1 < argc && argv[i + 1][0] != '-') i++;  // value-flag: skip
     } else {
       fprintf(stderr, "warning: ignoring unknown flag %s\n", f.c_str());
     }
+  }
+  if (a.users == 0 && a.synthetic) {  // boundary is implicit in the spec
+    long long nu, ni, ne;
+    if (sscanf(a.synthetic, "bipartite:%lld:%lld:%lld", &nu, &ni, &ne) == 3)
+      a.users = (lux::V_ID)nu;
   }
   return a;
 }

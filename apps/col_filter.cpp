@@ -23,7 +23,7 @@ int main(int argc, char** argv) {
     return col_filter_multi_child(g, atoi(mr),
                                   atoi(getenv("LUX_MULTI_WORLD")),
                                   getenv("LUX_MULTI_IDFILE"), a.k,
-                                  a.num_iter, a.als, a.dump);
+                                  a.num_iter, a.als, a.dump, a.users);
   }
   if (a.num_gpu > 1 && getenv("LUX_TORCHRUN"))
     // escape hatch: the torchrun RCCL engine (same CLI, Python driver)
@@ -72,7 +72,9 @@ int main(int argc, char** argv) {
   } else {
     dg = DeviceGraph::upload(g, arena, s);
   }
-  SingleGpuCF engine(dg, a.k, arena, s, a.als);
+  // a.users (from -users or the -synthetic bipartite spec) enables true
+  // Gauss-Seidel ALS alternation; 0 falls back to simultaneous Jacobi
+  SingleGpuCF engine(dg, a.k, arena, s, a.als, a.users);
 
   auto t0 = std::chrono::steady_clock::now();
   engine.iterate(a.num_iter);

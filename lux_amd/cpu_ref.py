@@ -85,23 +85,27 @@ def sssp_partitioned(g: Graph, nparts: int, source: int):
 
 
 def cf_als(g: Graph, K: int, iters: int, lam: float = 0.001,
-           init: "np.ndarray | None" = None) -> np.ndarray:
+           init: "np.ndarray | None" = None,
+           n_users: "int | None" = None) -> np.ndarray:
     """Plain-numpy ALS reference: per sweep, for each vertex with in-edges,
-    solve (S^T S + lam I) d = S^T w exactly against the OLD vectors (S =
-    src vectors of the in-edges). Vertices with no in-edges keep their old
-    vector. Same fixed point as the reference SGD sweep (cf.hip docstring);
-    solved in float64 here, compared with tolerance against the fp32 GPU
-    Cholesky path (src/gpu/cf_als.hip)."""
+    solve (S^T S + lam I) d = S^T w exactly (S = src vectors of the
+    in-edges). Vertices with no in-edges keep their old vector. With
+    n_users given (bipartite user/item boundary) the sweep ALTERNATES:
+    user rows [0, n_users) solve against the old item factors, then item
+    rows solve against the UPDATED users — true Gauss-Seidel ALS, matching
+    CFALSEngine. Without it, every row solves against old values
+    (simultaneous Jacobi). Solved in float64 here, compared with tolerance
+    against the fp32 GPU Cholesky path (src/gpu/cf_als.hip)."""
     import math as _math
     if init is not None:
         vec = np.array(init, dtype=np.float32).reshape(g.nv, K).copy()
     else:
         vec = np.full((g.nv, K), _math.sqrt(1.0 / K), dtype=np.float32)
     eye = lam * np.eye(K, dtype=np.float64)
-    for _ in range(iters):
-        new = vec.copy()
-        b = 0
-        for v in range(g.nv):
+
+    def _solve_rows(vec, new, lo, hi):
+        b = 0 if lo == 0 else int(g.col_end[lo - 1])
+        for v in range(lo, hi):
             e = int(g.col_end[v])
             if e > b:
                 S = vec[g.src[b:e]].astype(np.float64)
@@ -109,5 +113,15 @@ def cf_als(g: Graph, K: int, iters: int, lam: float = 0.001,
                 G = S.T @ S + eye
                 new[v] = np.linalg.solve(G, S.T @ w).astype(np.float32)
             b = e
-        vec = new
+
+    for _ in range(iters):
+        if n_users is None:
+            new = vec.copy()
+            _solve_rows(vec, new, 0, g.nv)
+            vec = new
+        else:
+            # in-place is exact per phase: a bipartite row only reads the
+            # other side's factors
+            _solve_rows(vec, vec, 0, n_users)
+            _solve_rows(vec, vec, n_users, g.nv)
     return vec

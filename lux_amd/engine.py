@@ -129,6 +129,8 @@ class DeviceCSC:
     """Full CSC graph resident on one GPU (col_end u64[nv], src u32[ne],
     optional weight i32[ne])."""
 
+    n_users = None  # set by bipartite(): user/item boundary
+
     def __init__(self, nv, ne, col_end, src, weight=None):
         self.nv, self.ne = nv, ne
         self.col_end, self.src, self.weight = col_end, src, weight
@@ -170,7 +172,9 @@ class DeviceCSC:
         edst = torch.empty(ne, dtype=U32, device=device)
         ew = torch.empty(ne, dtype=U32, device=device)
         ng.bipartite_edges(s, seed, n_users, n_items, ne, esrc, edst, ew)
-        return cls._from_device_edges(nv, ne, esrc, edst, ew, device)
+        g = cls._from_device_edges(nv, ne, esrc, edst, ew, device)
+        g.n_users = n_users  # user/item boundary for ALS alternation
+        return g
 
     @classmethod
     def _from_device_edges(cls, nv, ne, esrc, edst, ew, device):
@@ -203,9 +207,12 @@ class GraphPart:
     """This rank's partition: local row_ptr (u64[vp+1], 0-based), local col
     slice, plus the global partition table (all ranks' bounds)."""
 
+    n_users = None  # bipartite user/item boundary (ALS alternation)
+
     def __init__(self, full: DeviceCSC, nparts, my_part, keep_full=False):
         device = full.col_end.device
         self.nv, self.ne = full.nv, full.ne
+        self.n_users = full.n_users
         self.nparts, self.p = nparts, my_part
         rl, rr = partition_bounds(full.col_end, full.ne, nparts)
         self.row_left_all, self.row_right_all = rl, rr
@@ -269,8 +276,10 @@ class GraphPart:
         def gen(e0, n, src, dst, w):
             ng.bipartite_edges_chunk(_stream(), seed, n_users, n_items, e0,
                                      n, src, dst, w)
-        return cls._sliced_build(n_users + n_items, ne, nparts, my_part,
+        part = cls._sliced_build(n_users + n_items, ne, nparts, my_part,
                                  device, False, gen, weighted=True)
+        part.n_users = n_users
+        return part
 
     @classmethod
     def _sliced_build(cls, nv, ne, nparts, my_part, device, sym, gen,

@@ -430,7 +430,7 @@ int components_multi_child(const HostCSC& g, int rank, int ngpus,
 
 int col_filter_multi_child(const HostCSC& g, int rank, int ngpus,
                            const char* idfile, int K, int iters, bool als,
-                           const char* dump) {
+                           const char* dump, V_ID n_users) {
   MultiCtx c;
   int rc = multi_join(g, ngpus, rank, idfile, &c);
   if (rc) return rc;
@@ -450,6 +450,17 @@ int col_filter_multi_child(const HostCSC& g, int rank, int ngpus,
   }
   int* hubidx = nullptr;
   float *gram = nullptr, *rhs = nullptr;
+  // Gauss-Seidel alternation (n_users > 0): split my bin lists at the
+  // LOCAL user/item boundary; hub scratch slots become phase-relative
+  // (see SingleGpuCF). A rank whose partition is all-items still joins
+  // both publish collectives with an empty user half-sweep.
+  bool alt = als && n_users > 0;
+  BinSplit split;
+  if (alt && c.vp) {
+    V_ID lb = n_users <= c.rl ? 0
+              : (n_users - c.rl < c.vp ? n_users - c.rl : c.vp);
+    split = split_bins_at(c.bins, lb, c.s);
+  }
   if (als && c.bins.nbig) {
     hubidx = arena.alloc_n<int>(c.vp ? c.vp : 1);
     gram = arena.alloc_n<float>((uint64_t)c.bins.nbig * 64 * 64);
@@ -459,7 +470,8 @@ int col_filter_multi_child(const HostCSC& g, int rank, int ngpus,
                           hipMemcpyDeviceToHost, c.s));
     LUX_OK(hipStreamSynchronize(c.s));
     std::vector<int> hidx(c.vp, -1);
-    for (uint32_t i = 0; i < c.bins.nbig; i++) hidx[hv[i]] = (int)i;
+    for (uint32_t i = 0; i < c.bins.nbig; i++)
+      hidx[hv[i]] = (int)(alt && i >= split.nbigu ? i - split.nbigu : i);
     LUX_OK(hipMemcpyAsync(hubidx, hidx.data(), 4ull * c.vp,
                           hipMemcpyHostToDevice, c.s));
     LUX_OK(hipStreamSynchronize(c.s));  // hidx leaves scope
@@ -467,32 +479,8 @@ int col_filter_multi_child(const HostCSC& g, int rank, int ngpus,
   float* bar = arena.alloc_n<float>(1);
   multi_barrier(c, bar);
   double t0 = now_seconds();
-  for (int it = 0; it < iters; it++) {
-    if (c.vp) {
-      if (als) {
-        LUX_OK(hipMemcpyAsync(new_, old_ + (uint64_t)c.rl * K,
-                              4ull * c.vp * K, hipMemcpyDeviceToDevice,
-                              c.s));
-        if (c.bins.nbig) {
-          LUX_OK(hipMemsetAsync(gram, 0, 4ull * c.bins.nbig * 64 * 64,
-                                c.s));
-          LUX_OK(hipMemsetAsync(rhs, 0, 4ull * c.bins.nbig * 64, c.s));
-        }
-        lux_gpu_cf_als_iter((uint64_t)c.s, c.bins.n0, c.bins.bin0,
-                            c.bins.n1, c.bins.bin1, c.bins.n2, c.bins.bin2,
-                            c.bins.nbig, c.bins.bin2v, hubidx, gram, rhs,
-                            c.row_ptr, c.col, c.w, old_, nullptr, new_,
-                            c.rl, K);
-      } else {
-        lux_gpu_cf_seed((uint64_t)c.s, (uint64_t)c.vp * K,
-                        old_ + (uint64_t)c.rl * K, new_);
-        lux_gpu_cf_iter((uint64_t)c.s, c.bins.n0, c.bins.bin0, c.bins.n1,
-                        c.bins.bin1, c.bins.n2, c.bins.bin2, c.bins.nbig,
-                        c.bins.bin2v, c.row_ptr, c.col, c.w, old_, new_,
-                        c.rl, K);
-      }
-    }
-    // all-gather(v) of K-dim vector slices (direct pairwise over xGMI)
+  // all-gather(v) of K-dim vector slices (direct pairwise over xGMI)
+  auto publish = [&]() {
     LUX_NCCL(ncclGroupStart());
     for (int r = 0; r < ngpus; r++) {
       if (r == rank) continue;
@@ -508,6 +496,57 @@ int col_filter_multi_child(const HostCSC& g, int rank, int ngpus,
     if (c.vp)
       LUX_OK(hipMemcpyAsync(old_ + (uint64_t)c.rl * K, new_,
                             4ull * c.vp * K, hipMemcpyDeviceToDevice, c.s));
+  };
+  for (int it = 0; it < iters; it++) {
+    if (c.vp && als) {
+      LUX_OK(hipMemcpyAsync(new_, old_ + (uint64_t)c.rl * K,
+                            4ull * c.vp * K, hipMemcpyDeviceToDevice,
+                            c.s));
+      if (c.bins.nbig) {
+        LUX_OK(hipMemsetAsync(gram, 0, 4ull * c.bins.nbig * 64 * 64,
+                              c.s));
+        LUX_OK(hipMemsetAsync(rhs, 0, 4ull * c.bins.nbig * 64, c.s));
+      }
+    }
+    if (alt) {
+      // user half-sweep against old item factors, publish, then item
+      // half-sweep against the globally UPDATED users in old_
+      if (c.vp)
+        lux_gpu_cf_als_iter((uint64_t)c.s, split.n0u, c.bins.bin0,
+                            split.n1u, c.bins.bin1, split.n2u, c.bins.bin2,
+                            split.nbigu, c.bins.bin2v, hubidx, gram, rhs,
+                            c.row_ptr, c.col, c.w, old_, nullptr, new_,
+                            c.rl, K);
+      publish();
+      if (c.vp)
+        lux_gpu_cf_als_iter(
+            (uint64_t)c.s, c.bins.n0 - split.n0u, c.bins.bin0 + split.n0u,
+            c.bins.n1 - split.n1u, c.bins.bin1 + split.n1u,
+            c.bins.n2 - split.n2u, c.bins.bin2 + split.n2u,
+            c.bins.nbig - split.nbigu, c.bins.bin2v + split.nbigu, hubidx,
+            gram ? gram + (uint64_t)split.nbigu * 64 * 64 : nullptr,
+            rhs ? rhs + (uint64_t)split.nbigu * 64 : nullptr, c.row_ptr,
+            c.col, c.w, old_, nullptr, new_, c.rl, K);
+      publish();
+      continue;
+    }
+    if (c.vp) {
+      if (als) {
+        lux_gpu_cf_als_iter((uint64_t)c.s, c.bins.n0, c.bins.bin0,
+                            c.bins.n1, c.bins.bin1, c.bins.n2, c.bins.bin2,
+                            c.bins.nbig, c.bins.bin2v, hubidx, gram, rhs,
+                            c.row_ptr, c.col, c.w, old_, nullptr, new_,
+                            c.rl, K);
+      } else {
+        lux_gpu_cf_seed((uint64_t)c.s, (uint64_t)c.vp * K,
+                        old_ + (uint64_t)c.rl * K, new_);
+        lux_gpu_cf_iter((uint64_t)c.s, c.bins.n0, c.bins.bin0, c.bins.n1,
+                        c.bins.bin1, c.bins.n2, c.bins.bin2, c.bins.nbig,
+                        c.bins.bin2v, c.row_ptr, c.col, c.w, old_, new_,
+                        c.rl, K);
+      }
+    }
+    publish();
   }
   multi_barrier(c, bar);
   double secs = now_seconds() - t0;

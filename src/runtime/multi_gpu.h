@@ -24,9 +24,11 @@ int pagerank_multi_child(const HostCSC& g, int rank, int ngpus,
 int components_multi_child(const HostCSC& g, int rank, int ngpus,
                            const char* idfile, bool check, const char* dump,
                            bool verbose);
+// n_users > 0 (bipartite boundary) makes each ALS sweep alternate:
+// user half-sweep, publish, item half-sweep against updated users.
 int col_filter_multi_child(const HostCSC& g, int rank, int ngpus,
                            const char* idfile, int K, int iters, bool als,
-                           const char* dump);
+                           const char* dump, lux::V_ID n_users = 0);
 int push_multi_child(const HostCSC& g, int rank, int ngpus, bool is_min,
                      lux::V_ID source, const char* idfile, bool check,
                      const char* dump, bool verbose);

@@ -571,25 +571,61 @@ uint64_t SingleGpuCCUnionFind::check() {
 
 // ---------------- CF ----------------
 
+// Stable-partition the bin lists at local row boundary lb (rows < lb
+// first). Host roundtrip; init-time only.
+BinSplit split_bins_at(Bins& b, V_ID lb, hipStream_t s) {
+  LUX_OK(hipStreamSynchronize(s));
+  BinSplit out;
+  auto part_v = [&](V_ID* dev, uint32_t n, uint32_t* cnt) {
+    if (!n) return;
+    std::vector<V_ID> h(n), lo, hi;
+    LUX_OK(hipMemcpy(h.data(), dev, sizeof(V_ID) * n,
+                     hipMemcpyDeviceToHost));
+    for (V_ID v : h) (v < lb ? lo : hi).push_back(v);
+    *cnt = (uint32_t)lo.size();
+    lo.insert(lo.end(), hi.begin(), hi.end());
+    LUX_OK(hipMemcpy(dev, lo.data(), sizeof(V_ID) * n,
+                     hipMemcpyHostToDevice));
+  };
+  part_v(b.bin0, b.n0, &out.n0u);
+  part_v(b.bin1, b.n1, &out.n1u);
+  part_v(b.bin2v, b.nbig, &out.nbigu);
+  if (b.n2) {
+    std::vector<lux_uint2> h(b.n2), lo, hi;
+    LUX_OK(hipMemcpy(h.data(), b.bin2, sizeof(lux_uint2) * b.n2,
+                     hipMemcpyDeviceToHost));
+    for (auto& e : h) (e.x < lb ? lo : hi).push_back(e);
+    out.n2u = (uint32_t)lo.size();
+    lo.insert(lo.end(), hi.begin(), hi.end());
+    LUX_OK(hipMemcpy(b.bin2, lo.data(), sizeof(lux_uint2) * b.n2,
+                     hipMemcpyHostToDevice));
+  }
+  return out;
+}
+
 SingleGpuCF::SingleGpuCF(const DeviceGraph& g, int K, DeviceArena& arena,
-                         hipStream_t s, bool als)
-    : g_(g), s_(s), K_(K), als_(als) {
+                         hipStream_t s, bool als, V_ID n_users)
+    : g_(g), s_(s), K_(K), als_(als), nu_(als ? n_users : 0) {
   row_ptr_ = arena.alloc_n<E_ID>(g.nv + 1);
   lux_gpu_local_row_ptr((uint64_t)s, g.nv, 0, g.col_end, row_ptr_);
   bins_.build(row_ptr_, g.nv, g.ne, arena, s);
+  if (nu_) split_ = split_bins_at(bins_, nu_ < g.nv ? nu_ : g.nv, s);
   old_ = arena.alloc_n<float>((size_t)g.nv * K);
   new_ = arena.alloc_n<float>((size_t)g.nv * K);
   std::vector<float> h((size_t)g.nv * K, sqrtf(1.0f / K));
   LUX_OK(hipMemcpyAsync(old_, h.data(), sizeof(float) * h.size(),
                            hipMemcpyHostToDevice, s));
   if (als_ && bins_.nbig) {
-    // hub scratch: slot map (local id -> [0, nbig)), Gram + rhs
+    // hub scratch: slot map (local id -> scratch slot), Gram + rhs.
+    // With alternation, slots are PHASE-relative: the item half-sweep
+    // passes gram/rhs offset by nbigu, so item hub i maps to i - nbigu.
     hubidx_ = arena.alloc_n<int>(g.nv);
     std::vector<V_ID> hubs(bins_.nbig);
     LUX_OK(hipMemcpy(hubs.data(), bins_.bin2v,
                      sizeof(V_ID) * bins_.nbig, hipMemcpyDeviceToHost));
     std::vector<int> hidx(g.nv, -1);
-    for (uint32_t i = 0; i < bins_.nbig; i++) hidx[hubs[i]] = (int)i;
+    for (uint32_t i = 0; i < bins_.nbig; i++)
+      hidx[hubs[i]] = (int)(nu_ && i >= split_.nbigu ? i - split_.nbigu : i);
     LUX_OK(hipMemcpyAsync(hubidx_, hidx.data(), sizeof(int) * g.nv,
                           hipMemcpyHostToDevice, s));
     LUX_OK(hipStreamSynchronize(s));  // hidx leaves scope
@@ -613,10 +649,29 @@ void SingleGpuCF::iterate(int iters) {
         LUX_OK(hipMemsetAsync(rhs_, 0,
                               sizeof(float) * (size_t)bins_.nbig * 64, s_));
       }
-      lux_gpu_cf_als_iter((uint64_t)s_, bins_.n0, bins_.bin0, bins_.n1,
-                          bins_.bin1, bins_.n2, bins_.bin2, bins_.nbig,
-                          bins_.bin2v, hubidx_, gram_, rhs_, row_ptr_,
-                          g_.src, g_.weight, old_, nullptr, new_, 0, K_);
+      if (nu_) {
+        // Gauss-Seidel alternation: users from old_, then items from the
+        // UPDATED users already in new_ (item rows of new_ still hold the
+        // seeded old values, which no item row reads)
+        lux_gpu_cf_als_iter((uint64_t)s_, split_.n0u, bins_.bin0,
+                            split_.n1u, bins_.bin1, split_.n2u, bins_.bin2,
+                            split_.nbigu, bins_.bin2v, hubidx_, gram_, rhs_,
+                            row_ptr_, g_.src, g_.weight, old_, nullptr,
+                            new_, 0, K_);
+        lux_gpu_cf_als_iter(
+            (uint64_t)s_, bins_.n0 - split_.n0u, bins_.bin0 + split_.n0u,
+            bins_.n1 - split_.n1u, bins_.bin1 + split_.n1u,
+            bins_.n2 - split_.n2u, bins_.bin2 + split_.n2u,
+            bins_.nbig - split_.nbigu, bins_.bin2v + split_.nbigu, hubidx_,
+            gram_ ? gram_ + (size_t)split_.nbigu * 64 * 64 : nullptr,
+            rhs_ ? rhs_ + (size_t)split_.nbigu * 64 : nullptr, row_ptr_,
+            g_.src, g_.weight, new_, nullptr, new_, 0, K_);
+      } else {
+        lux_gpu_cf_als_iter((uint64_t)s_, bins_.n0, bins_.bin0, bins_.n1,
+                            bins_.bin1, bins_.n2, bins_.bin2, bins_.nbig,
+                            bins_.bin2v, hubidx_, gram_, rhs_, row_ptr_,
+                            g_.src, g_.weight, old_, nullptr, new_, 0, K_);
+      }
     } else {
       // SGD contract (cf.hip): output pre-seeded old*(1-GAMMA*LAMBDA),
       // sweeps add GAMMA*acc

@@ -72,6 +72,15 @@ struct Bins {
              hipStream_t s);
 };
 
+// Stable-partition counts after splitting every bin list at a local row
+// boundary (rows < lb first). Used by ALS alternation: the user
+// half-sweep runs the first n*u entries of each list, the item half-sweep
+// the rest via offset pointers. Init-time host roundtrip.
+struct BinSplit {
+  uint32_t n0u = 0, n1u = 0, n2u = 0, nbigu = 0;
+};
+BinSplit split_bins_at(Bins& b, V_ID lb, hipStream_t s);
+
 // src-blocked CSC + per-block compacted bins — the native twin of
 // engine.py build_blocked: regroups edges by 32 MB src window so the
 // random old-property gather stays Infinity-Cache-resident, with
@@ -166,9 +175,13 @@ class SingleGpuCCUnionFind {
 
 class SingleGpuCF {
  public:
-  // als: exact MFMA ALS sweeps (cf_als.hip, K <= 64) instead of SGD
+  // als: exact MFMA ALS sweeps (cf_als.hip, K <= 64) instead of SGD.
+  // n_users > 0 (bipartite boundary) turns each ALS sweep into true
+  // Gauss-Seidel alternation: user rows solve against old item factors,
+  // item rows against the UPDATED users (exact in place — a bipartite
+  // row only reads the other side). 0 = unknown -> simultaneous Jacobi.
   SingleGpuCF(const DeviceGraph& g, int K, DeviceArena& arena,
-              hipStream_t s, bool als = false);
+              hipStream_t s, bool als = false, V_ID n_users = 0);
   void iterate(int iters);
   const float* vectors() const { return old_; }
 
@@ -183,6 +196,8 @@ class SingleGpuCF {
   int* hubidx_ = nullptr;       // ALS hub scratch slot map
   float* gram_ = nullptr;       // ALS: nbig x 64 x 64
   float* rhs_ = nullptr;        // ALS: nbig x 64
+  V_ID nu_ = 0;                 // bipartite boundary (0 = Jacobi)
+  BinSplit split_;              // bin counts below nu_ (alternation)
 };
 
 }  // namespace lux

@@ -174,7 +174,7 @@ def test_cf_als_reference_beats_sgd():
     g = Graph.bipartite(300, 60, 6000, seed=21)
     K, sweeps = 16, 2
     sgd = cpu_ref.cf(g, K, sweeps)
-    als = cpu_ref.cf_als(g, K, sweeps)
+    als = cpu_ref.cf_als(g, K, sweeps, n_users=300)  # alternating sweeps
     assert cpu_ref.cf_loss(g, K, als) < cpu_ref.cf_loss(g, K, sgd)
 
 

@@ -122,8 +122,9 @@ def test_native_col_filter_als(tmp_path):
           "300", "-ne", "60000", "-o", lux])
     _run([f"{BIN}/col_filter", "-file", lux, "-ni", "3", "-k", "32",
           "-dump", o1])
+    # -users gives -als the bipartite boundary -> Gauss-Seidel alternation
     _run([f"{BIN}/col_filter", "-file", lux, "-ni", "3", "-k", "32", "-als",
-          "-dump", o2])
+          "-users", "2000", "-dump", o2])
     g = Graph.load(lux, want_weights=True)
     sgd, _ = ck.load_state(o1)
     als, _ = ck.load_state(o2)
@@ -235,7 +236,8 @@ def test_native_col_filter_multi_rccl_world1(tmp_path):
     want = cpu_ref.cf(g, 20, 3)
     np.testing.assert_allclose(got, want, rtol=2e-4, atol=1e-7)
     r = subprocess.run([f"{BIN}/col_filter", "-file", lux, "-ni", "3",
-                        "-k", "32", "-als", "-dump", o2], cwd=ROOT,
+                        "-k", "32", "-als", "-users", "800", "-dump", o2],
+                       cwd=ROOT,
                        capture_output=True, text=True, timeout=300, env=env)
     assert r.returncode == 0, r.stdout + r.stderr
     als, _ = ck.load_state(o2)
