@@ -17,6 +17,26 @@ def _stream():
     return torch.cuda.current_stream().cuda_stream
 
 
+def als_init(nv, K):
+    """Deterministic jittered ALS init: sqrt(1/K) * [0.5, 1.5) per
+    component (splitmix64 hash of the flat index; identical on every
+    rank). The reference's constant sqrt(1/K) init makes the FIRST
+    alternating half-sweep rank-1 degenerate — every user lands on the
+    same line, the item solve then amplifies null-space noise by 1/lambda
+    (measured: max|item| ~270 after one fp32 sweep, divergent feedback in
+    bf16) — so the exact-solve optimizer needs the degeneracy broken at
+    init. SGD keeps the reference's constant init."""
+    import math
+    import numpy as np
+    i = np.arange(nv * K, dtype=np.uint64)
+    z = (i + np.uint64(0x9E3779B97F4A7C15)) * np.uint64(0xBF58476D1CE4E5B9)
+    z ^= z >> np.uint64(30)
+    z *= np.uint64(0x94D049BB133111EB)
+    z ^= z >> np.uint64(27)
+    u = (z >> np.uint64(11)).astype(np.float64) * (2.0 ** -53)
+    return (math.sqrt(1.0 / K) * (0.5 + u)).astype(np.float32)
+
+
 class CFEngine:
     def __init__(self, part: GraphPart, K=64):
         assert part.weight is not None, "CF needs a weighted graph"
@@ -71,8 +91,7 @@ class CFALSEngine:
         self.K = K
         part.build_bins()
         device = part.device
-        v0 = math.sqrt(1.0 / K)
-        self.old = torch.full((part.nv * K,), v0, dtype=F32, device=device)
+        self.old = torch.from_numpy(als_init(part.nv, K)).to(device)
         self.new_part = torch.empty(part.vp * K, dtype=F32, device=device)
         self.verts_elems = [v * K for v in part.verts_all]
         self.left_elems = [l * K for l in part.row_left_all]

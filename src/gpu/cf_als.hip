@@ -591,14 +591,26 @@ void lux_gpu_cf_als_iter(uint64_t stream, uint32_t n0, const V_ID* bin0,
   hipStream_t s = (hipStream_t)stream;
   CFAlsArgs a{row_ptr, col, w, oldv, (const __bf16*)oldv_bf, newv, row_left,
               K};
-  // bf16 Gram (16x16x32 MFMA, fp32 accumulate) is the default; exact
-  // fp32 (16x16x4) via LUX_ALS_F32=1; mode 2 = dword-pair bf16 gathers
-  // (only when the engine passed a bf16 replica)
-  int mode = getenv("LUX_ALS_F32") ? 0 : (oldv_bf ? 2 : 1);
+  // Mode defaults are PER KERNEL CLASS. Fused per-row solves: exact fp32
+  // Gram (16x16x4). bf16 Gram noise (~0.4% of ||G||) swamps sigma_min for
+  // moderate-degree rows during alternation — measured: item factors jump
+  // 0.32 -> 409 after one bf16 half-sweep and the feedback diverges to
+  // 1e13 loss. Chunk/hub Gram (deg >= T2 = 2048): bf16 16x16x32 MFMA
+  // stays the default — averaging over >= 2048 sources keeps ||G|| /
+  // sigma_min well above the bf16 noise floor, and that is where the
+  // NetFlix-shaped item volume (avg item degree ~5.7K) lives.
+  // LUX_ALS_F32=1 forces fp32 everywhere; LUX_ALS_BF16=1 forces bf16 for
+  // the fused solves too; mode 2 = dword-pair bf16 gathers (only when
+  // the engine passed a bf16 replica; LUX_ALS_BF_GATHER experiment).
+  int mode = getenv("LUX_ALS_F32")
+                 ? 0
+                 : (oldv_bf ? 2 : (getenv("LUX_ALS_BF16") ? 1 : 0));
+  int chunk_mode = getenv("LUX_ALS_F32") ? 0 : (oldv_bf ? 2 : 1);
 #define LUX_ALS_LAUNCH(K_, GRID_, N_, LIST_)                                  do {                                                                          if (mode == 2)                                                                hipLaunchKernelGGL(K_<2>, GRID_, dim3(ALS_TB), 0, s, N_, LIST_, a);       else if (mode == 1)                                                           hipLaunchKernelGGL(K_<1>, GRID_, dim3(ALS_TB), 0, s, N_, LIST_, a);       else                                                                          hipLaunchKernelGGL(K_<0>, GRID_, dim3(ALS_TB), 0, s, N_, LIST_, a);     } while (0)
   if (nbig) {
     uint32_t gw = (n2 + 1) / 2;
     dim3 grid(gw > MAX_GRID ? MAX_GRID : gw);
+    int mode = chunk_mode;  // hub Gram keeps bf16 (see above)
     if (mode == 2)
       hipLaunchKernelGGL(cf_als_gram_chunk_kernel<2>, grid,
                          dim3(ALS_CHUNK_TB), 0, s, n2, bin2, (V_ID)8192,

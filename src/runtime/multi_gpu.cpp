@@ -444,6 +444,8 @@ int col_filter_multi_child(const HostCSC& g, int rank, int ngpus,
   float* new_ = arena.alloc_n<float>((uint64_t)(c.vp ? c.vp : 1) * K);
   {
     std::vector<float> h((uint64_t)g.nv * K, sqrtf(1.0f / K));
+    if (als)  // jittered init (see als_init_val); SGD keeps the constant
+      for (size_t i = 0; i < h.size(); i++) h[i] = als_init_val(i, K);
     LUX_OK(hipMemcpyAsync(old_, h.data(), 4ull * g.nv * K,
                           hipMemcpyHostToDevice, c.s));
     LUX_OK(hipStreamSynchronize(c.s));  // h leaves scope

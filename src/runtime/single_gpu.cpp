@@ -613,6 +613,8 @@ SingleGpuCF::SingleGpuCF(const DeviceGraph& g, int K, DeviceArena& arena,
   old_ = arena.alloc_n<float>((size_t)g.nv * K);
   new_ = arena.alloc_n<float>((size_t)g.nv * K);
   std::vector<float> h((size_t)g.nv * K, sqrtf(1.0f / K));
+  if (als_)  // jittered init (see als_init_val); SGD keeps the constant
+    for (size_t i = 0; i < h.size(); i++) h[i] = als_init_val(i, K);
   LUX_OK(hipMemcpyAsync(old_, h.data(), sizeof(float) * h.size(),
                            hipMemcpyHostToDevice, s));
   if (als_ && bins_.nbig) {

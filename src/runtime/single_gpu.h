@@ -81,6 +81,22 @@ struct BinSplit {
 };
 BinSplit split_bins_at(Bins& b, V_ID lb, hipStream_t s);
 
+// Deterministic jittered ALS init: sqrt(1/K) * [0.5, 1.5) per component
+// (splitmix64 of the flat index; identical on every rank — matches
+// lux_amd/cf_engine.py als_init). The constant sqrt(1/K) init makes the
+// first alternating half-sweep rank-1 degenerate (all users land on one
+// line; the item solve amplifies null-space noise by 1/lambda and the
+// bf16 path diverges), so the exact-solve optimizer breaks the
+// degeneracy at init. SGD keeps the reference's constant init.
+inline float als_init_val(uint64_t i, int K) {
+  uint64_t z = (i + 0x9E3779B97F4A7C15ull) * 0xBF58476D1CE4E5B9ull;
+  z ^= z >> 30;
+  z *= 0x94D049BB133111EBull;
+  z ^= z >> 27;
+  double u = (double)(z >> 11) * (1.0 / 9007199254740992.0);
+  return (float)(sqrt(1.0 / K) * (0.5 + u));
+}
+
 // src-blocked CSC + per-block compacted bins — the native twin of
 // engine.py build_blocked: regroups edges by 32 MB src window so the
 // random old-property gather stays Infinity-Cache-resident, with
