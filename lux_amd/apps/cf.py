@@ -36,6 +36,8 @@ def main(argv=None):
     torch.cuda.set_device(local)
     device = f"cuda:{local}"
     part = load_part(a, device, weighted=True)
+    if a.users and part.n_users is None:  # -file graphs don't carry the
+        part.n_users = a.users            # boundary; -users supplies it
     if dx.rank() == 0:
         print_memory_estimate(part.nv, part.ne, dx.world_size(),
                               weighted=True, k=a.k)

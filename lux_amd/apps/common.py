@@ -20,6 +20,9 @@ class AppArgs:
         self.check = False
         self.k = 64
         self.als = False  # col_filter: ALS (MFMA) optimizer instead of SGD
+        self.users = 0  # col_filter -als: bipartite user/item boundary for
+        #                 Gauss-Seidel alternation on -file graphs
+        #                 (synthetic bipartite specs carry it implicitly)
         self.labelprop = False  # components: reference-parity label prop
         #                         (default is the union-find fast path)
         self.synthetic = None  # e.g. "rmat:20:1000000"
@@ -46,6 +49,8 @@ def parse_input_args(argv):
             a.k = int(argv[i + 1]); i += 2
         elif f == "-als":
             a.als = True; i += 1
+        elif f == "-users":
+            a.users = int(argv[i + 1]); i += 2
         elif f == "-labelprop":
             a.labelprop = True; i += 1
         elif f == "-synthetic":

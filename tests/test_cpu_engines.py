@@ -204,3 +204,50 @@ def test_config1_pagerank_rmat16_cpu():
     deg = g.out_degrees().astype(np.float64)
     undiv = pr * np.maximum(deg, 1.0)
     assert 0.2 < undiv.sum() <= 1.0 + 1e-6
+
+
+def test_cf_als_alternation_beats_jacobi():
+    """The r2.15 negative result pinned as a regression: on a TWO-SIDED
+    bipartite graph the simultaneous-Jacobi ALS update (every row solved
+    from old values) oscillates, while Gauss-Seidel alternation converges
+    well past SGD. Ordering contract: alternating < sgd < jacobi."""
+    from lux_amd.graph import Graph
+    g = Graph.bipartite(500, 80, 16000, seed=7)
+    K, sweeps = 16, 3
+    sgd = cpu_ref.cf(g, K, sweeps)
+    jac = cpu_ref.cf_als(g, K, sweeps)              # no boundary -> Jacobi
+    alt = cpu_ref.cf_als(g, K, sweeps, n_users=500)
+    l_sgd = cpu_ref.cf_loss(g, K, sgd)
+    l_jac = cpu_ref.cf_loss(g, K, jac)
+    l_alt = cpu_ref.cf_loss(g, K, alt)
+    assert l_alt < l_sgd < l_jac
+
+
+def test_als_init_jitter():
+    """als_init: deterministic, componentwise in [0.5, 1.5)*sqrt(1/K),
+    and actually spread (the constant parity init makes the first
+    alternating half-sweep rank-1 degenerate — cf_engine docstring)."""
+    import math
+    from lux_amd.cf_engine import als_init
+    a = als_init(1000, 16)
+    b = als_init(1000, 16)
+    np.testing.assert_array_equal(a, b)
+    v0 = math.sqrt(1.0 / 16)
+    assert a.min() >= 0.5 * v0 and a.max() < 1.5 * v0
+    assert a.std() > 0.2 * v0  # genuinely jittered, not near-constant
+    # the native engines use the same splitmix64 formula (als_init_val,
+    # src/runtime/single_gpu.h) — spot-check values against a direct
+    # python transcription of that C++ helper
+    for i in (0, 1, 7, 999):
+        z = (i + 0x9E3779B97F4A7C15) * 0xBF58476D1CE4E5B9 % (1 << 64)
+        z ^= z >> 30
+        z = z * 0x94D049BB133111EB % (1 << 64)
+        z ^= z >> 27
+        u = (z >> 11) * 2.0 ** -53
+        assert abs(a[i] - v0 * (0.5 + u)) < 1e-6
+
+
+def test_cf_app_parse_users_flag():
+    from lux_amd.apps.common import parse_input_args
+    a = parse_input_args(["-als", "-users", "480189", "-k", "32"])
+    assert a.als and a.users == 480189 and a.k == 32
