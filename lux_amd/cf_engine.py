@@ -167,6 +167,11 @@ class CFALSEngine:
         and solves user rows against old item factors; 'items' solves item
         rows against the (globally published) updated users. Tests drive
         these directly to phase-lock multiple partitions in one process."""
+        if self.lb is None:
+            raise RuntimeError("half_step needs the bipartite boundary: "
+                               "part.n_users is unset (load a graph with "
+                               "-users / use the bipartite builders)")
+        assert phase in ("users", "items")
         p = self.part
         if phase == "users":
             self._begin_sweep()
