@@ -251,3 +251,16 @@ def test_cf_app_parse_users_flag():
     from lux_amd.apps.common import parse_input_args
     a = parse_input_args(["-als", "-users", "480189", "-k", "32"])
     assert a.als and a.users == 480189 and a.k == 32
+
+
+def test_cf_als_alternation_deg0_keeps_vector():
+    """Alternating sweeps leave rating-less rows untouched too (the
+    half-sweep solves only rows with in-edges on its side)."""
+    from lux_amd.graph import Graph
+    g = Graph.bipartite(50, 10, 300, seed=5)
+    K = 8
+    init = np.full((g.nv, K), 0.25, dtype=np.float32)
+    out = cpu_ref.cf_als(g, K, 2, init=init, n_users=50)
+    deg = np.diff(np.concatenate([[0], g.col_end]))
+    for v in np.nonzero(deg == 0)[0][:5]:
+        np.testing.assert_array_equal(out[v], init[v])

@@ -58,3 +58,22 @@ def test_bipartite_pair_symmetry():
     assert np.array_equal(w[0::2], w[1::2])
     # direction check: even edges go user->item, odd item->user
     assert src[0::2].max() < 500 and dst[0::2].min() >= 500
+
+
+def test_bipartite_pairs_fuzz():
+    """Pair invariants over random shapes: endpoints reversed, shared
+    weight, ids in range, and the dst histogram splits exactly ne/2 per
+    side (every rating contributes one in-edge to each side)."""
+    rng = np.random.default_rng(11)
+    for _ in range(5):
+        nu = int(rng.integers(10, 3000))
+        ni = int(rng.integers(2, 500))
+        ne = int(rng.integers(2, 20000)) * 2
+        seed = int(rng.integers(1, 1 << 30))
+        src, dst, w = nat.bipartite_edges(seed, nu, ni, ne)
+        assert np.array_equal(src[0::2], dst[1::2])
+        assert np.array_equal(dst[0::2], src[1::2])
+        assert np.array_equal(w[0::2], w[1::2])
+        assert src[0::2].max() < nu and dst[1::2].max() < nu
+        assert dst[0::2].min() >= nu and dst[0::2].max() < nu + ni
+        assert (dst < nu).sum() == ne // 2  # item->user half
