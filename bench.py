@@ -186,6 +186,11 @@ def main():
         }
         if labelprop is not None:
             out["config"]["cc_labelprop"] = labelprop
+        if args.app == "cf_als":
+            out["config"]["optimizer"] = (
+                "als-gauss-seidel-alternating (two-sided: each sweep "
+                "solves users then items; one sweep beats SGD loss >20x "
+                "- BENCHLOG r2.15)")
         print(json.dumps(out), flush=True)
 
 
